@@ -1,0 +1,277 @@
+"""Continuous-batching inference engine for one (logical) MI355X backend.
+
+This is the compute side of what the reference merely proxies: where
+ollamaMQ forwards a request over HTTP and pumps bytes back
+(reference src/dispatcher.rs:742-778), this engine admits the request into a
+resident batch, runs prefill, then emits one token per decode step from the
+hand-written HIP path, streaming tokens through a callback that the server
+layer turns into Ollama JSON-lines / OpenAI SSE chunks.
+
+Scheduling inside the engine: new sequences are prefilled as a varlen batch
+(chunked), then join the decode batch; one decode step advances every
+running sequence by one token (the reference's one-request-per-backend cap,
+src/dispatcher.rs:589, is the *scheduler's* default; the engine itself
+batches whatever it is given).
+"""
+from __future__ import annotations
+
+import itertools
+import time
+from dataclasses import dataclass, field
+from typing import Callable, Dict, List, Optional
+
+import torch
+
+from ..ops.interface import AttnMeta
+from ..ops import interface as ops
+from .kvcache import PagedKVCache
+
+
+@dataclass
+class GenParams:
+    max_tokens: int = 128
+    temperature: float = 0.0
+    top_k: int = 0
+    top_p: float = 1.0
+    stop_token: Optional[int] = None
+    seed: Optional[int] = None
+
+
+@dataclass
+class Sequence:
+    seq_id: int
+    prompt: List[int]
+    params: GenParams
+    on_token: Optional[Callable[[int, bool], None]] = None
+    slot: int = -1
+    generated: List[int] = field(default_factory=list)
+    state: str = "waiting"            # waiting -> prefill -> running -> done
+    prefill_done: int = 0             # tokens of prompt already prefilled
+    submitted_at: float = field(default_factory=time.monotonic)
+    first_token_at: Optional[float] = None
+    finished_at: Optional[float] = None
+    finish_reason: Optional[str] = None
+    cancelled: bool = False
+
+    @property
+    def total_len(self) -> int:
+        return len(self.prompt) + len(self.generated)
+
+
+class LlamaEngine:
+    def __init__(
+        self,
+        model,
+        kv_cache: PagedKVCache,
+        max_batch: int = 64,
+        prefill_chunk: int = 2048,
+    ):
+        self.model = model
+        self.kv = kv_cache
+        self.max_batch = max_batch
+        self.prefill_chunk = prefill_chunk
+        self.waiting: List[Sequence] = []
+        self.running: List[Sequence] = []
+        self.seqs: Dict[int, Sequence] = {}
+        self._ids = itertools.count(1)
+        self.dev = model.device
+        self._gen = None
+        if self.dev.type == "cuda":
+            self._gen = torch.Generator(device=self.dev)
+            self._gen.manual_seed(0xC0FFEE)
+        self.steps = 0
+        self.tokens_out = 0
+
+    # -- submission --------------------------------------------------------
+    def submit(self, prompt: List[int], params: GenParams,
+               on_token=None) -> int:
+        if len(prompt) == 0:
+            prompt = [0]
+        sid = next(self._ids)
+        seq = Sequence(sid, list(prompt), params, on_token)
+        self.seqs[sid] = seq
+        self.waiting.append(seq)
+        return sid
+
+    def cancel(self, seq_id: int) -> None:
+        seq = self.seqs.get(seq_id)
+        if seq is not None and seq.state != "done":
+            seq.cancelled = True
+
+    def has_work(self) -> bool:
+        return bool(self.waiting or self.running)
+
+    def n_active(self) -> int:
+        return len(self.waiting) + len(self.running)
+
+    # -- stepping ----------------------------------------------------------
+    def step(self) -> List[Sequence]:
+        """Advance the engine by one iteration; returns sequences finished."""
+        self._reap_cancelled()
+        admitted = self._admit()
+        if admitted:
+            self._prefill_step(admitted)
+            finished = self._postprocess(admitted)
+        elif self.running:
+            self._decode_step()
+            finished = self._postprocess(self.running)
+        else:
+            return []
+        self.steps += 1
+        for seq in finished:
+            self._finish(seq)
+        return finished
+
+    # -- internals ---------------------------------------------------------
+    def _reap_cancelled(self):
+        for lst in (self.waiting, self.running):
+            for seq in [s for s in lst if s.cancelled]:
+                lst.remove(seq)
+                seq.finish_reason = "cancelled"
+                self._finish(seq, emit=False)
+
+    def _admit(self) -> List[Sequence]:
+        """Move waiting sequences into a prefill batch (chunked varlen)."""
+        batch: List[Sequence] = []
+        budget = self.prefill_chunk
+        room = self.max_batch - len(self.running)
+        for seq in list(self.waiting):
+            if room <= 0 or budget <= 0:
+                break
+            todo = len(seq.prompt) - seq.prefill_done
+            take = min(todo, budget)
+            if take <= 0:
+                break
+            new_len = seq.prefill_done + take
+            if not self.kv.can_fit(new_len - self.kv.seq_lens[seq.slot]
+                                   if seq.slot >= 0 else new_len):
+                break
+            if seq.slot < 0:
+                try:
+                    seq.slot = self.kv.alloc_slot()
+                except RuntimeError:
+                    break
+            seq.state = "prefill"
+            seq._chunk = take  # tokens of prompt this step
+            batch.append(seq)
+            budget -= take
+            if new_len == len(seq.prompt):
+                room -= 1
+        return batch
+
+    def _build_meta(self, seqs, q_lens, mode) -> AttnMeta:
+        dev = self.dev
+        slot_ids = torch.tensor([s.slot for s in seqs], dtype=torch.int32,
+                                device=dev)
+        seq_lens = torch.tensor([self.kv.seq_lens[s.slot] for s in seqs],
+                                dtype=torch.int32, device=dev)
+        cu = [0]
+        for ql in q_lens:
+            cu.append(cu[-1] + ql)
+        cu_q = torch.tensor(cu, dtype=torch.int32, device=dev)
+        logits_idx = torch.tensor([c - 1 for c in cu[1:]], dtype=torch.long,
+                                  device=dev)
+        return AttnMeta(
+            mode=mode, slot_ids=slot_ids, seq_lens=seq_lens, cu_q=cu_q,
+            logits_idx=logits_idx, max_q=max(q_lens),
+            max_kv=int(max(self.kv.seq_lens[s.slot] for s in seqs)),
+        )
+
+    def _forward(self, seqs, token_list, pos_list, q_lens, mode):
+        dev = self.dev
+        tokens = torch.tensor(token_list, dtype=torch.int32, device=dev)
+        positions = torch.tensor(pos_list, dtype=torch.int32, device=dev)
+        slot_per_tok = torch.tensor(
+            [s.slot for s, ql in zip(seqs, q_lens) for _ in range(ql)],
+            dtype=torch.int32, device=dev)
+        for s, ql in zip(seqs, q_lens):
+            self.kv.ensure(s.slot, self.kv.seq_lens[s.slot] + ql)
+        meta = self._build_meta(seqs, q_lens, mode)
+        logits = self.model.forward(tokens, positions, self.kv, slot_per_tok,
+                                    meta)
+        return logits
+
+    def _sample(self, seqs, logits):
+        temps = torch.tensor([s.params.temperature for s in seqs],
+                             dtype=torch.float32, device=logits.device)
+        # engine-wide top_k/top_p: use per-seq params of the first seq's
+        # class; mixed params fall back to the max (conservative).
+        top_k = max((s.params.top_k or 0) for s in seqs)
+        top_p = min((s.params.top_p if s.params.top_p else 1.0) for s in seqs)
+        toks = ops.sample(logits, temps, top_k, top_p, self._gen)
+        return toks.tolist()
+
+    def _prefill_step(self, batch: List[Sequence]):
+        token_list, pos_list, q_lens = [], [], []
+        for s in batch:
+            take = s._chunk
+            start = s.prefill_done
+            token_list.extend(s.prompt[start:start + take])
+            pos_list.extend(range(start, start + take))
+            q_lens.append(take)
+        logits = self._forward(batch, token_list, pos_list, q_lens, "prefill")
+        # Only sequences whose whole prompt is now in KV emit a token.
+        rows = [i for i, s in enumerate(batch)
+                if s.prefill_done + s._chunk == len(s.prompt)]
+        done_prefill = [batch[i] for i in rows]
+        toks: List[int] = []
+        if done_prefill:
+            sub = logits if len(rows) == len(batch) else logits[rows]
+            toks = self._sample(done_prefill, sub)
+        for s in batch:
+            s.prefill_done += s._chunk
+            if s.prefill_done == len(s.prompt):
+                tok = toks[done_prefill.index(s)]
+                s.generated.append(int(tok))
+                s.first_token_at = time.monotonic()
+                s.state = "running"
+                self.waiting.remove(s)
+                self.running.append(s)
+                self.tokens_out += 1
+                if s.on_token:
+                    s.on_token(int(tok), False)
+
+    def _decode_step(self):
+        seqs = self.running
+        token_list = [s.generated[-1] for s in seqs]
+        pos_list = [s.total_len - 1 for s in seqs]
+        q_lens = [1] * len(seqs)
+        logits = self._forward(seqs, token_list, pos_list, q_lens, "decode")
+        toks = self._sample(seqs, logits)
+        now = time.monotonic()
+        for s, tok in zip(seqs, toks):
+            s.generated.append(int(tok))
+            self.tokens_out += 1
+            if s.first_token_at is None:
+                s.first_token_at = now
+            if s.on_token:
+                s.on_token(int(tok), False)
+
+    def _postprocess(self, seqs) -> List[Sequence]:
+        finished = []
+        for s in seqs:
+            if s.state != "running":
+                continue
+            p = s.params
+            if len(s.generated) >= p.max_tokens:
+                s.finish_reason = "length"
+                finished.append(s)
+            elif p.stop_token is not None and s.generated \
+                    and s.generated[-1] == p.stop_token:
+                s.finish_reason = "stop"
+                finished.append(s)
+        return finished
+
+    def _finish(self, seq: Sequence, emit: bool = True):
+        if seq in self.running:
+            self.running.remove(seq)
+        if seq in self.waiting:
+            self.waiting.remove(seq)
+        if seq.slot >= 0:
+            self.kv.free_slot(seq.slot)
+            seq.slot = -1
+        seq.state = "done"
+        seq.finished_at = time.monotonic()
+        if emit and seq.on_token:
+            seq.on_token(-1, True)  # done marker
+        self.seqs.pop(seq.seq_id, None)

@@ -1,0 +1,220 @@
+"""Llama-3 architecture, MI355X-first implementation.
+
+The reference (Chleba/ollamaMQ) delegates all model math to external Ollama
+servers over HTTP (reference src/dispatcher.rs:742-747); this module is the
+in-process compute contract those backends implicitly satisfied
+(SURVEY.md §2 "CUDA kernels ... There are none").
+
+Design notes (MI355X):
+* weights live resident in HBM3E as bf16 (8B = ~16 GB of 288 GB);
+* projection GEMMs go through torch F.linear (hipBLASLt/rocBLAS) — plain
+  library GEMMs; every fused hot op (fused residual+RMSNorm, RoPE,
+  prefill/decode attention over the paged KV pool, SwiGLU, sampler) is a
+  hand-written gfx950 HIP kernel behind ollamamq_amd.ops;
+* tensor parallelism: column-parallel QKV/gate-up, row-parallel o/down with
+  one RCCL all-reduce each (2/layer), vocab-parallel logits with all-gather.
+"""
+from __future__ import annotations
+
+import math
+from dataclasses import dataclass, field
+from typing import Optional
+
+import torch
+
+from ..ops import interface as ops
+
+
+@dataclass
+class LlamaConfig:
+    name: str = "llama3-8b"
+    n_layers: int = 32
+    hidden: int = 4096
+    n_heads: int = 32
+    n_kv_heads: int = 8
+    head_dim: int = 128
+    ffn: int = 14336
+    vocab: int = 128256
+    norm_eps: float = 1e-5
+    rope_theta: float = 500000.0
+    max_ctx: int = 8192
+
+    @property
+    def q_dim(self) -> int:
+        return self.n_heads * self.head_dim
+
+    @property
+    def kv_dim(self) -> int:
+        return self.n_kv_heads * self.head_dim
+
+
+PRESETS = {
+    # Llama-3 8B: 32 layers, hidden 4096, 32Q/8KV, ffn 14336, vocab 128256
+    "llama3-8b": LlamaConfig(),
+    # Llama-3 70B: 80 layers, hidden 8192, 64Q/8KV, ffn 28672
+    "llama3-70b": LlamaConfig(
+        name="llama3-70b", n_layers=80, hidden=8192, n_heads=64, n_kv_heads=8,
+        ffn=28672,
+    ),
+    # Tiny config for CPU tests and fast GPU smoke: same head_dim=128 the
+    # kernels are specialized for.
+    "tiny": LlamaConfig(
+        name="tiny", n_layers=2, hidden=512, n_heads=4, n_kv_heads=2,
+        head_dim=128, ffn=1024, vocab=512, max_ctx=512, rope_theta=10000.0,
+    ),
+    # Extra-small for scheduler tests (CPU-fast).
+    "tiny-cpu": LlamaConfig(
+        name="tiny-cpu", n_layers=2, hidden=256, n_heads=2, n_kv_heads=1,
+        head_dim=128, ffn=512, vocab=256, max_ctx=256, rope_theta=10000.0,
+    ),
+}
+
+
+def _randn(shape, dev, dtype, gen, scale):
+    w = torch.empty(shape, device=dev, dtype=torch.float32)
+    w.normal_(0.0, scale, generator=gen)
+    return w.to(dtype)
+
+
+class LlamaLayer:
+    """One transformer block's resident weights (possibly TP-sharded)."""
+
+    __slots__ = (
+        "wqkv", "wo", "wgate_up", "wdown", "attn_norm", "mlp_norm",
+    )
+
+    def __init__(self, cfg: LlamaConfig, dev, dtype, gen, tp: int, rank: int):
+        h = cfg.hidden
+        scale = 1.0 / math.sqrt(h)
+        nh, nkv = cfg.n_heads // tp, max(1, cfg.n_kv_heads // tp)
+        d = cfg.head_dim
+        # Fused column-parallel QKV: [q_shard + k_shard + v_shard, hidden]
+        self.wqkv = _randn((nh * d + 2 * nkv * d, h), dev, dtype, gen, scale)
+        # Row-parallel output projection: [hidden, q_shard]
+        self.wo = _randn((h, nh * d), dev, dtype, gen, scale)
+        # Fused column-parallel gate+up: [2*ffn_shard, hidden]
+        f = cfg.ffn // tp
+        self.wgate_up = _randn((2 * f, h), dev, dtype, gen, scale)
+        # Row-parallel down: [hidden, ffn_shard]
+        self.wdown = _randn((h, f), dev, dtype, gen, 1.0 / math.sqrt(cfg.ffn))
+        self.attn_norm = torch.ones(h, device=dev, dtype=dtype)
+        self.mlp_norm = torch.ones(h, device=dev, dtype=dtype)
+
+
+class LlamaModel:
+    """Random-init resident Llama model, forward built on ollamamq_amd.ops.
+
+    Holds no KV state: attention reads/writes the engine's paged KV pool.
+    """
+
+    def __init__(
+        self,
+        cfg: LlamaConfig,
+        device: str = "cpu",
+        dtype: torch.dtype = torch.float32,
+        seed: int = 1234,
+        tp_rank: int = 0,
+        tp_size: int = 1,
+        process_group=None,
+    ):
+        assert cfg.n_heads % tp_size == 0, "n_heads must divide TP"
+        self.cfg = cfg
+        self.device = torch.device(device)
+        self.dtype = dtype
+        self.tp_rank = tp_rank
+        self.tp_size = tp_size
+        self.group = process_group
+        # KV-head replication when tp > n_kv_heads is not supported; shard.
+        self.n_local_heads = cfg.n_heads // tp_size
+        self.n_local_kv_heads = max(1, cfg.n_kv_heads // tp_size)
+
+        gen = torch.Generator(device="cpu")
+        gen.manual_seed(seed + tp_rank)
+        dev = self.device
+        scale = 1.0 / math.sqrt(cfg.hidden)
+        self.embed = _randn((cfg.vocab, cfg.hidden), dev, dtype, gen, scale)
+        self.layers = [
+            LlamaLayer(cfg, dev, dtype, gen, tp_size, tp_rank)
+            for _ in range(cfg.n_layers)
+        ]
+        self.final_norm = torch.ones(cfg.hidden, device=dev, dtype=dtype)
+        # Vocab-parallel LM head shard: [vocab/tp, hidden]
+        vshard = cfg.vocab // tp_size
+        self.lm_head = _randn((vshard, cfg.hidden), dev, dtype, gen, scale)
+        # RoPE cos/sin tables precomputed on host (guide: trig on device
+        # turns memory-bound RoPE into VALU-bound).
+        pos = torch.arange(cfg.max_ctx, dtype=torch.float32)
+        inv = 1.0 / (
+            cfg.rope_theta
+            ** (torch.arange(0, cfg.head_dim, 2, dtype=torch.float32) / cfg.head_dim)
+        )
+        ang = torch.outer(pos, inv)  # [max_ctx, head_dim/2]
+        self.rope_cos = ang.cos().to(dev)
+        self.rope_sin = ang.sin().to(dev)
+
+    # -- helpers -----------------------------------------------------------
+    def _allreduce(self, x: torch.Tensor) -> torch.Tensor:
+        if self.tp_size > 1:
+            torch.distributed.all_reduce(x, group=self.group)
+        return x
+
+    def weight_bytes(self) -> int:
+        n = self.embed.numel() + self.lm_head.numel() + self.final_norm.numel()
+        for l in self.layers:
+            n += (
+                l.wqkv.numel() + l.wo.numel() + l.wgate_up.numel()
+                + l.wdown.numel() + l.attn_norm.numel() + l.mlp_norm.numel()
+            )
+        return n * self.embed.element_size()
+
+    # -- forward -----------------------------------------------------------
+    def forward(
+        self,
+        tokens: torch.Tensor,      # [T] int32/int64 flat token ids
+        positions: torch.Tensor,   # [T] int32 position of each token in its seq
+        kv_cache,                  # engine.kvcache.PagedKVCache
+        slot_ids: torch.Tensor,    # [T] int32 kv slot (sequence) of each token
+        attn_meta,                 # ops.AttnMeta (prefill/decode metadata)
+    ) -> torch.Tensor:
+        """Returns logits [T_last, vocab_full] for the tokens attn_meta
+        selects as "last" (decode: all; prefill: final token per seq)."""
+        cfg = self.cfg
+        x = ops.embedding(tokens, self.embed)
+        residual = None
+        for li, layer in enumerate(self.layers):
+            normed, residual = ops.rmsnorm_residual(
+                x, residual, layer.attn_norm, cfg.norm_eps
+            )
+            qkv = torch.nn.functional.linear(normed, layer.wqkv)
+            nl, nkl, d = self.n_local_heads, self.n_local_kv_heads, cfg.head_dim
+            q, k, v = qkv.split([nl * d, nkl * d, nkl * d], dim=-1)
+            q = q.view(-1, nl, d)
+            k = k.view(-1, nkl, d)
+            v = v.view(-1, nkl, d)
+            ops.rope(q, k, positions, self.rope_cos, self.rope_sin)
+            ops.kv_append(kv_cache, li, k, v, slot_ids, positions)
+            attn = ops.attention(q, kv_cache, li, attn_meta)
+            x = torch.nn.functional.linear(attn.view(-1, nl * d), layer.wo)
+            self._allreduce(x)  # RCCL all-reduce #1 (TP)
+            normed, residual = ops.rmsnorm_residual(
+                x, residual, layer.mlp_norm, cfg.norm_eps
+            )
+            gate_up = torch.nn.functional.linear(normed, layer.wgate_up)
+            act = ops.swiglu(gate_up)
+            x = torch.nn.functional.linear(act, layer.wdown)
+            self._allreduce(x)  # RCCL all-reduce #2 (TP)
+
+        # Only the last token of each sequence needs logits.
+        idx = attn_meta.logits_idx
+        if idx is not None:
+            h = x[idx] + residual[idx]
+        else:
+            h = x + residual
+        h, _ = ops.rmsnorm_residual(h, None, self.final_norm, cfg.norm_eps)
+        logits = torch.nn.functional.linear(h, self.lm_head)
+        if self.tp_size > 1:
+            # vocab-parallel logits: all-gather shards on the last dim
+            shards = [torch.empty_like(logits) for _ in range(self.tp_size)]
+            torch.distributed.all_gather(shards, logits, group=self.group)
+            logits = torch.cat(shards, dim=-1)
+        return logits

@@ -1,0 +1,97 @@
+"""Op dispatch: hand-written gfx950 HIP kernels on GPU, fp32 torch on CPU.
+
+Policy (task requirement): on a GPU box the HIP extension MUST be present —
+ops fail loudly rather than falling back to eager PyTorch, so a passing GPU
+test is evidence the native path ran.  Set OLLAMAMQ_FORCE_REF=1 to force the
+reference path explicitly (used only by numerics tests to produce oracles).
+"""
+from __future__ import annotations
+
+import os
+from dataclasses import dataclass
+from typing import Optional
+
+import torch
+
+from . import reference as ref
+
+
+def _use_hip(t: torch.Tensor) -> bool:
+    if not t.is_cuda:
+        return False
+    if os.environ.get("OLLAMAMQ_FORCE_REF") == "1":
+        return False
+    from . import hip  # deferred: imports/builds the extension
+    hip.require()      # raises if the extension is missing on a GPU box
+    return True
+
+
+@dataclass
+class AttnMeta:
+    """Per-forward attention metadata (flat varlen layout).
+
+    mode: "prefill" (multi-token queries) or "decode" (1 token/seq).
+    slot_ids : [S] int32  kv-cache slot of each sequence in batch order
+    seq_lens : [S] int32  total KV length per slot AFTER the kv_append
+    cu_q     : [S+1] int32 exclusive prefix sum of per-seq query counts
+    logits_idx: [S] int64 flat indices of each sequence's last token
+                (None = all tokens get logits)
+    """
+    mode: str
+    slot_ids: torch.Tensor
+    seq_lens: torch.Tensor
+    cu_q: torch.Tensor
+    logits_idx: Optional[torch.Tensor]
+    max_q: int
+    max_kv: int
+
+
+def embedding(tokens: torch.Tensor, table: torch.Tensor) -> torch.Tensor:
+    # index_select is already a single gather kernel on ROCm; no custom op.
+    return table.index_select(0, tokens.long())
+
+
+def rmsnorm_residual(x, residual, weight, eps):
+    if _use_hip(x):
+        from . import hip
+        return hip.rmsnorm_residual(x, residual, weight, eps)
+    return ref.rmsnorm_residual(x, residual, weight, eps)
+
+
+def rope(q, k, positions, cos, sin):
+    if _use_hip(q):
+        from . import hip
+        return hip.rope(q, k, positions, cos, sin)
+    return ref.rope(q, k, positions, cos, sin)
+
+
+def kv_append(cache, layer, k, v, slot_ids, positions):
+    if _use_hip(k):
+        from . import hip
+        return hip.kv_append(cache, layer, k, v, slot_ids, positions)
+    return ref.kv_append(cache, layer, k, v, slot_ids, positions)
+
+
+def attention(q, cache, layer, meta: AttnMeta):
+    if _use_hip(q):
+        from . import hip
+        if meta.mode == "decode":
+            return hip.attention_decode(q, cache, layer, meta)
+        return hip.attention_prefill(q, cache, layer, meta)
+    return ref.attention(q, cache, layer, meta)
+
+
+def swiglu(gate_up):
+    if _use_hip(gate_up):
+        from . import hip
+        return hip.swiglu(gate_up)
+    return ref.swiglu(gate_up)
+
+
+def sample(logits, temperature, top_k, top_p, generator=None):
+    # Sampler: torch ops compose on-GPU; custom fused kernel in ops/hip.py
+    # handles the greedy + temperature paths.
+    if _use_hip(logits):
+        from . import hip
+        return hip.sample(logits, temperature, top_k, top_p, generator)
+    return ref.sample(logits, temperature, top_k, top_p, generator)
