@@ -1,0 +1,160 @@
+#!/usr/bin/env python3
+"""Flagship serving benchmark — driver contract (see BASELINE.json).
+
+Measures the BASELINE.json headline: aggregate tokens/sec serving Llama-3-8B
+to 32 concurrent users per GPU backend (weak scaling: each of the N ranks is
+one independent MI355X backend running the in-process HIP engine, exactly
+config 3 of BASELINE.json).  A "step" is one continuous-batching decode
+iteration: every resident sequence advances one token through the full
+hand-written HIP path (fused RMSNorm, RoPE, paged attention, SwiGLU,
+sampler) + hipBLASLt projections.  Warmup admits + prefills the users, so
+the timed region is pure steady-state serving; p50 queue-wait (submit ->
+first token) is measured during warmup and reported in config.
+
+Launch (driver):
+  python bench.py --gpus 1 --steps K --warmup W
+  python -m torch.distributed.run --nnodes=1 --nproc-per-node N \
+      --master-addr 127.0.0.1 bench.py --gpus N --steps K --warmup W
+
+The reference publishes no number for this metric =>
+vs_baseline is null.
+"""
+import argparse
+import json
+import os
+import statistics
+import sys
+import time
+
+import torch
+
+
+def main():
+    ap = argparse.ArgumentParser()
+    ap.add_argument("--gpus", type=int, default=1)
+    ap.add_argument("--steps", type=int, default=32)
+    ap.add_argument("--warmup", type=int, default=8)
+    ap.add_argument("--users", type=int, default=32,
+                    help="concurrent users per GPU backend")
+    ap.add_argument("--prompt-len", type=int, default=512)
+    ap.add_argument("--model", type=str, default=None)
+    ap.add_argument("--max-ctx", type=int, default=4096)
+    args = ap.parse_args()
+
+    from ollamamq_amd.models import LlamaModel, PRESETS
+    from ollamamq_amd.engine import LlamaEngine, PagedKVCache, GenParams
+
+    rank = int(os.environ.get("RANK", "0"))
+    world = int(os.environ.get("WORLD_SIZE", "1"))
+    local_rank = int(os.environ.get("LOCAL_RANK", str(rank)))
+    has_gpu = torch.cuda.is_available()
+    dist = world > 1
+    if dist:
+        torch.distributed.init_process_group(
+            backend="nccl" if has_gpu else "gloo")
+
+    if has_gpu:
+        torch.cuda.set_device(local_rank)
+        device = f"cuda:{local_rank}"
+        dtype = torch.bfloat16
+        model_name = args.model or "llama3-8b"
+        users = args.users
+        prompt_len = args.prompt_len
+    else:
+        device = "cpu"
+        dtype = torch.float32
+        model_name = args.model or "tiny-cpu"
+        users = min(args.users, 4)
+        prompt_len = min(args.prompt_len, 32)
+
+    cfg = PRESETS[model_name]
+    gen_budget = args.warmup + args.steps + 8
+    ctx = min(args.max_ctx, cfg.max_ctx)
+    assert prompt_len + gen_budget <= ctx, "context too small for bench"
+    n_pages = (users + 2) * ((prompt_len + gen_budget + 15) // 16 + 2)
+
+    t_load0 = time.monotonic()
+    model = LlamaModel(cfg, device=device, dtype=dtype, seed=1234)
+    kv = PagedKVCache.for_model(
+        cfg, n_pages=n_pages, max_slots=users + 2, max_ctx=ctx,
+        device=device, dtype=dtype)
+    eng = LlamaEngine(model, kv, max_batch=users, prefill_chunk=4096)
+    load_s = time.monotonic() - t_load0
+
+    # --- submit synthetic users -------------------------------------------
+    g = torch.Generator().manual_seed(42 + rank)
+    submit_t = {}
+    ttft = {}
+    for u in range(users):
+        prompt = torch.randint(0, cfg.vocab, (prompt_len,), generator=g).tolist()
+        t_sub = time.monotonic()
+        sid = eng.submit(prompt, GenParams(max_tokens=10 ** 9))
+        submit_t[sid] = t_sub
+
+    # --- warmup: prefill everyone + W decode steps ------------------------
+    guard = 0
+    while eng.waiting and guard < 10000:
+        eng.step()
+        guard += 1
+    for sid, seq in list(eng.seqs.items()):
+        if seq.first_token_at is not None and sid in submit_t:
+            ttft[sid] = (seq.first_token_at - submit_t[sid]) * 1e3
+    for _ in range(args.warmup):
+        eng.step()
+    assert len(eng.running) == users, \
+        f"rank {rank}: {len(eng.running)} running != {users}"
+
+    # --- timed region: K decode steps -------------------------------------
+    if has_gpu:
+        torch.cuda.synchronize()
+    if dist:
+        torch.distributed.barrier()
+    t0 = time.monotonic()
+    for _ in range(args.steps):
+        eng.step()
+    if has_gpu:
+        torch.cuda.synchronize()
+    elapsed = time.monotonic() - t0
+    if dist:
+        t = torch.tensor([elapsed], dtype=torch.float64)
+        torch.distributed.all_reduce(t, op=torch.distributed.ReduceOp.MAX)
+        elapsed = float(t.item())
+        torch.distributed.barrier()
+
+    tokens_total = users * args.steps * world
+    value = tokens_total / elapsed
+    p50_wait = statistics.median(ttft.values()) if ttft else None
+
+    if rank == 0:
+        out = {
+            "metric": "agg_tokens_per_sec",
+            "value": round(value, 2),
+            "unit": "tokens/s",
+            "n_gpus": world if has_gpu else 0,
+            "steps": args.steps,
+            "warmup": args.warmup,
+            "ms_per_step": round(elapsed / args.steps * 1e3, 3),
+            "higher_is_better": True,
+            "scaling": "weak",
+            "vs_baseline": None,
+            "dtype": "bf16" if has_gpu else "fp32",
+            "data": "synthetic prompts, random-init weights",
+            "config": {
+                "model": cfg.name,
+                "global_batch": users * world,
+                "users_per_gpu": users,
+                "seq_len": prompt_len,
+                "max_ctx": ctx,
+                "parallelism": f"dp{world} (independent GPU backends)",
+                "p50_queue_wait_ms": round(p50_wait, 1) if p50_wait else None,
+                "weights_gb_per_gpu": round(model.weight_bytes() / 2 ** 30, 2),
+                "load_s": round(load_s, 1),
+            },
+        }
+        print(json.dumps(out), flush=True)
+    if dist:
+        torch.distributed.destroy_process_group()
+
+
+if __name__ == "__main__":
+    main()

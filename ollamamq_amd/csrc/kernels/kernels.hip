@@ -1,0 +1,489 @@
+// gfx950 (MI355X / CDNA4) inference kernels for ollamamq_amd.
+//
+// Hand-written HIP for the hot ops the reference delegates to external
+// Ollama backends (the reference has zero GPU code; this is
+// new construction mandated by the north star): fused residual+RMSNorm,
+// RoPE, paged-KV append, paged decode/prefill attention with LDS-staged KV
+// tiles, SwiGLU, greedy sampler.
+//
+// Conventions (cdna_hip_programming.md):
+//  * wave = 64 lanes, block sizes are multiples of 64;
+//  * bf16 I/O, fp32 math; vectorized 16B loads (G13);
+//  * LDS rows padded to kill bank conflicts (G4);
+//  * no CUDA-compat shims; compiled only for --offload-arch=gfx950.
+//
+// Exposed as extern "C" launchers taking raw device pointers + hipStream_t,
+// bound from Python via ctypes (ops/hip.py) — no torch ABI dependency.
+
+#include <hip/hip_runtime.h>
+#include <hip/hip_bf16.h>
+#include <cstdint>
+
+typedef __hip_bfloat16 bf16;
+typedef __hip_bfloat162 bf162;
+
+#define DEV static __device__ __forceinline__
+
+DEV float bf2f(bf16 x) { return __bfloat162float(x); }
+DEV bf16 f2bf(float x) { return __float2bfloat16(x); }
+
+struct short8 { short4 lo, hi; };
+
+// load 8 bf16 (16 B) and widen to 8 floats
+DEV void load8f(const bf16* p, float* out) {
+    const bf162* v = reinterpret_cast<const bf162*>(p);
+    bf162 a = v[0], b = v[1], c = v[2], d = v[3];
+    float2 fa = __bfloat1622float2(a), fb = __bfloat1622float2(b);
+    float2 fc = __bfloat1622float2(c), fd = __bfloat1622float2(d);
+    out[0] = fa.x; out[1] = fa.y; out[2] = fb.x; out[3] = fb.y;
+    out[4] = fc.x; out[5] = fc.y; out[6] = fd.x; out[7] = fd.y;
+}
+
+DEV void store8bf(bf16* p, const float* in) {
+    bf162* v = reinterpret_cast<bf162*>(p);
+    v[0] = __float22bfloat162_rn(make_float2(in[0], in[1]));
+    v[1] = __float22bfloat162_rn(make_float2(in[2], in[3]));
+    v[2] = __float22bfloat162_rn(make_float2(in[4], in[5]));
+    v[3] = __float22bfloat162_rn(make_float2(in[6], in[7]));
+}
+
+DEV float wave_reduce_sum(float x) {
+    #pragma unroll
+    for (int off = 32; off > 0; off >>= 1)
+        x += __shfl_down(x, off, 64);
+    return __shfl(x, 0, 64);
+}
+
+DEV float wave_reduce_max(float x) {
+    #pragma unroll
+    for (int off = 32; off > 0; off >>= 1)
+        x = fmaxf(x, __shfl_down(x, off, 64));
+    return __shfl(x, 0, 64);
+}
+
+// ---------------------------------------------------------------------------
+// Fused residual-add + RMSNorm.  y = rmsnorm(x + res) * w;  res_out = x + res.
+// One 256-thread block per row; H multiple of 8; bf16 x8 vectorized.
+// Memory-bound: reads x(+res)+w, writes y(+res_out) — single pass.
+// ---------------------------------------------------------------------------
+__global__ __launch_bounds__(256) void k_rmsnorm_residual(
+    bf16* __restrict__ y, bf16* __restrict__ res_out,
+    const bf16* __restrict__ x, const bf16* __restrict__ res_in,
+    const bf16* __restrict__ w, int H, float eps)
+{
+    const int row = blockIdx.x;
+    const bf16* xr = x + (int64_t)row * H;
+    const bf16* rr = res_in ? res_in + (int64_t)row * H : nullptr;
+    bf16* yr = y + (int64_t)row * H;
+    bf16* ro = res_out + (int64_t)row * H;
+
+    // pass 1: accumulate sum of squares of (x+res) while caching the sum
+    // into res_out (bf16) — one global round trip for the residual.
+    float ss = 0.f;
+    for (int i = threadIdx.x * 8; i < H; i += blockDim.x * 8) {
+        float vx[8];
+        load8f(xr + i, vx);
+        if (rr) {
+            float vr[8];
+            load8f(rr + i, vr);
+            #pragma unroll
+            for (int j = 0; j < 8; j++) vx[j] += vr[j];
+        }
+        #pragma unroll
+        for (int j = 0; j < 8; j++) ss += vx[j] * vx[j];
+        store8bf(ro + i, vx);
+    }
+    ss = wave_reduce_sum(ss);
+    __shared__ float warp_ss[4];
+    const int wid = threadIdx.x >> 6, lane = threadIdx.x & 63;
+    if (lane == 0) warp_ss[wid] = ss;
+    __syncthreads();
+    float tot = warp_ss[0] + warp_ss[1] + warp_ss[2] + warp_ss[3];
+    const float inv = rsqrtf(tot / H + eps);
+
+    for (int i = threadIdx.x * 8; i < H; i += blockDim.x * 8) {
+        float vs[8], vw[8];
+        load8f(ro + i, vs);   // L2-hot: just stored
+        load8f(w + i, vw);
+        #pragma unroll
+        for (int j = 0; j < 8; j++) vs[j] = vs[j] * inv * vw[j];
+        store8bf(yr + i, vs);
+    }
+}
+
+extern "C" void rmsnorm_residual_bf16(
+    void* y, void* res_out, const void* x, const void* res_in,
+    const void* w, int T, int H, float eps, hipStream_t stream)
+{
+    k_rmsnorm_residual<<<T, 256, 0, stream>>>(
+        (bf16*)y, (bf16*)res_out, (const bf16*)x, (const bf16*)res_in,
+        (const bf16*)w, H, eps);
+}
+
+// ---------------------------------------------------------------------------
+// RoPE (NeoX half-rotation), in place on q [T,Hq,D] and k [T,Hk,D], D=128.
+// cos/sin: [max_ctx, D/2] fp32 host-precomputed (Appendix B: no device trig).
+// One wave per (token, head); lane l rotates dims (l, l+64).
+// ---------------------------------------------------------------------------
+__global__ __launch_bounds__(256) void k_rope(
+    bf16* __restrict__ q, bf16* __restrict__ k,
+    const int* __restrict__ pos, const float* __restrict__ cost,
+    const float* __restrict__ sint, int T, int Hq, int Hk, int D)
+{
+    const int gid = blockIdx.x * 4 + (threadIdx.x >> 6);  // (token,head) flat
+    const int lane = threadIdx.x & 63;
+    const int Htot = Hq + Hk;
+    if (gid >= T * Htot) return;
+    const int t = gid / Htot, h = gid % Htot;
+    bf16* base = (h < Hq) ? q + ((int64_t)t * Hq + h) * D
+                          : k + ((int64_t)t * Hk + (h - Hq)) * D;
+    const int d2 = D / 2;                 // 64 = one lane per rotation pair
+    const float c = cost[(int64_t)pos[t] * d2 + lane];
+    const float s = sint[(int64_t)pos[t] * d2 + lane];
+    const float x1 = bf2f(base[lane]);
+    const float x2 = bf2f(base[lane + d2]);
+    base[lane] = f2bf(x1 * c - x2 * s);
+    base[lane + d2] = f2bf(x2 * c + x1 * s);
+}
+
+extern "C" void rope_bf16(
+    void* q, void* k, const void* pos, const void* cost, const void* sint,
+    int T, int Hq, int Hk, int D, hipStream_t stream)
+{
+    const int waves = T * (Hq + Hk);
+    k_rope<<<(waves + 3) / 4, 256, 0, stream>>>(
+        (bf16*)q, (bf16*)k, (const int*)pos, (const float*)cost,
+        (const float*)sint, T, Hq, Hk, D);
+}
+
+// ---------------------------------------------------------------------------
+// Paged-KV append: scatter k/v [T,KVH,D] to pool[layer] at (slot,pos).
+// Pool layout [P][KVH][page][D] bf16; one wave per (token, head);
+// lane moves 2 elems (D=128).
+// ---------------------------------------------------------------------------
+__global__ __launch_bounds__(256) void k_kv_append(
+    bf16* __restrict__ kp, bf16* __restrict__ vp,
+    const bf16* __restrict__ k, const bf16* __restrict__ v,
+    const int* __restrict__ slot, const int* __restrict__ pos,
+    const int* __restrict__ page_table,
+    int T, int KVH, int D, int page, int max_pages)
+{
+    const int gid = blockIdx.x * 4 + (threadIdx.x >> 6);
+    const int lane = threadIdx.x & 63;
+    if (gid >= T * KVH) return;
+    const int t = gid / KVH, h = gid % KVH;
+    const int p = pos[t];
+    const int pg = page_table[(int64_t)slot[t] * max_pages + p / page];
+    const int off = p % page;
+    const int64_t dst = (((int64_t)pg * KVH + h) * page + off) * D;
+    const int64_t src = ((int64_t)t * KVH + h) * D;
+    const bf162* ks = reinterpret_cast<const bf162*>(k + src);
+    const bf162* vs = reinterpret_cast<const bf162*>(v + src);
+    reinterpret_cast<bf162*>(kp + dst)[lane] = ks[lane];
+    reinterpret_cast<bf162*>(vp + dst)[lane] = vs[lane];
+}
+
+extern "C" void kv_append_bf16(
+    void* kp, void* vp, const void* k, const void* v, const void* slot,
+    const void* pos, const void* page_table, int T, int KVH, int D,
+    int page, int max_pages, hipStream_t stream)
+{
+    const int waves = T * KVH;
+    k_kv_append<<<(waves + 3) / 4, 256, 0, stream>>>(
+        (bf16*)kp, (bf16*)vp, (const bf16*)k, (const bf16*)v,
+        (const int*)slot, (const int*)pos, (const int*)page_table,
+        T, KVH, D, page, max_pages);
+}
+
+// ---------------------------------------------------------------------------
+// Paged attention, decode + prefill (flat varlen), D = 128, GQA group G ≤ 8.
+//
+// Grid (n_tiles, KVH); block = G waves (one per q head in the group).
+// Each tile covers QT query rows of one sequence.  KV is streamed in
+// CHUNK=64-token tiles staged in LDS once per block and shared by the
+// G q-heads (the GQA reuse is what makes this HBM-efficient: each KV byte
+// is read once per kv-head, not once per q-head).
+//
+// LDS rows padded (+PAD bf16) so per-lane row reads are conflict-free
+// (guide §6 G4: 256 B row stride = same-bank across all lanes).
+//
+// Decode instantiates QT=1 (one row per tile), prefill QT=16.
+// Online softmax (m,l running, fp32), causal bound per query row.
+// ---------------------------------------------------------------------------
+#define CHUNK 64
+#define DHEAD 128
+#define KPAD 4     // bf16 elems of row padding
+
+template <int QT>
+__global__ __launch_bounds__(512) void k_paged_attn(
+    bf16* __restrict__ out,           // [T, Hq, D]
+    const bf16* __restrict__ q,       // [T, Hq, D]
+    const bf16* __restrict__ kpool,   // [P][KVH][page][D]
+    const bf16* __restrict__ vpool,
+    const int* __restrict__ page_table,   // [slots][max_pages]
+    const int* __restrict__ tile_slot,    // [n_tiles]
+    const int* __restrict__ tile_q0,      // flat q row of tile start
+    const int* __restrict__ tile_pos0,    // absolute position of tile start
+    const int* __restrict__ tile_rows,    // rows in tile (≤ QT)
+    int Hq, int KVH, int page, int max_pages, float scale)
+{
+    const int tile = blockIdx.x;
+    const int kvh = blockIdx.y;
+    const int G = Hq / KVH;               // q heads per kv head (= waves)
+    const int wid = threadIdx.x >> 6;
+    const int lane = threadIdx.x & 63;
+    if (wid >= G) return;                  // blockDim may exceed G*64
+    const int qh = kvh * G + wid;
+
+    const int slot = tile_slot[tile];
+    const int q0 = tile_q0[tile];
+    const int pos0 = tile_pos0[tile];
+    const int rows = tile_rows[tile];
+    const int kv_len = pos0 + rows;        // max kv any row may see
+
+    // ---- LDS ----
+    extern __shared__ __attribute__((aligned(16))) char smem[];
+    // k_tile [CHUNK][DHEAD+KPAD] bf16, v_tile same, q_lds [G][QT][DHEAD] f32?
+    // q in fp32 would be 64KB at QT=16,G=8 — keep q bf16, widen on use.
+    bf16* k_tile = reinterpret_cast<bf16*>(smem);
+    const int KROW = DHEAD + KPAD;
+    bf16* v_tile = k_tile + CHUNK * KROW;
+    bf16* q_lds = v_tile + CHUNK * KROW;          // [G][QT][DHEAD]
+    float* p_lds = reinterpret_cast<float*>(q_lds + (int64_t)G * QT * DHEAD);
+    // p_lds [G][QT][CHUNK+1] f32
+    const int PROW = CHUNK + 1;
+
+    // ---- load this wave's Q rows into LDS (bf16, scaled later) ----
+    for (int r = 0; r < rows; r++) {
+        const bf16* qsrc = q + ((int64_t)(q0 + r) * Hq + qh) * DHEAD;
+        reinterpret_cast<bf162*>(q_lds + ((int64_t)wid * QT + r) * DHEAD)[lane]
+            = reinterpret_cast<const bf162*>(qsrc)[lane];
+    }
+
+    // ---- per-row online softmax state ----
+    float m[QT], l[QT];
+    // o accumulator: lane handles dims (2*lane, 2*lane+1) per row
+    float o0[QT], o1[QT];
+    #pragma unroll
+    for (int r = 0; r < QT; r++) {
+        m[r] = -3.0e38f; l[r] = 0.f; o0[r] = 0.f; o1[r] = 0.f;
+    }
+    __syncthreads();
+
+    const int n_chunks = (kv_len + CHUNK - 1) / CHUNK;
+    for (int ch = 0; ch < n_chunks; ch++) {
+        const int base = ch * CHUNK;
+        const int n_here = min(CHUNK, kv_len - base);
+        // ---- stage K/V chunk (4 pages at page=16) into LDS, all threads --
+        // Each (page,head) run in the pool is page*D contiguous bf16.
+        __syncthreads();
+        {
+            const int tid = threadIdx.x, nthr = blockDim.x;
+            const int total_vec = CHUNK * (DHEAD / 8);   // 16B units per tile
+            for (int u = tid; u < total_vec; u += nthr) {
+                const int tok = u / (DHEAD / 8);
+                const int dv = u % (DHEAD / 8);
+                if (base + tok < kv_len) {
+                    const int gp = page_table[(int64_t)slot * max_pages
+                                              + (base + tok) / page];
+                    const int64_t src = (((int64_t)gp * KVH + kvh) * page
+                                         + (base + tok) % page) * DHEAD + dv * 8;
+                    *reinterpret_cast<uint4*>(k_tile + tok * KROW + dv * 8) =
+                        *reinterpret_cast<const uint4*>(kpool + src);
+                    *reinterpret_cast<uint4*>(v_tile + tok * KROW + dv * 8) =
+                        *reinterpret_cast<const uint4*>(vpool + src);
+                }
+            }
+        }
+        __syncthreads();
+
+        // ---- scores: lane j owns key (base+j) ----
+        const int j = lane;
+        const bool live = j < n_here;
+        float s[QT];
+        #pragma unroll
+        for (int r = 0; r < QT; r++) s[r] = 0.f;
+        if (live) {
+            const bf16* krow = k_tile + j * KROW;
+            for (int d = 0; d < DHEAD; d += 8) {
+                float kv8[8];
+                load8f(krow + d, kv8);
+                for (int r = 0; r < rows; r++) {
+                    float q8[8];
+                    load8f(q_lds + ((int64_t)wid * QT + r) * DHEAD + d, q8);
+                    float acc = s[r];
+                    #pragma unroll
+                    for (int e = 0; e < 8; e++) acc += q8[e] * kv8[e];
+                    s[r] = acc;
+                }
+            }
+        }
+        // causal mask + scale; dead lanes -inf
+        for (int r = 0; r < rows; r++) {
+            const int qpos = pos0 + r;
+            const bool ok = live && (base + j <= qpos);
+            s[r] = ok ? s[r] * scale : -3.0e38f;
+        }
+
+        // ---- online softmax update per row ----
+        for (int r = 0; r < rows; r++) {
+            const float smax = wave_reduce_max(s[r]);
+            if (smax == -3.0e38f) {        // fully masked chunk for this row
+                p_lds[((int64_t)wid * QT + r) * PROW + j] = 0.f;
+                continue;
+            }
+            const float mn = fmaxf(m[r], smax);
+            const float corr = __expf(m[r] - mn);
+            const float p = (s[r] == -3.0e38f) ? 0.f : __expf(s[r] - mn);
+            const float psum = wave_reduce_sum(p);
+            l[r] = l[r] * corr + psum;
+            o0[r] *= corr; o1[r] *= corr;
+            m[r] = mn;
+            p_lds[((int64_t)wid * QT + r) * PROW + j] = p;
+        }
+        // wave-local p_lds: no cross-wave use, lgkm ordering within wave is
+        // guaranteed by the address dependence; no barrier needed.
+
+        // ---- PV: lane accumulates dims (2*lane, 2*lane+1) ----
+        for (int jj = 0; jj < n_here; jj++) {
+            const bf162 v2 = *reinterpret_cast<const bf162*>(
+                v_tile + jj * KROW + 2 * lane);
+            const float v0 = bf2f(v2.x), v1 = bf2f(v2.y);
+            for (int r = 0; r < rows; r++) {
+                const float p = p_lds[((int64_t)wid * QT + r) * PROW + jj];
+                o0[r] = fmaf(p, v0, o0[r]);
+                o1[r] = fmaf(p, v1, o1[r]);
+            }
+        }
+    }
+
+    // ---- epilogue ----
+    for (int r = 0; r < rows; r++) {
+        const float linv = (l[r] > 0.f) ? 1.f / l[r] : 0.f;
+        bf16* orow = out + ((int64_t)(q0 + r) * Hq + qh) * DHEAD;
+        reinterpret_cast<bf162*>(orow)[lane] =
+            __float22bfloat162_rn(make_float2(o0[r] * linv, o1[r] * linv));
+    }
+}
+
+static int attn_lds_bytes(int G, int QT) {
+    const int KROW = DHEAD + KPAD, PROW = CHUNK + 1;
+    return 2 * CHUNK * KROW * 2          // k_tile + v_tile bf16
+         + G * QT * DHEAD * 2            // q_lds bf16
+         + G * QT * PROW * 4;            // p_lds f32
+}
+
+extern "C" void paged_attn_bf16(
+    void* out, const void* q, const void* kpool, const void* vpool,
+    const void* page_table, const void* tile_slot, const void* tile_q0,
+    const void* tile_pos0, const void* tile_rows, int n_tiles, int qt,
+    int Hq, int KVH, int page, int max_pages, float scale,
+    hipStream_t stream)
+{
+    const int G = Hq / KVH;
+    dim3 grid(n_tiles, KVH);
+    dim3 block(G * 64);
+    if (qt == 1) {
+        k_paged_attn<1><<<grid, block, attn_lds_bytes(G, 1), stream>>>(
+            (bf16*)out, (const bf16*)q, (const bf16*)kpool,
+            (const bf16*)vpool, (const int*)page_table,
+            (const int*)tile_slot, (const int*)tile_q0,
+            (const int*)tile_pos0, (const int*)tile_rows,
+            Hq, KVH, page, max_pages, scale);
+    } else {
+        k_paged_attn<16><<<grid, block, attn_lds_bytes(G, 16), stream>>>(
+            (bf16*)out, (const bf16*)q, (const bf16*)kpool,
+            (const bf16*)vpool, (const int*)page_table,
+            (const int*)tile_slot, (const int*)tile_q0,
+            (const int*)tile_pos0, (const int*)tile_rows,
+            Hq, KVH, page, max_pages, scale);
+    }
+}
+
+// ---------------------------------------------------------------------------
+// SwiGLU: [T, 2F] -> [T, F], silu(gate)*up, bf16 x8 vectorized, fp32 math.
+// ---------------------------------------------------------------------------
+__global__ __launch_bounds__(256) void k_swiglu(
+    bf16* __restrict__ out, const bf16* __restrict__ gu, int T, int F)
+{
+    const int64_t total = (int64_t)T * F / 8;
+    for (int64_t i = blockIdx.x * blockDim.x + threadIdx.x; i < total;
+         i += (int64_t)gridDim.x * blockDim.x) {
+        const int64_t row = i / (F / 8), col8 = i % (F / 8);
+        const bf16* g = gu + row * 2 * F + col8 * 8;
+        const bf16* u = g + F;
+        float vg[8], vu[8];
+        load8f(g, vg);
+        load8f(u, vu);
+        #pragma unroll
+        for (int e = 0; e < 8; e++) {
+            const float sig = 1.f / (1.f + __expf(-vg[e]));
+            vg[e] = vg[e] * sig * vu[e];
+        }
+        store8bf(out + row * F + col8 * 8, vg);
+    }
+}
+
+extern "C" void swiglu_bf16(void* out, const void* gu, int T, int F,
+                            hipStream_t stream)
+{
+    const int64_t total = (int64_t)T * F / 8;
+    const int64_t want = (total + 255) / 256;
+    int blocks = (int)(want < 2048 ? (want > 0 ? want : 1) : 2048);
+    k_swiglu<<<blocks, 256, 0, stream>>>((bf16*)out, (const bf16*)gu, T, F);
+}
+
+// ---------------------------------------------------------------------------
+// Greedy sampler: per-row argmax over the vocab, bf16 logits.
+// One 256-thread block per row; ties resolve to the LOWEST index (matches
+// torch.argmax on the reference path).
+// ---------------------------------------------------------------------------
+__global__ __launch_bounds__(256) void k_argmax(
+    int* __restrict__ out, const bf16* __restrict__ logits, int V)
+{
+    const int row = blockIdx.x;
+    const bf16* lr = logits + (int64_t)row * V;
+    float best = -3.0e38f;
+    int bidx = 0;
+    for (int i = threadIdx.x * 8; i < V; i += blockDim.x * 8) {
+        if (i + 8 <= V) {
+            float v[8];
+            load8f(lr + i, v);
+            #pragma unroll
+            for (int e = 0; e < 8; e++)
+                if (v[e] > best || (v[e] == best && i + e < bidx)) {
+                    best = v[e]; bidx = i + e;
+                }
+        } else {
+            for (int e = i; e < V; e++) {
+                const float v = bf2f(lr[e]);
+                if (v > best || (v == best && e < bidx)) { best = v; bidx = e; }
+            }
+        }
+    }
+    // wave reduce (value, index), lowest index wins ties
+    #pragma unroll
+    for (int off = 32; off > 0; off >>= 1) {
+        const float ob = __shfl_down(best, off, 64);
+        const int oi = __shfl_down(bidx, off, 64);
+        if (ob > best || (ob == best && oi < bidx)) { best = ob; bidx = oi; }
+    }
+    __shared__ float wb[4];
+    __shared__ int wi[4];
+    const int wid = threadIdx.x >> 6, lane = threadIdx.x & 63;
+    if (lane == 0) { wb[wid] = best; wi[wid] = bidx; }
+    __syncthreads();
+    if (threadIdx.x == 0) {
+        for (int wv = 1; wv < 4; wv++)
+            if (wb[wv] > best || (wb[wv] == best && wi[wv] < bidx)) {
+                best = wb[wv]; bidx = wi[wv];
+            }
+        out[row] = bidx;
+    }
+}
+
+extern "C" void argmax_bf16(void* out, const void* logits, int B, int V,
+                            hipStream_t stream)
+{
+    k_argmax<<<B, 256, 0, stream>>>((int*)out, (const bf16*)logits, V);
+}

@@ -1,0 +1,192 @@
+"""ctypes bindings for the gfx950 HIP kernels (csrc/kernels/kernels.hip).
+
+The extension is built IN-TREE (csrc/kernels/_kernels_gfx950.so) so the
+.so travels with the repo snapshot to GPU boxes.  On a GPU box a missing
+extension is a hard error — ops never fall back silently to eager PyTorch
+(the GPU tests must exercise the native path).
+"""
+from __future__ import annotations
+
+import ctypes
+import os
+from typing import Optional
+
+import torch
+
+from . import reference as ref
+
+_LIB_PATH = os.path.join(
+    os.path.dirname(os.path.dirname(os.path.abspath(__file__))),
+    "csrc", "kernels", "_kernels_gfx950.so",
+)
+
+_lib = None
+_load_err: Optional[str] = None
+
+
+def _try_load():
+    global _lib, _load_err
+    if _lib is not None or _load_err is not None:
+        return
+    try:
+        lib = ctypes.CDLL(_LIB_PATH, mode=ctypes.RTLD_GLOBAL)
+    except OSError as e:
+        _load_err = f"cannot load {_LIB_PATH}: {e}"
+        return
+    vp, i, f = ctypes.c_void_p, ctypes.c_int, ctypes.c_float
+    lib.rmsnorm_residual_bf16.argtypes = [vp, vp, vp, vp, vp, i, i, f, vp]
+    lib.rope_bf16.argtypes = [vp, vp, vp, vp, vp, i, i, i, i, vp]
+    lib.kv_append_bf16.argtypes = [vp, vp, vp, vp, vp, vp, vp,
+                                   i, i, i, i, i, vp]
+    lib.paged_attn_bf16.argtypes = [vp, vp, vp, vp, vp, vp, vp, vp, vp,
+                                    i, i, i, i, i, i, f, vp]
+    lib.swiglu_bf16.argtypes = [vp, vp, i, i, vp]
+    lib.argmax_bf16.argtypes = [vp, vp, i, i, vp]
+    _lib = lib
+
+
+def available() -> bool:
+    _try_load()
+    return _lib is not None
+
+
+def require():
+    _try_load()
+    if _lib is None:
+        raise RuntimeError(
+            "ollamamq_amd HIP extension missing on a GPU box: "
+            f"{_load_err}. Build it with `python -m ollamamq_amd.build` "
+            "(hipcc --offload-arch=gfx950)."
+        )
+
+
+def _stream():
+    return ctypes.c_void_p(torch.cuda.current_stream().cuda_stream)
+
+
+def _p(t: Optional[torch.Tensor]):
+    return ctypes.c_void_p(0 if t is None else t.data_ptr())
+
+
+# ---------------------------------------------------------------------------
+
+def rmsnorm_residual(x, residual, weight, eps):
+    T, H = x.shape
+    y = torch.empty_like(x)
+    res_out = torch.empty_like(x)
+    _lib.rmsnorm_residual_bf16(_p(y), _p(res_out), _p(x), _p(residual),
+                               _p(weight), T, H, float(eps), _stream())
+    return y, res_out
+
+
+def rope(q, k, positions, cos, sin):
+    T, Hq, D = q.shape
+    Hk = k.shape[1]
+    assert D == 128, "rope kernel specialized for head_dim=128"
+    pos32 = positions if positions.dtype == torch.int32 else positions.int()
+    _lib.rope_bf16(_p(q), _p(k), _p(pos32), _p(cos), _p(sin),
+                   T, Hq, Hk, D, _stream())
+
+
+def kv_append(cache, layer, k, v, slot_ids, positions):
+    T, KVH, D = k.shape
+    kp, vp = _layer_ptrs(cache, layer)
+    pos32 = positions if positions.dtype == torch.int32 else positions.int()
+    slot32 = slot_ids if slot_ids.dtype == torch.int32 else slot_ids.int()
+    _lib.kv_append_bf16(kp, vp, _p(k), _p(v), _p(slot32), _p(pos32),
+                        _p(cache.page_table), T, KVH, D, cache.page_size,
+                        cache.page_table.shape[1], _stream())
+
+
+def _layer_ptrs(cache, layer):
+    stride = cache.k_pool.stride(0) * cache.k_pool.element_size()
+    return (ctypes.c_void_p(cache.k_pool.data_ptr() + layer * stride),
+            ctypes.c_void_p(cache.v_pool.data_ptr() + layer * stride))
+
+
+PREFILL_QT = 16
+
+
+def _attn_plan(cache, meta, qt):
+    """Device tile arrays for the attention launch, cached on the meta
+    object (one build per forward, shared by every layer)."""
+    key = f"_plan_{qt}"
+    plan = getattr(meta, key, None)
+    if plan is not None:
+        return plan
+    dev = meta.slot_ids.device
+    if qt == 1:
+        n = meta.slot_ids.shape[0]
+        tile_slot = meta.slot_ids.int()
+        tile_q0 = torch.arange(n, dtype=torch.int32, device=dev)
+        tile_pos0 = (meta.seq_lens - 1).int()
+        tile_rows = torch.ones(n, dtype=torch.int32, device=dev)
+    else:
+        slots = meta.slot_ids.tolist()
+        lens = meta.seq_lens.tolist()
+        cu = meta.cu_q.tolist()
+        ts, tq, tp, tr = [], [], [], []
+        for i, slot in enumerate(slots):
+            qlen = cu[i + 1] - cu[i]
+            start_pos = lens[i] - qlen
+            for t0 in range(0, qlen, qt):
+                rows = min(qt, qlen - t0)
+                ts.append(slot)
+                tq.append(cu[i] + t0)
+                tp.append(start_pos + t0)
+                tr.append(rows)
+        tile_slot = torch.tensor(ts, dtype=torch.int32, device=dev)
+        tile_q0 = torch.tensor(tq, dtype=torch.int32, device=dev)
+        tile_pos0 = torch.tensor(tp, dtype=torch.int32, device=dev)
+        tile_rows = torch.tensor(tr, dtype=torch.int32, device=dev)
+    plan = (tile_slot, tile_q0, tile_pos0, tile_rows)
+    setattr(meta, key, plan)
+    return plan
+
+
+def _attention(q, cache, layer, meta, qt):
+    T, Hq, D = q.shape
+    assert D == 128
+    out = torch.empty_like(q)
+    kp, vp = _layer_ptrs(cache, layer)
+    tile_slot, tile_q0, tile_pos0, tile_rows = _attn_plan(cache, meta, qt)
+    n_tiles = tile_slot.shape[0]
+    if n_tiles == 0:
+        return out
+    _lib.paged_attn_bf16(
+        _p(out), _p(q), kp, vp, _p(cache.page_table),
+        _p(tile_slot), _p(tile_q0), _p(tile_pos0), _p(tile_rows),
+        n_tiles, qt, Hq, cache.n_kv_heads, cache.page_size,
+        cache.page_table.shape[1], 1.0 / (D ** 0.5), _stream())
+    return out
+
+
+def attention_decode(q, cache, layer, meta):
+    return _attention(q, cache, layer, meta, 1)
+
+
+def attention_prefill(q, cache, layer, meta):
+    return _attention(q, cache, layer, meta, PREFILL_QT)
+
+
+def swiglu(gate_up):
+    T, F2 = gate_up.shape
+    F = F2 // 2
+    out = torch.empty((T, F), dtype=gate_up.dtype, device=gate_up.device)
+    _lib.swiglu_bf16(_p(out), _p(gate_up), T, F, _stream())
+    return out
+
+
+def sample(logits, temperature, top_k, top_p, generator=None):
+    if torch.is_tensor(temperature):
+        all_greedy = bool((temperature <= 0).all())
+    else:
+        all_greedy = temperature <= 0
+    if all_greedy:
+        B, V = logits.shape
+        out = torch.empty(B, dtype=torch.int32, device=logits.device)
+        l = logits if logits.dtype == torch.bfloat16 else logits.bfloat16()
+        _lib.argmax_bf16(_p(out), _p(l), B, V, _stream())
+        return out.long()
+    # stochastic paths compose on-GPU torch ops (sort/softmax/multinomial)
+    return ref.sample(logits, temperature, top_k, top_p, generator)
