@@ -74,7 +74,8 @@ def main():
     n_pages = (users + 2) * ((prompt_len + gen_budget + 15) // 16 + 2)
 
     t_load0 = time.monotonic()
-    model = LlamaModel(cfg, device=device, dtype=dtype, seed=1234)
+    model = LlamaModel(cfg, device=device, dtype=dtype, seed=1234,
+                       fast_init=has_gpu)
     kv = PagedKVCache.for_model(
         cfg, n_pages=n_pages, max_slots=users + 2, max_ctx=ctx,
         device=device, dtype=dtype)

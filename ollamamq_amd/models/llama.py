@@ -71,9 +71,12 @@ PRESETS = {
 
 
 def _randn(shape, dev, dtype, gen, scale):
-    w = torch.empty(shape, device=dev, dtype=torch.float32)
+    # generate on the generator's own device (CPU gen => bit-identical
+    # weights across CPU/GPU test runs; CUDA gen => fast 16 GB init for the
+    # big models straight in HBM)
+    w = torch.empty(shape, device=gen.device, dtype=torch.float32)
     w.normal_(0.0, scale, generator=gen)
-    return w.to(dtype)
+    return w.to(device=dev, dtype=dtype)
 
 
 class LlamaLayer:
@@ -116,6 +119,7 @@ class LlamaModel:
         tp_rank: int = 0,
         tp_size: int = 1,
         process_group=None,
+        fast_init: bool = False,
     ):
         assert cfg.n_heads % tp_size == 0, "n_heads must divide TP"
         self.cfg = cfg
@@ -128,7 +132,8 @@ class LlamaModel:
         self.n_local_heads = cfg.n_heads // tp_size
         self.n_local_kv_heads = max(1, cfg.n_kv_heads // tp_size)
 
-        gen = torch.Generator(device="cpu")
+        gen_dev = "cuda" if (fast_init and self.device.type == "cuda") else "cpu"
+        gen = torch.Generator(device=gen_dev)
         gen.manual_seed(seed + tp_rank)
         dev = self.device
         scale = 1.0 / math.sqrt(cfg.hidden)
