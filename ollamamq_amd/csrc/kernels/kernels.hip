@@ -111,13 +111,14 @@ __global__ __launch_bounds__(256) void k_rmsnorm_residual(
     }
 }
 
-extern "C" void rmsnorm_residual_bf16(
+extern "C" int rmsnorm_residual_bf16(
     void* y, void* res_out, const void* x, const void* res_in,
     const void* w, int T, int H, float eps, hipStream_t stream)
 {
     k_rmsnorm_residual<<<T, 256, 0, stream>>>(
         (bf16*)y, (bf16*)res_out, (const bf16*)x, (const bf16*)res_in,
         (const bf16*)w, H, eps);
+    return (int)hipGetLastError();
 }
 
 // ---------------------------------------------------------------------------
@@ -146,7 +147,7 @@ __global__ __launch_bounds__(256) void k_rope(
     base[lane + d2] = f2bf(x2 * c + x1 * s);
 }
 
-extern "C" void rope_bf16(
+extern "C" int rope_bf16(
     void* q, void* k, const void* pos, const void* cost, const void* sint,
     int T, int Hq, int Hk, int D, hipStream_t stream)
 {
@@ -154,6 +155,7 @@ extern "C" void rope_bf16(
     k_rope<<<(waves + 3) / 4, 256, 0, stream>>>(
         (bf16*)q, (bf16*)k, (const int*)pos, (const float*)cost,
         (const float*)sint, T, Hq, Hk, D);
+    return (int)hipGetLastError();
 }
 
 // ---------------------------------------------------------------------------
@@ -183,7 +185,7 @@ __global__ __launch_bounds__(256) void k_kv_append(
     reinterpret_cast<bf162*>(vp + dst)[lane] = vs[lane];
 }
 
-extern "C" void kv_append_bf16(
+extern "C" int kv_append_bf16(
     void* kp, void* vp, const void* k, const void* v, const void* slot,
     const void* pos, const void* page_table, int T, int KVH, int D,
     int page, int max_pages, hipStream_t stream)
@@ -193,6 +195,7 @@ extern "C" void kv_append_bf16(
         (bf16*)kp, (bf16*)vp, (const bf16*)k, (const bf16*)v,
         (const int*)slot, (const int*)pos, (const int*)page_table,
         T, KVH, D, page, max_pages);
+    return (int)hipGetLastError();
 }
 
 // ---------------------------------------------------------------------------
@@ -373,7 +376,19 @@ static int attn_lds_bytes(int G, int QT) {
          + G * QT * PROW * 4;            // p_lds f32
 }
 
-extern "C" void paged_attn_bf16(
+// Dynamic-LDS requests above the 64 KiB default (prefill tiles at G=8)
+// need an explicit opt-in; CDNA4 hardware allows up to 160 KiB/workgroup.
+static void allow_big_lds(const void* fn, int bytes) {
+    static int done_16 = 0, done_1 = 0;
+    int* flag = (fn == (const void*)k_paged_attn<16>) ? &done_16 : &done_1;
+    if (!*flag && bytes > 64 * 1024) {
+        (void)hipFuncSetAttribute(
+            fn, hipFuncAttributeMaxDynamicSharedMemorySize, 160 * 1024);
+        *flag = 1;
+    }
+}
+
+extern "C" int paged_attn_bf16(
     void* out, const void* q, const void* kpool, const void* vpool,
     const void* page_table, const void* tile_slot, const void* tile_q0,
     const void* tile_pos0, const void* tile_rows, int n_tiles, int qt,
@@ -384,6 +399,7 @@ extern "C" void paged_attn_bf16(
     dim3 grid(n_tiles, KVH);
     dim3 block(G * 64);
     if (qt == 1) {
+        allow_big_lds((const void*)k_paged_attn<1>, attn_lds_bytes(G, 1));
         k_paged_attn<1><<<grid, block, attn_lds_bytes(G, 1), stream>>>(
             (bf16*)out, (const bf16*)q, (const bf16*)kpool,
             (const bf16*)vpool, (const int*)page_table,
@@ -391,6 +407,7 @@ extern "C" void paged_attn_bf16(
             (const int*)tile_pos0, (const int*)tile_rows,
             Hq, KVH, page, max_pages, scale);
     } else {
+        allow_big_lds((const void*)k_paged_attn<16>, attn_lds_bytes(G, 16));
         k_paged_attn<16><<<grid, block, attn_lds_bytes(G, 16), stream>>>(
             (bf16*)out, (const bf16*)q, (const bf16*)kpool,
             (const bf16*)vpool, (const int*)page_table,
@@ -398,6 +415,7 @@ extern "C" void paged_attn_bf16(
             (const int*)tile_pos0, (const int*)tile_rows,
             Hq, KVH, page, max_pages, scale);
     }
+    return (int)hipGetLastError();
 }
 
 // ---------------------------------------------------------------------------
@@ -424,13 +442,14 @@ __global__ __launch_bounds__(256) void k_swiglu(
     }
 }
 
-extern "C" void swiglu_bf16(void* out, const void* gu, int T, int F,
+extern "C" int swiglu_bf16(void* out, const void* gu, int T, int F,
                             hipStream_t stream)
 {
     const int64_t total = (int64_t)T * F / 8;
     const int64_t want = (total + 255) / 256;
     int blocks = (int)(want < 2048 ? (want > 0 ? want : 1) : 2048);
     k_swiglu<<<blocks, 256, 0, stream>>>((bf16*)out, (const bf16*)gu, T, F);
+    return (int)hipGetLastError();
 }
 
 // ---------------------------------------------------------------------------
@@ -482,8 +501,9 @@ __global__ __launch_bounds__(256) void k_argmax(
     }
 }
 
-extern "C" void argmax_bf16(void* out, const void* logits, int B, int V,
+extern "C" int argmax_bf16(void* out, const void* logits, int B, int V,
                             hipStream_t stream)
 {
     k_argmax<<<B, 256, 0, stream>>>((int*)out, (const bf16*)logits, V);
+    return (int)hipGetLastError();
 }

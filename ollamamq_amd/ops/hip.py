@@ -42,7 +42,15 @@ def _try_load():
                                     i, i, i, i, i, i, f, vp]
     lib.swiglu_bf16.argtypes = [vp, vp, i, i, vp]
     lib.argmax_bf16.argtypes = [vp, vp, i, i, vp]
+    for fn in ("rmsnorm_residual_bf16", "rope_bf16", "kv_append_bf16",
+               "paged_attn_bf16", "swiglu_bf16", "argmax_bf16"):
+        getattr(lib, fn).restype = ctypes.c_int
     _lib = lib
+
+
+def _check(err: int, op: str):
+    if err != 0:
+        raise RuntimeError(f"HIP kernel launch failed in {op}: hipError {err}")
 
 
 def available() -> bool:
@@ -74,8 +82,9 @@ def rmsnorm_residual(x, residual, weight, eps):
     T, H = x.shape
     y = torch.empty_like(x)
     res_out = torch.empty_like(x)
-    _lib.rmsnorm_residual_bf16(_p(y), _p(res_out), _p(x), _p(residual),
-                               _p(weight), T, H, float(eps), _stream())
+    _check(_lib.rmsnorm_residual_bf16(_p(y), _p(res_out), _p(x),
+                                      _p(residual), _p(weight), T, H,
+                                      float(eps), _stream()), "rmsnorm")
     return y, res_out
 
 
@@ -84,8 +93,8 @@ def rope(q, k, positions, cos, sin):
     Hk = k.shape[1]
     assert D == 128, "rope kernel specialized for head_dim=128"
     pos32 = positions if positions.dtype == torch.int32 else positions.int()
-    _lib.rope_bf16(_p(q), _p(k), _p(pos32), _p(cos), _p(sin),
-                   T, Hq, Hk, D, _stream())
+    _check(_lib.rope_bf16(_p(q), _p(k), _p(pos32), _p(cos), _p(sin),
+                          T, Hq, Hk, D, _stream()), "rope")
 
 
 def kv_append(cache, layer, k, v, slot_ids, positions):
@@ -93,9 +102,10 @@ def kv_append(cache, layer, k, v, slot_ids, positions):
     kp, vp = _layer_ptrs(cache, layer)
     pos32 = positions if positions.dtype == torch.int32 else positions.int()
     slot32 = slot_ids if slot_ids.dtype == torch.int32 else slot_ids.int()
-    _lib.kv_append_bf16(kp, vp, _p(k), _p(v), _p(slot32), _p(pos32),
-                        _p(cache.page_table), T, KVH, D, cache.page_size,
-                        cache.page_table.shape[1], _stream())
+    _check(_lib.kv_append_bf16(kp, vp, _p(k), _p(v), _p(slot32), _p(pos32),
+                               _p(cache.page_table), T, KVH, D,
+                               cache.page_size, cache.page_table.shape[1],
+                               _stream()), "kv_append")
 
 
 def _layer_ptrs(cache, layer):
@@ -153,11 +163,12 @@ def _attention(q, cache, layer, meta, qt):
     n_tiles = tile_slot.shape[0]
     if n_tiles == 0:
         return out
-    _lib.paged_attn_bf16(
+    _check(_lib.paged_attn_bf16(
         _p(out), _p(q), kp, vp, _p(cache.page_table),
         _p(tile_slot), _p(tile_q0), _p(tile_pos0), _p(tile_rows),
         n_tiles, qt, Hq, cache.n_kv_heads, cache.page_size,
-        cache.page_table.shape[1], 1.0 / (D ** 0.5), _stream())
+        cache.page_table.shape[1], 1.0 / (D ** 0.5), _stream()),
+        "paged_attn")
     return out
 
 
@@ -173,7 +184,8 @@ def swiglu(gate_up):
     T, F2 = gate_up.shape
     F = F2 // 2
     out = torch.empty((T, F), dtype=gate_up.dtype, device=gate_up.device)
-    _lib.swiglu_bf16(_p(out), _p(gate_up), T, F, _stream())
+    _check(_lib.swiglu_bf16(_p(out), _p(gate_up), T, F, _stream()),
+           "swiglu")
     return out
 
 
@@ -186,7 +198,7 @@ def sample(logits, temperature, top_k, top_p, generator=None):
         B, V = logits.shape
         out = torch.empty(B, dtype=torch.int32, device=logits.device)
         l = logits if logits.dtype == torch.bfloat16 else logits.bfloat16()
-        _lib.argmax_bf16(_p(out), _p(l), B, V, _stream())
+        _check(_lib.argmax_bf16(_p(out), _p(l), B, V, _stream()), "argmax")
         return out.long()
     # stochastic paths compose on-GPU torch ops (sort/softmax/multinomial)
     return ref.sample(logits, temperature, top_k, top_p, generator)
