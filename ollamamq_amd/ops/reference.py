@@ -34,10 +34,14 @@ def rope(q, k, positions, cos, sin):
     s = sin[positions.long()].unsqueeze(1)
     for t in (q, k):
         d2 = t.shape[-1] // 2
+        # NB: .float() on an fp32 tensor is a VIEW — compute both halves
+        # before mutating t, or the lower-half store corrupts x1.
         x1 = t[..., :d2].float()
         x2 = t[..., d2:].float()
-        t[..., :d2] = (x1 * c - x2 * s).to(t.dtype)
-        t[..., d2:] = (x2 * c + x1 * s).to(t.dtype)
+        lo = x1 * c - x2 * s
+        hi = x2 * c + x1 * s
+        t[..., :d2] = lo.to(t.dtype)
+        t[..., d2:] = hi.to(t.dtype)
 
 
 def kv_append(cache, layer, k, v, slot_ids, positions):
