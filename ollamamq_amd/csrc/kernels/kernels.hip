@@ -129,15 +129,16 @@ extern "C" int rmsnorm_residual_bf16(
 __global__ __launch_bounds__(256) void k_rope(
     bf16* __restrict__ q, bf16* __restrict__ k,
     const int* __restrict__ pos, const float* __restrict__ cost,
-    const float* __restrict__ sint, int T, int Hq, int Hk, int D)
+    const float* __restrict__ sint, int T, int Hq, int Hk, int D,
+    int64_t qs, int64_t ks)   // row strides in elements (fused-QKV views)
 {
     const int gid = blockIdx.x * 4 + (threadIdx.x >> 6);  // (token,head) flat
     const int lane = threadIdx.x & 63;
     const int Htot = Hq + Hk;
     if (gid >= T * Htot) return;
     const int t = gid / Htot, h = gid % Htot;
-    bf16* base = (h < Hq) ? q + ((int64_t)t * Hq + h) * D
-                          : k + ((int64_t)t * Hk + (h - Hq)) * D;
+    bf16* base = (h < Hq) ? q + (int64_t)t * qs + (int64_t)h * D
+                          : k + (int64_t)t * ks + (int64_t)(h - Hq) * D;
     const int d2 = D / 2;                 // 64 = one lane per rotation pair
     const float c = cost[(int64_t)pos[t] * d2 + lane];
     const float s = sint[(int64_t)pos[t] * d2 + lane];
@@ -149,12 +150,13 @@ __global__ __launch_bounds__(256) void k_rope(
 
 extern "C" int rope_bf16(
     void* q, void* k, const void* pos, const void* cost, const void* sint,
-    int T, int Hq, int Hk, int D, hipStream_t stream)
+    int T, int Hq, int Hk, int D, int64_t qs, int64_t ks,
+    hipStream_t stream)
 {
     const int waves = T * (Hq + Hk);
     k_rope<<<(waves + 3) / 4, 256, 0, stream>>>(
         (bf16*)q, (bf16*)k, (const int*)pos, (const float*)cost,
-        (const float*)sint, T, Hq, Hk, D);
+        (const float*)sint, T, Hq, Hk, D, qs, ks);
     return (int)hipGetLastError();
 }
 
@@ -168,7 +170,7 @@ __global__ __launch_bounds__(256) void k_kv_append(
     const bf16* __restrict__ k, const bf16* __restrict__ v,
     const int* __restrict__ slot, const int* __restrict__ pos,
     const int* __restrict__ page_table,
-    int T, int KVH, int D, int page, int max_pages)
+    int T, int KVH, int D, int page, int max_pages, int64_t src_stride)
 {
     const int gid = blockIdx.x * 4 + (threadIdx.x >> 6);
     const int lane = threadIdx.x & 63;
@@ -178,7 +180,7 @@ __global__ __launch_bounds__(256) void k_kv_append(
     const int pg = page_table[(int64_t)slot[t] * max_pages + p / page];
     const int off = p % page;
     const int64_t dst = (((int64_t)pg * KVH + h) * page + off) * D;
-    const int64_t src = ((int64_t)t * KVH + h) * D;
+    const int64_t src = (int64_t)t * src_stride + (int64_t)h * D;
     const bf162* ks = reinterpret_cast<const bf162*>(k + src);
     const bf162* vs = reinterpret_cast<const bf162*>(v + src);
     reinterpret_cast<bf162*>(kp + dst)[lane] = ks[lane];
@@ -188,13 +190,13 @@ __global__ __launch_bounds__(256) void k_kv_append(
 extern "C" int kv_append_bf16(
     void* kp, void* vp, const void* k, const void* v, const void* slot,
     const void* pos, const void* page_table, int T, int KVH, int D,
-    int page, int max_pages, hipStream_t stream)
+    int page, int max_pages, int64_t src_stride, hipStream_t stream)
 {
     const int waves = T * KVH;
     k_kv_append<<<(waves + 3) / 4, 256, 0, stream>>>(
         (bf16*)kp, (bf16*)vp, (const bf16*)k, (const bf16*)v,
         (const int*)slot, (const int*)pos, (const int*)page_table,
-        T, KVH, D, page, max_pages);
+        T, KVH, D, page, max_pages, src_stride);
     return (int)hipGetLastError();
 }
 
@@ -228,7 +230,7 @@ __global__ __launch_bounds__(512) void k_paged_attn(
     const int* __restrict__ tile_q0,      // flat q row of tile start
     const int* __restrict__ tile_pos0,    // absolute position of tile start
     const int* __restrict__ tile_rows,    // rows in tile (≤ QT)
-    int Hq, int KVH, int page, int max_pages, float scale)
+    int Hq, int KVH, int page, int max_pages, float scale, int64_t q_stride)
 {
     const int tile = blockIdx.x;
     const int kvh = blockIdx.y;
@@ -258,7 +260,8 @@ __global__ __launch_bounds__(512) void k_paged_attn(
 
     // ---- load this wave's Q rows into LDS (bf16, scaled later) ----
     for (int r = 0; r < rows; r++) {
-        const bf16* qsrc = q + ((int64_t)(q0 + r) * Hq + qh) * DHEAD;
+        const bf16* qsrc = q + (int64_t)(q0 + r) * q_stride
+                           + (int64_t)qh * DHEAD;
         reinterpret_cast<bf162*>(q_lds + ((int64_t)wid * QT + r) * DHEAD)[lane]
             = reinterpret_cast<const bf162*>(qsrc)[lane];
     }
@@ -392,7 +395,7 @@ extern "C" int paged_attn_bf16(
     void* out, const void* q, const void* kpool, const void* vpool,
     const void* page_table, const void* tile_slot, const void* tile_q0,
     const void* tile_pos0, const void* tile_rows, int n_tiles, int qt,
-    int Hq, int KVH, int page, int max_pages, float scale,
+    int Hq, int KVH, int page, int max_pages, float scale, int64_t q_stride,
     hipStream_t stream)
 {
     const int G = Hq / KVH;
@@ -405,7 +408,7 @@ extern "C" int paged_attn_bf16(
             (const bf16*)vpool, (const int*)page_table,
             (const int*)tile_slot, (const int*)tile_q0,
             (const int*)tile_pos0, (const int*)tile_rows,
-            Hq, KVH, page, max_pages, scale);
+            Hq, KVH, page, max_pages, scale, q_stride);
     } else {
         allow_big_lds((const void*)k_paged_attn<16>, attn_lds_bytes(G, 16));
         k_paged_attn<16><<<grid, block, attn_lds_bytes(G, 16), stream>>>(
@@ -413,7 +416,7 @@ extern "C" int paged_attn_bf16(
             (const bf16*)vpool, (const int*)page_table,
             (const int*)tile_slot, (const int*)tile_q0,
             (const int*)tile_pos0, (const int*)tile_rows,
-            Hq, KVH, page, max_pages, scale);
+            Hq, KVH, page, max_pages, scale, q_stride);
     }
     return (int)hipGetLastError();
 }

@@ -35,11 +35,12 @@ def _try_load():
         return
     vp, i, f = ctypes.c_void_p, ctypes.c_int, ctypes.c_float
     lib.rmsnorm_residual_bf16.argtypes = [vp, vp, vp, vp, vp, i, i, f, vp]
-    lib.rope_bf16.argtypes = [vp, vp, vp, vp, vp, i, i, i, i, vp]
+    i64 = ctypes.c_int64
+    lib.rope_bf16.argtypes = [vp, vp, vp, vp, vp, i, i, i, i, i64, i64, vp]
     lib.kv_append_bf16.argtypes = [vp, vp, vp, vp, vp, vp, vp,
-                                   i, i, i, i, i, vp]
+                                   i, i, i, i, i, i64, vp]
     lib.paged_attn_bf16.argtypes = [vp, vp, vp, vp, vp, vp, vp, vp, vp,
-                                    i, i, i, i, i, i, f, vp]
+                                    i, i, i, i, i, i, f, i64, vp]
     lib.swiglu_bf16.argtypes = [vp, vp, i, i, vp]
     lib.argmax_bf16.argtypes = [vp, vp, i, i, vp]
     for fn in ("rmsnorm_residual_bf16", "rope_bf16", "kv_append_bf16",
@@ -88,13 +89,20 @@ def rmsnorm_residual(x, residual, weight, eps):
     return y, res_out
 
 
+def _row_stride(t, D):
+    # tolerate fused-QKV views: [T, H, D] with stride (S, D, 1)
+    assert t.stride(2) == 1 and t.stride(1) == D, "head dim must be packed"
+    return t.stride(0)
+
+
 def rope(q, k, positions, cos, sin):
     T, Hq, D = q.shape
     Hk = k.shape[1]
     assert D == 128, "rope kernel specialized for head_dim=128"
     pos32 = positions if positions.dtype == torch.int32 else positions.int()
     _check(_lib.rope_bf16(_p(q), _p(k), _p(pos32), _p(cos), _p(sin),
-                          T, Hq, Hk, D, _stream()), "rope")
+                          T, Hq, Hk, D, _row_stride(q, D), _row_stride(k, D),
+                          _stream()), "rope")
 
 
 def kv_append(cache, layer, k, v, slot_ids, positions):
@@ -102,10 +110,11 @@ def kv_append(cache, layer, k, v, slot_ids, positions):
     kp, vp = _layer_ptrs(cache, layer)
     pos32 = positions if positions.dtype == torch.int32 else positions.int()
     slot32 = slot_ids if slot_ids.dtype == torch.int32 else slot_ids.int()
+    assert _row_stride(k, D) == _row_stride(v, D)
     _check(_lib.kv_append_bf16(kp, vp, _p(k), _p(v), _p(slot32), _p(pos32),
                                _p(cache.page_table), T, KVH, D,
                                cache.page_size, cache.page_table.shape[1],
-                               _stream()), "kv_append")
+                               _row_stride(k, D), _stream()), "kv_append")
 
 
 def _layer_ptrs(cache, layer):
@@ -167,8 +176,8 @@ def _attention(q, cache, layer, meta, qt):
         _p(out), _p(q), kp, vp, _p(cache.page_table),
         _p(tile_slot), _p(tile_q0), _p(tile_pos0), _p(tile_rows),
         n_tiles, qt, Hq, cache.n_kv_heads, cache.page_size,
-        cache.page_table.shape[1], 1.0 / (D ** 0.5), _stream()),
-        "paged_attn")
+        cache.page_table.shape[1], 1.0 / (D ** 0.5), _row_stride(q, D),
+        _stream()), "paged_attn")
     return out
 
 
