@@ -2,6 +2,8 @@
 // (reference src/dispatcher.rs:355-397, src/control.rs:450-510).
 #include "matching.h"
 
+#include "core.h"
+
 #include <algorithm>
 
 namespace omq {

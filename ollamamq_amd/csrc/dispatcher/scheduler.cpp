@@ -1,0 +1,224 @@
+#include "scheduler.h"
+
+#include <algorithm>
+#include <chrono>
+
+#include "json.h"
+#include "matching.h"
+
+namespace omq {
+
+int64_t now_ms() {
+    using namespace std::chrono;
+    return duration_cast<milliseconds>(
+               steady_clock::now().time_since_epoch())
+        .count();
+}
+
+void AppState::notify() {
+    {
+        std::lock_guard<std::mutex> g(wake_mu);
+        wake_flag = true;
+    }
+    wake_cv.notify_all();
+}
+
+void AppState::wait_work(int timeout_ms) {
+    std::unique_lock<std::mutex> g(wake_mu);
+    wake_cv.wait_for(g, std::chrono::milliseconds(timeout_ms),
+                     [&] { return wake_flag || shutting_down.load(); });
+    wake_flag = false;
+}
+
+// ---------------------------------------------------------------- ordering
+std::vector<std::string> candidate_order(
+    const std::vector<std::pair<std::string, int64_t>>& active_users,
+    const std::string& vip, const std::string& boost, uint64_t counter) {
+    // stable sort by total processed ascending (least-served first),
+    // reference src/dispatcher.rs:518-534
+    std::vector<std::pair<std::string, int64_t>> rest;
+    bool has_vip = false, has_boost = false;
+    for (const auto& [u, n] : active_users) {
+        if (u == vip) { has_vip = true; continue; }
+        if (u == boost) { has_boost = true; continue; }
+        rest.emplace_back(u, n);
+    }
+    std::stable_sort(rest.begin(), rest.end(),
+                     [](const auto& a, const auto& b) {
+                         return a.second < b.second;
+                     });
+    // rotation: advance the start point each pass (src/dispatcher.rs:551-564)
+    std::vector<std::string> order;
+    if (!rest.empty()) {
+        const size_t rot = counter % rest.size();
+        for (size_t i = 0; i < rest.size(); i++)
+            order.push_back(rest[(rot + i) % rest.size()].first);
+    }
+    // boost: prepended on every 2nd tick (src/dispatcher.rs:538-549)
+    if (has_boost && counter % 2 == 0)
+        order.insert(order.begin(), boost);
+    else if (has_boost)
+        order.push_back(boost);
+    // VIP: absolute head (src/dispatcher.rs:507,538)
+    if (has_vip) order.insert(order.begin(), vip);
+    return order;
+}
+
+// ------------------------------------------------------------- eligibility
+static bool family_supported(ApiType t, const std::string& path) {
+    if (t == ApiType::Both) return true;
+    const bool openai = path.rfind("/v1/", 0) == 0;
+    if (openai) return t == ApiType::OpenAi;
+    return t == ApiType::Ollama || t == ApiType::Unknown;
+}
+
+bool backend_eligible(const BackendStatus& b, bool has_control_op,
+                      const std::string& requested_model,
+                      const std::string& path) {
+    if (!b.is_online) return false;
+    if (b.active_requests >= b.max_concurrency) return false;
+    if (has_control_op) return false;
+    if (!family_supported(b.api_type, path)) return false;
+    if (!requested_model.empty() &&
+        !model_routable(requested_model, b.available_models))
+        return false;
+    return true;
+}
+
+size_t pick_backend(const std::vector<BackendStatus>& backends,
+                    const std::vector<size_t>& eligible, size_t last_idx) {
+    // min active_requests, then first index strictly after last_idx
+    // (reference src/dispatcher.rs:664-681)
+    int min_active = INT32_MAX;
+    for (size_t i : eligible)
+        min_active = std::min(min_active, backends[i].active_requests);
+    std::vector<size_t> least;
+    for (size_t i : eligible)
+        if (backends[i].active_requests == min_active) least.push_back(i);
+    for (size_t i : least)
+        if (i > last_idx) return i;
+    return least.front();
+}
+
+// --------------------------------------------------------------- the pass
+bool schedule_once(AppState& st, Dispatch* out) {
+    // lock order: control before backends (control.rs:963-965), then queues
+    std::scoped_lock lk(st.control_mu, st.backends_mu, st.queues_mu,
+                        st.prio_mu);
+
+    // enforce stuck-timeout 503 (declared-but-latent in the reference —
+    // config.rs:54-57; we implement the documented behavior)
+    const int64_t now = now_ms();
+    const int64_t stuck_ms = st.settings.stuck_timeout_s * 1000;
+    for (auto& [user, us] : st.users) {
+        for (auto it = us.queue.begin(); it != us.queue.end();) {
+            if (stuck_ms > 0 && now - it->queued_at_ms > stuck_ms) {
+                if (it->resp) {
+                    it->resp->send_status(
+                        503, {{"Content-Type", "application/json"}});
+                    it->resp->send_chunk(
+                        "{\"error\":\"no backend available for this request "
+                        "(stuck timeout)\"}");
+                    it->resp->finish();
+                }
+                us.dropped++;
+                st.log.push("OUT", "drop(stuck) user=" + user +
+                                       " path=" + it->path);
+                it = us.queue.erase(it);
+            } else {
+                ++it;
+            }
+        }
+    }
+
+    std::vector<std::pair<std::string, int64_t>> active;
+    for (auto& [user, us] : st.users)
+        if (!us.queue.empty()) active.emplace_back(user, us.processed);
+    if (active.empty()) return false;
+
+    const uint64_t counter = st.sched_counter.load();
+    auto order = candidate_order(active, st.vip_user, st.boost_user, counter);
+
+    for (const auto& user : order) {
+        auto& us = st.users[user];
+        // scan for the FIRST routable task in this user's queue
+        for (auto it = us.queue.begin(); it != us.queue.end(); ++it) {
+            std::vector<size_t> eligible;
+            for (size_t bi = 0; bi < st.backends.size(); bi++) {
+                const bool op = st.control_ops.count(bi) > 0;
+                if (backend_eligible(st.backends[bi], op,
+                                     it->requested_model, it->path))
+                    eligible.push_back(bi);
+            }
+            if (eligible.empty()) {
+                if (!it->stuck_warned) {
+                    it->stuck_warned = true;
+                    st.log.push("OUT", "no backend available yet for user=" +
+                                           user + " model=" +
+                                           it->requested_model);
+                }
+                continue;
+            }
+            // prefer backends that already have the model loaded
+            // (src/dispatcher.rs:622-640)
+            if (!it->requested_model.empty()) {
+                std::vector<size_t> loaded;
+                for (size_t bi : eligible)
+                    if (smart_model_match(it->requested_model,
+                                          st.backends[bi].loaded_models) ||
+                        fuzzy_model_match(it->requested_model,
+                                          st.backends[bi].loaded_models))
+                        loaded.push_back(bi);
+                if (!loaded.empty()) eligible = std::move(loaded);
+            }
+            const size_t bi =
+                pick_backend(st.backends, eligible, st.last_backend_idx);
+            st.last_backend_idx = bi;
+            st.sched_counter.fetch_add(1);
+            out->task = std::move(*it);
+            us.queue.erase(it);
+            out->backend_idx = bi;
+            out->user = user;
+            us.processing++;
+            st.backends[bi].active_requests++;
+            st.backends[bi].current_model = out->task.requested_model;
+            return true;
+        }
+    }
+    return false;
+}
+
+void finish_dispatch(AppState& st, const Dispatch& d, bool ok,
+                     const std::string& outcome) {
+    {
+        std::scoped_lock lk(st.backends_mu, st.queues_mu);
+        auto& b = st.backends[d.backend_idx];
+        b.active_requests = std::max(0, b.active_requests - 1);
+        b.processed_count++;
+        b.current_model.clear();
+        auto& us = st.users[d.user];
+        us.processing = std::max<int64_t>(0, us.processing - 1);
+        if (ok)
+            us.processed++;
+        else
+            us.dropped++;
+    }
+    st.log.push("OUT", (ok ? "done " : "drop(") + outcome +
+                           (ok ? " " : ") ") + "user=" + d.user + " -> b" +
+                           std::to_string(d.backend_idx));
+    st.notify();  // backend freed: wake the scheduler
+}
+
+// ------------------------------------------------------------- log ring
+void LogRing::push(const std::string& kind, const std::string& text) {
+    std::lock_guard<std::mutex> g(mu_);
+    ring_.push_back({now_ms(), kind, text});
+    while (ring_.size() > cap_) ring_.pop_front();
+}
+
+std::vector<LogEvent> LogRing::snapshot() const {
+    std::lock_guard<std::mutex> g(mu_);
+    return {ring_.begin(), ring_.end()};
+}
+
+}  // namespace omq
