@@ -32,14 +32,15 @@ def _newer(target, *sources):
 
 
 def build_kernels(force=False):
-    src = os.path.join(KDIR, "kernels.hip")
+    import glob
+    srcs = sorted(glob.glob(os.path.join(KDIR, "*.hip")))
     out = os.path.join(KDIR, "_kernels_gfx950.so")
-    if not force and _newer(out, src):
+    if not force and _newer(out, *srcs):
         print(f"kernels up to date: {out}")
         return out
     _run([
         "hipcc", "--offload-arch=gfx950", "-O3", "-std=c++17",
-        "-shared", "-fPIC", src, "-o", out,
+        "-shared", "-fPIC", *srcs, "-o", out,
     ])
     return out
 

@@ -1,0 +1,234 @@
+// gfx950 decode attention with KV-length splitting (flash-decoding style).
+//
+// Problem shape: B sequences × Hq heads × 1 query token, paged KV pool,
+// GQA group G = Hq/KVH.  Memory-bound: the whole job is streaming each
+// sequence's K+V (seq_len × 512 B per kv-head) once.  A (seq, kv-head)
+// grid alone gives B×KVH blocks (256 at B=32 ⇒ 1 workgroup/CU, far too
+// few to hide HBM latency — measured 0.65 TB/s).  Splitting the KV length
+// into SPLIT segments multiplies parallelism: each block computes a
+// partial (o, m, l) over its segment; a small combine kernel merges the
+// online-softmax partials exactly.
+//
+// LDS: K/V chunk tiles shared by the group's G waves (each KV byte read
+// once per kv-head, not per q-head); rows padded +4 bf16 to keep per-lane
+// row reads conflict-free (guide §6 G4).
+#include <hip/hip_runtime.h>
+#include <hip/hip_bf16.h>
+#include <cstdint>
+
+typedef __hip_bfloat16 bf16;
+typedef __hip_bfloat162 bf162;
+
+#define DEV static __device__ __forceinline__
+DEV float bf2f(bf16 x) { return __bfloat162float(x); }
+
+DEV void load8f_lds(const bf16* p, float* out) {
+    const bf162* v = reinterpret_cast<const bf162*>(p);
+    float2 f0 = __bfloat1622float2(v[0]), f1 = __bfloat1622float2(v[1]);
+    float2 f2 = __bfloat1622float2(v[2]), f3 = __bfloat1622float2(v[3]);
+    out[0] = f0.x; out[1] = f0.y; out[2] = f1.x; out[3] = f1.y;
+    out[4] = f2.x; out[5] = f2.y; out[6] = f3.x; out[7] = f3.y;
+}
+
+DEV float wave_max(float x) {
+    #pragma unroll
+    for (int off = 32; off > 0; off >>= 1)
+        x = fmaxf(x, __shfl_down(x, off, 64));
+    return __shfl(x, 0, 64);
+}
+DEV float wave_sum(float x) {
+    #pragma unroll
+    for (int off = 32; off > 0; off >>= 1)
+        x += __shfl_down(x, off, 64);
+    return __shfl(x, 0, 64);
+}
+
+#define DCHUNK 64
+#define DHEAD 128
+#define DKPAD 4
+
+// grid: (n_seqs * split, KVH); block: G*64 threads.
+// Partials: o_part [S][Hq][split][128] f32, ml_part [S][Hq][split][2] f32.
+// When split == 1, writes normalized bf16 straight to out.
+__global__ __launch_bounds__(512) void k_decode_attn(
+    bf16* __restrict__ out,               // [S, Hq, D] (split==1)
+    float* __restrict__ o_part,           // split>1 partials
+    float* __restrict__ ml_part,
+    const bf16* __restrict__ q,           // [S, Hq, D] (row stride qs)
+    const bf16* __restrict__ kpool,       // [P][KVH][page][D]
+    const bf16* __restrict__ vpool,
+    const int* __restrict__ page_table,   // [slots][max_pages]
+    const int* __restrict__ slot_ids,     // [S]
+    const int* __restrict__ seq_lens,     // [S] (kv length incl. this tok)
+    int Hq, int KVH, int page, int max_pages, float scale,
+    int64_t qs, int split)
+{
+    const int S_idx = blockIdx.x / split;
+    const int seg = blockIdx.x % split;
+    const int kvh = blockIdx.y;
+    const int G = Hq / KVH;
+    const int wid = threadIdx.x >> 6;
+    const int lane = threadIdx.x & 63;
+    const int qh = kvh * G + wid;
+
+    const int slot = slot_ids[S_idx];
+    const int kv_len = seq_lens[S_idx];
+    // segment bounds in whole chunks so pages don't straddle segments
+    const int n_chunks = (kv_len + DCHUNK - 1) / DCHUNK;
+    const int per_seg = (n_chunks + split - 1) / split;
+    const int c0 = seg * per_seg;
+    const int c1 = min(n_chunks, c0 + per_seg);
+
+    extern __shared__ __attribute__((aligned(16))) char smem[];
+    const int KROW = DHEAD + DKPAD;
+    bf16* k_tile = reinterpret_cast<bf16*>(smem);
+    bf16* v_tile = k_tile + DCHUNK * KROW;
+    float* q_lds = reinterpret_cast<float*>(v_tile + DCHUNK * KROW); // [G][128]
+    float* p_lds = q_lds + (int64_t)G * DHEAD;                       // [G][65]
+
+    // stage q (fp32 in LDS: repeated broadcast reads in the score loop)
+    if (wid < G) {
+        const bf16* qsrc = q + (int64_t)S_idx * qs + (int64_t)qh * DHEAD;
+        q_lds[wid * DHEAD + lane] = bf2f(qsrc[lane]);
+        q_lds[wid * DHEAD + lane + 64] = bf2f(qsrc[lane + 64]);
+    }
+    __syncthreads();
+
+    float m = -3.0e38f, l = 0.f, o0 = 0.f, o1 = 0.f;
+
+    for (int ch = c0; ch < c1; ch++) {
+        const int base = ch * DCHUNK;
+        const int n_here = min(DCHUNK, kv_len - base);
+        // ---- cooperative K/V chunk stage (4 pages @ page=16) ----
+        {
+            const int tid = threadIdx.x, nthr = blockDim.x;
+            for (int u = tid; u < DCHUNK * (DHEAD / 8); u += nthr) {
+                const int tok = u / (DHEAD / 8), dv = u % (DHEAD / 8);
+                if (base + tok < kv_len) {
+                    const int gp = page_table[(int64_t)slot * max_pages
+                                              + (base + tok) / page];
+                    const int64_t src = (((int64_t)gp * KVH + kvh) * page
+                                         + (base + tok) % page) * DHEAD
+                                        + dv * 8;
+                    *reinterpret_cast<uint4*>(k_tile + tok * KROW + dv * 8) =
+                        *reinterpret_cast<const uint4*>(kpool + src);
+                    *reinterpret_cast<uint4*>(v_tile + tok * KROW + dv * 8) =
+                        *reinterpret_cast<const uint4*>(vpool + src);
+                }
+            }
+        }
+        __syncthreads();
+
+        // ---- score for key j = lane ----
+        float s = 0.f;
+        const bool live = lane < n_here;
+        if (live) {
+            const bf16* krow = k_tile + lane * KROW;
+            const float* qrow = q_lds + wid * DHEAD;
+            #pragma unroll 4
+            for (int d = 0; d < DHEAD; d += 8) {
+                float k8[8];
+                load8f_lds(krow + d, k8);
+                s += k8[0] * qrow[d] + k8[1] * qrow[d + 1]
+                   + k8[2] * qrow[d + 2] + k8[3] * qrow[d + 3]
+                   + k8[4] * qrow[d + 4] + k8[5] * qrow[d + 5]
+                   + k8[6] * qrow[d + 6] + k8[7] * qrow[d + 7];
+            }
+            s *= scale;
+        } else {
+            s = -3.0e38f;
+        }
+
+        // ---- online softmax ----
+        const float smax = wave_max(s);
+        float p = 0.f;
+        if (smax > -3.0e38f) {
+            const float mn = fmaxf(m, smax);
+            const float corr = __expf(m - mn);
+            p = live ? __expf(s - mn) : 0.f;
+            l = l * corr + wave_sum(p);
+            o0 *= corr; o1 *= corr;
+            m = mn;
+        }
+        p_lds[wid * (DCHUNK + 1) + lane] = p;
+
+        // ---- PV: lane accumulates dims (2*lane, 2*lane+1) ----
+        const float* prow = p_lds + wid * (DCHUNK + 1);
+        for (int j = 0; j < n_here; j++) {
+            const bf162 v2 = *reinterpret_cast<const bf162*>(
+                v_tile + j * KROW + 2 * lane);
+            const float pj = prow[j];
+            o0 = fmaf(pj, bf2f(v2.x), o0);
+            o1 = fmaf(pj, bf2f(v2.y), o1);
+        }
+        __syncthreads();  // protect k/v tiles before next stage
+    }
+
+    // ---- emit ----
+    if (split == 1) {
+        const float linv = l > 0.f ? 1.f / l : 0.f;
+        bf16* orow = out + ((int64_t)S_idx * Hq + qh) * DHEAD;
+        reinterpret_cast<bf162*>(orow)[lane] =
+            __float22bfloat162_rn(make_float2(o0 * linv, o1 * linv));
+    } else {
+        const int64_t pi = (((int64_t)S_idx * Hq + qh) * split + seg);
+        float* op = o_part + pi * DHEAD;
+        op[2 * lane] = o0;
+        op[2 * lane + 1] = o1;
+        ml_part[pi * 2] = m;
+        ml_part[pi * 2 + 1] = l;
+    }
+}
+
+// combine: one wave per (seq, head); merges the split online-softmax
+// partials exactly: m* = max m_i; o = Σ e^{m_i-m*} o_i; l = Σ e^{m_i-m*} l_i
+__global__ __launch_bounds__(256) void k_decode_combine(
+    bf16* __restrict__ out, const float* __restrict__ o_part,
+    const float* __restrict__ ml_part, int total, int split)
+{
+    const int sh = blockIdx.x * 4 + (threadIdx.x >> 6);  // seq*Hq + head
+    const int lane = threadIdx.x & 63;
+    if (sh >= total) return;
+    float m = -3.0e38f;
+    for (int i = 0; i < split; i++)
+        m = fmaxf(m, ml_part[((int64_t)sh * split + i) * 2]);
+    float l = 0.f, a0 = 0.f, a1 = 0.f;
+    for (int i = 0; i < split; i++) {
+        const int64_t pi = (int64_t)sh * split + i;
+        const float mi = ml_part[pi * 2];
+        if (mi <= -3.0e38f) continue;
+        const float w = __expf(mi - m);
+        l += w * ml_part[pi * 2 + 1];
+        const float* op = o_part + pi * DHEAD;
+        a0 = fmaf(w, op[2 * lane], a0);
+        a1 = fmaf(w, op[2 * lane + 1], a1);
+    }
+    const float linv = l > 0.f ? 1.f / l : 0.f;
+    reinterpret_cast<bf162*>(out + (int64_t)sh * DHEAD)[lane] =
+        __float22bfloat162_rn(make_float2(a0 * linv, a1 * linv));
+}
+
+extern "C" int decode_attn_bf16(
+    void* out, void* o_part, void* ml_part, const void* q,
+    const void* kpool, const void* vpool, const void* page_table,
+    const void* slot_ids, const void* seq_lens, int S, int Hq, int KVH,
+    int page, int max_pages, float scale, int64_t q_stride, int split,
+    hipStream_t stream)
+{
+    const int G = Hq / KVH;
+    const int lds = 2 * DCHUNK * (DHEAD + DKPAD) * 2 + G * DHEAD * 4
+                  + G * (DCHUNK + 1) * 4;
+    dim3 grid(S * split, KVH);
+    k_decode_attn<<<grid, G * 64, lds, stream>>>(
+        (bf16*)out, (float*)o_part, (float*)ml_part, (const bf16*)q,
+        (const bf16*)kpool, (const bf16*)vpool, (const int*)page_table,
+        (const int*)slot_ids, (const int*)seq_lens, Hq, KVH, page,
+        max_pages, scale, q_stride, split);
+    if (split > 1) {
+        const int waves = S * Hq;
+        k_decode_combine<<<(waves + 3) / 4, 256, 0, stream>>>(
+            (bf16*)out, (const float*)o_part, (const float*)ml_part, waves,
+            split);
+    }
+    return (int)hipGetLastError();
+}

@@ -1,0 +1,60 @@
+"""Microbench: decode attention old (unsplit) vs flash-decoding path.
+
+Run on a GPU box:  python tools/perf_decode.py [ctx] [batch]
+"""
+import sys
+import time
+
+import torch
+
+from ollamamq_amd.engine.kvcache import PagedKVCache
+from ollamamq_amd.ops.interface import AttnMeta
+from ollamamq_amd.ops import hip
+
+
+def main():
+    ctx = int(sys.argv[1]) if len(sys.argv) > 1 else 1024
+    B = int(sys.argv[2]) if len(sys.argv) > 2 else 32
+    Hq, KVH, D, L = 32, 8, 128, 1
+    dev = "cuda"
+    n_pages = B * (ctx // 16 + 2)
+    cache = PagedKVCache(L, KVH, D, page_size=16, n_pages=n_pages,
+                         max_slots=B, max_ctx=ctx + 64, device=dev,
+                         dtype=torch.bfloat16)
+    for i in range(B):
+        s = cache.alloc_slot()
+        cache.ensure(s, ctx)
+    cache.k_pool.normal_()
+    cache.v_pool.normal_()
+    q = torch.randn(B, Hq, D, device=dev).bfloat16()
+    meta = AttnMeta(
+        mode="decode",
+        slot_ids=torch.arange(B, dtype=torch.int32, device=dev),
+        seq_lens=torch.full((B,), ctx, dtype=torch.int32, device=dev),
+        cu_q=torch.arange(B + 1, dtype=torch.int32, device=dev),
+        logits_idx=None, max_q=1, max_kv=ctx)
+
+    kv_bytes = B * KVH * ctx * D * 2 * 2
+
+    def bench(fn, label):
+        for _ in range(5):
+            out = fn()
+        torch.cuda.synchronize()
+        t0 = time.perf_counter()
+        n = 50
+        for _ in range(n):
+            out = fn()
+        torch.cuda.synchronize()
+        dt = (time.perf_counter() - t0) / n
+        print(f"{label}: {dt*1e6:8.1f} us  {kv_bytes/dt/1e12:6.2f} TB/s")
+        return out
+
+    o_new = bench(lambda: hip.attention_decode(q, cache, 0, meta), "split ")
+    o_old = bench(lambda: hip._attention(q, cache, 0, meta, 1), "legacy")
+    diff = (o_new.float() - o_old.float()).abs().max().item()
+    print("max |new-old| =", diff)
+    assert diff < 3e-2, "split path disagrees with legacy"
+
+
+if __name__ == "__main__":
+    main()
