@@ -81,6 +81,13 @@ class LlamaEngine:
             self._gen.manual_seed(0xC0FFEE)
         self.steps = 0
         self.tokens_out = 0
+        # hipGraph-captured decode steps (one graph per exact batch size):
+        # decode is ~7 kernels × n_layers of launches; replay collapses the
+        # launch gaps (MI355X_MICROARCH "launches-baseline": ≈1.2 µs/boundary)
+        import os as _os
+        self.use_graphs = (self.dev.type == "cuda"
+                           and _os.environ.get("OLLAMAMQ_NO_GRAPH") != "1")
+        self._graphs: Dict[int, tuple] = {}
 
     # -- submission --------------------------------------------------------
     def submit(self, prompt: List[int], params: GenParams,
@@ -231,12 +238,70 @@ class LlamaEngine:
                 if s.on_token:
                     s.on_token(int(tok), False)
 
+    def _graph_entry(self, B: int):
+        """Capture (once per batch size) a full decode forward as a hipGraph
+        reading its inputs from static device buffers."""
+        entry = self._graphs.get(B)
+        if entry is not None:
+            return entry
+        dev = self.dev
+        bufs = {
+            "tok": torch.zeros(B, dtype=torch.int32, device=dev),
+            "pos": torch.zeros(B, dtype=torch.int32, device=dev),
+            "slot": torch.zeros(B, dtype=torch.int32, device=dev),
+            "lens": torch.zeros(B, dtype=torch.int32, device=dev),
+        }
+        meta = AttnMeta(
+            mode="decode", slot_ids=bufs["slot"], seq_lens=bufs["lens"],
+            cu_q=torch.arange(B + 1, dtype=torch.int32, device=dev),
+            logits_idx=torch.arange(B, dtype=torch.long, device=dev),
+            max_q=1, max_kv=self.kv.max_ctx)
+        entry = {"bufs": bufs, "meta": meta, "graph": None, "logits": None}
+        self._graphs[B] = entry
+        return entry
+
+    def _decode_forward_graphed(self, seqs, token_list, pos_list):
+        B = len(seqs)
+        entry = self._graph_entry(B)
+        bufs = entry["bufs"]
+        host = torch.tensor(
+            [token_list, pos_list, [s.slot for s in seqs],
+             [self.kv.seq_lens[s.slot] for s in seqs]], dtype=torch.int32)
+        staged = host.to(self.dev, non_blocking=True)
+        bufs["tok"].copy_(staged[0])
+        bufs["pos"].copy_(staged[1])
+        bufs["slot"].copy_(staged[2])
+        bufs["lens"].copy_(staged[3])
+        meta = entry["meta"]
+        if entry["graph"] is None:
+            # warmup twice on a side stream, then capture
+            s = torch.cuda.Stream()
+            s.wait_stream(torch.cuda.current_stream())
+            with torch.cuda.stream(s):
+                for _ in range(2):
+                    self.model.forward(bufs["tok"], bufs["pos"], self.kv,
+                                       bufs["slot"], meta)
+            torch.cuda.current_stream().wait_stream(s)
+            g = torch.cuda.CUDAGraph()
+            with torch.cuda.graph(g):
+                entry["logits"] = self.model.forward(
+                    bufs["tok"], bufs["pos"], self.kv, bufs["slot"], meta)
+            entry["graph"] = g
+        entry["graph"].replay()
+        return entry["logits"]
+
     def _decode_step(self):
         seqs = self.running
         token_list = [s.generated[-1] for s in seqs]
         pos_list = [s.total_len - 1 for s in seqs]
         q_lens = [1] * len(seqs)
-        logits = self._forward(seqs, token_list, pos_list, q_lens, "decode")
+        if self.use_graphs:
+            for s in seqs:
+                self.kv.ensure(s.slot, self.kv.seq_lens[s.slot] + 1)
+            logits = self._decode_forward_graphed(seqs, token_list, pos_list)
+        else:
+            logits = self._forward(seqs, token_list, pos_list, q_lens,
+                                   "decode")
         toks = self._sample(seqs, logits)
         now = time.monotonic()
         for s, tok in zip(seqs, toks):

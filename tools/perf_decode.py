@@ -2,8 +2,11 @@
 
 Run on a GPU box:  python tools/perf_decode.py [ctx] [batch]
 """
+import os
 import sys
 import time
+
+sys.path.insert(0, os.path.dirname(os.path.dirname(os.path.abspath(__file__))))
 
 import torch
 
