@@ -78,6 +78,7 @@ struct ResponseChannel {
 struct Task {
     std::string method;
     std::string path;
+    std::string query;
     std::vector<std::pair<std::string, std::string>> headers;  // Host stripped
     std::string body;
     std::string user_id;

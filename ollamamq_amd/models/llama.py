@@ -190,7 +190,7 @@ class LlamaModel:
             normed, residual = ops.rmsnorm_residual(
                 x, residual, layer.attn_norm, cfg.norm_eps
             )
-            qkv = torch.nn.functional.linear(normed, layer.wqkv)
+            qkv = ops.linear(normed, layer.wqkv)
             nl, nkl, d = self.n_local_heads, self.n_local_kv_heads, cfg.head_dim
             q, k, v = qkv.split([nl * d, nkl * d, nkl * d], dim=-1)
             q = q.view(-1, nl, d)
@@ -199,14 +199,14 @@ class LlamaModel:
             ops.rope(q, k, positions, self.rope_cos, self.rope_sin)
             ops.kv_append(kv_cache, li, k, v, slot_ids, positions)
             attn = ops.attention(q, kv_cache, li, attn_meta)
-            x = torch.nn.functional.linear(attn.view(-1, nl * d), layer.wo)
+            x = ops.linear(attn.view(-1, nl * d), layer.wo)
             self._allreduce(x)  # RCCL all-reduce #1 (TP)
             normed, residual = ops.rmsnorm_residual(
                 x, residual, layer.mlp_norm, cfg.norm_eps
             )
-            gate_up = torch.nn.functional.linear(normed, layer.wgate_up)
+            gate_up = ops.linear(normed, layer.wgate_up)
             act = ops.swiglu(gate_up)
-            x = torch.nn.functional.linear(act, layer.wdown)
+            x = ops.linear(act, layer.wdown)
             self._allreduce(x)  # RCCL all-reduce #2 (TP)
 
         # Only the last token of each sequence needs logits.
@@ -216,7 +216,7 @@ class LlamaModel:
         else:
             h = x + residual
         h, _ = ops.rmsnorm_residual(h, None, self.final_norm, cfg.norm_eps)
-        logits = torch.nn.functional.linear(h, self.lm_head)
+        logits = ops.linear(h, self.lm_head)
         if self.tp_size > 1:
             # vocab-parallel logits: all-gather shards on the last dim
             shards = [torch.empty_like(logits) for _ in range(self.tp_size)]

@@ -45,9 +45,10 @@ def _try_load():
     lib.argmax_bf16.argtypes = [vp, vp, i, i, vp]
     lib.decode_attn_bf16.argtypes = [vp, vp, vp, vp, vp, vp, vp, vp, vp,
                                      i, i, i, i, i, f, i64, i, vp]
+    lib.skinny_gemm_bf16.argtypes = [vp, vp, vp, i, i, i, i64, vp]
     for fn in ("rmsnorm_residual_bf16", "rope_bf16", "kv_append_bf16",
                "paged_attn_bf16", "swiglu_bf16", "argmax_bf16",
-               "decode_attn_bf16"):
+               "decode_attn_bf16", "skinny_gemm_bf16"):
         getattr(lib, fn).restype = ctypes.c_int
     _lib = lib
 
@@ -233,6 +234,19 @@ def attention_decode(q, cache, layer, meta):
 
 def attention_prefill(q, cache, layer, meta):
     return _attention(q, cache, layer, meta, PREFILL_QT)
+
+
+def linear(x, weight):
+    M, K = x.shape
+    N = weight.shape[0]
+    if M <= 32 and K % 128 == 0 and N % 32 == 0 \
+            and x.dtype == torch.bfloat16 and weight.stride(1) == 1:
+        y = torch.empty((M, N), dtype=x.dtype, device=x.device)
+        assert x.stride(1) == 1
+        _check(_lib.skinny_gemm_bf16(_p(y), _p(x), _p(weight), M, N, K,
+                                     x.stride(0), _stream()), "skinny_gemm")
+        return y
+    return torch.nn.functional.linear(x, weight)
 
 
 def swiglu(gate_up):

@@ -81,6 +81,16 @@ def attention(q, cache, layer, meta: AttnMeta):
     return ref.attention(q, cache, layer, meta)
 
 
+def linear(x, weight):
+    """y = x @ weight.T — library GEMM normally; hand-written MFMA skinny
+    GEMM on the decode path (M<=32), where hipBLASLt is far off the
+    weights-streaming roofline."""
+    if _use_hip(x):
+        from . import hip
+        return hip.linear(x, weight)
+    return torch.nn.functional.linear(x, weight)
+
+
 def swiglu(gate_up):
     if _use_hip(gate_up):
         from . import hip

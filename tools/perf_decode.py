@@ -16,6 +16,7 @@ from ollamamq_amd.ops import hip
 
 
 def main():
+    hip.require()
     ctx = int(sys.argv[1]) if len(sys.argv) > 1 else 1024
     B = int(sys.argv[2]) if len(sys.argv) > 2 else 32
     Hq, KVH, D, L = 32, 8, 128, 1
