@@ -1,0 +1,299 @@
+// C19: terminal dashboard — native ANSI-escape renderer mirroring the
+// reference's panels and keymap (reference src/tui.rs: Backends / Users /
+// Blocked / Logs panels, 100 ms frame cadence, snapshot-per-frame so no
+// state lock is held while rendering; keys: j/k nav, Tab panel cycle,
+// L/U load/unload (typed model name), r config reload, p VIP, b Boost,
+// x/X block user/IP, u unblock, ? help, q/Esc quit).
+#include "tui.h"
+
+#include <poll.h>
+#include <termios.h>
+#include <unistd.h>
+
+#include <algorithm>
+#include <cstdio>
+#include <sstream>
+#include <vector>
+
+#include "control.h"
+#include "core.h"
+#include "scheduler.h"
+#include "server.h"
+
+namespace omq {
+
+namespace {
+
+struct Snapshot {
+    struct B {
+        std::string url;
+        bool online;
+        std::string api;
+        int active;
+        int64_t processed;
+        std::string loaded;
+        std::string op;
+    };
+    struct U {
+        std::string name;
+        size_t queued;
+        int64_t processing, processed, dropped;
+        bool vip, boost;
+    };
+    std::vector<B> backends;
+    std::vector<U> users;
+    std::vector<std::string> blocked;
+    std::vector<LogEvent> logs;
+    uint64_t counter;
+};
+
+Snapshot capture(AppState& st) {
+    Snapshot s;
+    {
+        std::scoped_lock lk(st.control_mu, st.backends_mu);
+        for (size_t i = 0; i < st.backends.size(); i++) {
+            const auto& b = st.backends[i];
+            Snapshot::B e;
+            e.url = b.url;
+            e.online = b.is_online;
+            e.api = api_type_name(b.api_type);
+            e.active = b.active_requests;
+            e.processed = b.processed_count;
+            for (const auto& m : b.loaded_models) {
+                if (!e.loaded.empty()) e.loaded += ",";
+                e.loaded += m;
+            }
+            auto op = st.control_ops.find(i);
+            if (op != st.control_ops.end())
+                e.op = (op->second.action == ControlAction::Load ? "load "
+                                                                 : "unload ") +
+                       op->second.model;
+            s.backends.push_back(std::move(e));
+        }
+    }
+    {
+        std::scoped_lock lk(st.queues_mu, st.prio_mu);
+        for (const auto& [name, us] : st.users) {
+            Snapshot::U u;
+            u.name = name;
+            u.queued = us.queue.size();
+            u.processing = us.processing;
+            u.processed = us.processed;
+            u.dropped = us.dropped;
+            u.vip = name == st.vip_user;
+            u.boost = name == st.boost_user;
+            s.users.push_back(std::move(u));
+        }
+    }
+    {
+        std::lock_guard<std::mutex> g(st.blocked_mu);
+        for (const auto& u : st.blocked_users) s.blocked.push_back("user " + u);
+        for (const auto& ip : st.blocked_ips) s.blocked.push_back("ip " + ip);
+    }
+    s.logs = st.log.snapshot();
+    s.counter = st.sched_counter.load();
+    return s;
+}
+
+struct RawTerm {
+    termios orig{};
+    bool ok = false;
+    RawTerm() {
+        if (tcgetattr(STDIN_FILENO, &orig) == 0) {
+            termios raw = orig;
+            raw.c_lflag &= ~(ECHO | ICANON);
+            raw.c_cc[VMIN] = 0;
+            raw.c_cc[VTIME] = 0;
+            tcsetattr(STDIN_FILENO, TCSANOW, &raw);
+            ok = true;
+        }
+        printf("\x1b[?1049h\x1b[?25l");  // alt screen, hide cursor
+        fflush(stdout);
+    }
+    ~RawTerm() {
+        printf("\x1b[?1049l\x1b[?25h");
+        fflush(stdout);
+        if (ok) tcsetattr(STDIN_FILENO, TCSANOW, &orig);
+    }
+};
+
+std::string pad(std::string s, size_t w) {
+    if (s.size() > w) return s.substr(0, w - 1) + "…";
+    s.resize(w, ' ');
+    return s;
+}
+
+}  // namespace
+
+void run_tui(Server& server) {
+    AppState& st = server.state();
+    RawTerm term;
+    int sel_user = 0;
+    std::string input;        // typed model name for L/U
+    char input_mode = 0;      // 'L' or 'U' when typing
+    std::string status_msg;
+
+    while (true) {
+        Snapshot s = capture(st);
+        std::ostringstream out;
+        out << "\x1b[H\x1b[2J";
+        out << "\x1b[1m ollamamq-amd dispatcher — MI355X \x1b[0m"
+            << "  backends:" << s.backends.size()
+            << "  users:" << s.users.size()
+            << "  sched:" << s.counter << "\r\n";
+        out << "\x1b[7m" << pad(" Backends", 90) << "\x1b[0m\r\n";
+        for (const auto& b : s.backends) {
+            out << (b.online ? " \x1b[32m●\x1b[0m " : " \x1b[31m○\x1b[0m ")
+                << pad(b.url, 32) << pad(b.api, 8)
+                << "act:" << b.active << " done:" << b.processed << " "
+                << pad(b.loaded, 24)
+                << (b.op.empty() ? "" : " [" + b.op + "]") << "\r\n";
+        }
+        out << "\x1b[7m" << pad(" Users (j/k nav, p VIP, b Boost, x block)",
+                                90)
+            << "\x1b[0m\r\n";
+        for (int i = 0; i < (int)s.users.size(); i++) {
+            const auto& u = s.users[i];
+            out << (i == sel_user ? " >" : "  ")
+                << pad(u.name + (u.vip ? " ★" : "") + (u.boost ? " ⚡" : ""),
+                       24)
+                << " q:" << u.queued << " run:" << u.processing
+                << " done:" << u.processed << " drop:" << u.dropped
+                << "\r\n";
+        }
+        if (!s.blocked.empty()) {
+            out << "\x1b[7m" << pad(" Blocked (u unblock)", 90)
+                << "\x1b[0m\r\n";
+            for (const auto& b : s.blocked) out << "  " << b << "\r\n";
+        }
+        out << "\x1b[7m" << pad(" Logs", 90) << "\x1b[0m\r\n";
+        const size_t n0 = s.logs.size() > 9 ? s.logs.size() - 9 : 0;
+        for (size_t i = n0; i < s.logs.size(); i++)
+            out << "  " << pad(s.logs[i].kind, 4) << s.logs[i].text
+                << "\r\n";
+        if (input_mode)
+            out << "\r\n " << (input_mode == 'L' ? "load" : "unload")
+                << " model: " << input << "_\r\n";
+        else
+            out << "\r\n " << status_msg
+                << "  [L]oad [U]nload [r]eload-cfg [p]VIP [b]Boost "
+                   "[x]block [u]unblock [q]uit\r\n";
+        fputs(out.str().c_str(), stdout);
+        fflush(stdout);
+
+        pollfd pfd{STDIN_FILENO, POLLIN, 0};
+        if (poll(&pfd, 1, 100) > 0) {
+            char c;
+            if (read(STDIN_FILENO, &c, 1) == 1) {
+                if (input_mode) {
+                    if (c == 27) {  // Esc
+                        input_mode = 0;
+                        input.clear();
+                    } else if (c == '\n' || c == '\r') {
+                        ControlRequest req;
+                        req.action = input_mode == 'L'
+                                         ? ControlAction::Load
+                                         : ControlAction::Unload;
+                        req.model = input;
+                        req.backend_idx = 0;  // first backend
+                        auto r = start_model_control(st, req);
+                        status_msg = r.http_status == 202
+                                         ? "accepted"
+                                         : r.body.get_str("error");
+                        input_mode = 0;
+                        input.clear();
+                    } else if (c == 127 || c == 8) {
+                        if (!input.empty()) input.pop_back();
+                    } else if (c >= 32) {
+                        input += c;
+                    }
+                    continue;
+                }
+                auto sel_name = [&]() -> std::string {
+                    if (sel_user < (int)s.users.size())
+                        return s.users[sel_user].name;
+                    return "";
+                };
+                switch (c) {
+                    case 'q':
+                    case 27:
+                        return;
+                    case 'j':
+                        sel_user = std::min<int>(sel_user + 1,
+                                                 (int)s.users.size() - 1);
+                        break;
+                    case 'k':
+                        sel_user = std::max(sel_user - 1, 0);
+                        break;
+                    case 'L':
+                    case 'U':
+                        input_mode = c;
+                        input.clear();
+                        break;
+                    case 'r':
+                        reload_model_config(st, server.config_path);
+                        status_msg = "config reloaded";
+                        break;
+                    case 'p': {
+                        std::lock_guard<std::mutex> g(st.prio_mu);
+                        const std::string u = sel_name();
+                        st.vip_user = st.vip_user == u ? "" : u;
+                        if (st.vip_user == st.boost_user)
+                            st.boost_user.clear();
+                        break;
+                    }
+                    case 'b': {
+                        std::lock_guard<std::mutex> g(st.prio_mu);
+                        const std::string u = sel_name();
+                        st.boost_user = st.boost_user == u ? "" : u;
+                        if (st.boost_user == st.vip_user)
+                            st.vip_user.clear();
+                        break;
+                    }
+                    case 'x': {
+                        const std::string u = sel_name();
+                        if (!u.empty()) {
+                            {
+                                std::lock_guard<std::mutex> g(st.blocked_mu);
+                                st.blocked_users.insert(u);
+                            }
+                            st.save_blocked();
+                        }
+                        break;
+                    }
+                    case 'X': {
+                        const std::string u = sel_name();
+                        std::string ip;
+                        {
+                            std::lock_guard<std::mutex> g(st.queues_mu);
+                            auto it = st.user_ips.find(u);
+                            if (it != st.user_ips.end()) ip = it->second;
+                        }
+                        if (!ip.empty()) {
+                            {
+                                std::lock_guard<std::mutex> g(st.blocked_mu);
+                                st.blocked_ips.insert(ip);
+                            }
+                            st.save_blocked();
+                        }
+                        break;
+                    }
+                    case 'u': {
+                        {
+                            std::lock_guard<std::mutex> g(st.blocked_mu);
+                            st.blocked_users.clear();
+                            st.blocked_ips.clear();
+                        }
+                        st.save_blocked();
+                        status_msg = "unblocked all";
+                        break;
+                    }
+                    default:
+                        break;
+                }
+            }
+        }
+    }
+}
+
+}  // namespace omq

@@ -1,0 +1,232 @@
+// WorkerBackend: an in-process MI355X GPU worker as a scheduler backend.
+//
+// This replaces the reference's process/network boundary (an HTTP hop to an
+// external Ollama server, reference src/dispatcher.rs:742) with a direct
+// submit to a framework-owned per-GPU engine process over a unix domain
+// socket: one engine process per GPU (the torch.distributed/RCCL process
+// model), newline-framed JSON control messages, raw byte streaming for
+// token output, cancellation by socket close (the worker aborts the
+// sequence and frees its KV when the dispatcher hangs up — the reference
+// merely dropped bytes, SURVEY.md §7 hard-part 4).
+#include <sys/socket.h>
+#include <sys/un.h>
+#include <unistd.h>
+
+#include <cstring>
+
+#include "control.h"
+#include "json.h"
+#include "server.h"
+#include "tui.h"
+
+namespace omq {
+
+namespace {
+
+int uds_connect(const std::string& path, int timeout_ms) {
+    int fd = ::socket(AF_UNIX, SOCK_STREAM, 0);
+    if (fd < 0) return -1;
+    timeval tv{timeout_ms / 1000, (timeout_ms % 1000) * 1000};
+    setsockopt(fd, SOL_SOCKET, SO_RCVTIMEO, &tv, sizeof tv);
+    setsockopt(fd, SOL_SOCKET, SO_SNDTIMEO, &tv, sizeof tv);
+    sockaddr_un addr{};
+    addr.sun_family = AF_UNIX;
+    strncpy(addr.sun_path, path.c_str(), sizeof(addr.sun_path) - 1);
+    if (connect(fd, (sockaddr*)&addr, sizeof addr) != 0) {
+        ::close(fd);
+        return -1;
+    }
+    return fd;
+}
+
+bool send_all(int fd, const std::string& s) {
+    const char* p = s.data();
+    size_t n = s.size();
+    while (n) {
+        ssize_t w = ::send(fd, p, n, MSG_NOSIGNAL);
+        if (w <= 0) return false;
+        p += w;
+        n -= (size_t)w;
+    }
+    return true;
+}
+
+// read one newline-terminated JSON line
+bool recv_line(int fd, std::string& line) {
+    line.clear();
+    char c;
+    while (true) {
+        ssize_t r = ::recv(fd, &c, 1, 0);
+        if (r <= 0) return false;
+        if (c == '\n') return true;
+        line += c;
+        if (line.size() > (1 << 20)) return false;
+    }
+}
+
+class WorkerBackend : public Backend {
+public:
+    WorkerBackend(std::string sock, int64_t timeout_s)
+        : sock_(std::move(sock)), timeout_s_(timeout_s) {}
+
+    ProbeResult probe(const std::set<std::string>&) override {
+        ProbeResult r;
+        int fd = uds_connect(sock_, 2000);
+        if (fd < 0) return r;
+        Json req = Json::object();
+        req.set("cmd", Json::string("probe"));
+        if (!send_all(fd, req.dump() + "\n")) {
+            ::close(fd);
+            return r;
+        }
+        std::string line;
+        if (recv_line(fd, line)) {
+            auto j = Json::parse(line);
+            if (j) {
+                r.online = j->get_bool("online", true);
+                r.api_type = ApiType::Both;  // workers speak Ollama + OpenAI
+                if (const Json* ms = j->find("models"))
+                    for (const auto& m : ms->arr)
+                        if (m.is_str()) r.available_models.push_back(m.str);
+                if (const Json* ls = j->find("loaded"))
+                    for (const auto& m : ls->arr)
+                        if (m.is_str()) r.loaded_models.push_back(m.str);
+                if (const Json* cx = j->find("ctx"))
+                    for (const auto& [k, v] : cx->obj)
+                        if (v.is_num()) r.loaded_ctx[k] = (int64_t)v.num;
+                max_conc_ = (int)j->get_num("max_concurrency", 1);
+            }
+        }
+        ::close(fd);
+        return r;
+    }
+
+    int execute(const Task& task) override {
+        auto resp = task.resp;
+        int fd = uds_connect(sock_, 5000);
+        if (fd < 0) {
+            if (resp) {
+                resp->send_status(502,
+                                  {{"Content-Type", "application/json"}});
+                resp->send_chunk("{\"error\":\"worker unavailable\"}");
+                resp->finish();
+            }
+            return -1;
+        }
+        Json req = Json::object();
+        req.set("cmd", Json::string("request"));
+        req.set("method", Json::string(task.method));
+        req.set("path", Json::string(task.path));
+        req.set("user", Json::string(task.user_id));
+        req.set("body", Json::string(task.body));
+        bool ok = send_all(fd, req.dump() + "\n");
+        int status = -1;
+        std::string line;
+        if (ok && recv_line(fd, line)) {
+            auto j = Json::parse(line);
+            if (j) {
+                status = (int)j->get_num("status", 200);
+                std::string ct = j->get_str("content_type",
+                                            "application/x-ndjson");
+                if (resp)
+                    resp->send_status(status, {{"Content-Type", ct}});
+                // stream raw bytes until worker closes
+                char buf[65536];
+                while (true) {
+                    ssize_t r = ::recv(fd, buf, sizeof buf, 0);
+                    if (r <= 0) break;
+                    if (resp && !resp->send_chunk(std::string(buf, r))) {
+                        // client gone: hang up so the worker cancels
+                        break;
+                    }
+                }
+            }
+        }
+        if (status < 0 && resp) {
+            resp->send_status(502, {{"Content-Type", "application/json"}});
+            resp->send_chunk("{\"error\":\"worker failed\"}");
+        }
+        if (resp) resp->finish();
+        ::close(fd);
+        return status;
+    }
+
+    std::string load_model(const std::string& model, int64_t num_ctx,
+                           int64_t keep_alive,
+                           const BackendStatus&) override {
+        Json req = Json::object();
+        req.set("cmd", Json::string("load"));
+        req.set("model", Json::string(model));
+        if (num_ctx > 0) req.set("num_ctx", Json::number((double)num_ctx));
+        req.set("keep_alive", Json::number((double)keep_alive));
+        return control(req);
+    }
+
+    std::string unload_model(const std::string& model,
+                             const BackendStatus&) override {
+        Json req = Json::object();
+        req.set("cmd", Json::string("unload"));
+        req.set("model", Json::string(model));
+        return control(req);
+    }
+
+    bool supports_control(const BackendStatus&) const override {
+        return true;
+    }
+
+    int max_concurrency() const { return max_conc_; }
+
+private:
+    std::string control(const Json& req) {
+        int fd = uds_connect(sock_, 5000);
+        if (fd < 0) return "worker unavailable";
+        std::string err = "worker did not answer";
+        if (send_all(fd, req.dump() + "\n")) {
+            std::string line;
+            // loads allocate ~16 GB in HBM: allow minutes
+            timeval tv{600, 0};
+            setsockopt(fd, SOL_SOCKET, SO_RCVTIMEO, &tv, sizeof tv);
+            if (recv_line(fd, line)) {
+                auto j = Json::parse(line);
+                if (j) err = j->get_bool("ok") ? "" : j->get_str(
+                                 "error", "worker error");
+            }
+        }
+        ::close(fd);
+        return err;
+    }
+
+    std::string sock_;
+    int64_t timeout_s_;
+    int max_conc_ = 1;
+};
+
+}  // namespace
+
+void add_worker_backend(Server& server, const std::string& spec) {
+    std::string path = spec;
+    int max_conc = 0;
+    auto q = spec.find('?');
+    if (q != std::string::npos) {
+        path = spec.substr(0, q);
+        const std::string opt = spec.substr(q + 1);
+        if (opt.rfind("max_conc=", 0) == 0)
+            max_conc = atoi(opt.c_str() + 9);
+    }
+    auto impl = std::make_shared<WorkerBackend>(
+        path, server.state().settings.timeout_s);
+    server.add_backend(impl, "worker:" + path);
+    // a worker advertises its own concurrency (continuous batching);
+    // an explicit ?max_conc= pins it
+    auto& st = server.state();
+    std::lock_guard<std::mutex> g(st.backends_mu);
+    auto& b = st.backends.back();
+    ProbeResult p = impl->probe({});
+    apply_probe(b, p);
+    b.max_concurrency = max_conc > 0 ? max_conc
+                        : (impl->max_concurrency() > 0
+                               ? impl->max_concurrency()
+                               : 1);
+}
+
+}  // namespace omq

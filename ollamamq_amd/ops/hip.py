@@ -236,10 +236,17 @@ def attention_prefill(q, cache, layer, meta):
     return _attention(q, cache, layer, meta, PREFILL_QT)
 
 
+SKINNY_MAX_N = int(os.environ.get("OLLAMAMQ_SKINNY_MAX_N", "4096"))
+SKINNY_MAX_K = int(os.environ.get("OLLAMAMQ_SKINNY_MAX_K", "4096"))
+
+
 def linear(x, weight):
     M, K = x.shape
     N = weight.shape[0]
+    # gated to the shapes where the hand-written kernel beats hipBLASLt
+    # (measured tools/perf_gemm.py); widen via env as the kernel improves
     if M <= 32 and K % 128 == 0 and N % 32 == 0 \
+            and N <= SKINNY_MAX_N and K <= SKINNY_MAX_K \
             and x.dtype == torch.bfloat16 and weight.stride(1) == 1:
         y = torch.empty((M, N), dtype=x.dtype, device=x.device)
         assert x.stride(1) == 1
