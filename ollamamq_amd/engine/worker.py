@@ -1,0 +1,442 @@
+"""Per-GPU engine worker process.
+
+One worker owns one MI355X (or a TP group of them) and serves the
+dispatcher over a unix domain socket — the in-process replacement for the
+reference's HTTP hop to an external Ollama server (reference
+src/dispatcher.rs:742; SURVEY.md §5 "Distributed communication backend":
+the hop becomes a direct submit into the engine's continuous batch).
+
+Protocol (newline-framed JSON in, framed stream out):
+  {"cmd":"probe"}                  -> one JSON line of state
+  {"cmd":"load","model":M,...}     -> {"ok":true} | {"error":...}
+  {"cmd":"unload","model":M}       -> same
+  {"cmd":"request","method":...,"path":...,"body":"<raw http body>"}
+                                   -> {"status":...,"content_type":...}\n
+                                      then raw response bytes, EOF = done.
+Cancellation: the dispatcher closes the socket; the worker notices the
+write failure and cancels the sequence (KV pages freed) — the reference
+only dropped bytes, we stop the compute (SURVEY.md §7 hard-part 4).
+
+Run:  python -m ollamamq_amd.engine.worker --socket /tmp/omq0.sock \
+          --gpu 0 [--model llama3-8b --max-ctx 4096 --preload]
+"""
+from __future__ import annotations
+
+import argparse
+import json
+import os
+import queue
+import socket
+import threading
+import time
+from typing import Dict, Optional
+
+import torch
+
+from ..models import LlamaModel, PRESETS
+from .engine import GenParams, LlamaEngine
+from .kvcache import PagedKVCache
+from .tokenizer import ByteTokenizer
+
+VERSION = "0.1.0-ollamamq-amd"
+
+
+class Worker:
+    def __init__(self, gpu: int, max_batch: int = 32,
+                 default_ctx: int = 4096):
+        self.gpu = gpu
+        self.max_batch = max_batch
+        self.default_ctx = default_ctx
+        self.device = f"cuda:{gpu}" if torch.cuda.is_available() else "cpu"
+        self.dtype = (torch.bfloat16 if torch.cuda.is_available()
+                      else torch.float32)
+        self.engines: Dict[str, LlamaEngine] = {}
+        self.tokenizers: Dict[str, ByteTokenizer] = {}
+        self.loaded_ctx: Dict[str, int] = {}
+        self.lock = threading.RLock()          # engine-state lock
+        self.work_ev = threading.Event()
+        self.started = time.time()
+        threading.Thread(target=self._engine_loop, daemon=True).start()
+
+    # ---------------------------------------------------------- control
+    def available_models(self):
+        if self.device == "cpu":
+            return ["tiny", "tiny-cpu"]
+        return [n for n in PRESETS if n != "tiny-cpu"]
+
+    def load(self, model: str, num_ctx: int = 0) -> Optional[str]:
+        with self.lock:
+            if model in self.engines:
+                if num_ctx and self.loaded_ctx.get(model) != num_ctx:
+                    err = self.unload(model)
+                    if err:
+                        return err
+                else:
+                    return None
+            cfg = PRESETS.get(model)
+            if cfg is None:
+                return f"unknown model: {model}"
+            ctx = min(num_ctx or self.default_ctx, cfg.max_ctx)
+            try:
+                if torch.cuda.is_available():
+                    torch.cuda.set_device(self.gpu)
+                m = LlamaModel(cfg, device=self.device, dtype=self.dtype,
+                               seed=1234, fast_init=self.device != "cpu")
+                pages = (self.max_batch + 2) * ((ctx + 15) // 16 + 2)
+                kv = PagedKVCache.for_model(
+                    cfg, n_pages=pages, max_slots=self.max_batch + 2,
+                    max_ctx=ctx, device=self.device, dtype=self.dtype)
+                self.engines[model] = LlamaEngine(m, kv,
+                                                 max_batch=self.max_batch)
+                self.tokenizers[model] = ByteTokenizer(cfg.vocab)
+                self.loaded_ctx[model] = ctx
+                return None
+            except torch.cuda.OutOfMemoryError:
+                return "out of HBM: model + KV pool do not fit"
+            except Exception as e:  # pragma: no cover
+                return f"load failed: {e}"
+
+    def unload(self, model: str) -> Optional[str]:
+        with self.lock:
+            eng = self.engines.pop(model, None)
+            if eng is None:
+                return f"model not loaded: {model}"
+            self.tokenizers.pop(model, None)
+            self.loaded_ctx.pop(model, None)
+            del eng
+            if torch.cuda.is_available():
+                torch.cuda.empty_cache()
+            return None
+
+    def resolve(self, requested: str) -> Optional[str]:
+        with self.lock:
+            if requested in self.engines:
+                return requested
+            for name in self.engines:
+                if requested.split(":")[0].lower() == name.lower():
+                    return name
+            # auto-load when a known preset is requested
+            if requested in PRESETS:
+                return requested
+            base = requested.split(":")[0].lower()
+            for name in PRESETS:
+                if name.lower() == base:
+                    return name
+        return None
+
+    # ----------------------------------------------------------- engine
+    def _engine_loop(self):
+        while True:
+            busy = False
+            with self.lock:
+                engines = list(self.engines.values())
+            for eng in engines:
+                with self.lock:
+                    if eng.has_work():
+                        eng.step()
+                        busy = eng.has_work() or busy
+            if not busy:
+                self.work_ev.wait(timeout=0.02)
+                self.work_ev.clear()
+
+    def generate(self, model: str, prompt_tokens, params: GenParams,
+                 on_token) -> int:
+        with self.lock:
+            if model not in self.engines:
+                err = self.load(model)
+                if err:
+                    raise RuntimeError(err)
+            sid = self.engines[model].submit(prompt_tokens, params, on_token)
+        self.work_ev.set()
+        return sid
+
+    def cancel(self, model: str, sid: int):
+        with self.lock:
+            eng = self.engines.get(model)
+            if eng:
+                eng.cancel(sid)
+
+
+# ------------------------------------------------------------------ wire
+def _params_from_body(body: dict, tok: ByteTokenizer,
+                      openai: bool) -> GenParams:
+    opts = body.get("options") or {}
+    max_t = (body.get("max_tokens") or body.get("max_completion_tokens")
+             or opts.get("num_predict") or 64)
+    return GenParams(
+        max_tokens=int(max_t),
+        temperature=float(body.get("temperature",
+                                   opts.get("temperature", 0.0)) or 0.0),
+        top_k=int(opts.get("top_k", body.get("top_k", 0)) or 0),
+        top_p=float(body.get("top_p", opts.get("top_p", 1.0)) or 1.0),
+        stop_token=tok.stop_token,
+        seed=opts.get("seed") or body.get("seed"),
+    )
+
+
+def _prompt_text(body: dict, path: str) -> str:
+    if "prompt" in body and isinstance(body["prompt"], str):
+        return body["prompt"]
+    parts = []
+    for m in body.get("messages") or []:
+        parts.append(f"<{m.get('role', 'user')}>{m.get('content', '')}")
+    return "\n".join(parts) or " "
+
+
+class Conn:
+    """One dispatcher request served over one accepted UDS connection."""
+
+    def __init__(self, sock: socket.socket, worker: Worker):
+        self.sock = sock
+        self.worker = worker
+
+    def run(self):
+        try:
+            line = self._recv_line()
+            if line is None:
+                return
+            msg = json.loads(line)
+            cmd = msg.get("cmd")
+            if cmd == "probe":
+                self._probe()
+            elif cmd == "load":
+                err = self.worker.load(msg.get("model", ""),
+                                       int(msg.get("num_ctx", 0) or 0))
+                self._line({"ok": err is None, **({"error": err} if err
+                                                 else {})})
+            elif cmd == "unload":
+                err = self.worker.unload(msg.get("model", ""))
+                self._line({"ok": err is None, **({"error": err} if err
+                                                 else {})})
+            elif cmd == "request":
+                self._request(msg)
+            else:
+                self._line({"error": f"unknown cmd {cmd}"})
+        except (BrokenPipeError, ConnectionResetError, OSError):
+            pass
+        except Exception as e:
+            try:
+                self._line({"error": str(e)})
+            except OSError:
+                pass
+        finally:
+            try:
+                self.sock.close()
+            except OSError:
+                pass
+
+    # -- framing helpers --
+    def _recv_line(self):
+        buf = b""
+        while b"\n" not in buf:
+            chunk = self.sock.recv(65536)
+            if not chunk:
+                return None
+            buf += chunk
+        return buf.split(b"\n", 1)[0].decode()
+
+    def _line(self, obj):
+        self.sock.sendall((json.dumps(obj) + "\n").encode())
+
+    def _probe(self):
+        w = self.worker
+        with w.lock:
+            loaded = list(w.engines.keys())
+            ctx = dict(w.loaded_ctx)
+        self._line({
+            "online": True,
+            "models": w.available_models(),
+            "loaded": loaded,
+            "ctx": ctx,
+            "max_concurrency": w.max_batch,
+            "device": w.device,
+        })
+
+    # -- the request path --
+    def _request(self, msg):
+        path = msg.get("path", "/")
+        body_raw = msg.get("body") or "{}"
+        try:
+            body = json.loads(body_raw) if body_raw.strip() else {}
+        except json.JSONDecodeError:
+            body = {}
+        w = self.worker
+
+        if path in ("/api/tags", "/v1/models", "/api/ps", "/api/version",
+                    "/", "/api/show"):
+            return self._meta(path, body)
+
+        openai = path.startswith("/v1/")
+        model_req = body.get("model") or ""
+        model = w.resolve(model_req) if model_req else \
+            (next(iter(w.engines), None) or
+             ("tiny" if w.device == "cpu" else "llama3-8b"))
+        if model is None:
+            self._line({"status": 404, "content_type": "application/json"})
+            self.sock.sendall(json.dumps(
+                {"error": f"model not found: {model_req}"}).encode())
+            return
+
+        tok = w.tokenizers.get(model) or ByteTokenizer(
+            PRESETS[model].vocab if model in PRESETS else 512)
+        params = _params_from_body(body, tok, openai)
+        stream = body.get("stream", not openai)
+        prompt = tok.encode(_prompt_text(body, path))
+
+        q: "queue.Queue" = queue.Queue()
+        t0 = time.time()
+        sid = w.generate(model, prompt, params,
+                         lambda t, done: q.put((t, done)))
+
+        if openai:
+            ct = "text/event-stream" if stream else "application/json"
+        else:
+            ct = "application/x-ndjson" if stream else "application/json"
+        self._line({"status": 200, "content_type": ct})
+
+        pieces = []
+        n_out = 0
+        try:
+            while True:
+                t, done = q.get(timeout=600)
+                if done:
+                    break
+                n_out += 1
+                piece = tok.decode_one(t)
+                pieces.append(piece)
+                if stream:
+                    self._stream_piece(path, model, piece, openai)
+            self._final(path, model, pieces, n_out, t0, openai, stream,
+                        len(prompt))
+        except (BrokenPipeError, ConnectionResetError, OSError):
+            w.cancel(model, sid)
+        except queue.Empty:
+            w.cancel(model, sid)
+
+    def _stream_piece(self, path, model, piece, openai):
+        if openai:
+            if path == "/v1/completions":
+                obj = {"object": "text_completion", "model": model,
+                       "choices": [{"index": 0, "text": piece}]}
+            else:
+                obj = {"object": "chat.completion.chunk", "model": model,
+                       "choices": [{"index": 0,
+                                    "delta": {"content": piece}}]}
+            self.sock.sendall(f"data: {json.dumps(obj)}\n\n".encode())
+        else:
+            if path == "/api/generate":
+                obj = {"model": model, "response": piece, "done": False}
+            else:
+                obj = {"model": model,
+                       "message": {"role": "assistant", "content": piece},
+                       "done": False}
+            self.sock.sendall((json.dumps(obj) + "\n").encode())
+
+    def _final(self, path, model, pieces, n_out, t0, openai, stream,
+               n_prompt):
+        dur_ns = int((time.time() - t0) * 1e9)
+        text = "".join(pieces)
+        if openai:
+            if stream:
+                obj = {"object": "chat.completion.chunk", "model": model,
+                       "choices": [{"index": 0, "delta": {},
+                                    "finish_reason": "stop"}]}
+                self.sock.sendall(f"data: {json.dumps(obj)}\n\n".encode())
+                self.sock.sendall(b"data: [DONE]\n\n")
+            else:
+                key = ("text" if path == "/v1/completions" else "message")
+                choice = {"index": 0, "finish_reason": "stop"}
+                if key == "text":
+                    choice["text"] = text
+                else:
+                    choice["message"] = {"role": "assistant",
+                                         "content": text}
+                obj = {"object": ("text_completion"
+                                  if path == "/v1/completions"
+                                  else "chat.completion"),
+                       "model": model, "choices": [choice],
+                       "usage": {"prompt_tokens": n_prompt,
+                                 "completion_tokens": n_out,
+                                 "total_tokens": n_prompt + n_out}}
+                self.sock.sendall(json.dumps(obj).encode())
+        else:
+            obj = {"model": model, "done": True, "done_reason": "stop",
+                   "total_duration": dur_ns,
+                   "prompt_eval_count": n_prompt,
+                   "eval_count": n_out}
+            if not stream:
+                if path == "/api/generate":
+                    obj["response"] = text
+                else:
+                    obj["message"] = {"role": "assistant", "content": text}
+            self.sock.sendall((json.dumps(obj) + "\n").encode())
+
+    def _meta(self, path, body):
+        w = self.worker
+        with w.lock:
+            loaded = list(w.engines.keys())
+            ctx = dict(w.loaded_ctx)
+        if path == "/api/tags":
+            obj = {"models": [{"name": m, "model": m}
+                              for m in w.available_models()]}
+        elif path == "/v1/models":
+            obj = {"object": "list",
+                   "data": [{"id": m, "object": "model"}
+                            for m in w.available_models()]}
+        elif path == "/api/ps":
+            obj = {"models": [{"name": m, "model": m,
+                               "context_length": ctx.get(m, 0)}
+                              for m in loaded]}
+        elif path == "/api/version":
+            obj = {"version": VERSION}
+        elif path == "/api/show":
+            m = body.get("model", "")
+            cfg = PRESETS.get(m)
+            obj = ({"error": "model not found"} if cfg is None else
+                   {"details": {"family": "llama",
+                                "parameter_size": m},
+                    "model_info": {"n_layers": cfg.n_layers,
+                                   "hidden": cfg.hidden,
+                                   "context_length": cfg.max_ctx}})
+        else:
+            obj = {"status": "ollamamq-amd worker",
+                   "device": w.device, "loaded": loaded}
+        self._line({"status": 200, "content_type": "application/json"})
+        self.sock.sendall(json.dumps(obj).encode())
+
+
+def serve(sock_path: str, worker: Worker):
+    try:
+        os.unlink(sock_path)
+    except FileNotFoundError:
+        pass
+    srv = socket.socket(socket.AF_UNIX, socket.SOCK_STREAM)
+    srv.bind(sock_path)
+    srv.listen(64)
+    print(f"worker ready on {sock_path} ({worker.device})", flush=True)
+    while True:
+        conn, _ = srv.accept()
+        threading.Thread(target=Conn(conn, worker).run,
+                         daemon=True).start()
+
+
+def main():
+    ap = argparse.ArgumentParser()
+    ap.add_argument("--socket", required=True)
+    ap.add_argument("--gpu", type=int, default=0)
+    ap.add_argument("--model", type=str, default=None,
+                    help="preload this model")
+    ap.add_argument("--max-ctx", type=int, default=4096)
+    ap.add_argument("--max-batch", type=int, default=32)
+    args = ap.parse_args()
+
+    w = Worker(args.gpu, max_batch=args.max_batch, default_ctx=args.max_ctx)
+    if args.model:
+        err = w.load(args.model, args.max_ctx)
+        if err:
+            raise SystemExit(f"preload failed: {err}")
+    serve(args.socket, w)
+
+
+if __name__ == "__main__":
+    main()

@@ -1,0 +1,172 @@
+"""Full stack on CPU: C++ dispatcher -> UDS -> Python engine worker (tiny
+model, reference-free wire compatibility checks for Ollama + OpenAI)."""
+import json
+import os
+import socket
+import subprocess
+import sys
+import time
+
+import httpx
+import pytest
+
+REPO = os.path.dirname(os.path.dirname(os.path.abspath(__file__)))
+BIN = os.path.join(REPO, "ollamamq_amd", "csrc", "dispatcher",
+                   "ollamamq-server")
+
+
+def _wait_socket(path, timeout=60):
+    t0 = time.time()
+    while time.time() - t0 < timeout:
+        if os.path.exists(path):
+            s = socket.socket(socket.AF_UNIX, socket.SOCK_STREAM)
+            try:
+                s.connect(path)
+                s.close()
+                return True
+            except OSError:
+                pass
+        time.sleep(0.2)
+    return False
+
+
+@pytest.fixture(scope="module")
+def stack(tmp_path_factory):
+    if not os.path.exists(BIN):
+        subprocess.run([sys.executable, "-m", "ollamamq_amd.build"],
+                       check=True)
+    tmp = tmp_path_factory.mktemp("stack")
+    sock = os.path.join(str(tmp), "w0.sock")
+    worker = subprocess.Popen(
+        [sys.executable, "-m", "ollamamq_amd.engine.worker",
+         "--socket", sock, "--model", "tiny-cpu", "--max-ctx", "256",
+         "--max-batch", "4"],
+        cwd=REPO, stdout=subprocess.PIPE, stderr=subprocess.STDOUT,
+        text=True)
+    assert _wait_socket(sock), "worker did not come up"
+    server = subprocess.Popen(
+        [BIN, "--no-tui", "-p", "0", "-w", sock,
+         "-c", os.path.join(str(tmp), "absent.yaml")],
+        stderr=subprocess.PIPE, cwd=str(tmp), text=True)
+    line = server.stderr.readline()
+    port = int(line.rsplit(":", 1)[1].split()[0])
+    base = f"http://127.0.0.1:{port}"
+    # wait for probe
+    deadline = time.time() + 20
+    while time.time() < deadline:
+        try:
+            r = httpx.get(base + "/admin/models").json()
+            if r["backends"] and r["backends"][0]["online"]:
+                break
+        except Exception:
+            pass
+        time.sleep(0.2)
+    yield base
+    server.terminate()
+    worker.terminate()
+
+
+def test_worker_probed(stack):
+    r = httpx.get(stack + "/admin/models").json()
+    b = r["backends"][0]
+    assert b["online"]
+    assert b["api"] == "both"
+    assert "tiny-cpu" in b["available_models"]
+    assert "tiny-cpu" in b["loaded_models"]
+
+
+def test_ollama_chat_streams_through_stack(stack):
+    r = httpx.post(stack + "/api/chat",
+                   json={"model": "tiny-cpu",
+                         "messages": [{"role": "user", "content": "hi"}],
+                         "options": {"num_predict": 6}},
+                   headers={"X-User-ID": "u1"}, timeout=120.0)
+    assert r.status_code == 200, r.text
+    lines = [json.loads(l) for l in r.text.strip().split("\n")]
+    assert lines[-1]["done"] is True
+    assert lines[-1]["eval_count"] >= 1
+    assert all("message" in l for l in lines[:-1])
+
+
+def test_ollama_generate_nonstream(stack):
+    r = httpx.post(stack + "/api/generate",
+                   json={"model": "tiny-cpu", "prompt": "abc",
+                         "stream": False, "options": {"num_predict": 4}},
+                   headers={"X-User-ID": "u2"}, timeout=120.0)
+    assert r.status_code == 200, r.text
+    obj = json.loads(r.text.strip())
+    assert obj["done"] is True
+    assert isinstance(obj["response"], str)
+
+
+def test_openai_chat_completion(stack):
+    r = httpx.post(stack + "/v1/chat/completions",
+                   json={"model": "tiny-cpu", "max_tokens": 5,
+                         "messages": [{"role": "user", "content": "yo"}]},
+                   headers={"X-User-ID": "u3"}, timeout=120.0)
+    assert r.status_code == 200, r.text
+    obj = r.json()
+    assert obj["choices"][0]["message"]["role"] == "assistant"
+    assert obj["usage"]["completion_tokens"] >= 1
+
+
+def test_openai_stream_sse(stack):
+    with httpx.stream(
+            "POST", stack + "/v1/chat/completions",
+            json={"model": "tiny-cpu", "max_tokens": 5, "stream": True,
+                  "messages": [{"role": "user", "content": "yo"}]},
+            headers={"X-User-ID": "u4"}, timeout=120.0) as r:
+        assert r.status_code == 200
+        body = "".join(r.iter_text())
+    assert "data: " in body
+    assert body.strip().endswith("data: [DONE]")
+
+
+def test_tags_and_ps_served_by_worker(stack):
+    r = httpx.get(stack + "/api/tags", timeout=30.0)
+    assert r.status_code == 200
+    names = [m["name"] for m in r.json()["models"]]
+    assert "tiny-cpu" in names
+    r = httpx.get(stack + "/api/version", timeout=30.0)
+    assert "ollamamq-amd" in r.json()["version"]
+
+
+def test_concurrent_users_fair_share(stack):
+    import concurrent.futures as cf
+
+    def one(user):
+        r = httpx.post(stack + "/api/generate",
+                       json={"model": "tiny-cpu", "prompt": "x" * 5,
+                             "stream": False,
+                             "options": {"num_predict": 3}},
+                       headers={"X-User-ID": user}, timeout=120.0)
+        return r.status_code
+
+    with cf.ThreadPoolExecutor(8) as ex:
+        codes = list(ex.map(one, [f"user{i}" for i in range(8)]))
+    assert codes == [200] * 8
+
+
+def test_admin_load_unload_on_worker(stack):
+    r = httpx.post(stack + "/admin/models/load",
+                   json={"model": "tiny", "backend": 0, "num_ctx": 256})
+    assert r.status_code == 202, r.text
+    deadline = time.time() + 60
+    ok = False
+    while time.time() < deadline:
+        st = httpx.get(stack + "/admin/models").json()["backends"][0]
+        if "tiny" in st["loaded_models"] and st["operation"] is None:
+            ok = True
+            break
+        time.sleep(0.3)
+    assert ok, "tiny did not load"
+    r = httpx.post(stack + "/admin/models/unload",
+                   json={"model": "tiny", "backend": 0})
+    assert r.status_code == 202, r.text
+    deadline = time.time() + 30
+    while time.time() < deadline:
+        st = httpx.get(stack + "/admin/models").json()["backends"][0]
+        if "tiny" not in st["loaded_models"]:
+            return
+        time.sleep(0.3)
+    assert False, "tiny did not unload"
