@@ -80,7 +80,11 @@ def _randn(shape, dev, dtype, gen, scale):
 
 
 class LlamaLayer:
-    """One transformer block's resident weights (possibly TP-sharded)."""
+    """One transformer block's resident weights (possibly TP-sharded).
+
+    Every rank generates the SAME seeded global matrix and keeps its own
+    shard, so a TP=N model computes the same function as TP=1 (the gloo
+    CPU tests assert logits parity across degrees)."""
 
     __slots__ = (
         "wqkv", "wo", "wgate_up", "wdown", "attn_norm", "mlp_norm",
@@ -91,15 +95,31 @@ class LlamaLayer:
         scale = 1.0 / math.sqrt(h)
         nh, nkv = cfg.n_heads // tp, max(1, cfg.n_kv_heads // tp)
         d = cfg.head_dim
-        # Fused column-parallel QKV: [q_shard + k_shard + v_shard, hidden]
-        self.wqkv = _randn((nh * d + 2 * nkv * d, h), dev, dtype, gen, scale)
-        # Row-parallel output projection: [hidden, q_shard]
-        self.wo = _randn((h, nh * d), dev, dtype, gen, scale)
-        # Fused column-parallel gate+up: [2*ffn_shard, hidden]
-        f = cfg.ffn // tp
-        self.wgate_up = _randn((2 * f, h), dev, dtype, gen, scale)
-        # Row-parallel down: [hidden, ffn_shard]
-        self.wdown = _randn((h, f), dev, dtype, gen, 1.0 / math.sqrt(cfg.ffn))
+        Hq, KVH = cfg.n_heads, cfg.n_kv_heads
+        # global fused QKV [Hq*d + 2*KVH*d, h]; column-parallel: this rank
+        # keeps its q-head rows + its kv-head rows
+        wqkv_g = _randn(((Hq + 2 * KVH) * d, h), dev, dtype, gen, scale)
+        q = wqkv_g[rank * nh * d:(rank + 1) * nh * d]
+        k = wqkv_g[Hq * d + rank * nkv * d: Hq * d + (rank + 1) * nkv * d]
+        v = wqkv_g[(Hq + KVH) * d + rank * nkv * d:
+                   (Hq + KVH) * d + (rank + 1) * nkv * d]
+        self.wqkv = torch.cat([q, k, v], dim=0).contiguous()
+        del wqkv_g
+        # row-parallel output projection: columns of the global [h, Hq*d]
+        wo_g = _randn((h, Hq * d), dev, dtype, gen, scale)
+        self.wo = wo_g[:, rank * nh * d:(rank + 1) * nh * d].contiguous()
+        del wo_g
+        # column-parallel gate+up: global [2F, h], shard gate + up rows
+        F, f = cfg.ffn, cfg.ffn // tp
+        wgu_g = _randn((2 * F, h), dev, dtype, gen, scale)
+        self.wgate_up = torch.cat(
+            [wgu_g[rank * f:(rank + 1) * f],
+             wgu_g[F + rank * f:F + (rank + 1) * f]], dim=0).contiguous()
+        del wgu_g
+        # row-parallel down: columns of the global [h, F]
+        wd_g = _randn((h, F), dev, dtype, gen, 1.0 / math.sqrt(cfg.ffn))
+        self.wdown = wd_g[:, rank * f:(rank + 1) * f].contiguous()
+        del wd_g
         self.attn_norm = torch.ones(h, device=dev, dtype=dtype)
         self.mlp_norm = torch.ones(h, device=dev, dtype=dtype)
 
@@ -134,7 +154,8 @@ class LlamaModel:
 
         gen_dev = "cuda" if (fast_init and self.device.type == "cuda") else "cpu"
         gen = torch.Generator(device=gen_dev)
-        gen.manual_seed(seed + tp_rank)
+        gen.manual_seed(seed)  # same seed on every rank: shards slice one
+                               # consistent global weight set
         dev = self.device
         scale = 1.0 / math.sqrt(cfg.hidden)
         self.embed = _randn((cfg.vocab, cfg.hidden), dev, dtype, gen, scale)
@@ -143,9 +164,12 @@ class LlamaModel:
             for _ in range(cfg.n_layers)
         ]
         self.final_norm = torch.ones(cfg.hidden, device=dev, dtype=dtype)
-        # Vocab-parallel LM head shard: [vocab/tp, hidden]
+        # Vocab-parallel LM head shard: rows of the global [vocab, hidden]
         vshard = cfg.vocab // tp_size
-        self.lm_head = _randn((vshard, cfg.hidden), dev, dtype, gen, scale)
+        lm_g = _randn((cfg.vocab, cfg.hidden), dev, dtype, gen, scale)
+        self.lm_head = lm_g[tp_rank * vshard:(tp_rank + 1) * vshard]\
+            .contiguous()
+        del lm_g
         # RoPE cos/sin tables precomputed on host (guide: trig on device
         # turns memory-bound RoPE into VALU-bound).
         pos = torch.arange(cfg.max_ctx, dtype=torch.float32)
