@@ -1,21 +1,27 @@
 // gfx950 skinny GEMM for the decode path: y[M,N] = x[M,K] @ W[N,K]^T,
-// M ≤ 32 (the decode batch), bf16 in/out, fp32 MFMA accumulation.
+// M ≤ 32 (decode batch), bf16 in/out, fp32 MFMA accumulation.
 //
 // Why hand-written: decode projections are weights-streaming bound
-// (N×K×2 B read once per step; x is L2-resident), and the library GEMM
-// measured ~1.5-4.5 TB/s on these shapes (M=32) — far off the ≈6.3 TB/s
-// HBM roofline.  This kernel maps the whole problem onto
-// v_mfma_f32_32x32x16_bf16 tiles whose B-fragment is a 16-byte contiguous
-// run of a W row, so every lane issues one dwordx4 per MFMA and the wave
-// streams 1 KiB per instruction.
+// (N×K×2 B read once per step; x is L2/L3-resident) and the library GEMM
+// measured 1.5-2.7 TB/s on the small-N shapes (qkv/o at M=32) — far off
+// the ≈6.3 TB/s HBM roofline.
 //
-// Geometry: one block = 8 waves = one 32-column tile of y; wave w
-// accumulates the k-segment [w*K/8, (w+1)*K/8) (in-block split-K), then
-// the partials are reduced through LDS — no global partial slabs, no
-// second kernel.  Grid = N/32 blocks → N/32 × 8 waves (1536 waves for the
-// 8B qkv projection: enough to keep every SIMD streaming).
+// v2 design (v1 read W fragment-shaped, 16 B per lane at 8 KB row stride:
+// TA/L2-request bound at ~2 TB/s — the guide's M=256 projection-GEMM
+// analysis: fragment-shaped operand loads cost +18-45% with TA_BUSY 2×;
+// full-line LDS staging wins):
+//  * one block = 4 waves = one 32-column tile of y;
+//  * W is streamed in [32 rows][256 k] chunks staged into LDS by ALL
+//    threads with fully-coalesced 16 B lane-consecutive loads (whole
+//    128 B lines), double-buffered, loads for chunk c+1 issued before the
+//    MFMAs of chunk c (T14 issue-early/write-late: HBM latency hides
+//    under compute);
+//  * wave w computes the k-quarter [w*64,(w+1)*64) of each chunk from LDS
+//    (ds_read_b128, row pad +8 bf16 keeps the b128 lane groups
+//    conflict-free) and accumulates its partial in AGPRs;
+//  * epilogue reduces the 4 wave partials through LDS.
 //
-// Constraints: K % 128 == 0, N % 32 == 0, M ≤ 32.
+// Constraints: K % 256 == 0, N % 32 == 0, M ≤ 32 (hip.py gates shapes).
 #include <hip/hip_runtime.h>
 #include <hip/hip_bf16.h>
 #include <cstdint>
@@ -24,73 +30,100 @@ typedef __hip_bfloat16 bf16;
 typedef __attribute__((ext_vector_type(8))) __bf16 bf16x8;
 typedef __attribute__((ext_vector_type(16))) float f32x16;
 
-// A-fragment: lane l -> x[row = l&31][k0 + (l>>5)*8 .. +8)  (zero if row>=M)
-// B-fragment: lane l -> W[n0 + (l&31)][k0 + (l>>5)*8 .. +8)
-__global__ __launch_bounds__(512) void k_skinny_gemm(
+#define KCH 256                  // staged k elems per chunk
+#define BN 32                    // output columns per block
+#define LROW (KCH + 8)           // LDS row stride (bf16), +16 B pad
+
+__global__ __launch_bounds__(256) void k_skinny_gemm(
     bf16* __restrict__ y,            // [M, N]
-    const bf16* __restrict__ x,      // [M, K] row stride xs
+    const bf16* __restrict__ x,      // [M, K], row stride xs
     const bf16* __restrict__ w,      // [N, K] row-major
     int M, int N, int K, int64_t xs)
 {
-    const int n0 = blockIdx.x * 32;
-    const int wid = threadIdx.x >> 6;
-    const int lane = threadIdx.x & 63;
-    const int row = lane & 31;         // x row (M) / w col (N tile)
+    const int n0 = blockIdx.x * BN;
+    const int tid = threadIdx.x;
+    const int wid = tid >> 6;
+    const int lane = tid & 63;
+    const int row = lane & 31;       // x row (M) / fragment row
     const int khalf = (lane >> 5) * 8;
 
-    const int kseg = K >> 3;           // per-wave k extent
-    const int k0 = wid * kseg;
-    const int k1 = k0 + kseg;
+    extern __shared__ __attribute__((aligned(16))) char smem[];
+    bf16* wt = reinterpret_cast<bf16*>(smem);          // [2][BN][LROW]
+    float* red = reinterpret_cast<float*>(wt + 2 * BN * LROW);  // [4][32][32]
 
-    const bf16* xrow = x + (int64_t)row * xs;      // row < 32; masked below
-    const bf16* wrow = w + (int64_t)(n0 + row) * K;
+    // staging geometry: 16 KiB chunk = BN rows × 512 B; 256 threads × 16 B
+    // per round × 4 rounds.  thread t covers (row, 16 B-offset):
+    const int s_row = tid >> 3;            // 32 threads rows? no: 256/8
+    const int s_off = (tid & 7) * 16;      // byte offset within 128 B seg
+    // each round r covers bytes [r*128, r*128+128) of every row via 8
+    // threads; rows covered by tid>>3 (32 rows)
+    const int64_t wbase = (int64_t)(n0 + s_row) * K;
+
+    const int nch = K / KCH;
     const bool live_a = row < M;
+    const bf16* xrow = live_a ? x + (int64_t)row * xs : x;
+
+    uint4 stage[4];
+    auto load_chunk = [&](int c) {
+        const int64_t base = wbase + (int64_t)c * KCH;  // elems
+        #pragma unroll
+        for (int r = 0; r < 4; r++) {
+            // byte address: row start + r*128 + s_off
+            stage[r] = *reinterpret_cast<const uint4*>(
+                reinterpret_cast<const char*>(w + base) + r * 128 + s_off);
+        }
+    };
+    auto write_chunk = [&](int buf) {
+        bf16* dst = wt + buf * BN * LROW + s_row * LROW;
+        #pragma unroll
+        for (int r = 0; r < 4; r++) {
+            *reinterpret_cast<uint4*>(
+                reinterpret_cast<char*>(dst) + r * 128 + s_off) = stage[r];
+        }
+    };
 
     f32x16 acc = {};
-    // 1-deep software prefetch; the uniform-branch-free body lets hipcc
-    // keep several dwordx4 loads in flight across the MFMAs (guide §5
-    // "Three .s-level traps": no per-element runtime condition on loads).
-    if (live_a) {
-        bf16x8 a_n = *reinterpret_cast<const bf16x8*>(xrow + k0 + khalf);
-        bf16x8 b_n = *reinterpret_cast<const bf16x8*>(wrow + k0 + khalf);
-        #pragma unroll 2
-        for (int k = k0; k < k1 - 16; k += 16) {
-            const bf16x8 a = a_n, b = b_n;
-            a_n = *reinterpret_cast<const bf16x8*>(xrow + k + 16 + khalf);
-            b_n = *reinterpret_cast<const bf16x8*>(wrow + k + 16 + khalf);
+    load_chunk(0);
+    write_chunk(0);
+    __syncthreads();
+
+    for (int c = 0; c < nch; c++) {
+        // issue next chunk's loads early (clamped on the last chunk: a
+        // branch around loads would de-pipeline — guide §5 traps (c))
+        load_chunk(c + 1 < nch ? c + 1 : c);
+        // compute this chunk: wave w owns k in [w*64, (w+1)*64)
+        const bf16* wrow = wt + (c & 1) * BN * LROW + row * LROW
+                           + wid * 64 + khalf;
+        const bf16* xk = xrow + (int64_t)c * KCH + wid * 64 + khalf;
+        #pragma unroll
+        for (int s = 0; s < 4; s++) {
+            const bf16x8 b = *reinterpret_cast<const bf16x8*>(wrow + s * 16);
+            const bf16x8 a = live_a
+                ? *reinterpret_cast<const bf16x8*>(xk + s * 16)
+                : bf16x8{};
             acc = __builtin_amdgcn_mfma_f32_32x32x16_bf16(a, b, acc, 0, 0, 0);
         }
-        acc = __builtin_amdgcn_mfma_f32_32x32x16_bf16(a_n, b_n, acc, 0, 0, 0);
-    } else {
-        const bf16x8 a = {};
-        bf16x8 b_n = *reinterpret_cast<const bf16x8*>(wrow + k0 + khalf);
-        #pragma unroll 2
-        for (int k = k0; k < k1 - 16; k += 16) {
-            const bf16x8 b = b_n;
-            b_n = *reinterpret_cast<const bf16x8*>(wrow + k + 16 + khalf);
-            acc = __builtin_amdgcn_mfma_f32_32x32x16_bf16(a, b, acc, 0, 0, 0);
-        }
-        acc = __builtin_amdgcn_mfma_f32_32x32x16_bf16(a, b_n, acc, 0, 0, 0);
+        __syncthreads();
+        if (c + 1 < nch) write_chunk((c + 1) & 1);
+        __syncthreads();
     }
 
-    // ---- in-block split-K reduction through LDS ----
-    // C/D layout (32x32x16): col = lane&31, row = (r&3) + 8*(r>>2) + 4*(lane>>5)
-    __shared__ float red[8][32][32];
+    // ---- reduce the 4 wave partials through LDS ----
+    // C/D layout (32x32x16): col=lane&31, row=(r&3)+8*(r>>2)+4*(lane>>5)
     #pragma unroll
     for (int r = 0; r < 16; r++) {
         const int crow = (r & 3) + 8 * (r >> 2) + 4 * (lane >> 5);
-        red[wid][crow][lane & 31] = acc[r];
+        red[(wid * 32 + crow) * 32 + (lane & 31)] = acc[r];
     }
     __syncthreads();
-    // 512 threads reduce 1024 outputs: 2 per thread
-    const int tid = threadIdx.x;
     #pragma unroll
-    for (int e = tid; e < 1024; e += 512) {
+    for (int e = tid; e < 1024; e += 256) {
         const int m = e >> 5, n = e & 31;
         if (m < M) {
-            float s = red[0][m][n] + red[1][m][n] + red[2][m][n]
-                    + red[3][m][n] + red[4][m][n] + red[5][m][n]
-                    + red[6][m][n] + red[7][m][n];
+            const float s = red[(0 * 32 + m) * 32 + n]
+                          + red[(1 * 32 + m) * 32 + n]
+                          + red[(2 * 32 + m) * 32 + n]
+                          + red[(3 * 32 + m) * 32 + n];
             y[(int64_t)m * N + n0 + n] = __float2bfloat16(s);
         }
     }
@@ -100,7 +133,8 @@ extern "C" int skinny_gemm_bf16(
     void* y, const void* x, const void* w, int M, int N, int K,
     int64_t xs, hipStream_t stream)
 {
-    k_skinny_gemm<<<N / 32, 512, 0, stream>>>(
+    const int lds = 2 * BN * LROW * 2 + 4 * 32 * 32 * 4;
+    k_skinny_gemm<<<N / BN, 256, lds, stream>>>(
         (bf16*)y, (const bf16*)x, (const bf16*)w, M, N, K, xs);
     return (int)hipGetLastError();
 }
