@@ -34,13 +34,17 @@ typedef __attribute__((ext_vector_type(16))) float f32x16;
 #define BN 32                    // output columns per block
 #define LROW (KCH + 8)           // LDS row stride (bf16), +16 B pad
 
+// grid (N/BN, ksplit); block ks owns k-range [ks*K/ksplit, ...).
+// ksplit>1 writes fp32 partials to part[ks][M][N]; k_skinny_combine sums.
 __global__ __launch_bounds__(256) void k_skinny_gemm(
-    bf16* __restrict__ y,            // [M, N]
+    bf16* __restrict__ y,            // [M, N] (ksplit == 1)
+    float* __restrict__ part,        // [ksplit, M, N] (ksplit > 1)
     const bf16* __restrict__ x,      // [M, K], row stride xs
     const bf16* __restrict__ w,      // [N, K] row-major
-    int M, int N, int K, int64_t xs)
+    int M, int N, int K, int64_t xs, int ksplit)
 {
     const int n0 = blockIdx.x * BN;
+    const int ks = blockIdx.y;
     const int tid = threadIdx.x;
     const int wid = tid >> 6;
     const int lane = tid & 63;
@@ -51,35 +55,33 @@ __global__ __launch_bounds__(256) void k_skinny_gemm(
     bf16* wt = reinterpret_cast<bf16*>(smem);          // [2][BN][LROW]
     float* red = reinterpret_cast<float*>(wt + 2 * BN * LROW);  // [4][32][32]
 
-    // staging geometry: 16 KiB chunk = BN rows × 512 B; 256 threads × 16 B
-    // per round × 4 rounds.  thread t covers (row, 16 B-offset):
-    const int s_row = tid >> 3;            // 32 threads rows? no: 256/8
-    const int s_off = (tid & 7) * 16;      // byte offset within 128 B seg
-    // each round r covers bytes [r*128, r*128+128) of every row via 8
-    // threads; rows covered by tid>>3 (32 rows)
+    // staging geometry: 16 KiB chunk = BN rows × 512 B; thread t covers
+    // (row = t>>3, 16 B at (t&7)*16) × 4 rounds of 128 B — consecutive
+    // threads hit consecutive 16 B of one row: full-line coalescing.
+    const int s_row = tid >> 3;
+    const int s_off = (tid & 7) * 16;
     const int64_t wbase = (int64_t)(n0 + s_row) * K;
 
-    const int nch = K / KCH;
+    const int kseg = K / ksplit;          // this block's k extent
+    const int k0 = ks * kseg;
+    const int nch = kseg / KCH;
     const bool live_a = row < M;
     const bf16* xrow = live_a ? x + (int64_t)row * xs : x;
 
     uint4 stage[4];
     auto load_chunk = [&](int c) {
-        const int64_t base = wbase + (int64_t)c * KCH;  // elems
+        const int64_t base = wbase + k0 + (int64_t)c * KCH;  // elems
         #pragma unroll
-        for (int r = 0; r < 4; r++) {
-            // byte address: row start + r*128 + s_off
+        for (int r = 0; r < 4; r++)
             stage[r] = *reinterpret_cast<const uint4*>(
                 reinterpret_cast<const char*>(w + base) + r * 128 + s_off);
-        }
     };
     auto write_chunk = [&](int buf) {
         bf16* dst = wt + buf * BN * LROW + s_row * LROW;
         #pragma unroll
-        for (int r = 0; r < 4; r++) {
+        for (int r = 0; r < 4; r++)
             *reinterpret_cast<uint4*>(
                 reinterpret_cast<char*>(dst) + r * 128 + s_off) = stage[r];
-        }
     };
 
     f32x16 acc = {};
@@ -88,20 +90,24 @@ __global__ __launch_bounds__(256) void k_skinny_gemm(
     __syncthreads();
 
     for (int c = 0; c < nch; c++) {
-        // issue next chunk's loads early (clamped on the last chunk: a
-        // branch around loads would de-pipeline — guide §5 traps (c))
-        load_chunk(c + 1 < nch ? c + 1 : c);
+        // ORDER MATTERS (guide §5 traps (b)): the A fragments are ordinary
+        // global loads — issue them BEFORE the next chunk's staging loads,
+        // or hipcc's wait at the first MFMA drains the staging prefetch.
+        const bf16* xk = xrow + k0 + (int64_t)c * KCH + wid * 64 + khalf;
+        bf16x8 a[4];
+        #pragma unroll
+        for (int s = 0; s < 4; s++)
+            a[s] = live_a ? *reinterpret_cast<const bf16x8*>(xk + s * 16)
+                          : bf16x8{};
+        load_chunk(c + 1 < nch ? c + 1 : c);  // clamped: no branch on loads
         // compute this chunk: wave w owns k in [w*64, (w+1)*64)
         const bf16* wrow = wt + (c & 1) * BN * LROW + row * LROW
                            + wid * 64 + khalf;
-        const bf16* xk = xrow + (int64_t)c * KCH + wid * 64 + khalf;
         #pragma unroll
         for (int s = 0; s < 4; s++) {
             const bf16x8 b = *reinterpret_cast<const bf16x8*>(wrow + s * 16);
-            const bf16x8 a = live_a
-                ? *reinterpret_cast<const bf16x8*>(xk + s * 16)
-                : bf16x8{};
-            acc = __builtin_amdgcn_mfma_f32_32x32x16_bf16(a, b, acc, 0, 0, 0);
+            acc = __builtin_amdgcn_mfma_f32_32x32x16_bf16(a[s], b, acc,
+                                                          0, 0, 0);
         }
         __syncthreads();
         if (c + 1 < nch) write_chunk((c + 1) & 1);
@@ -124,17 +130,41 @@ __global__ __launch_bounds__(256) void k_skinny_gemm(
                           + red[(1 * 32 + m) * 32 + n]
                           + red[(2 * 32 + m) * 32 + n]
                           + red[(3 * 32 + m) * 32 + n];
-            y[(int64_t)m * N + n0 + n] = __float2bfloat16(s);
+            if (ksplit == 1)
+                y[(int64_t)m * N + n0 + n] = __float2bfloat16(s);
+            else
+                part[((int64_t)ks * M + m) * N + n0 + n] = s;
         }
     }
 }
 
+__global__ __launch_bounds__(256) void k_skinny_combine(
+    bf16* __restrict__ y, const float* __restrict__ part,
+    int64_t total, int64_t mn, int ksplit)
+{
+    for (int64_t i = blockIdx.x * blockDim.x + threadIdx.x; i < total;
+         i += (int64_t)gridDim.x * blockDim.x) {
+        float s = 0.f;
+        for (int k = 0; k < ksplit; k++) s += part[k * mn + i];
+        y[i] = __float2bfloat16(s);
+    }
+}
+
 extern "C" int skinny_gemm_bf16(
-    void* y, const void* x, const void* w, int M, int N, int K,
-    int64_t xs, hipStream_t stream)
+    void* y, void* part, const void* x, const void* w, int M, int N, int K,
+    int64_t xs, int ksplit, hipStream_t stream)
 {
     const int lds = 2 * BN * LROW * 2 + 4 * 32 * 32 * 4;
-    k_skinny_gemm<<<N / BN, 256, lds, stream>>>(
-        (bf16*)y, (const bf16*)x, (const bf16*)w, M, N, K, xs);
+    dim3 grid(N / BN, ksplit);
+    k_skinny_gemm<<<grid, 256, lds, stream>>>(
+        (bf16*)y, (float*)part, (const bf16*)x, (const bf16*)w, M, N, K,
+        xs, ksplit);
+    if (ksplit > 1) {
+        const int64_t mn = (int64_t)M * N;
+        const int64_t want = (mn + 255) / 256;
+        const int blocks = (int)(want < 1024 ? want : 1024);
+        k_skinny_combine<<<blocks, 256, 0, stream>>>(
+            (bf16*)y, (const float*)part, mn, mn, ksplit);
+    }
     return (int)hipGetLastError();
 }

@@ -45,7 +45,7 @@ def _try_load():
     lib.argmax_bf16.argtypes = [vp, vp, i, i, vp]
     lib.decode_attn_bf16.argtypes = [vp, vp, vp, vp, vp, vp, vp, vp, vp,
                                      i, i, i, i, i, f, i64, i, vp]
-    lib.skinny_gemm_bf16.argtypes = [vp, vp, vp, i, i, i, i64, vp]
+    lib.skinny_gemm_bf16.argtypes = [vp, vp, vp, vp, i, i, i, i64, i, vp]
     for fn in ("rmsnorm_residual_bf16", "rope_bf16", "kv_append_bf16",
                "paged_attn_bf16", "swiglu_bf16", "argmax_bf16",
                "decode_attn_bf16", "skinny_gemm_bf16"):
@@ -239,19 +239,41 @@ def attention_prefill(q, cache, layer, meta):
 SKINNY_MAX_N = int(os.environ.get("OLLAMAMQ_SKINNY_MAX_N", "4096"))
 SKINNY_MAX_K = int(os.environ.get("OLLAMAMQ_SKINNY_MAX_K", "4096"))
 
+_gemm_scratch = {}
+
+
+def _skinny_ksplit(N, K):
+    """grid k-split so small-N shapes still fill the 256-CU chip."""
+    blocks = N // 32
+    ks = 1
+    while blocks * ks < 256 and ks < 8 and K % (256 * 2 * ks) == 0:
+        ks *= 2
+    return ks
+
 
 def linear(x, weight):
     M, K = x.shape
     N = weight.shape[0]
     # gated to the shapes where the hand-written kernel beats hipBLASLt
     # (measured tools/perf_gemm.py); widen via env as the kernel improves
-    if M <= 32 and K % 128 == 0 and N % 32 == 0 \
+    if M <= 32 and K % 256 == 0 and N % 32 == 0 \
             and N <= SKINNY_MAX_N and K <= SKINNY_MAX_K \
             and x.dtype == torch.bfloat16 and weight.stride(1) == 1:
         y = torch.empty((M, N), dtype=x.dtype, device=x.device)
         assert x.stride(1) == 1
-        _check(_lib.skinny_gemm_bf16(_p(y), _p(x), _p(weight), M, N, K,
-                                     x.stride(0), _stream()), "skinny_gemm")
+        ks = _skinny_ksplit(N, K)
+        part = ctypes.c_void_p(0)
+        if ks > 1:
+            key = (M, N, ks, str(x.device))
+            t = _gemm_scratch.get(key)
+            if t is None:
+                t = torch.empty(ks * M * N, dtype=torch.float32,
+                                device=x.device)
+                _gemm_scratch[key] = t
+            part = _p(t)
+        _check(_lib.skinny_gemm_bf16(_p(y), part, _p(x), _p(weight),
+                                     M, N, K, x.stride(0), ks, _stream()),
+               "skinny_gemm")
         return y
     return torch.nn.functional.linear(x, weight)
 
