@@ -104,17 +104,19 @@ __global__ __launch_bounds__(512) void k_decode_attn(
             const int tid = threadIdx.x, nthr = blockDim.x;
             for (int u = tid; u < DCHUNK * (DHEAD / 8); u += nthr) {
                 const int tok = u / (DHEAD / 8), dv = u % (DHEAD / 8);
-                if (base + tok < kv_len) {
-                    const int gp = page_table[(int64_t)slot * max_pages
-                                              + (base + tok) / page];
-                    const int64_t src = (((int64_t)gp * KVH + kvh) * page
-                                         + (base + tok) % page) * DHEAD
-                                        + dv * 8;
-                    *reinterpret_cast<uint4*>(k_tile + tok * KROW + dv * 8) =
-                        *reinterpret_cast<const uint4*>(kpool + src);
-                    *reinterpret_cast<uint4*>(v_tile + tok * KROW + dv * 8) =
-                        *reinterpret_cast<const uint4*>(vpool + src);
-                }
+                // clamp instead of guarding: a conditional load would
+                // de-pipeline the whole stage (guide §5 traps (c)); the
+                // duplicated tail rows are masked to p=0 by the softmax
+                const int tk = base + tok < kv_len ? base + tok
+                                                   : kv_len - 1;
+                const int gp = page_table[(int64_t)slot * max_pages
+                                          + tk / page];
+                const int64_t src = (((int64_t)gp * KVH + kvh) * page
+                                     + tk % page) * DHEAD + dv * 8;
+                *reinterpret_cast<uint4*>(k_tile + tok * KROW + dv * 8) =
+                    *reinterpret_cast<const uint4*>(kpool + src);
+                *reinterpret_cast<uint4*>(v_tile + tok * KROW + dv * 8) =
+                    *reinterpret_cast<const uint4*>(vpool + src);
             }
         }
         __syncthreads();

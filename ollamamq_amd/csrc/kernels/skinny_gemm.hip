@@ -65,8 +65,12 @@ __global__ __launch_bounds__(256) void k_skinny_gemm(
     const int kseg = K / ksplit;          // this block's k extent
     const int k0 = ks * kseg;
     const int nch = kseg / KCH;
-    const bool live_a = row < M;
-    const bf16* xrow = live_a ? x + (int64_t)row * xs : x;
+    // clamp the row so EVERY lane loads valid memory: a per-lane
+    // conditional load makes hipcc branch around each load with a
+    // vmcnt(0) (guide §5 traps (c): +11k cycles/block).  C rows >= M are
+    // duplicates of row M-1 and are never stored.
+    const int arow = row < M ? row : (M - 1);
+    const bf16* xrow = x + (int64_t)arow * xs;
 
     uint4 stage[4];
     auto load_chunk = [&](int c) {
@@ -97,8 +101,7 @@ __global__ __launch_bounds__(256) void k_skinny_gemm(
         bf16x8 a[4];
         #pragma unroll
         for (int s = 0; s < 4; s++)
-            a[s] = live_a ? *reinterpret_cast<const bf16x8*>(xk + s * 16)
-                          : bf16x8{};
+            a[s] = *reinterpret_cast<const bf16x8*>(xk + s * 16);
         load_chunk(c + 1 < nch ? c + 1 : c);  // clamped: no branch on loads
         // compute this chunk: wave w owns k in [w*64, (w+1)*64)
         const bf16* wrow = wt + (c & 1) * BN * LROW + row * LROW

@@ -46,9 +46,12 @@ def _try_load():
     lib.decode_attn_bf16.argtypes = [vp, vp, vp, vp, vp, vp, vp, vp, vp,
                                      i, i, i, i, i, f, i64, i, vp]
     lib.skinny_gemm_bf16.argtypes = [vp, vp, vp, vp, i, i, i, i64, i, vp]
+    lib.prefill_attn_bf16.argtypes = [vp, vp, vp, vp, vp, vp, vp, vp, vp,
+                                      i, i, i, i, i, f, i64, vp]
     for fn in ("rmsnorm_residual_bf16", "rope_bf16", "kv_append_bf16",
                "paged_attn_bf16", "swiglu_bf16", "argmax_bf16",
-               "decode_attn_bf16", "skinny_gemm_bf16"):
+               "decode_attn_bf16", "skinny_gemm_bf16",
+               "prefill_attn_bf16"):
         getattr(lib, fn).restype = ctypes.c_int
     _lib = lib
 
@@ -127,7 +130,7 @@ def _layer_ptrs(cache, layer):
             ctypes.c_void_p(cache.v_pool.data_ptr() + layer * stride))
 
 
-PREFILL_QT = 16
+PREFILL_QT = 32
 
 
 def _attn_plan(cache, meta, qt):
@@ -233,7 +236,26 @@ def attention_decode(q, cache, layer, meta):
 
 
 def attention_prefill(q, cache, layer, meta):
-    return _attention(q, cache, layer, meta, PREFILL_QT)
+    """MFMA flash prefill (prefill_attn.hip); set OLLAMAMQ_VALU_PREFILL=1
+    to fall back to the VALU paged_attn path (A/B + debugging)."""
+    if os.environ.get("OLLAMAMQ_VALU_PREFILL") == "1":
+        return _attention(q, cache, layer, meta, 16)
+    T, Hq, D = q.shape
+    assert D == 128
+    out = torch.empty((T, Hq, D), dtype=q.dtype, device=q.device)
+    kp, vp = _layer_ptrs(cache, layer)
+    tile_slot, tile_q0, tile_pos0, tile_rows = _attn_plan(cache, meta,
+                                                          PREFILL_QT)
+    n_tiles = tile_slot.shape[0]
+    if n_tiles == 0:
+        return out
+    _check(_lib.prefill_attn_bf16(
+        _p(out), _p(q), kp, vp, _p(cache.page_table),
+        _p(tile_slot), _p(tile_q0), _p(tile_pos0), _p(tile_rows),
+        n_tiles, Hq, cache.n_kv_heads, cache.page_size,
+        cache.page_table.shape[1], 1.0 / (D ** 0.5), _row_stride(q, D),
+        _stream()), "prefill_attn")
+    return out
 
 
 SKINNY_MAX_N = int(os.environ.get("OLLAMAMQ_SKINNY_MAX_N", "4096"))
