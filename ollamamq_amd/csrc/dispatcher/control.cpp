@@ -168,6 +168,66 @@ ControlOutcome start_model_control(AppState& st, const ControlRequest& req) {
 }
 
 // ------------------------------------------------------------- admin API
+ControlOutcome admin_stats(AppState& st) {
+    Json out = Json::object();
+    out.set("uptime_s", Json::number(
+        st.started_ms ? (now_ms() - st.started_ms) / 1000.0 : 0));
+    int64_t processed = 0, dropped = 0, queued = 0, processing = 0;
+    Json users = Json::array();
+    {
+        std::lock_guard<std::mutex> g(st.queues_mu);
+        for (const auto& [name, us] : st.users) {
+            processed += us.processed;
+            dropped += us.dropped;
+            queued += (int64_t)us.queue.size();
+            processing += us.processing;
+            Json u = Json::object();
+            u.set("user", Json::string(name));
+            u.set("queued", Json::number((double)us.queue.size()));
+            u.set("processing", Json::number((double)us.processing));
+            u.set("processed", Json::number((double)us.processed));
+            u.set("dropped", Json::number((double)us.dropped));
+            users.arr.push_back(std::move(u));
+        }
+    }
+    out.set("processed", Json::number((double)processed));
+    out.set("dropped", Json::number((double)dropped));
+    out.set("queued", Json::number((double)queued));
+    out.set("processing", Json::number((double)processing));
+    out.set("users", std::move(users));
+    {
+        std::lock_guard<std::mutex> g(st.backends_mu);
+        Json bs = Json::array();
+        for (const auto& b : st.backends) {
+            Json e = Json::object();
+            e.set("url", Json::string(b.url));
+            e.set("online", Json::boolean(b.is_online));
+            e.set("active_requests", Json::number(b.active_requests));
+            e.set("processed_count", Json::number((double)b.processed_count));
+            bs.arr.push_back(std::move(e));
+        }
+        out.set("backends", std::move(bs));
+    }
+    {
+        std::lock_guard<std::mutex> g(st.waits_mu);
+        std::vector<int64_t> w(st.wait_samples_ms.begin(),
+                               st.wait_samples_ms.end());
+        std::sort(w.begin(), w.end());
+        Json qw = Json::object();
+        auto pct = [&](double p) -> double {
+            if (w.empty()) return 0;
+            size_t i = (size_t)(p * (w.size() - 1));
+            return (double)w[i];
+        };
+        qw.set("samples", Json::number((double)w.size()));
+        qw.set("p50_ms", Json::number(pct(0.50)));
+        qw.set("p90_ms", Json::number(pct(0.90)));
+        qw.set("p99_ms", Json::number(pct(0.99)));
+        out.set("queue_wait", std::move(qw));
+    }
+    return {200, std::move(out)};
+}
+
 ControlOutcome admin_models_state(AppState& st) {
     Json out = Json::object();
     Json arr = Json::array();

@@ -222,6 +222,13 @@ public:
 
     LogRing log;
 
+    // metrics surface beyond the reference (tokens/sec and
+    // queue-wait percentiles are the BASELINE headline): ring of recent
+    // queue waits (enqueue -> dispatch), sampled by the scheduler
+    mutable std::mutex waits_mu;
+    std::deque<int64_t> wait_samples_ms;   // cap 2048
+    int64_t started_ms = 0;
+
     // scheduler wakeups (reference notify + backend_freed)
     std::mutex wake_mu;
     std::condition_variable wake_cv;

@@ -173,6 +173,12 @@ bool schedule_once(AppState& st, Dispatch* out) {
             }
             const size_t bi =
                 pick_backend(st.backends, eligible, st.last_backend_idx);
+            {
+                std::lock_guard<std::mutex> wg(st.waits_mu);
+                st.wait_samples_ms.push_back(now - it->queued_at_ms);
+                while (st.wait_samples_ms.size() > 2048)
+                    st.wait_samples_ms.pop_front();
+            }
             st.last_backend_idx = bi;
             st.sched_counter.fetch_add(1);
             out->task = std::move(*it);

@@ -52,6 +52,7 @@ void Server::add_backend(std::shared_ptr<Backend> impl,
 int Server::port() const { return http_ ? http_->port() : 0; }
 
 bool Server::start(std::string* err) {
+    st_.started_ms = now_ms();
     st_.load_blocked();
     http_ = std::make_unique<HttpServer>(
         st_.settings.host, st_.settings.port,
@@ -112,6 +113,11 @@ void Server::handle(const HttpRequest& req, HttpConn& conn) {
                   {{"Content-Type", "application/json"},
                    {"WWW-Authenticate", "Bearer"}},
                   "{\"error\":\"unauthorized\"}");
+        return;
+    }
+    if (req.path == "/admin/stats" && req.method == "GET") {
+        auto out = admin_stats(st_);
+        conn.send(out.http_status, {}, out.body.dump());
         return;
     }
     if (req.path == "/admin/models" && req.method == "GET") {
