@@ -129,6 +129,36 @@ class LlamaEngine:
             self._finish(seq)
         return finished
 
+    def embed(self, prompt: List[int]):
+        """Synchronous embedding: final-norm hidden state of the last
+        prompt token (no generation, KV slot freed immediately)."""
+        if not prompt:
+            prompt = [0]
+        slot = self.kv.alloc_slot()
+        try:
+            self.kv.ensure(slot, len(prompt))
+            dev = self.dev
+            tokens = torch.tensor(prompt, dtype=torch.int32, device=dev)
+            positions = torch.arange(len(prompt), dtype=torch.int32,
+                                     device=dev)
+            slot_t = torch.full((len(prompt),), slot, dtype=torch.int32,
+                                device=dev)
+            meta = AttnMeta(
+                mode="prefill",
+                slot_ids=torch.tensor([slot], dtype=torch.int32, device=dev),
+                seq_lens=torch.tensor([len(prompt)], dtype=torch.int32,
+                                      device=dev),
+                cu_q=torch.tensor([0, len(prompt)], dtype=torch.int32,
+                                  device=dev),
+                logits_idx=torch.tensor([len(prompt) - 1], dtype=torch.long,
+                                        device=dev),
+                max_q=len(prompt), max_kv=len(prompt))
+            h = self.model.forward(tokens, positions, self.kv, slot_t, meta,
+                                   return_hidden=True)
+            return h[0].float().tolist()
+        finally:
+            self.kv.free_slot(slot)
+
     # -- internals ---------------------------------------------------------
     def _reap_cancelled(self):
         for lst in (self.waiting, self.running):

@@ -265,6 +265,8 @@ class Conn:
         if path in ("/api/tags", "/v1/models", "/api/ps", "/api/version",
                     "/", "/api/show"):
             return self._meta(path, body)
+        if path in ("/api/embed", "/api/embeddings", "/v1/embeddings"):
+            return self._embed(path, body)
 
         openai = path.startswith("/v1/")
         model_req = body.get("model") or ""
@@ -370,6 +372,38 @@ class Conn:
                 else:
                     obj["message"] = {"role": "assistant", "content": text}
             self.sock.sendall((json.dumps(obj) + "\n").encode())
+
+    def _embed(self, path, body):
+        w = self.worker
+        model_req = body.get("model") or ""
+        model = w.resolve(model_req) if model_req else \
+            next(iter(w.engines), "tiny" if w.device == "cpu"
+                 else "llama3-8b")
+        if model not in w.engines:
+            err = w.load(model)
+            if err:
+                self._line({"status": 404,
+                            "content_type": "application/json"})
+                self.sock.sendall(json.dumps({"error": err}).encode())
+                return
+        raw = body.get("input", body.get("prompt", ""))
+        inputs = raw if isinstance(raw, list) else [raw]
+        tok = w.tokenizers[model]
+        vecs = []
+        with w.lock:
+            eng = w.engines[model]
+            for text in inputs:
+                vecs.append(eng.embed(tok.encode(str(text))))
+        self._line({"status": 200, "content_type": "application/json"})
+        if path == "/v1/embeddings":
+            obj = {"object": "list", "model": model,
+                   "data": [{"object": "embedding", "index": i,
+                             "embedding": v} for i, v in enumerate(vecs)]}
+        elif path == "/api/embed":
+            obj = {"model": model, "embeddings": vecs}
+        else:  # legacy /api/embeddings
+            obj = {"model": model, "embedding": vecs[0] if vecs else []}
+        self.sock.sendall(json.dumps(obj).encode())
 
     def _meta(self, path, body):
         w = self.worker

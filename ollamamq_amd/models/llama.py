@@ -204,6 +204,7 @@ class LlamaModel:
         kv_cache,                  # engine.kvcache.PagedKVCache
         slot_ids: torch.Tensor,    # [T] int32 kv slot (sequence) of each token
         attn_meta,                 # ops.AttnMeta (prefill/decode metadata)
+        return_hidden: bool = False,
     ) -> torch.Tensor:
         """Returns logits [T_last, vocab_full] for the tokens attn_meta
         selects as "last" (decode: all; prefill: final token per seq)."""
@@ -240,6 +241,8 @@ class LlamaModel:
         else:
             h = x + residual
         h, _ = ops.rmsnorm_residual(h, None, self.final_norm, cfg.norm_eps)
+        if return_hidden:
+            return h
         logits = ops.linear(h, self.lm_head)
         if self.tp_size > 1:
             # vocab-parallel logits: all-gather shards on the last dim

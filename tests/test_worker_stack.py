@@ -170,3 +170,23 @@ def test_admin_load_unload_on_worker(stack):
             return
         time.sleep(0.3)
     assert False, "tiny did not unload"
+
+
+def test_embeddings_endpoints(stack):
+    r = httpx.post(stack + "/api/embed",
+                   json={"model": "tiny-cpu", "input": ["hello", "world"]},
+                   headers={"X-User-ID": "e1"}, timeout=120.0)
+    assert r.status_code == 200, r.text
+    obj = r.json()
+    assert len(obj["embeddings"]) == 2
+    assert len(obj["embeddings"][0]) == 256     # tiny-cpu hidden
+    r = httpx.post(stack + "/v1/embeddings",
+                   json={"model": "tiny-cpu", "input": "hello"},
+                   headers={"X-User-ID": "e2"}, timeout=120.0)
+    assert r.status_code == 200, r.text
+    assert r.json()["data"][0]["object"] == "embedding"
+    r = httpx.post(stack + "/api/embeddings",
+                   json={"model": "tiny-cpu", "prompt": "hello"},
+                   headers={"X-User-ID": "e3"}, timeout=120.0)
+    assert r.status_code == 200, r.text
+    assert isinstance(r.json()["embedding"], list)
