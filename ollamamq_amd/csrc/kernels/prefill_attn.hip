@@ -146,6 +146,10 @@ __global__ __launch_bounds__(512) void k_prefill_attn(
                 s[t][r] = ok ? s[t][r] * scale : -3.0e38f;
                 pmax = fmaxf(pmax, s[t][r]);
             }
+        // lanes l and l+32 each hold HALF of q-row (l&31)'s kv scores
+        // (the C layout's 4*(lane>>5) row offset): combine the pair's
+        // running max and sum so both halves share one softmax state
+        pmax = fmaxf(pmax, __shfl_xor(pmax, 32, 64));
         const float mn = fmaxf(m, pmax);
         float corr = 1.f, psum = 0.f;
         if (mn > -3.0e38f) {
@@ -159,6 +163,7 @@ __global__ __launch_bounds__(512) void k_prefill_attn(
                     s[t][r] = p;
                     psum += p;
                 }
+            psum += __shfl_xor(psum, 32, 64);
             m = mn;
         } else {
             #pragma unroll

@@ -30,9 +30,10 @@ typedef __hip_bfloat16 bf16;
 typedef __attribute__((ext_vector_type(8))) __bf16 bf16x8;
 typedef __attribute__((ext_vector_type(16))) float f32x16;
 
-#define KCH 256                  // staged k elems per chunk
+#define KCH 128                  // staged k elems per chunk
 #define BN 32                    // output columns per block
 #define LROW (KCH + 8)           // LDS row stride (bf16), +16 B pad
+#define NBUF 3                   // LDS ring depth (2-deep load prefetch)
 
 // grid (N/BN, ksplit); block ks owns k-range [ks*K/ksplit, ...).
 // ksplit>1 writes fp32 partials to part[ks][M][N]; k_skinny_combine sums.
@@ -52,11 +53,11 @@ __global__ __launch_bounds__(256) void k_skinny_gemm(
     const int khalf = (lane >> 5) * 8;
 
     extern __shared__ __attribute__((aligned(16))) char smem[];
-    bf16* wt = reinterpret_cast<bf16*>(smem);          // [2][BN][LROW]
-    float* red = reinterpret_cast<float*>(wt + 2 * BN * LROW);  // [4][32][32]
+    bf16* wt = reinterpret_cast<bf16*>(smem);      // [NBUF][BN][LROW]
+    float* red = reinterpret_cast<float*>(wt + NBUF * BN * LROW);
 
-    // staging geometry: 16 KiB chunk = BN rows × 512 B; thread t covers
-    // (row = t>>3, 16 B at (t&7)*16) × 4 rounds of 128 B — consecutive
+    // staging geometry: 8 KiB chunk = BN rows × 256 B; thread t covers
+    // (row = t>>3, 16 B at (t&7)*16) × 2 rounds of 128 B — consecutive
     // threads hit consecutive 16 B of one row: full-line coalescing.
     const int s_row = tid >> 3;
     const int s_off = (tid & 7) * 16;
@@ -64,7 +65,7 @@ __global__ __launch_bounds__(256) void k_skinny_gemm(
 
     const int kseg = K / ksplit;          // this block's k extent
     const int k0 = ks * kseg;
-    const int nch = kseg / KCH;
+    const int nch = kseg / KCH;           // hip.py guarantees even, >= 2
     // clamp the row so EVERY lane loads valid memory: a per-lane
     // conditional load makes hipcc branch around each load with a
     // vmcnt(0) (guide §5 traps (c): +11k cycles/block).  C rows >= M are
@@ -72,49 +73,76 @@ __global__ __launch_bounds__(256) void k_skinny_gemm(
     const int arow = row < M ? row : (M - 1);
     const bf16* xrow = x + (int64_t)arow * xs;
 
-    uint4 stage[4];
-    auto load_chunk = [&](int c) {
+    // 2-deep staging prefetch: two register sets (sA even chunks, sB odd)
+    // and an LDS ring of 3 — a chunk's loads stay in flight across a FULL
+    // loop iteration (the depth-1 version exposed ~the whole HBM latency
+    // per chunk: PMC showed 87% SQ_WAIT_ANY).
+    uint4 sA[2], sB[2];
+    auto load_to = [&](uint4* dst, int c) {
         const int64_t base = wbase + k0 + (int64_t)c * KCH;  // elems
         #pragma unroll
-        for (int r = 0; r < 4; r++)
-            stage[r] = *reinterpret_cast<const uint4*>(
+        for (int r = 0; r < 2; r++)
+            dst[r] = *reinterpret_cast<const uint4*>(
                 reinterpret_cast<const char*>(w + base) + r * 128 + s_off);
     };
-    auto write_chunk = [&](int buf) {
+    auto write_from = [&](const uint4* src, int buf) {
         bf16* dst = wt + buf * BN * LROW + s_row * LROW;
         #pragma unroll
-        for (int r = 0; r < 4; r++)
+        for (int r = 0; r < 2; r++)
             *reinterpret_cast<uint4*>(
-                reinterpret_cast<char*>(dst) + r * 128 + s_off) = stage[r];
+                reinterpret_cast<char*>(dst) + r * 128 + s_off) = src[r];
     };
 
     f32x16 acc = {};
-    load_chunk(0);
-    write_chunk(0);
+    const int cmax = nch - 1;
+    load_to(sA, 0);
+    write_from(sA, 0);
+    load_to(sB, 1 < cmax ? 1 : cmax);
     __syncthreads();
 
-    for (int c = 0; c < nch; c++) {
-        // ORDER MATTERS (guide §5 traps (b)): the A fragments are ordinary
-        // global loads — issue them BEFORE the next chunk's staging loads,
-        // or hipcc's wait at the first MFMA drains the staging prefetch.
-        const bf16* xk = xrow + k0 + (int64_t)c * KCH + wid * 64 + khalf;
-        bf16x8 a[4];
-        #pragma unroll
-        for (int s = 0; s < 4; s++)
-            a[s] = *reinterpret_cast<const bf16x8*>(xk + s * 16);
-        load_chunk(c + 1 < nch ? c + 1 : c);  // clamped: no branch on loads
-        // compute this chunk: wave w owns k in [w*64, (w+1)*64)
-        const bf16* wrow = wt + (c & 1) * BN * LROW + row * LROW
-                           + wid * 64 + khalf;
-        #pragma unroll
-        for (int s = 0; s < 4; s++) {
-            const bf16x8 b = *reinterpret_cast<const bf16x8*>(wrow + s * 16);
-            acc = __builtin_amdgcn_mfma_f32_32x32x16_bf16(a[s], b, acc,
+    for (int c = 0; c < nch; c += 2) {
+        // --- even chunk c: compute buf[c%3], stage chunk c+1 (sB) into
+        //     buf[(c+1)%3], refill sA with chunk c+2 ---
+        {
+            const bf16* xk = xrow + k0 + (int64_t)c * KCH + wid * 32 + khalf;
+            bf16x8 a[2];
+            a[0] = *reinterpret_cast<const bf16x8*>(xk);
+            a[1] = *reinterpret_cast<const bf16x8*>(xk + 16);
+            load_to(sA, c + 2 < cmax ? c + 2 : cmax);
+            const bf16* wrow = wt + (c % NBUF) * BN * LROW + row * LROW
+                               + wid * 32 + khalf;
+            const bf16x8 b0 = *reinterpret_cast<const bf16x8*>(wrow);
+            const bf16x8 b1 = *reinterpret_cast<const bf16x8*>(wrow + 16);
+            acc = __builtin_amdgcn_mfma_f32_32x32x16_bf16(a[0], b0, acc,
                                                           0, 0, 0);
+            acc = __builtin_amdgcn_mfma_f32_32x32x16_bf16(a[1], b1, acc,
+                                                          0, 0, 0);
+            __syncthreads();
+            write_from(sB, (c + 1) % NBUF);
+            __syncthreads();
         }
-        __syncthreads();
-        if (c + 1 < nch) write_chunk((c + 1) & 1);
-        __syncthreads();
+        // --- odd chunk c+1: compute buf[(c+1)%3], stage chunk c+2 (sA)
+        //     into buf[(c+2)%3], refill sB with chunk c+3 ---
+        if (c + 1 < nch) {
+            const int c1 = c + 1;
+            const bf16* xk = xrow + k0 + (int64_t)c1 * KCH + wid * 32
+                             + khalf;
+            bf16x8 a[2];
+            a[0] = *reinterpret_cast<const bf16x8*>(xk);
+            a[1] = *reinterpret_cast<const bf16x8*>(xk + 16);
+            load_to(sB, c1 + 2 < cmax ? c1 + 2 : cmax);
+            const bf16* wrow = wt + (c1 % NBUF) * BN * LROW + row * LROW
+                               + wid * 32 + khalf;
+            const bf16x8 b0 = *reinterpret_cast<const bf16x8*>(wrow);
+            const bf16x8 b1 = *reinterpret_cast<const bf16x8*>(wrow + 16);
+            acc = __builtin_amdgcn_mfma_f32_32x32x16_bf16(a[0], b0, acc,
+                                                          0, 0, 0);
+            acc = __builtin_amdgcn_mfma_f32_32x32x16_bf16(a[1], b1, acc,
+                                                          0, 0, 0);
+            __syncthreads();
+            write_from(sA, (c1 + 1) % NBUF);
+            __syncthreads();
+        }
     }
 
     // ---- reduce the 4 wave partials through LDS ----
@@ -157,7 +185,7 @@ extern "C" int skinny_gemm_bf16(
     void* y, void* part, const void* x, const void* w, int M, int N, int K,
     int64_t xs, int ksplit, hipStream_t stream)
 {
-    const int lds = 2 * BN * LROW * 2 + 4 * 32 * 32 * 4;
+    const int lds = NBUF * BN * LROW * 2 + 4 * 32 * 32 * 4;
     dim3 grid(N / BN, ksplit);
     k_skinny_gemm<<<grid, 256, lds, stream>>>(
         (bf16*)y, (float*)part, (const bf16*)x, (const bf16*)w, M, N, K,

@@ -113,15 +113,30 @@ class LlamaEngine:
 
     # -- stepping ----------------------------------------------------------
     def step(self) -> List[Sequence]:
-        """Advance the engine by one iteration; returns sequences finished."""
+        """Advance the engine by one iteration; returns sequences finished.
+
+        When both new prompts and running sequences exist, prefill and
+        decode steps alternate so a long prompt cannot starve the decode
+        batch (inter-token latency stays bounded by ~one prefill chunk).
+        """
         self._reap_cancelled()
-        admitted = self._admit()
+        admitted = []
+        do_prefill = bool(self.waiting) and (not self.running
+                                             or self.steps % 2 == 0)
+        if do_prefill:
+            admitted = self._admit()
         if admitted:
             self._prefill_step(admitted)
             finished = self._postprocess(admitted)
         elif self.running:
             self._decode_step()
             finished = self._postprocess(self.running)
+        elif self.waiting:
+            admitted = self._admit()
+            if not admitted:
+                return []   # waiting but nothing admissible (kv full)
+            self._prefill_step(admitted)
+            finished = self._postprocess(admitted)
         else:
             return []
         self.steps += 1

@@ -267,6 +267,14 @@ class Conn:
             return self._meta(path, body)
         if path in ("/api/embed", "/api/embeddings", "/v1/embeddings"):
             return self._embed(path, body)
+        if path in ("/api/create", "/api/copy", "/api/delete", "/api/pull",
+                    "/api/push") or path.startswith("/api/blobs"):
+            self._line({"status": 501, "content_type": "application/json"})
+            self.sock.sendall(json.dumps(
+                {"error": f"{path} is not supported by the in-process GPU "
+                          "worker (models are resident presets; use "
+                          "/admin/models/load)"}).encode())
+            return
 
         openai = path.startswith("/v1/")
         model_req = body.get("model") or ""

@@ -265,10 +265,12 @@ _gemm_scratch = {}
 
 
 def _skinny_ksplit(N, K):
-    """grid k-split so small-N shapes still fill the 256-CU chip."""
+    """grid k-split: target >=768 blocks (~3/CU) so independent blocks
+    hide each other's staging latency; kseg must stay a multiple of 256
+    (even chunk count for the 2-deep pipeline)."""
     blocks = N // 32
     ks = 1
-    while blocks * ks < 256 and ks < 8 and K % (256 * 2 * ks) == 0:
+    while blocks * ks < 768 and ks < 8 and K % (256 * 2 * ks) == 0:
         ks *= 2
     return ks
 
