@@ -177,6 +177,7 @@ struct Settings {
     bool allow_all_routes = false;
     int64_t stuck_timeout_s = 60;        // reference declares but never
                                          // enforces; we DO enforce (503)
+    int probe_interval_ms = 10000;       // health loop cadence (ref: 10 s)
     std::string api_key;                 // empty = auth off
 };
 

@@ -43,6 +43,7 @@ int main(int argc, char** argv) {
     std::string cli_host, cli_backends, cli_workers;
     int cli_port = -1;
     int64_t cli_timeout = -1, cli_keep = -1, cli_stuck = -1;
+    int cli_probe = -1;
     bool cli_allow_all = false, no_tui = false;
 
     for (int i = 1; i < argc; i++) {
@@ -58,6 +59,7 @@ int main(int argc, char** argv) {
         else if (a == "-t" || a == "--timeout") cli_timeout = atoll(next());
         else if (a == "--load-keep-alive") cli_keep = atoll(next());
         else if (a == "--stuck-timeout") cli_stuck = atoll(next());
+        else if (a == "--probe-interval-ms") cli_probe = atoi(next());
         else if (a == "--allow-all-routes") cli_allow_all = true;
         else if (a == "--no-tui") no_tui = true;
         else if (a == "-c" || a == "--model-config") config_path = next();
@@ -85,6 +87,7 @@ int main(int argc, char** argv) {
     if (cli_timeout >= 0) st.settings.timeout_s = cli_timeout;
     if (cli_keep >= 0) st.settings.load_keep_alive_s = cli_keep;
     if (cli_stuck >= 0) st.settings.stuck_timeout_s = cli_stuck;
+    if (cli_probe > 0) st.settings.probe_interval_ms = cli_probe;
     if (cli_allow_all) st.settings.allow_all_routes = true;
     if (const char* k = getenv("OLLAMA_MQ_API_KEY")) st.settings.api_key = k;
     {

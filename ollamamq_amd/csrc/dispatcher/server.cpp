@@ -258,10 +258,13 @@ void Server::run_worker() {
 void Server::health_loop() {
     int cycle = 0;
     while (!st_.shutting_down) {
-        probe_all(st_, cycle % 6 == 0);
+        probe_all(st_, cycle % 6 == 0);   // full reprobe every 6th (~60 s)
         cycle++;
-        for (int i = 0; i < 100 && !st_.shutting_down; i++)
-            std::this_thread::sleep_for(std::chrono::milliseconds(100));
+        const int step = 50;
+        for (int waited = 0;
+             waited < st_.settings.probe_interval_ms && !st_.shutting_down;
+             waited += step)
+            std::this_thread::sleep_for(std::chrono::milliseconds(step));
     }
 }
 

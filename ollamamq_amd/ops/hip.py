@@ -258,8 +258,10 @@ def attention_prefill(q, cache, layer, meta):
     return out
 
 
-SKINNY_MAX_N = int(os.environ.get("OLLAMAMQ_SKINNY_MAX_N", "4096"))
-SKINNY_MAX_K = int(os.environ.get("OLLAMAMQ_SKINNY_MAX_K", "4096"))
+# default 0: hipBLASLt currently wins every decode shape (perf_gemm.py);
+# re-enable by env once the hand-written kernel beats it per shape
+SKINNY_MAX_N = int(os.environ.get("OLLAMAMQ_SKINNY_MAX_N", "0"))
+SKINNY_MAX_K = int(os.environ.get("OLLAMAMQ_SKINNY_MAX_K", "0"))
 
 _gemm_scratch = {}
 
