@@ -199,3 +199,101 @@ extern "C" int skinny_gemm_bf16(
     }
     return (int)hipGetLastError();
 }
+
+// ---------------------------------------------------------------------
+// Fragment-direct variant: no LDS staging, no barriers — every wave
+// streams its own (32-col tile, k-slice) with a 4-deep register
+// prefetch ring, so latency is hidden by wave count alone (grid k-split
+// pushes occupancy to ~16+ waves/CU).  B loads are 16 B/lane at 8 KB row
+// stride; rows are re-touched every k-step so L1/L2 serve the interior
+// of each 128 B line.
+// grid (N/BN, ksplit); block = 8 waves, wave w takes k-slice w of 8.
+__global__ __launch_bounds__(512) void k_skinny_direct(
+    bf16* __restrict__ y, float* __restrict__ part,
+    const bf16* __restrict__ x, const bf16* __restrict__ w,
+    int M, int N, int K, int64_t xs, int ksplit)
+{
+    const int n0 = blockIdx.x * BN;
+    const int ks = blockIdx.y;
+    const int tid = threadIdx.x;
+    const int wid = tid >> 6;
+    const int lane = tid & 63;
+    const int row = lane & 31;
+    const int khalf = (lane >> 5) * 8;
+
+    const int kseg = K / (ksplit * 8);       // per-wave k extent
+    const int k0 = (ks * 8 + wid) * kseg;
+    const int k1 = k0 + kseg;                // kseg % 64 == 0 (hip.py)
+
+    const int arow = row < M ? row : (M - 1);
+    const bf16* xrow = x + (int64_t)arow * xs + khalf;
+    const bf16* wrow = w + (int64_t)(n0 + row) * K + khalf;
+
+    f32x16 acc = {};
+    bf16x8 a0, a1, a2, a3, b0, b1, b2, b3;
+    a0 = *reinterpret_cast<const bf16x8*>(xrow + k0);
+    b0 = *reinterpret_cast<const bf16x8*>(wrow + k0);
+    a1 = *reinterpret_cast<const bf16x8*>(xrow + k0 + 16);
+    b1 = *reinterpret_cast<const bf16x8*>(wrow + k0 + 16);
+    a2 = *reinterpret_cast<const bf16x8*>(xrow + k0 + 32);
+    b2 = *reinterpret_cast<const bf16x8*>(wrow + k0 + 32);
+    a3 = *reinterpret_cast<const bf16x8*>(xrow + k0 + 48);
+    b3 = *reinterpret_cast<const bf16x8*>(wrow + k0 + 48);
+    for (int k = k0; k < k1 - 64; k += 64) {
+        acc = __builtin_amdgcn_mfma_f32_32x32x16_bf16(a0, b0, acc, 0, 0, 0);
+        a0 = *reinterpret_cast<const bf16x8*>(xrow + k + 64);
+        b0 = *reinterpret_cast<const bf16x8*>(wrow + k + 64);
+        acc = __builtin_amdgcn_mfma_f32_32x32x16_bf16(a1, b1, acc, 0, 0, 0);
+        a1 = *reinterpret_cast<const bf16x8*>(xrow + k + 80);
+        b1 = *reinterpret_cast<const bf16x8*>(wrow + k + 80);
+        acc = __builtin_amdgcn_mfma_f32_32x32x16_bf16(a2, b2, acc, 0, 0, 0);
+        a2 = *reinterpret_cast<const bf16x8*>(xrow + k + 96);
+        b2 = *reinterpret_cast<const bf16x8*>(wrow + k + 96);
+        acc = __builtin_amdgcn_mfma_f32_32x32x16_bf16(a3, b3, acc, 0, 0, 0);
+        a3 = *reinterpret_cast<const bf16x8*>(xrow + k + 112);
+        b3 = *reinterpret_cast<const bf16x8*>(wrow + k + 112);
+    }
+    acc = __builtin_amdgcn_mfma_f32_32x32x16_bf16(a0, b0, acc, 0, 0, 0);
+    acc = __builtin_amdgcn_mfma_f32_32x32x16_bf16(a1, b1, acc, 0, 0, 0);
+    acc = __builtin_amdgcn_mfma_f32_32x32x16_bf16(a2, b2, acc, 0, 0, 0);
+    acc = __builtin_amdgcn_mfma_f32_32x32x16_bf16(a3, b3, acc, 0, 0, 0);
+
+    __shared__ float red8[8][32][32];
+    #pragma unroll
+    for (int r = 0; r < 16; r++) {
+        const int crow = (r & 3) + 8 * (r >> 2) + 4 * (lane >> 5);
+        red8[wid][crow][lane & 31] = acc[r];
+    }
+    __syncthreads();
+    #pragma unroll
+    for (int e = tid; e < 1024; e += 512) {
+        const int m = e >> 5, n = e & 31;
+        if (m < M) {
+            float s = 0.f;
+            #pragma unroll
+            for (int wv = 0; wv < 8; wv++) s += red8[wv][m][n];
+            if (ksplit == 1)
+                y[(int64_t)m * N + n0 + n] = __float2bfloat16(s);
+            else
+                part[((int64_t)ks * M + m) * N + n0 + n] = s;
+        }
+    }
+}
+
+extern "C" int skinny_direct_bf16(
+    void* y, void* part, const void* x, const void* w, int M, int N, int K,
+    int64_t xs, int ksplit, hipStream_t stream)
+{
+    dim3 grid(N / BN, ksplit);
+    k_skinny_direct<<<grid, 512, 0, stream>>>(
+        (bf16*)y, (float*)part, (const bf16*)x, (const bf16*)w, M, N, K,
+        xs, ksplit);
+    if (ksplit > 1) {
+        const int64_t mn = (int64_t)M * N;
+        const int64_t want = (mn + 255) / 256;
+        const int blocks = (int)(want < 1024 ? want : 1024);
+        k_skinny_combine<<<blocks, 256, 0, stream>>>(
+            (bf16*)y, (const float*)part, mn, mn, ksplit);
+    }
+    return (int)hipGetLastError();
+}

@@ -46,12 +46,13 @@ def _try_load():
     lib.decode_attn_bf16.argtypes = [vp, vp, vp, vp, vp, vp, vp, vp, vp,
                                      i, i, i, i, i, f, i64, i, vp]
     lib.skinny_gemm_bf16.argtypes = [vp, vp, vp, vp, i, i, i, i64, i, vp]
+    lib.skinny_direct_bf16.argtypes = [vp, vp, vp, vp, i, i, i, i64, i, vp]
     lib.prefill_attn_bf16.argtypes = [vp, vp, vp, vp, vp, vp, vp, vp, vp,
                                       i, i, i, i, i, f, i64, vp]
     for fn in ("rmsnorm_residual_bf16", "rope_bf16", "kv_append_bf16",
                "paged_attn_bf16", "swiglu_bf16", "argmax_bf16",
                "decode_attn_bf16", "skinny_gemm_bf16",
-               "prefill_attn_bf16"):
+               "skinny_direct_bf16", "prefill_attn_bf16"):
         getattr(lib, fn).restype = ctypes.c_int
     _lib = lib
 
@@ -287,7 +288,17 @@ def linear(x, weight):
             and x.dtype == torch.bfloat16 and weight.stride(1) == 1:
         y = torch.empty((M, N), dtype=x.dtype, device=x.device)
         assert x.stride(1) == 1
-        ks = _skinny_ksplit(N, K)
+        direct = os.environ.get("OLLAMAMQ_SKINNY_STAGED") != "1"
+        if direct:
+            # per-wave k extent must be a multiple of 64
+            ks = 1
+            blocks = N // 32
+            while blocks * ks < 768 and ks < 8 and K % (ks * 2 * 512) == 0:
+                ks *= 2
+            if K % (ks * 512) != 0:
+                return torch.nn.functional.linear(x, weight)
+        else:
+            ks = _skinny_ksplit(N, K)
         part = ctypes.c_void_p(0)
         if ks > 1:
             key = (M, N, ks, str(x.device))
@@ -297,9 +308,9 @@ def linear(x, weight):
                                 device=x.device)
                 _gemm_scratch[key] = t
             part = _p(t)
-        _check(_lib.skinny_gemm_bf16(_p(y), part, _p(x), _p(weight),
-                                     M, N, K, x.stride(0), ks, _stream()),
-               "skinny_gemm")
+        fn = _lib.skinny_direct_bf16 if direct else _lib.skinny_gemm_bf16
+        _check(fn(_p(y), part, _p(x), _p(weight),
+                  M, N, K, x.stride(0), ks, _stream()), "skinny_gemm")
         return y
     return torch.nn.functional.linear(x, weight)
 
