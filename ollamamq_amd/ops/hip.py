@@ -259,10 +259,11 @@ def attention_prefill(q, cache, layer, meta):
     return out
 
 
-# default 0: hipBLASLt currently wins every decode shape (perf_gemm.py);
-# re-enable by env once the hand-written kernel beats it per shape
-SKINNY_MAX_N = int(os.environ.get("OLLAMAMQ_SKINNY_MAX_N", "0"))
-SKINNY_MAX_K = int(os.environ.get("OLLAMAMQ_SKINNY_MAX_K", "0"))
+# defaults pick the shapes where the fragment-direct kernel beats
+# hipBLASLt (measured, tools/perf_gemm.py: qkv 2.9 vs 2.7 TB/s, o-proj
+# 2.4 vs 1.8); the large-N/large-K shapes stay on the library
+SKINNY_MAX_N = int(os.environ.get("OLLAMAMQ_SKINNY_MAX_N", "6144"))
+SKINNY_MAX_K = int(os.environ.get("OLLAMAMQ_SKINNY_MAX_K", "4096"))
 
 _gemm_scratch = {}
 
