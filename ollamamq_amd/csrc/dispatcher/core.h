@@ -42,11 +42,17 @@ public:
     explicit LogRing(size_t cap = 300) : cap_(cap) {}
     void push(const std::string& kind, const std::string& text);
     std::vector<LogEvent> snapshot() const;
+    // L8 parity (reference src/main.rs:199-218): TUI mode appends to
+    // ./ollamamq.log; headless logs to stderr.
+    void set_file_sink(const std::string& path);
+    void set_stderr_sink(bool on) { stderr_sink_ = on; }
 
 private:
     mutable std::mutex mu_;
     size_t cap_;
     std::deque<LogEvent> ring_;
+    std::string file_path_;
+    bool stderr_sink_ = false;
 };
 
 // ------------------------------------------------------------------ request

@@ -113,6 +113,11 @@ int main(int argc, char** argv) {
             add_worker_backend(server, w);
     }
 
+    if (no_tui)
+        st.log.set_stderr_sink(true);
+    else
+        st.log.set_file_sink("ollamamq.log");
+
     if (!server.start(&err)) {
         std::cerr << "failed to start: " << err << "\n";
         return 1;

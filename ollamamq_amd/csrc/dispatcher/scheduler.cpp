@@ -216,10 +216,25 @@ void finish_dispatch(AppState& st, const Dispatch& d, bool ok,
 }
 
 // ------------------------------------------------------------- log ring
+void LogRing::set_file_sink(const std::string& path) {
+    std::lock_guard<std::mutex> g(mu_);
+    file_path_ = path;
+}
+
 void LogRing::push(const std::string& kind, const std::string& text) {
     std::lock_guard<std::mutex> g(mu_);
     ring_.push_back({now_ms(), kind, text});
     while (ring_.size() > cap_) ring_.pop_front();
+    if (!file_path_.empty()) {
+        FILE* f = fopen(file_path_.c_str(), "a");
+        if (f) {
+            fprintf(f, "%lld %s %s\n", (long long)now_ms(), kind.c_str(),
+                    text.c_str());
+            fclose(f);
+        }
+    }
+    if (stderr_sink_)
+        fprintf(stderr, "[%s] %s\n", kind.c_str(), text.c_str());
 }
 
 std::vector<LogEvent> LogRing::snapshot() const {
