@@ -259,11 +259,13 @@ def attention_prefill(q, cache, layer, meta):
     return out
 
 
-# defaults pick the shapes where the fragment-direct kernel beats
-# hipBLASLt (measured, tools/perf_gemm.py: qkv 2.9 vs 2.7 TB/s, o-proj
-# 2.4 vs 1.8); the large-N/large-K shapes stay on the library
-SKINNY_MAX_N = int(os.environ.get("OLLAMAMQ_SKINNY_MAX_N", "6144"))
-SKINNY_MAX_K = int(os.environ.get("OLLAMAMQ_SKINNY_MAX_K", "4096"))
+# default 0 = library GEMMs everywhere: the fragment-direct kernel beats
+# hipBLASLt on qkv/o in MICROBENCHES (L3-warm weights re-read 100x) but
+# loses ~0.3 ms/step in the real serving loop where weights stream cold
+# from HBM — honest A/B in bench.py decided this (5473 vs 5221 tok/s).
+# The hand-written variants stay available via these env gates.
+SKINNY_MAX_N = int(os.environ.get("OLLAMAMQ_SKINNY_MAX_N", "0"))
+SKINNY_MAX_K = int(os.environ.get("OLLAMAMQ_SKINNY_MAX_K", "0"))
 
 _gemm_scratch = {}
 
