@@ -124,10 +124,10 @@ __global__ __launch_bounds__(512) void k_decode_attn(
         // ---- score for key j = lane ----
         float s = 0.f;
         const bool live = lane < n_here;
-        if (live) {
+        {
             const bf16* krow = k_tile + lane * KROW;
             const float* qrow = q_lds + wid * DHEAD;
-            #pragma unroll 4
+            #pragma unroll
             for (int d = 0; d < DHEAD; d += 8) {
                 float k8[8];
                 load8f_lds(krow + d, k8);
@@ -136,9 +136,7 @@ __global__ __launch_bounds__(512) void k_decode_attn(
                    + k8[4] * qrow[d + 4] + k8[5] * qrow[d + 5]
                    + k8[6] * qrow[d + 6] + k8[7] * qrow[d + 7];
             }
-            s *= scale;
-        } else {
-            s = -3.0e38f;
+            s = live ? s * scale : -3.0e38f;
         }
 
         // ---- online softmax ----
@@ -155,8 +153,12 @@ __global__ __launch_bounds__(512) void k_decode_attn(
         p_lds[wid * (DCHUNK + 1) + lane] = p;
 
         // ---- PV: lane accumulates dims (2*lane, 2*lane+1) ----
+        // fixed bound + full unroll: a runtime bound left this loop a
+        // serial ds_read latency chain (PMC: 25% SQ_WAIT_INST_ANY);
+        // dead keys contribute p=0 so looping to DCHUNK is exact
         const float* prow = p_lds + wid * (DCHUNK + 1);
-        for (int j = 0; j < n_here; j++) {
+        #pragma unroll 8
+        for (int j = 0; j < DCHUNK; j++) {
             const bf162 v2 = *reinterpret_cast<const bf162*>(
                 v_tile + j * KROW + 2 * lane);
             const float pj = prow[j];
