@@ -6,6 +6,7 @@
 #include <pybind11/pybind11.h>
 #include <pybind11/stl.h>
 
+#include "config.h"
 #include "core.h"
 #include "json.h"
 #include "matching.h"
@@ -144,6 +145,37 @@ PYBIND11_MODULE(_dispatch, m) {
           py::arg("native_display") = std::map<std::string, std::string>{});
 
     m.def("candidate_order", &candidate_order);
+
+    m.def("load_config", [](const std::string& path) -> py::object {
+        AppConfig cfg;
+        std::string err;
+        if (!load_config(path, &cfg, &err))
+            throw std::runtime_error(err);
+        py::dict d;
+        d["backends"] = cfg.backends;
+        py::dict st;
+        st["port"] = cfg.settings.port;
+        st["host"] = cfg.settings.host;
+        st["timeout"] = cfg.settings.timeout_s;
+        st["load_keep_alive"] = cfg.settings.load_keep_alive_s;
+        st["allow_all_routes"] = cfg.settings.allow_all_routes;
+        st["stuck_timeout"] = cfg.settings.stuck_timeout_s;
+        d["settings"] = st;
+        py::list models;
+        for (const auto& mdl : cfg.models) {
+            py::dict e;
+            e["name"] = mdl.name;
+            e["identifier"] = mdl.identifier;
+            e["max_ctx"] = mdl.max_ctx;
+            e["keep_alive"] = mdl.keep_alive;
+            e["max_concurrent_requests"] = mdl.max_concurrent_requests;
+            e["backends"] = mdl.backends;
+            models.append(e);
+        }
+        d["models"] = models;
+        return d;
+    });
+    m.def("normalize_backend_url", &normalize_backend_url);
 
     m.def("json_roundtrip", [](const std::string& s) -> py::object {
         auto j = Json::parse(s);

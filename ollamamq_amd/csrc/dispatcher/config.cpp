@@ -49,6 +49,7 @@ bool load_config(const std::string& path, AppConfig* out, std::string* err) {
     std::string section;            // backends | settings | models
     ModelConfigEntry* cur_model = nullptr;
     bool in_model_backends = false;
+    size_t model_indent = SIZE_MAX;  // indent of the "- name:" model items
 
     auto parse_i64 = [](const std::string& v, int64_t dflt) {
         try {
@@ -95,8 +96,10 @@ bool load_config(const std::string& path, AppConfig* out, std::string* err) {
             continue;
         }
         if (section == "models") {
-            if (t.rfind("- ", 0) == 0 && !in_model_backends &&
-                t.find(':') != std::string::npos && indent <= 2) {
+            const bool dash = t.rfind("- ", 0) == 0;
+            if (dash && model_indent == SIZE_MAX) model_indent = indent;
+            if (dash && indent == model_indent &&
+                t.find(':') != std::string::npos) {
                 // new model entry: "- name: x" or "- key: v"
                 out->models.emplace_back();
                 cur_model = &out->models.back();
