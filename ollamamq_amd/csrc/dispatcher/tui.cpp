@@ -25,6 +25,8 @@ namespace omq {
 namespace {
 
 struct Snapshot {
+    int64_t processed_total = 0, dropped_total = 0, queued_total = 0;
+    double p50_wait_ms = 0;
     struct B {
         std::string url;
         bool online;
@@ -74,6 +76,9 @@ Snapshot capture(AppState& st) {
     {
         std::scoped_lock lk(st.queues_mu, st.prio_mu);
         for (const auto& [name, us] : st.users) {
+            s.processed_total += us.processed;
+            s.dropped_total += us.dropped;
+            s.queued_total += (int64_t)us.queue.size();
             Snapshot::U u;
             u.name = name;
             u.queued = us.queue.size();
@@ -92,6 +97,15 @@ Snapshot capture(AppState& st) {
     }
     s.logs = st.log.snapshot();
     s.counter = st.sched_counter.load();
+    {
+        std::lock_guard<std::mutex> g(st.waits_mu);
+        std::vector<int64_t> w(st.wait_samples_ms.begin(),
+                               st.wait_samples_ms.end());
+        if (!w.empty()) {
+            std::sort(w.begin(), w.end());
+            s.p50_wait_ms = (double)w[w.size() / 2];
+        }
+    }
     return s;
 }
 
@@ -131,6 +145,7 @@ void run_tui(Server& server) {
     int sel_user = 0;
     std::string input;        // typed model name for L/U
     char input_mode = 0;      // 'L' or 'U' when typing
+    bool help = false;
     std::string status_msg;
 
     while (true) {
@@ -140,7 +155,33 @@ void run_tui(Server& server) {
         out << "\x1b[1m ollamamq-amd dispatcher — MI355X \x1b[0m"
             << "  backends:" << s.backends.size()
             << "  users:" << s.users.size()
+            << "  queued:" << s.queued_total
+            << "  done:" << s.processed_total
+            << "  dropped:" << s.dropped_total
+            << "  p50-wait:" << (int)s.p50_wait_ms << "ms"
             << "  sched:" << s.counter << "\r\n";
+        if (help) {
+            out << "\r\n \x1b[1mKeys\x1b[0m\r\n"
+                   "  j/k     select user\r\n"
+                   "  p / b   toggle VIP / Boost for selected user\r\n"
+                   "  x / X   block selected user / their IP\r\n"
+                   "  u       unblock all\r\n"
+                   "  L / U   load / unload model (type name, Enter)\r\n"
+                   "  r       reload models from appconf.yaml\r\n"
+                   "  ?       close help\r\n"
+                   "  q/Esc   quit\r\n";
+            fputs(out.str().c_str(), stdout);
+            fflush(stdout);
+            pollfd pfd{STDIN_FILENO, POLLIN, 0};
+            if (poll(&pfd, 1, 100) > 0) {
+                char c;
+                if (read(STDIN_FILENO, &c, 1) == 1) {
+                    if (c == 'q' || c == 27) return;
+                    help = false;
+                }
+            }
+            continue;
+        }
         out << "\x1b[7m" << pad(" Backends", 90) << "\x1b[0m\r\n";
         for (const auto& b : s.backends) {
             out << (b.online ? " \x1b[32m●\x1b[0m " : " \x1b[31m○\x1b[0m ")
@@ -278,6 +319,9 @@ void run_tui(Server& server) {
                         }
                         break;
                     }
+                    case '?':
+                        help = true;
+                        break;
                     case 'u': {
                         {
                             std::lock_guard<std::mutex> g(st.blocked_mu);
