@@ -510,3 +510,78 @@ extern "C" int argmax_bf16(void* out, const void* logits, int B, int V,
     k_argmax<<<B, 256, 0, stream>>>((int*)out, (const bf16*)logits, V);
     return (int)hipGetLastError();
 }
+
+// ---------------------------------------------------------------------
+// Fused RoPE + paged-KV append: one pass over the fresh q/k/v heads.
+// q heads: rotate in place.  k heads: rotate, then write BOTH the rotated
+// value in place and into the pool (k read once instead of twice).
+// v heads: copy to the pool.  One launch replaces rope_bf16 + kv_append.
+__global__ __launch_bounds__(256) void k_rope_append(
+    bf16* __restrict__ q, bf16* __restrict__ k, const bf16* __restrict__ v,
+    bf16* __restrict__ kp, bf16* __restrict__ vp,
+    const int* __restrict__ pos, const int* __restrict__ slot,
+    const int* __restrict__ page_table,
+    const float* __restrict__ cost, const float* __restrict__ sint,
+    int T, int Hq, int KVH, int D, int page, int max_pages,
+    int64_t qs, int64_t ks, int64_t vs)
+{
+    const int gid = blockIdx.x * 4 + (threadIdx.x >> 6);
+    const int lane = threadIdx.x & 63;
+    const int Htot = Hq + 2 * KVH;
+    if (gid >= T * Htot) return;
+    const int t = gid / Htot, h = gid % Htot;
+    const int d2 = D / 2;
+    const int p = pos[t];
+
+    if (h < Hq) {                        // q head: rope in place
+        bf16* base = q + (int64_t)t * qs + (int64_t)h * D;
+        const float c = cost[(int64_t)p * d2 + lane];
+        const float s = sint[(int64_t)p * d2 + lane];
+        const float x1 = bf2f(base[lane]);
+        const float x2 = bf2f(base[lane + d2]);
+        base[lane] = f2bf(x1 * c - x2 * s);
+        base[lane + d2] = f2bf(x2 * c + x1 * s);
+        return;
+    }
+    const int pg = page_table[(int64_t)slot[t] * max_pages + p / page];
+    const int off = p % page;
+    if (h < Hq + KVH) {                  // k head: rope + pool write
+        const int kh = h - Hq;
+        bf16* base = k + (int64_t)t * ks + (int64_t)kh * D;
+        bf16* dst = kp + (((int64_t)pg * KVH + kh) * page + off) * D;
+        const float c = cost[(int64_t)p * d2 + lane];
+        const float s = sint[(int64_t)p * d2 + lane];
+        const float x1 = bf2f(base[lane]);
+        const float x2 = bf2f(base[lane + d2]);
+        const bf16 lo = f2bf(x1 * c - x2 * s);
+        const bf16 hi = f2bf(x2 * c + x1 * s);
+        base[lane] = lo;
+        base[lane + d2] = hi;
+        dst[lane] = lo;
+        dst[lane + d2] = hi;
+        return;
+    }
+    {                                    // v head: pool copy
+        const int vh = h - Hq - KVH;
+        const bf162* src = reinterpret_cast<const bf162*>(
+            v + (int64_t)t * vs + (int64_t)vh * D);
+        bf162* dst = reinterpret_cast<bf162*>(
+            vp + (((int64_t)pg * KVH + vh) * page + off) * D);
+        dst[lane] = src[lane];
+    }
+}
+
+extern "C" int rope_append_bf16(
+    void* q, void* k, const void* v, void* kp, void* vp, const void* pos,
+    const void* slot, const void* page_table, const void* cost,
+    const void* sint, int T, int Hq, int KVH, int D, int page,
+    int max_pages, int64_t qs, int64_t ks, int64_t vs, hipStream_t stream)
+{
+    const int waves = T * (Hq + 2 * KVH);
+    k_rope_append<<<(waves + 3) / 4, 256, 0, stream>>>(
+        (bf16*)q, (bf16*)k, (const bf16*)v, (bf16*)kp, (bf16*)vp,
+        (const int*)pos, (const int*)slot, (const int*)page_table,
+        (const float*)cost, (const float*)sint,
+        T, Hq, KVH, D, page, max_pages, qs, ks, vs);
+    return (int)hipGetLastError();
+}

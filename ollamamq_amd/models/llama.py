@@ -221,8 +221,8 @@ class LlamaModel:
             q = q.view(-1, nl, d)
             k = k.view(-1, nkl, d)
             v = v.view(-1, nkl, d)
-            ops.rope(q, k, positions, self.rope_cos, self.rope_sin)
-            ops.kv_append(kv_cache, li, k, v, slot_ids, positions)
+            ops.rope_append(kv_cache, li, q, k, v, positions, slot_ids,
+                            self.rope_cos, self.rope_sin)
             attn = ops.attention(q, kv_cache, li, attn_meta)
             x = ops.linear(attn.view(-1, nl * d), layer.wo)
             self._allreduce(x)  # RCCL all-reduce #1 (TP)

@@ -47,12 +47,15 @@ def _try_load():
                                      i, i, i, i, i, f, i64, i, vp]
     lib.skinny_gemm_bf16.argtypes = [vp, vp, vp, vp, i, i, i, i64, i, vp]
     lib.skinny_direct_bf16.argtypes = [vp, vp, vp, vp, i, i, i, i64, i, vp]
+    lib.rope_append_bf16.argtypes = [vp, vp, vp, vp, vp, vp, vp, vp, vp, vp,
+                                     i, i, i, i, i, i, i64, i64, i64, vp]
     lib.prefill_attn_bf16.argtypes = [vp, vp, vp, vp, vp, vp, vp, vp, vp,
                                       i, i, i, i, i, f, i64, vp]
     for fn in ("rmsnorm_residual_bf16", "rope_bf16", "kv_append_bf16",
                "paged_attn_bf16", "swiglu_bf16", "argmax_bf16",
                "decode_attn_bf16", "skinny_gemm_bf16",
-               "skinny_direct_bf16", "prefill_attn_bf16"):
+               "skinny_direct_bf16", "prefill_attn_bf16",
+               "rope_append_bf16"):
         getattr(lib, fn).restype = ctypes.c_int
     _lib = lib
 
@@ -123,6 +126,22 @@ def kv_append(cache, layer, k, v, slot_ids, positions):
                                _p(cache.page_table), T, KVH, D,
                                cache.page_size, cache.page_table.shape[1],
                                _row_stride(k, D), _stream()), "kv_append")
+
+
+def rope_append(cache, layer, q, k, v, positions, slot_ids, cos, sin):
+    """Fused RoPE (q,k in place) + paged append of rotated k and v."""
+    T, Hq, D = q.shape
+    KVH = k.shape[1]
+    assert D == 128
+    kp, vp = _layer_ptrs(cache, layer)
+    pos32 = positions if positions.dtype == torch.int32 else positions.int()
+    slot32 = slot_ids if slot_ids.dtype == torch.int32 else slot_ids.int()
+    _check(_lib.rope_append_bf16(
+        _p(q), _p(k), _p(v), kp, vp, _p(pos32), _p(slot32),
+        _p(cache.page_table), _p(cos), _p(sin), T, Hq, KVH, D,
+        cache.page_size, cache.page_table.shape[1],
+        _row_stride(q, D), _row_stride(k, D), _row_stride(v, D),
+        _stream()), "rope_append")
 
 
 def _layer_ptrs(cache, layer):

@@ -65,6 +65,17 @@ def rope(q, k, positions, cos, sin):
     return ref.rope(q, k, positions, cos, sin)
 
 
+def rope_append(cache, layer, q, k, v, positions, slot_ids, cos, sin):
+    """Fused RoPE + KV append (one pass); falls back to the two reference
+    ops on CPU."""
+    if _use_hip(q):
+        from . import hip
+        return hip.rope_append(cache, layer, q, k, v, positions, slot_ids,
+                               cos, sin)
+    ref.rope(q, k, positions, cos, sin)
+    ref.kv_append(cache, layer, k, v, slot_ids, positions)
+
+
 def kv_append(cache, layer, k, v, slot_ids, positions):
     if _use_hip(k):
         from . import hip

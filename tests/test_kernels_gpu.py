@@ -215,3 +215,36 @@ def test_forward_e2e_vs_cpu():
     a, b = gpu_logits[0], cpu_logits[0]
     cos = torch.nn.functional.cosine_similarity(a, b, dim=-1)
     assert cos.min() > 0.995, f"prefill logits diverged: {cos.min()}"
+
+
+def test_fused_rope_append():
+    """Fused rope+append == reference rope then append."""
+    hip = _hip()
+    T, Hq, KVH, D = 13, 4, 2, 128
+    gc, cc = make_caches(KVH=KVH)
+    sg, sc = gc.alloc_slot(), cc.alloc_slot()
+    gc.ensure(sg, T)
+    cc.ensure(sc, T)
+    cc.page_table.copy_(gc.page_table.cpu())
+    g = torch.Generator().manual_seed(8)
+    qkv = torch.randn(T, (Hq + 2 * KVH) * D, generator=g).bfloat16().to(dev())
+    q = qkv[:, :Hq * D].view(T, Hq, D)
+    k = qkv[:, Hq * D:(Hq + KVH) * D].view(T, KVH, D)
+    v = qkv[:, (Hq + KVH) * D:].view(T, KVH, D)
+    pos = torch.arange(T, dtype=torch.int32, device=dev())
+    slot = torch.zeros(T, dtype=torch.int32, device=dev())
+    ang = torch.outer(torch.arange(64, dtype=torch.float32),
+                      1.0 / 10000 ** (torch.arange(0, D, 2) / D))
+    cos, sin = ang.cos().to(dev()), ang.sin().to(dev())
+    q_ref = q.float().cpu().clone()
+    k_ref = k.float().cpu().clone()
+    v_ref = v.float().cpu().clone()
+    ref.rope(q_ref, k_ref, pos.cpu(), cos.cpu(), sin.cpu())
+    ref.kv_append(cc, 0, k_ref, v_ref, slot.cpu(), pos.cpu())
+    hip.rope_append(gc, 0, q, k, v, pos, slot, cos, sin)
+    torch.testing.assert_close(q.float().cpu(), q_ref, atol=2e-2, rtol=2e-2)
+    torch.testing.assert_close(k.float().cpu(), k_ref, atol=2e-2, rtol=2e-2)
+    torch.testing.assert_close(gc.k_pool.float().cpu(), cc.k_pool,
+                               atol=2e-2, rtol=2e-2)
+    torch.testing.assert_close(gc.v_pool.float().cpu(), cc.v_pool,
+                               atol=0, rtol=0)
