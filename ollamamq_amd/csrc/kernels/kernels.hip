@@ -66,6 +66,12 @@ DEV float wave_reduce_max(float x) {
 // One 256-thread block per row; H multiple of 8; bf16 x8 vectorized.
 // Memory-bound: reads x(+res)+w, writes y(+res_out) — single pass.
 // ---------------------------------------------------------------------------
+// VPT = 8-element vectors per thread (ceil(H / 2048)); the whole row
+// stays in registers between the reduce and the scale, so each block does
+// ONE read of x(+res) and one write each of res_out and y — the two-pass
+// version re-read its own store and its second pass re-paid load latency
+// (decode rows are few: the grid is tiny and latency-bound).
+template <int VPT>
 __global__ __launch_bounds__(256) void k_rmsnorm_residual(
     bf16* __restrict__ y, bf16* __restrict__ res_out,
     const bf16* __restrict__ x, const bf16* __restrict__ res_in,
@@ -77,21 +83,23 @@ __global__ __launch_bounds__(256) void k_rmsnorm_residual(
     bf16* yr = y + (int64_t)row * H;
     bf16* ro = res_out + (int64_t)row * H;
 
-    // pass 1: accumulate sum of squares of (x+res) while caching the sum
-    // into res_out (bf16) — one global round trip for the residual.
+    float v[VPT][8];
     float ss = 0.f;
-    for (int i = threadIdx.x * 8; i < H; i += blockDim.x * 8) {
-        float vx[8];
-        load8f(xr + i, vx);
-        if (rr) {
-            float vr[8];
-            load8f(rr + i, vr);
+    #pragma unroll
+    for (int u = 0; u < VPT; u++) {
+        const int i = threadIdx.x * 8 + u * 2048;
+        if (i < H) {
+            load8f(xr + i, v[u]);
+            if (rr) {
+                float vr[8];
+                load8f(rr + i, vr);
+                #pragma unroll
+                for (int j = 0; j < 8; j++) v[u][j] += vr[j];
+            }
             #pragma unroll
-            for (int j = 0; j < 8; j++) vx[j] += vr[j];
+            for (int j = 0; j < 8; j++) ss += v[u][j] * v[u][j];
+            store8bf(ro + i, v[u]);
         }
-        #pragma unroll
-        for (int j = 0; j < 8; j++) ss += vx[j] * vx[j];
-        store8bf(ro + i, vx);
     }
     ss = wave_reduce_sum(ss);
     __shared__ float warp_ss[4];
@@ -101,13 +109,16 @@ __global__ __launch_bounds__(256) void k_rmsnorm_residual(
     float tot = warp_ss[0] + warp_ss[1] + warp_ss[2] + warp_ss[3];
     const float inv = rsqrtf(tot / H + eps);
 
-    for (int i = threadIdx.x * 8; i < H; i += blockDim.x * 8) {
-        float vs[8], vw[8];
-        load8f(ro + i, vs);   // L2-hot: just stored
-        load8f(w + i, vw);
-        #pragma unroll
-        for (int j = 0; j < 8; j++) vs[j] = vs[j] * inv * vw[j];
-        store8bf(yr + i, vs);
+    #pragma unroll
+    for (int u = 0; u < VPT; u++) {
+        const int i = threadIdx.x * 8 + u * 2048;
+        if (i < H) {
+            float vw[8];
+            load8f(w + i, vw);
+            #pragma unroll
+            for (int j = 0; j < 8; j++) v[u][j] = v[u][j] * inv * vw[j];
+            store8bf(yr + i, v[u]);
+        }
     }
 }
 
@@ -115,9 +126,16 @@ extern "C" int rmsnorm_residual_bf16(
     void* y, void* res_out, const void* x, const void* res_in,
     const void* w, int T, int H, float eps, hipStream_t stream)
 {
-    k_rmsnorm_residual<<<T, 256, 0, stream>>>(
-        (bf16*)y, (bf16*)res_out, (const bf16*)x, (const bf16*)res_in,
-        (const bf16*)w, H, eps);
+    #define RMS_LAUNCH(VPT)                                               \
+        k_rmsnorm_residual<VPT><<<T, 256, 0, stream>>>(                   \
+            (bf16*)y, (bf16*)res_out, (const bf16*)x,                     \
+            (const bf16*)res_in, (const bf16*)w, H, eps)
+    if (H <= 2048) RMS_LAUNCH(1);
+    else if (H <= 4096) RMS_LAUNCH(2);
+    else if (H <= 8192) RMS_LAUNCH(4);
+    else if (H <= 16384) RMS_LAUNCH(8);
+    else return (int)hipErrorInvalidValue;
+    #undef RMS_LAUNCH
     return (int)hipGetLastError();
 }
 
