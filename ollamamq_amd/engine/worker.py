@@ -293,6 +293,11 @@ class Conn:
         stream = body.get("stream", not openai)
         prompt = tok.encode(_prompt_text(body, path))
 
+        stops = body.get("stop") or (body.get("options") or {}).get("stop")
+        if isinstance(stops, str):
+            stops = [stops]
+        stops = [s for s in (stops or []) if s]
+
         q: "queue.Queue" = queue.Queue()
         t0 = time.time()
         sid = w.generate(model, prompt, params,
@@ -307,6 +312,7 @@ class Conn:
         pieces = []
         n_out = 0
         try:
+            stopped = False
             while True:
                 t, done = q.get(timeout=600)
                 if done:
@@ -314,7 +320,14 @@ class Conn:
                 n_out += 1
                 piece = tok.decode_one(t)
                 pieces.append(piece)
-                if stream:
+                # stop sequences (Ollama `stop` option): cancel generation
+                # when the decoded tail matches any stop string
+                if stops and not stopped:
+                    tail = "".join(pieces[-8:])
+                    if any(sp in tail for sp in stops):
+                        stopped = True
+                        w.cancel(model, sid)
+                if stream and not stopped:
                     self._stream_piece(path, model, piece, openai)
             self._final(path, model, pieces, n_out, t0, openai, stream,
                         len(prompt))
