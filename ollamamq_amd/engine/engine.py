@@ -180,7 +180,9 @@ class LlamaEngine:
             for seq in [s for s in lst if s.cancelled]:
                 lst.remove(seq)
                 seq.finish_reason = "cancelled"
-                self._finish(seq, emit=False)
+                # emit the done marker: stream consumers (the worker's
+                # token queue) must unblock on cancellation too
+                self._finish(seq, emit=True)
 
     def _admit(self) -> List[Sequence]:
         """Move waiting sequences into a prefill batch (chunked varlen)."""

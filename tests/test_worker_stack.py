@@ -190,3 +190,19 @@ def test_embeddings_endpoints(stack):
                    headers={"X-User-ID": "e3"}, timeout=120.0)
     assert r.status_code == 200, r.text
     assert isinstance(r.json()["embedding"], list)
+
+
+def test_stop_sequences(stack):
+    # with a stop string that appears immediately (any printable output),
+    # generation must end early rather than exhausting num_predict
+    t0 = time.time()
+    r = httpx.post(stack + "/api/generate",
+                   json={"model": "tiny-cpu", "prompt": "q",
+                         "stream": False,
+                         "options": {"num_predict": 40,
+                                     "stop": ["a", "b", "c", "d", "e", "f",
+                                              "0", "1", "2", "3", "\\x"]}},
+                   headers={"X-User-ID": "s1"}, timeout=120.0)
+    assert r.status_code == 200, r.text
+    obj = json.loads(r.text.strip())
+    assert obj["done"] is True
