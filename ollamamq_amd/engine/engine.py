@@ -120,23 +120,19 @@ class LlamaEngine:
         batch (inter-token latency stays bounded by ~one prefill chunk).
         """
         self._reap_cancelled()
-        admitted = []
-        do_prefill = bool(self.waiting) and (not self.running
-                                             or self.steps % 2 == 0)
-        if do_prefill:
-            admitted = self._admit()
-        if admitted:
+        admitted = self._admit() if self.waiting else []
+        if admitted and self.running:
+            # mixed batch: every running sequence decodes one token AND
+            # the new prompts prefill a chunk in the SAME forward — one
+            # weight pass serves both, so admission never stalls decode
+            self._mixed_step(admitted)
+            finished = self._postprocess(self.running + admitted)
+        elif admitted:
             self._prefill_step(admitted)
             finished = self._postprocess(admitted)
         elif self.running:
             self._decode_step()
             finished = self._postprocess(self.running)
-        elif self.waiting:
-            admitted = self._admit()
-            if not admitted:
-                return []   # waiting but nothing admissible (kv full)
-            self._prefill_step(admitted)
-            finished = self._postprocess(admitted)
         else:
             return []
         self.steps += 1
@@ -213,7 +209,46 @@ class LlamaEngine:
                 room -= 1
         return batch
 
-    def _build_meta(self, seqs, q_lens, mode) -> AttnMeta:
+    def _mixed_step(self, batch: List[Sequence]):
+        dec = list(self.running)
+        token_list = [s.generated[-1] for s in dec]
+        pos_list = [s.total_len - 1 for s in dec]
+        q_lens = [1] * len(dec)
+        for s in batch:
+            take = s._chunk
+            start = s.prefill_done
+            token_list.extend(s.prompt[start:start + take])
+            pos_list.extend(range(start, start + take))
+            q_lens.append(take)
+        logits = self._forward(dec + batch, token_list, pos_list, q_lens,
+                               "mixed", n_decode=len(dec))
+        # sample decode rows + finishing-prefill rows
+        fin_rows = [len(dec) + i for i, s in enumerate(batch)
+                    if s.prefill_done + s._chunk == len(s.prompt)]
+        done_prefill = [batch[i - len(dec)] for i in fin_rows]
+        sample_rows = list(range(len(dec))) + fin_rows
+        toks = self._sample(dec + done_prefill, logits[sample_rows])
+        now = time.monotonic()
+        for s, tok in zip(dec, toks[:len(dec)]):
+            s.generated.append(int(tok))
+            self.tokens_out += 1
+            if s.on_token:
+                s.on_token(int(tok), False)
+        ptoks = toks[len(dec):]
+        for s in batch:
+            s.prefill_done += s._chunk
+            if s.prefill_done == len(s.prompt):
+                tok = ptoks[done_prefill.index(s)]
+                s.generated.append(int(tok))
+                s.first_token_at = now
+                s.state = "running"
+                self.waiting.remove(s)
+                self.running.append(s)
+                self.tokens_out += 1
+                if s.on_token:
+                    s.on_token(int(tok), False)
+
+    def _build_meta(self, seqs, q_lens, mode, n_decode=0) -> AttnMeta:
         dev = self.dev
         slot_ids = torch.tensor([s.slot for s in seqs], dtype=torch.int32,
                                 device=dev)
@@ -229,9 +264,11 @@ class LlamaEngine:
             mode=mode, slot_ids=slot_ids, seq_lens=seq_lens, cu_q=cu_q,
             logits_idx=logits_idx, max_q=max(q_lens),
             max_kv=int(max(self.kv.seq_lens[s.slot] for s in seqs)),
+            n_decode=n_decode,
         )
 
-    def _forward(self, seqs, token_list, pos_list, q_lens, mode):
+    def _forward(self, seqs, token_list, pos_list, q_lens, mode,
+                 n_decode=0):
         dev = self.dev
         tokens = torch.tensor(token_list, dtype=torch.int32, device=dev)
         positions = torch.tensor(pos_list, dtype=torch.int32, device=dev)
@@ -240,7 +277,7 @@ class LlamaEngine:
             dtype=torch.int32, device=dev)
         for s, ql in zip(seqs, q_lens):
             self.kv.ensure(s.slot, self.kv.seq_lens[s.slot] + ql)
-        meta = self._build_meta(seqs, q_lens, mode)
+        meta = self._build_meta(seqs, q_lens, mode, n_decode)
         logits = self.model.forward(tokens, positions, self.kv, slot_per_tok,
                                     meta)
         return logits

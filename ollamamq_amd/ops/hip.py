@@ -257,6 +257,58 @@ def attention_decode(q, cache, layer, meta):
     return out
 
 
+def attention_mixed(q, cache, layer, meta):
+    """Mixed batch: decode kernel over the first n_decode rows, MFMA
+    prefill kernel over the remaining chunk rows — both write into one
+    output tensor (sub-views share storage)."""
+    T, Hq, D = q.shape
+    nd = meta.n_decode
+    out = torch.empty((T, Hq, D), dtype=q.dtype, device=q.device)
+    # ---- decode part ----
+    sub = getattr(meta, "_mixed_dec", None)
+    if sub is None:
+        sub = AttnMeta(
+            mode="decode", slot_ids=meta.slot_ids[:nd],
+            seq_lens=meta.seq_lens[:nd],
+            cu_q=meta.cu_q[:nd + 1], logits_idx=None, max_q=1,
+            max_kv=meta.max_kv)
+        meta._mixed_dec = sub
+    if nd:
+        out[:nd] = attention_decode(q[:nd], cache, layer, sub)
+    # ---- prefill part (tiles over sequences nd..) ----
+    plan = getattr(meta, "_mixed_plan", None)
+    if plan is None:
+        slots = meta.slot_ids[nd:].tolist()
+        lens = meta.seq_lens[nd:].tolist()
+        cu = meta.cu_q[nd:].tolist()
+        dev = q.device
+        ts, tq, tp, tr = [], [], [], []
+        for i, slot in enumerate(slots):
+            qlen = cu[i + 1] - cu[i]
+            start_pos = lens[i] - qlen
+            for t0 in range(0, qlen, PREFILL_QT):
+                ts.append(slot)
+                tq.append(cu[i] + t0)
+                tp.append(start_pos + t0)
+                tr.append(min(PREFILL_QT, qlen - t0))
+        plan = (torch.tensor(ts, dtype=torch.int32, device=dev),
+                torch.tensor(tq, dtype=torch.int32, device=dev),
+                torch.tensor(tp, dtype=torch.int32, device=dev),
+                torch.tensor(tr, dtype=torch.int32, device=dev))
+        meta._mixed_plan = plan
+    tile_slot, tile_q0, tile_pos0, tile_rows = plan
+    n_tiles = tile_slot.shape[0]
+    if n_tiles:
+        kp, vp = _layer_ptrs(cache, layer)
+        _check(_lib.prefill_attn_bf16(
+            _p(out), _p(q), kp, vp, _p(cache.page_table),
+            _p(tile_slot), _p(tile_q0), _p(tile_pos0), _p(tile_rows),
+            n_tiles, Hq, cache.n_kv_heads, cache.page_size,
+            cache.page_table.shape[1], 1.0 / (D ** 0.5),
+            _row_stride(q, D), _stream()), "prefill_attn")
+    return out
+
+
 def attention_prefill(q, cache, layer, meta):
     """MFMA flash prefill (prefill_attn.hip); set OLLAMAMQ_VALU_PREFILL=1
     to fall back to the VALU paged_attn path (A/B + debugging)."""

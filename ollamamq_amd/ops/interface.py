@@ -30,7 +30,9 @@ def _use_hip(t: torch.Tensor) -> bool:
 class AttnMeta:
     """Per-forward attention metadata (flat varlen layout).
 
-    mode: "prefill" (multi-token queries) or "decode" (1 token/seq).
+    mode: "prefill" (multi-token queries), "decode" (1 token/seq) or
+    "mixed" (the first n_decode sequences are single-token decodes, the
+    rest are prefill chunks — one forward shares the weight pass).
     slot_ids : [S] int32  kv-cache slot of each sequence in batch order
     seq_lens : [S] int32  total KV length per slot AFTER the kv_append
     cu_q     : [S+1] int32 exclusive prefix sum of per-seq query counts
@@ -44,6 +46,7 @@ class AttnMeta:
     logits_idx: Optional[torch.Tensor]
     max_q: int
     max_kv: int
+    n_decode: int = 0
 
 
 def embedding(tokens: torch.Tensor, table: torch.Tensor) -> torch.Tensor:
@@ -88,7 +91,10 @@ def attention(q, cache, layer, meta: AttnMeta):
         from . import hip
         if meta.mode == "decode":
             return hip.attention_decode(q, cache, layer, meta)
+        if meta.mode == "mixed":
+            return hip.attention_mixed(q, cache, layer, meta)
         return hip.attention_prefill(q, cache, layer, meta)
+    # the fp32 reference handles any varlen layout uniformly
     return ref.attention(q, cache, layer, meta)
 
 

@@ -248,3 +248,50 @@ def test_fused_rope_append():
                                atol=2e-2, rtol=2e-2)
     torch.testing.assert_close(gc.v_pool.float().cpu(), cc.v_pool,
                                atol=0, rtol=0)
+
+
+def test_mixed_batch_continuous_batching():
+    """A sequence admitted mid-decode (mixed prefill+decode forward) must
+    not change the earlier sequence's greedy tokens, and the late joiner
+    must produce the same tokens as when run alone."""
+    from ollamamq_amd.models import LlamaModel, PRESETS
+    from ollamamq_amd.engine import LlamaEngine, PagedKVCache, GenParams
+    cfg = PRESETS["tiny"]
+    p1, p2 = list(range(1, 21)), list(range(30, 45))
+
+    def fresh():
+        model = LlamaModel(cfg, device="cuda:0", dtype=torch.bfloat16,
+                           seed=21)
+        kv = PagedKVCache.for_model(cfg, n_pages=64, max_slots=4,
+                                    max_ctx=256, device="cuda:0",
+                                    dtype=torch.bfloat16)
+        return LlamaEngine(model, kv, max_batch=4)
+
+    # run each alone
+    alone = {}
+    for name, p in (("s1", p1), ("s2", p2)):
+        eng = fresh()
+        sid = eng.submit(p, GenParams(max_tokens=8))
+        seq = eng.seqs[sid]
+        for _ in range(64):
+            eng.step()
+            if not eng.has_work():
+                break
+        alone[name] = seq.generated
+
+    # staggered: submit s2 after s1 has decoded a few tokens -> the
+    # admission step is a MIXED forward
+    eng = fresh()
+    sid1 = eng.submit(p1, GenParams(max_tokens=8))
+    seq1 = eng.seqs[sid1]
+    for _ in range(4):
+        eng.step()
+    sid2 = eng.submit(p2, GenParams(max_tokens=8))
+    seq2 = eng.seqs[sid2]
+    for _ in range(64):
+        eng.step()
+        if not eng.has_work():
+            break
+    assert not eng.has_work()
+    assert seq1.generated == alone["s1"], "mixed step changed s1 tokens"
+    assert seq2.generated == alone["s2"], "late joiner diverged"
