@@ -58,7 +58,8 @@ def build_dispatcher(force=False):
         capture_output=True, text=True, check=True).stdout.strip()
     ext = os.path.join(DDIR, "_dispatch.so")
     core = [s for s in srcs if not s.endswith("main.cpp")
-            and not s.endswith("bindings.cpp")]
+            and not s.endswith("bindings.cpp")
+            and not s.endswith("tsan_stress.cpp")]
     bindings = os.path.join(DDIR, "bindings.cpp")
     if os.path.exists(bindings) and (force or not _newer(ext, *core, bindings)):
         _run(["g++", "-O2", "-std=c++17", "-shared", "-fPIC",
@@ -70,6 +71,26 @@ def build_dispatcher(force=False):
         _run(["g++", "-O2", "-std=c++17", "-I" + DDIR, main, *core,
               "-o", binary, "-lpthread"])
     return ext
+
+
+def build_tsan_stress(force=False):
+    """TSan-instrumented concurrency hammer (SURVEY §5 race detection)."""
+    import glob
+    srcs = sorted(glob.glob(os.path.join(DDIR, "*.cpp")))
+    core = [s for s in srcs if not s.endswith("main.cpp")
+            and not s.endswith("bindings.cpp")
+            and not s.endswith("tsan_stress.cpp")]
+    out = os.path.join(DDIR, "tsan_stress")
+    stress = os.path.join(DDIR, "tsan_stress.cpp")
+    if force or not _newer(out, stress, *core):
+        # clang's TSan runtime: gcc-11's libtsan false-positives on
+        # condition_variable timed waits ("double lock of a mutex")
+        cxx = "/opt/rocm/lib/llvm/bin/clang++"
+        if not os.path.exists(cxx):
+            cxx = "g++"
+        _run([cxx, "-O1", "-g", "-std=c++17", "-fsanitize=thread",
+              "-I" + DDIR, stress, *core, "-o", out, "-lpthread"])
+    return out
 
 
 def build_all(force=False):

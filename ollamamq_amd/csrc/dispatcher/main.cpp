@@ -4,6 +4,7 @@
 // wiring; defaults: port 11435, host 127.0.0.1, timeout 300 s,
 // load_keep_alive 86400 s, stuck_timeout 60 s, default backend
 // http://localhost:11434).  Auth key via env OLLAMA_MQ_API_KEY.
+#include <csignal>
 #include <unistd.h>
 
 #include <cstdlib>
@@ -141,8 +142,13 @@ int main(int argc, char** argv) {
     }).detach();
 
     if (no_tui) {
-        // headless: block forever (Ctrl-C to exit)
-        while (true) std::this_thread::sleep_for(std::chrono::seconds(3600));
+        // headless: run until SIGINT/SIGTERM, then stop cleanly
+        static std::atomic<bool> quit{false};
+        std::signal(SIGINT, [](int) { quit = true; });
+        std::signal(SIGTERM, [](int) { quit = true; });
+        while (!quit)
+            std::this_thread::sleep_for(std::chrono::milliseconds(200));
+        server.stop();
     } else {
         run_tui(server);  // returns on 'q'
         server.stop();

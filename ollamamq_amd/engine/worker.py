@@ -133,7 +133,21 @@ class Worker:
             for eng in engines:
                 with self.lock:
                     if eng.has_work():
-                        eng.step()
+                        try:
+                            eng.step()
+                        except Exception as e:
+                            # a failed step (e.g. OOM under load) must not
+                            # kill the loop: fail the in-flight sequences
+                            # so their streams terminate, keep serving
+                            print(f"engine step failed: {e!r}",
+                                  flush=True)
+                            for seq in (list(eng.waiting)
+                                        + list(eng.running)):
+                                seq.cancelled = True
+                            try:
+                                eng.step()   # reap + emit done markers
+                            except Exception:
+                                pass
                         busy = eng.has_work() or busy
             if not busy:
                 self.work_ev.wait(timeout=0.02)

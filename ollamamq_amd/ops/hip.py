@@ -14,6 +14,7 @@ from typing import Optional
 import torch
 
 from . import reference as ref
+from .interface import AttnMeta
 
 _LIB_PATH = os.path.join(
     os.path.dirname(os.path.dirname(os.path.abspath(__file__))),
