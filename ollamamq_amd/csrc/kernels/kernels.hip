@@ -227,6 +227,9 @@ extern "C" int kv_append_bf16(
 
 // ---------------------------------------------------------------------------
 // Paged attention, decode + prefill (flat varlen), D = 128, GQA group G ≤ 8.
+// STATUS: legacy VALU path — superseded by decode_attn.hip
+// (flash-decoding) and prefill_attn.hip (MFMA, 20x faster); kept as the
+// OLLAMAMQ_VALU_PREFILL fallback and as the perf A/B baseline.
 //
 // Grid (n_tiles, KVH); block = G waves (one per q head in the group).
 // Each tile covers QT query rows of one sequence.  KV is streamed in

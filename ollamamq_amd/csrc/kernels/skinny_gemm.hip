@@ -1,27 +1,25 @@
 // gfx950 skinny GEMM for the decode path: y[M,N] = x[M,K] @ W[N,K]^T,
 // M ≤ 32 (decode batch), bf16 in/out, fp32 MFMA accumulation.
 //
-// Why hand-written: decode projections are weights-streaming bound
-// (N×K×2 B read once per step; x is L2/L3-resident) and the library GEMM
-// measured 1.5-2.7 TB/s on the small-N shapes (qkv/o at M=32) — far off
-// the ≈6.3 TB/s HBM roofline.
+// STATUS (measured, tools/perf_gemm.py + bench.py A/B): kept in-tree as
+// opt-in (OLLAMAMQ_SKINNY_MAX_N/K env gates; default = library GEMMs).
+// Two variants below:
+//  * k_skinny_gemm ("staged"): W chunks staged through LDS in full
+//    coalesced 128 B lines, 3-buffer ring, 2-deep register prefetch,
+//    in-block + grid k-split.  ~2 TB/s — stall-bound (PMC: 87%
+//    SQ_WAIT_ANY at 1 block/CU before the ring; the ring recovered only
+//    part of it).
+//  * k_skinny_direct: fragment-direct (no LDS, no barriers), 4-deep
+//    register prefetch, 8 independent waves/block + grid k-split.
+//    Beats hipBLASLt on L3-WARM microbenches (qkv 2.9 vs 2.7 TB/s,
+//    o 2.4 vs 1.8) but loses ~0.3 ms/step in the real serving loop where
+//    weights stream COLD from HBM: its 16 B-per-lane loads at 8 KB row
+//    stride use 32 B per fetched 128 B line per pass, so cold misses
+//    over-fetch ~4x (the L3 hid this in the microbench).  Honest A/B in
+//    bench.py decided the default (5473 lib vs 5221 direct tok/s).
 //
-// v2 design (v1 read W fragment-shaped, 16 B per lane at 8 KB row stride:
-// TA/L2-request bound at ~2 TB/s — the guide's M=256 projection-GEMM
-// analysis: fragment-shaped operand loads cost +18-45% with TA_BUSY 2×;
-// full-line LDS staging wins):
-//  * one block = 4 waves = one 32-column tile of y;
-//  * W is streamed in [32 rows][256 k] chunks staged into LDS by ALL
-//    threads with fully-coalesced 16 B lane-consecutive loads (whole
-//    128 B lines), double-buffered, loads for chunk c+1 issued before the
-//    MFMAs of chunk c (T14 issue-early/write-late: HBM latency hides
-//    under compute);
-//  * wave w computes the k-quarter [w*64,(w+1)*64) of each chunk from LDS
-//    (ds_read_b128, row pad +8 bf16 keeps the b128 lane groups
-//    conflict-free) and accumulates its partial in AGPRs;
-//  * epilogue reduces the 4 wave partials through LDS.
-//
-// Constraints: K % 256 == 0, N % 32 == 0, M ≤ 32 (hip.py gates shapes).
+// Constraints: K % 256 == 0 (staged) / K % (8*64*ksplit) == 0 (direct),
+// N % 32 == 0, M ≤ 32.
 #include <hip/hip_runtime.h>
 #include <hip/hip_bf16.h>
 #include <cstdint>

@@ -284,3 +284,13 @@ def test_blocklist_persistence(fleet, tmp_path_factory):
         assert r.status_code == 200
     finally:
         p.stop()
+
+
+def test_admin_stats_surface(proxy):
+    r = httpx.get(proxy.base + "/admin/stats")
+    assert r.status_code == 200
+    st = r.json()
+    assert st["uptime_s"] > 0
+    assert "queue_wait" in st and "p50_ms" in st["queue_wait"]
+    assert any(u["processed"] > 0 for u in st["users"])
+    assert len(st["backends"]) == 2
