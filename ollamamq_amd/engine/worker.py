@@ -257,6 +257,9 @@ class Conn:
         with w.lock:
             loaded = list(w.engines.keys())
             ctx = dict(w.loaded_ctx)
+            tokens = sum(e.tokens_out for e in w.engines.values())
+            steps = sum(e.steps for e in w.engines.values())
+            active = sum(e.n_active() for e in w.engines.values())
         self._line({
             "online": True,
             "models": w.available_models(),
@@ -264,6 +267,9 @@ class Conn:
             "ctx": ctx,
             "max_concurrency": w.max_batch,
             "device": w.device,
+            "stats": {"tokens_out": tokens, "engine_steps": steps,
+                      "active_seqs": active,
+                      "uptime_s": round(time.time() - w.started, 1)},
         })
 
     # -- the request path --
