@@ -48,13 +48,15 @@ def main():
     world = int(os.environ.get("WORLD_SIZE", "1"))
     local_rank = int(os.environ.get("LOCAL_RANK", str(rank)))
     has_gpu = torch.cuda.is_available()
+    if has_gpu:
+        # bind the device BEFORE the NCCL/RCCL communicator is created
+        torch.cuda.set_device(local_rank)
     dist = world > 1
     if dist:
         torch.distributed.init_process_group(
             backend="nccl" if has_gpu else "gloo")
 
     if has_gpu:
-        torch.cuda.set_device(local_rank)
         device = f"cuda:{local_rank}"
         dtype = torch.bfloat16
         model_name = args.model or "llama3-8b"
