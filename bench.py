@@ -39,6 +39,7 @@ def main():
     ap.add_argument("--prompt-len", type=int, default=512)
     ap.add_argument("--model", type=str, default=None)
     ap.add_argument("--max-ctx", type=int, default=4096)
+    ap.add_argument("--prefill-chunk", type=int, default=4096)
     args = ap.parse_args()
 
     from ollamamq_amd.models import LlamaModel, PRESETS
@@ -81,7 +82,8 @@ def main():
     kv = PagedKVCache.for_model(
         cfg, n_pages=n_pages, max_slots=users + 2, max_ctx=ctx,
         device=device, dtype=dtype)
-    eng = LlamaEngine(model, kv, max_batch=users, prefill_chunk=4096)
+    eng = LlamaEngine(model, kv, max_batch=users,
+                      prefill_chunk=args.prefill_chunk)
     load_s = time.monotonic() - t_load0
 
     # --- submit synthetic users -------------------------------------------
