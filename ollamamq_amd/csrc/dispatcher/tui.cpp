@@ -34,6 +34,7 @@ struct Snapshot {
         int active;
         int64_t processed;
         std::string loaded;
+        std::string avail;
         std::string op;
     };
     struct U {
@@ -64,6 +65,10 @@ Snapshot capture(AppState& st) {
             for (const auto& m : b.loaded_models) {
                 if (!e.loaded.empty()) e.loaded += ",";
                 e.loaded += m;
+            }
+            for (const auto& m : b.available_models) {
+                if (!e.avail.empty()) e.avail += ",";
+                e.avail += m;
             }
             auto op = st.control_ops.find(i);
             if (op != st.control_ops.end())
@@ -143,6 +148,9 @@ void run_tui(Server& server) {
     AppState& st = server.state();
     RawTerm term;
     int sel_user = 0;
+    int sel_backend = 0;
+    int focus = 1;            // 0 = Backends panel, 1 = Users panel (Tab)
+    bool show_all = false;    // 'a': show available models per backend
     std::string input;        // typed model name for L/U
     char input_mode = 0;      // 'L' or 'U' when typing
     bool help = false;
@@ -162,7 +170,9 @@ void run_tui(Server& server) {
             << "  sched:" << s.counter << "\r\n";
         if (help) {
             out << "\r\n \x1b[1mKeys\x1b[0m\r\n"
-                   "  j/k     select user\r\n"
+                   "  Tab     switch panel focus (Backends/Users)\r\n"
+                   "  j/k     move selection in the focused panel\r\n"
+                   "  a       toggle available-models listing\r\n"
                    "  p / b   toggle VIP / Boost for selected user\r\n"
                    "  x / X   block selected user / their IP\r\n"
                    "  u       unblock all\r\n"
@@ -182,20 +192,28 @@ void run_tui(Server& server) {
             }
             continue;
         }
-        out << "\x1b[7m" << pad(" Backends", 90) << "\x1b[0m\r\n";
-        for (const auto& b : s.backends) {
-            out << (b.online ? " \x1b[32m●\x1b[0m " : " \x1b[31m○\x1b[0m ")
+        out << "\x1b[7m" << pad(focus == 0 ? " Backends [focused]"
+                                          : " Backends", 90)
+            << "\x1b[0m\r\n";
+        for (int i = 0; i < (int)s.backends.size(); i++) {
+            const auto& b = s.backends[i];
+            out << (focus == 0 && i == sel_backend ? ">" : " ")
+                << (b.online ? "\x1b[32m●\x1b[0m " : "\x1b[31m○\x1b[0m ")
                 << pad(b.url, 32) << pad(b.api, 8)
                 << "act:" << b.active << " done:" << b.processed << " "
                 << pad(b.loaded, 24)
                 << (b.op.empty() ? "" : " [" + b.op + "]") << "\r\n";
+            if (show_all && !b.avail.empty())
+                out << "    available: " << pad(b.avail, 80) << "\r\n";
         }
-        out << "\x1b[7m" << pad(" Users (j/k nav, p VIP, b Boost, x block)",
-                                90)
+        out << "\x1b[7m"
+            << pad(focus == 1 ? " Users [focused] (j/k, p VIP, b Boost, "
+                                "x block)"
+                              : " Users", 90)
             << "\x1b[0m\r\n";
         for (int i = 0; i < (int)s.users.size(); i++) {
             const auto& u = s.users[i];
-            out << (i == sel_user ? " >" : "  ")
+            out << (focus == 1 && i == sel_user ? " >" : "  ")
                 << pad(u.name + (u.vip ? " ★" : "") + (u.boost ? " ⚡" : ""),
                        24)
                 << " q:" << u.queued << " run:" << u.processing
@@ -217,8 +235,8 @@ void run_tui(Server& server) {
                 << " model: " << input << "_\r\n";
         else
             out << "\r\n " << status_msg
-                << "  [L]oad [U]nload [r]eload-cfg [p]VIP [b]Boost "
-                   "[x]block [u]unblock [q]uit\r\n";
+                << "  [Tab]panel [L]oad [U]nload [r]eload [p]VIP [b]Boost "
+                   "[x]block [u]unblock [a]ll [?]help [q]uit\r\n";
         fputs(out.str().c_str(), stdout);
         fflush(stdout);
 
@@ -236,7 +254,7 @@ void run_tui(Server& server) {
                                          ? ControlAction::Load
                                          : ControlAction::Unload;
                         req.model = input;
-                        req.backend_idx = 0;  // first backend
+                        req.backend_idx = (size_t)sel_backend;
                         auto r = start_model_control(st, req);
                         status_msg = r.http_status == 202
                                          ? "accepted"
@@ -259,12 +277,26 @@ void run_tui(Server& server) {
                     case 'q':
                     case 27:
                         return;
+                    case '\t':
+                        focus ^= 1;
+                        break;
+                    case 'a':
+                        show_all = !show_all;
+                        break;
                     case 'j':
-                        sel_user = std::min<int>(sel_user + 1,
-                                                 (int)s.users.size() - 1);
+                        if (focus == 1)
+                            sel_user = std::min<int>(
+                                sel_user + 1, (int)s.users.size() - 1);
+                        else
+                            sel_backend = std::min<int>(
+                                sel_backend + 1,
+                                (int)s.backends.size() - 1);
                         break;
                     case 'k':
-                        sel_user = std::max(sel_user - 1, 0);
+                        if (focus == 1)
+                            sel_user = std::max(sel_user - 1, 0);
+                        else
+                            sel_backend = std::max(sel_backend - 1, 0);
                         break;
                     case 'L':
                     case 'U':
