@@ -52,6 +52,7 @@ class Sequence:
     finished_at: Optional[float] = None
     finish_reason: Optional[str] = None
     cancelled: bool = False
+    _chunk: int = 0                   # prompt tokens taken this step
 
     @property
     def total_len(self) -> int:

@@ -23,6 +23,7 @@ Run:  python -m ollamamq_amd.engine.worker --socket /tmp/omq0.sock \
 from __future__ import annotations
 
 import argparse
+import datetime
 import json
 import os
 import queue
@@ -188,12 +189,22 @@ def _params_from_body(body: dict, tok: ByteTokenizer,
     )
 
 
+def _now_iso() -> str:
+    return datetime.datetime.now(datetime.timezone.utc).isoformat()
+
+
 def _prompt_text(body: dict, path: str) -> str:
     if "prompt" in body and isinstance(body["prompt"], str):
         return body["prompt"]
     parts = []
     for m in body.get("messages") or []:
-        parts.append(f"<{m.get('role', 'user')}>{m.get('content', '')}")
+        content = m.get("content", "")
+        if isinstance(content, list):   # OpenAI content-part arrays
+            content = " ".join(p.get("text", "") for p in content
+                               if isinstance(p, dict))
+        parts.append(f"<{m.get('role', 'user')}>{content}")
+        # multimodal fields (images) are accepted and ignored: random-init
+        # text models have no vision tower; the bytes still flow through
     return "\n".join(parts) or " "
 
 
@@ -368,9 +379,10 @@ class Conn:
             self.sock.sendall(f"data: {json.dumps(obj)}\n\n".encode())
         else:
             if path == "/api/generate":
-                obj = {"model": model, "response": piece, "done": False}
+                obj = {"model": model, "created_at": _now_iso(),
+                       "response": piece, "done": False}
             else:
-                obj = {"model": model,
+                obj = {"model": model, "created_at": _now_iso(),
                        "message": {"role": "assistant", "content": piece},
                        "done": False}
             self.sock.sendall((json.dumps(obj) + "\n").encode())
@@ -403,7 +415,8 @@ class Conn:
                                  "total_tokens": n_prompt + n_out}}
                 self.sock.sendall(json.dumps(obj).encode())
         else:
-            obj = {"model": model, "done": True, "done_reason": "stop",
+            obj = {"model": model, "created_at": _now_iso(),
+                   "done": True, "done_reason": "stop",
                    "total_duration": dur_ns,
                    "prompt_eval_count": n_prompt,
                    "eval_count": n_out}

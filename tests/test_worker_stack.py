@@ -206,3 +206,16 @@ def test_stop_sequences(stack):
     assert r.status_code == 200, r.text
     obj = json.loads(r.text.strip())
     assert obj["done"] is True
+
+
+def test_multimodal_fields_tolerated(stack):
+    import base64
+    img = base64.b64encode(b"\x89PNG fake image bytes" * 100).decode()
+    r = httpx.post(stack + "/api/generate",
+                   json={"model": "tiny-cpu", "prompt": "what is this?",
+                         "images": [img], "stream": False,
+                         "options": {"num_predict": 3}},
+                   headers={"X-User-ID": "mm1"}, timeout=120.0)
+    assert r.status_code == 200, r.text
+    obj = json.loads(r.text.strip())
+    assert obj["done"] is True and "created_at" in obj
