@@ -81,11 +81,8 @@ __global__ __launch_bounds__(512) void k_decode_attn(
 
     extern __shared__ __attribute__((aligned(16))) char smem[];
     const int KROW = DHEAD + DKPAD;
-    // K is NOT staged: lane j streams its own 256 B K row straight from
-    // the pool (rows are lane-private and read once, so L1 serves the
-    // interior of each line; staging K only added a round trip and LDS
-    // pressure).  V must land in LDS for the cross-lane PV broadcast.
-    bf16* v_tile = reinterpret_cast<bf16*>(smem);
+    bf16* k_tile = reinterpret_cast<bf16*>(smem);
+    bf16* v_tile = k_tile + DCHUNK * KROW;
     float* q_lds = reinterpret_cast<float*>(v_tile + DCHUNK * KROW); // [G][128]
     float* p_lds = q_lds + (int64_t)G * DHEAD;                       // [G][65]
 
@@ -102,7 +99,7 @@ __global__ __launch_bounds__(512) void k_decode_attn(
     for (int ch = c0; ch < c1; ch++) {
         const int base = ch * DCHUNK;
         const int n_here = min(DCHUNK, kv_len - base);
-        // ---- cooperative V chunk stage (4 pages @ page=16) ----
+        // ---- cooperative K/V chunk stage (4 pages @ page=16) ----
         {
             const int tid = threadIdx.x, nthr = blockDim.x;
             for (int u = tid; u < DCHUNK * (DHEAD / 8); u += nthr) {
@@ -116,21 +113,19 @@ __global__ __launch_bounds__(512) void k_decode_attn(
                                           + tk / page];
                 const int64_t src = (((int64_t)gp * KVH + kvh) * page
                                      + tk % page) * DHEAD + dv * 8;
+                *reinterpret_cast<uint4*>(k_tile + tok * KROW + dv * 8) =
+                    *reinterpret_cast<const uint4*>(kpool + src);
                 *reinterpret_cast<uint4*>(v_tile + tok * KROW + dv * 8) =
                     *reinterpret_cast<const uint4*>(vpool + src);
             }
         }
         __syncthreads();
 
-        // ---- score for key j = lane (K row streamed from the pool) ----
+        // ---- score for key j = lane ----
         float s = 0.f;
         const bool live = lane < n_here;
         {
-            const int tk = base + lane < kv_len ? base + lane : kv_len - 1;
-            const int gp = page_table[(int64_t)slot * max_pages
-                                      + tk / page];
-            const bf16* krow = kpool + (((int64_t)gp * KVH + kvh) * page
-                                        + tk % page) * DHEAD;
+            const bf16* krow = k_tile + lane * KROW;
             const float* qrow = q_lds + wid * DHEAD;
             #pragma unroll
             for (int d = 0; d < DHEAD; d += 8) {
@@ -225,7 +220,7 @@ extern "C" int decode_attn_bf16(
     hipStream_t stream)
 {
     const int G = Hq / KVH;
-    const int lds = DCHUNK * (DHEAD + DKPAD) * 2 + G * DHEAD * 4
+    const int lds = 2 * DCHUNK * (DHEAD + DKPAD) * 2 + G * DHEAD * 4
                   + G * (DCHUNK + 1) * 4;
     dim3 grid(S * split, KVH);
     k_decode_attn<<<grid, G * 64, lds, stream>>>(
