@@ -1,0 +1,89 @@
+"""HTTP-layer robustness: the native server must survive malformed and
+adversarial inputs (garbage bytes, bad content-lengths, huge headers,
+half-closed connections) without crashing or wedging."""
+import os
+import socket
+import subprocess
+import sys
+import time
+
+import httpx
+import pytest
+
+REPO = os.path.dirname(os.path.dirname(os.path.abspath(__file__)))
+BIN = os.path.join(REPO, "ollamamq_amd", "csrc", "dispatcher",
+                   "ollamamq-server")
+
+
+@pytest.fixture(scope="module")
+def srv(tmp_path_factory):
+    if not os.path.exists(BIN):
+        subprocess.run([sys.executable, "-m", "ollamamq_amd.build"],
+                       check=True)
+    tmp = tmp_path_factory.mktemp("robust")
+    p = subprocess.Popen(
+        [BIN, "--no-tui", "-p", "0",
+         "-c", os.path.join(str(tmp), "absent.yaml")],
+        stderr=subprocess.PIPE, cwd=str(tmp), text=True)
+    line = p.stderr.readline()
+    port = int(line.rsplit(":", 1)[1].split()[0])
+    yield port
+    p.terminate()
+
+
+def raw(port, data, expect_reply=False):
+    s = socket.create_connection(("127.0.0.1", port), timeout=5)
+    try:
+        s.sendall(data)
+        if expect_reply:
+            s.settimeout(5)
+            return s.recv(4096)
+        return b""
+    except OSError:
+        return b""
+    finally:
+        s.close()
+
+
+def alive(port):
+    r = httpx.get(f"http://127.0.0.1:{port}/health", timeout=5.0)
+    return r.status_code == 200
+
+
+def test_garbage_bytes(srv):
+    raw(srv, os.urandom(5000))
+    assert alive(srv)
+
+
+def test_bad_request_line(srv):
+    raw(srv, b"NOT A REQUEST\r\n\r\n")
+    assert alive(srv)
+
+
+def test_oversized_content_length(srv):
+    r = raw(srv, b"POST /api/chat HTTP/1.1\r\nHost: x\r\n"
+                 b"Content-Length: 99999999999\r\n\r\n{}",
+            expect_reply=True)
+    assert b"400" in r or r == b""
+    assert alive(srv)
+
+
+def test_header_flood(srv):
+    data = b"GET /health HTTP/1.1\r\n" + b"X-A: b\r\n" * 20000 + b"\r\n"
+    raw(srv, data)
+    assert alive(srv)
+
+
+def test_half_close_mid_body(srv):
+    s = socket.create_connection(("127.0.0.1", srv), timeout=5)
+    s.sendall(b"POST /api/chat HTTP/1.1\r\nContent-Length: 100\r\n\r\nhalf")
+    s.close()   # body never completes
+    time.sleep(0.2)
+    assert alive(srv)
+
+
+def test_many_rapid_connections(srv):
+    for _ in range(100):
+        s = socket.create_connection(("127.0.0.1", srv), timeout=5)
+        s.close()
+    assert alive(srv)
