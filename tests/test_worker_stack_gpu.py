@@ -1,0 +1,95 @@
+"""Full stack ON the GPU: C++ dispatcher -> UDS -> engine worker running
+the tiny preset on cuda:0 with the real HIP kernels (graphed decode).
+This is the round-end hardware check of the serving path itself."""
+import json
+import os
+import socket
+import subprocess
+import sys
+import time
+
+import httpx
+import pytest
+
+pytestmark = pytest.mark.gpu
+
+REPO = os.path.dirname(os.path.dirname(os.path.abspath(__file__)))
+BIN = os.path.join(REPO, "ollamamq_amd", "csrc", "dispatcher",
+                   "ollamamq-server")
+
+
+def _wait_socket(path, timeout=120):
+    t0 = time.time()
+    while time.time() - t0 < timeout:
+        if os.path.exists(path):
+            try:
+                s = socket.socket(socket.AF_UNIX, socket.SOCK_STREAM)
+                s.connect(path)
+                s.close()
+                return True
+            except OSError:
+                pass
+        time.sleep(0.2)
+    return False
+
+
+@pytest.fixture(scope="module")
+def gpu_stack(tmp_path_factory):
+    assert os.path.exists(BIN), "native server not built"
+    tmp = tmp_path_factory.mktemp("gstack")
+    sock = os.path.join(str(tmp), "wg.sock")
+    worker = subprocess.Popen(
+        [sys.executable, "-m", "ollamamq_amd.engine.worker",
+         "--socket", sock, "--gpu", "0", "--model", "tiny",
+         "--max-ctx", "512", "--max-batch", "8"],
+        cwd=REPO, stdout=subprocess.PIPE, stderr=subprocess.STDOUT,
+        text=True)
+    assert _wait_socket(sock), "GPU worker did not come up"
+    server = subprocess.Popen(
+        [BIN, "--no-tui", "-p", "0", "-w", sock,
+         "-c", os.path.join(str(tmp), "absent.yaml")],
+        stderr=subprocess.PIPE, cwd=str(tmp), text=True)
+    line = server.stderr.readline()
+    port = int(line.rsplit(":", 1)[1].split()[0])
+    base = f"http://127.0.0.1:{port}"
+    deadline = time.time() + 30
+    while time.time() < deadline:
+        try:
+            r = httpx.get(base + "/admin/models", timeout=2.0).json()
+            if r["backends"] and r["backends"][0]["online"]:
+                break
+        except Exception:
+            pass
+        time.sleep(0.2)
+    yield base
+    server.terminate()
+    worker.terminate()
+
+
+def test_gpu_worker_serves_ollama_chat(gpu_stack):
+    r = httpx.post(gpu_stack + "/api/chat",
+                   json={"model": "tiny",
+                         "messages": [{"role": "user", "content": "hi"}],
+                         "options": {"num_predict": 8}},
+                   headers={"X-User-ID": "g1"}, timeout=180.0)
+    assert r.status_code == 200, r.text
+    lines = [json.loads(l) for l in r.text.strip().split("\n")]
+    assert lines[-1]["done"] is True
+    assert lines[-1]["eval_count"] >= 1
+    st = httpx.get(gpu_stack + "/admin/models").json()["backends"][0]
+    assert st["api"] == "both" and "tiny" in st["loaded_models"]
+
+
+def test_gpu_worker_concurrent(gpu_stack):
+    import concurrent.futures as cf
+
+    def one(u):
+        r = httpx.post(gpu_stack + "/api/generate",
+                       json={"model": "tiny", "prompt": "abc" * 10,
+                             "stream": False,
+                             "options": {"num_predict": 4}},
+                       headers={"X-User-ID": f"gu{u}"}, timeout=180.0)
+        return r.status_code
+
+    with cf.ThreadPoolExecutor(6) as ex:
+        assert list(ex.map(one, range(6))) == [200] * 6
