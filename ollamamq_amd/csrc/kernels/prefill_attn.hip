@@ -34,10 +34,18 @@ DEV float bf2f(bf16 x) { return __bfloat162float(x); }
 #define PF_CHUNK 64          // kv tokens per staged chunk
 #define PF_ROWS 32           // q rows per tile
 #define DH 128
-#define KTROW (DH + 4)       // k_tile row stride (bf16)
+#define KTROW (DH + 8)       // k_tile row stride: 272 B rows stay 16-B
+                             // aligned for ds_read_b128 (G17) and the
+                             // 68-dword stride is conflict-free per group
 #define VTROW (PF_CHUNK + 8) // v_t row stride (bf16): [128 d][64 kv]
 #define PROW (PF_CHUNK + 8)  // p_lds row stride
+// v_t kv-BLOCK swizzle: the transposed V writes put every lane (d stride
+// 8) on one bank (8*VTROW/2 ≡ 0 mod 32 dwords → measured 42% of LDS
+// cycles were conflicts).  XORing the 8-element kv block index with
+// (d>>3)&7 spreads the banks; reads stay 16-B contiguous per block.
+#define VSWZ(d, kv) ((kv & 7) + 8 * (((kv) >> 3) ^ (((d) >> 3) & 7)))
 
+template <int G>
 __global__ __launch_bounds__(512) void k_prefill_attn(
     bf16* __restrict__ out,           // [T, Hq, D]
     const bf16* __restrict__ q,       // [T, Hq, D], row stride qs
@@ -52,7 +60,6 @@ __global__ __launch_bounds__(512) void k_prefill_attn(
 {
     const int tile = blockIdx.x;
     const int kvh = blockIdx.y;
-    const int G = Hq / KVH;
     const int wid = threadIdx.x >> 6;
     const int lane = threadIdx.x & 63;
     const int qh = kvh * G + wid;
@@ -66,9 +73,11 @@ __global__ __launch_bounds__(512) void k_prefill_attn(
     const int kv_len = pos0 + rows;
 
     extern __shared__ __attribute__((aligned(16))) char smem[];
-    bf16* k_tile = reinterpret_cast<bf16*>(smem);              // [64][KTROW]
-    bf16* v_t = k_tile + PF_CHUNK * KTROW;                     // [128][VTROW]
-    bf16* p_lds = v_t + DH * VTROW;       // [G][32][PROW]
+    // double-buffered K/V chunk tiles (T14: next chunk's loads issued
+    // before this chunk's MFMAs, written after the barrier)
+    bf16* k_tile0 = reinterpret_cast<bf16*>(smem);             // [64][KTROW]
+    bf16* v_t0 = k_tile0 + 2 * PF_CHUNK * KTROW;               // [128][VTROW]
+    bf16* p_lds = v_t0 + 2 * DH * VTROW;  // [G][32][PROW]
     bf16* my_p = p_lds + (int64_t)wid * PF_ROWS * PROW;
 
     // ---- load this wave's Q fragments into registers ----
@@ -90,34 +99,61 @@ __global__ __launch_bounds__(512) void k_prefill_attn(
     // O^T accumulators: 4 d-tiles of [32d x 32q]
     f32x16 o[4] = {};
 
+    // staging geometry: each thread owns N_PIECES 16-B pieces of the
+    // chunk; loads for chunk c+1 are issued before chunk c's MFMAs and
+    // written to the other buffer after the barrier.  Compile-time G
+    // keeps the piece arrays in registers (runtime indexing would go to
+    // scratch — guide §5.4 rule 20).
+    constexpr int NTHR = G * 64;
+    constexpr int N_PIECES = PF_CHUNK * (DH / 8) / NTHR;
+    uint4 stK[N_PIECES], stV[N_PIECES];
+    const int tid = threadIdx.x;
+
+    auto issue_chunk = [&](int base) {
+        #pragma unroll
+        for (int pi = 0; pi < N_PIECES; pi++) {
+            const int u = tid + pi * NTHR;
+            const int tok = u / (DH / 8);
+            const int dv = u % (DH / 8);
+            const int tk = base + tok < kv_len ? base + tok : kv_len - 1;
+            const int gp = page_table[(int64_t)slot * max_pages
+                                      + tk / page];
+            const int64_t src = (((int64_t)gp * KVH + kvh) * page
+                                 + tk % page) * DH + dv * 8;
+            stK[pi] = *reinterpret_cast<const uint4*>(kpool + src);
+            stV[pi] = *reinterpret_cast<const uint4*>(vpool + src);
+        }
+    };
+    auto write_chunk = [&](int buf) {
+        bf16* kt = k_tile0 + buf * PF_CHUNK * KTROW;
+        bf16* vt = v_t0 + buf * DH * VTROW;
+        #pragma unroll
+        for (int pi = 0; pi < N_PIECES; pi++) {
+            const int u = tid + pi * NTHR;
+            const int tok = u / (DH / 8);
+            const int dv = u % (DH / 8);
+            *reinterpret_cast<uint4*>(kt + tok * KTROW + dv * 8) = stK[pi];
+            const bf16* vsrc = reinterpret_cast<const bf16*>(&stV[pi]);
+            #pragma unroll
+            for (int e = 0; e < 8; e++) {
+                const int d = dv * 8 + e;
+                vt[d * VTROW + VSWZ(d, tok)] = vsrc[e];
+            }
+        }
+    };
+
     const int n_chunks = (kv_len + PF_CHUNK - 1) / PF_CHUNK;
+    issue_chunk(0);
+    write_chunk(0);
+    __syncthreads();
     for (int ch = 0; ch < n_chunks; ch++) {
         const int base = ch * PF_CHUNK;
         const int n_here = min(PF_CHUNK, kv_len - base);
-        __syncthreads();
-        // ---- stage K (row-major) and V (transposed) ----
-        {
-            const int tid = threadIdx.x, nthr = blockDim.x;
-            for (int u = tid; u < PF_CHUNK * (DH / 8); u += nthr) {
-                const int tok = u / (DH / 8), dv = u % (DH / 8);
-                // clamped, unconditional loads (guide §5 traps (c));
-                // dead kv rows are causally masked to p = 0 below
-                const int tk = base + tok < kv_len ? base + tok
-                                                   : kv_len - 1;
-                const int gp = page_table[(int64_t)slot * max_pages
-                                          + tk / page];
-                const int64_t src = (((int64_t)gp * KVH + kvh) * page
-                                     + tk % page) * DH + dv * 8;
-                const uint4 kv4 = *reinterpret_cast<const uint4*>(kpool + src);
-                const uint4 vv4 = *reinterpret_cast<const uint4*>(vpool + src);
-                *reinterpret_cast<uint4*>(k_tile + tok * KTROW + dv * 8) = kv4;
-                const bf16* vsrc = reinterpret_cast<const bf16*>(&vv4);
-                #pragma unroll
-                for (int e = 0; e < 8; e++)
-                    v_t[(dv * 8 + e) * VTROW + tok] = vsrc[e];
-            }
-        }
-        __syncthreads();
+        (void)n_here;
+        bf16* k_tile = k_tile0 + (ch & 1) * PF_CHUNK * KTROW;
+        bf16* v_t = v_t0 + (ch & 1) * DH * VTROW;
+        // issue next chunk's loads (clamped on the last chunk)
+        issue_chunk(ch + 1 < n_chunks ? (ch + 1) * PF_CHUNK : base);
 
         // ---- S^T[64kv, 32q] = K x Q^T, two 32-kv tiles ----
         f32x16 s[2] = {};
@@ -136,16 +172,32 @@ __global__ __launch_bounds__(512) void k_prefill_attn(
         // ---- causal mask + per-lane online softmax (q = col) ----
         const int qpos = pos0 + col;        // absolute position of this q
         float pmax = -3.0e38f;
-        #pragma unroll
-        for (int t = 0; t < 2; t++)
+        // interior chunks (base+63 <= every row's qpos, no tail) need no
+        // per-element masking — the mask VALU was ~1/3 of the kernel's
+        // non-MFMA instructions (uniform branch: fine)
+        const bool interior = (base + PF_CHUNK <= pos0 + 1)
+                              && (base + PF_CHUNK <= kv_len)
+                              && (rows == PF_ROWS);
+        if (interior) {
             #pragma unroll
-            for (int r = 0; r < 16; r++) {
-                const int kv = base + t * 32 + (r & 3) + 8 * (r >> 2)
-                               + 4 * (lane >> 5);
-                const bool ok = kv < kv_len && kv <= qpos && col < rows;
-                s[t][r] = ok ? s[t][r] * scale : -3.0e38f;
-                pmax = fmaxf(pmax, s[t][r]);
-            }
+            for (int t = 0; t < 2; t++)
+                #pragma unroll
+                for (int r = 0; r < 16; r++) {
+                    s[t][r] *= scale;
+                    pmax = fmaxf(pmax, s[t][r]);
+                }
+        } else {
+            #pragma unroll
+            for (int t = 0; t < 2; t++)
+                #pragma unroll
+                for (int r = 0; r < 16; r++) {
+                    const int kv = base + t * 32 + (r & 3) + 8 * (r >> 2)
+                                   + 4 * (lane >> 5);
+                    const bool ok = kv < kv_len && kv <= qpos && col < rows;
+                    s[t][r] = ok ? s[t][r] * scale : -3.0e38f;
+                    pmax = fmaxf(pmax, s[t][r]);
+                }
+        }
         // lanes l and l+32 each hold HALF of q-row (l&31)'s kv scores
         // (the C layout's 4*(lane>>5) row offset): combine the pair's
         // running max and sum so both halves share one softmax state
@@ -191,18 +243,23 @@ __global__ __launch_bounds__(512) void k_prefill_attn(
         // ---- O^T[d, q] += V^T x P : 4 d-tiles, kv = 64 ----
         #pragma unroll
         for (int t = 0; t < 4; t++) {
-            const bf16* vrow = v_t + (t * 32 + col) * VTROW + khalf;
+            const int d = t * 32 + col;
+            const bf16* vrow = v_t + d * VTROW;
             const bf16* prow = my_p + col * PROW + khalf;
             #pragma unroll
             for (int c = 0; c < 4; c++) {
-                const bf16x8 a =
-                    *reinterpret_cast<const bf16x8*>(vrow + c * 16);
+                const int kv0 = c * 16 + khalf;   // 8-aligned block base
+                const bf16x8 a = *reinterpret_cast<const bf16x8*>(
+                    vrow + VSWZ(d, kv0));
                 const bf16x8 b =
                     *reinterpret_cast<const bf16x8*>(prow + c * 16);
                 o[t] = __builtin_amdgcn_mfma_f32_32x32x16_bf16(
                     a, b, o[t], 0, 0, 0);
             }
         }
+        __syncthreads();
+        if (ch + 1 < n_chunks) write_chunk((ch + 1) & 1);
+        __syncthreads();
     }
 
     // ---- epilogue: normalize and write O (lane's q = col) ----
@@ -230,20 +287,33 @@ extern "C" int prefill_attn_bf16(
     hipStream_t stream)
 {
     const int G = Hq / KVH;
-    const int lds = PF_CHUNK * KTROW * 2 + DH * VTROW * 2
+    const int lds = 2 * PF_CHUNK * KTROW * 2 + 2 * DH * VTROW * 2
                     + G * PF_ROWS * PROW * 2;
-    static int allowed = 0;
-    if (!allowed && lds > 64 * 1024) {
-        (void)hipFuncSetAttribute(
-            (const void*)k_prefill_attn,
-            hipFuncAttributeMaxDynamicSharedMemorySize, 160 * 1024);
-        allowed = 1;
-    }
     dim3 grid(n_tiles, KVH);
-    k_prefill_attn<<<grid, G * 64, lds, stream>>>(
-        (bf16*)out, (const bf16*)q, (const bf16*)kpool, (const bf16*)vpool,
-        (const int*)page_table, (const int*)tile_slot,
-        (const int*)tile_q0, (const int*)tile_pos0, (const int*)tile_rows,
-        Hq, KVH, page, max_pages, scale, q_stride);
+    #define PF_LAUNCH(GG)                                                  \
+        do {                                                               \
+            static int allowed_##GG = 0;                                   \
+            if (!allowed_##GG && lds > 64 * 1024) {                        \
+                (void)hipFuncSetAttribute(                                 \
+                    (const void*)k_prefill_attn<GG>,                       \
+                    hipFuncAttributeMaxDynamicSharedMemorySize,            \
+                    160 * 1024);                                           \
+                allowed_##GG = 1;                                          \
+            }                                                              \
+            k_prefill_attn<GG><<<grid, GG * 64, lds, stream>>>(            \
+                (bf16*)out, (const bf16*)q, (const bf16*)kpool,            \
+                (const bf16*)vpool, (const int*)page_table,                \
+                (const int*)tile_slot, (const int*)tile_q0,                \
+                (const int*)tile_pos0, (const int*)tile_rows,              \
+                Hq, KVH, page, max_pages, scale, q_stride);                \
+        } while (0)
+    switch (G) {
+        case 1: PF_LAUNCH(1); break;
+        case 2: PF_LAUNCH(2); break;
+        case 4: PF_LAUNCH(4); break;
+        case 8: PF_LAUNCH(8); break;
+        default: return (int)hipErrorInvalidValue;
+    }
+    #undef PF_LAUNCH
     return (int)hipGetLastError();
 }
