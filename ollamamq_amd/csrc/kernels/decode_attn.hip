@@ -18,7 +18,6 @@
 
 typedef __hip_bfloat16 bf16;
 typedef __hip_bfloat162 bf162;
-typedef __attribute__((ext_vector_type(8))) __bf16 bf16x8v;
 
 #define DEV static __device__ __forceinline__
 DEV float bf2f(bf16 x) { return __bfloat162float(x); }
@@ -126,17 +125,16 @@ __global__ __launch_bounds__(512) void k_decode_attn(
         float s = 0.f;
         const bool live = lane < n_here;
         {
-            const bf16* krow = k_tile + lane * KROW;   // 16-B aligned row
+            const bf16* krow = k_tile + lane * KROW;
             const float* qrow = q_lds + wid * DHEAD;
             #pragma unroll
             for (int d = 0; d < DHEAD; d += 8) {
-                const bf16x8v kv8 =
-                    *reinterpret_cast<const bf16x8v*>(krow + d);
-                const bf16* k8 = reinterpret_cast<const bf16*>(&kv8);
-                s += bf2f(k8[0]) * qrow[d] + bf2f(k8[1]) * qrow[d + 1]
-                   + bf2f(k8[2]) * qrow[d + 2] + bf2f(k8[3]) * qrow[d + 3]
-                   + bf2f(k8[4]) * qrow[d + 4] + bf2f(k8[5]) * qrow[d + 5]
-                   + bf2f(k8[6]) * qrow[d + 6] + bf2f(k8[7]) * qrow[d + 7];
+                float k8[8];
+                load8f_lds(krow + d, k8);
+                s += k8[0] * qrow[d] + k8[1] * qrow[d + 1]
+                   + k8[2] * qrow[d + 2] + k8[3] * qrow[d + 3]
+                   + k8[4] * qrow[d + 4] + k8[5] * qrow[d + 5]
+                   + k8[6] * qrow[d + 6] + k8[7] * qrow[d + 7];
             }
             s = live ? s * scale : -3.0e38f;
         }
