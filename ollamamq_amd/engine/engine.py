@@ -89,6 +89,15 @@ class LlamaEngine:
         self.use_graphs = (self.dev.type == "cuda"
                            and _os.environ.get("OLLAMAMQ_NO_GRAPH") != "1")
         self._graphs: Dict[int, tuple] = {}
+        # Pipelined decode: the graph samples (greedy argmax) and advances
+        # tok/pos/lens on-device, so steady-state steps skip the H2D input
+        # fill AND the host token sync — the host consumes tokens one step
+        # late through a pinned-buffer/event ring while the GPU runs the
+        # next step.  Drained on any batch-composition change.
+        self.use_pipeline = (self.use_graphs
+                             and _os.environ.get("OLLAMAMQ_NO_PIPELINE")
+                             != "1")
+        self._pipe = None   # {"key": seq-id tuple, "inflight": (ev,buf,seqs)}
 
     # -- submission --------------------------------------------------------
     def submit(self, prompt: List[int], params: GenParams,
@@ -173,6 +182,8 @@ class LlamaEngine:
 
     # -- internals ---------------------------------------------------------
     def _reap_cancelled(self):
+        if any(s.cancelled for s in self.running):
+            self._drain_pipe()
         for lst in (self.waiting, self.running):
             for seq in [s for s in lst if s.cancelled]:
                 lst.remove(seq)
@@ -211,6 +222,7 @@ class LlamaEngine:
         return batch
 
     def _mixed_step(self, batch: List[Sequence]):
+        self._drain_pipe()   # decode rows need current host tokens
         dec = list(self.running)
         token_list = [s.generated[-1] for s in dec]
         pos_list = [s.total_len - 1 for s in dec]
@@ -294,6 +306,7 @@ class LlamaEngine:
         return toks.tolist()
 
     def _prefill_step(self, batch: List[Sequence]):
+        self._drain_pipe()
         token_list, pos_list, q_lens = [], [], []
         for s in batch:
             take = s._chunk
@@ -341,13 +354,15 @@ class LlamaEngine:
             cu_q=torch.arange(B + 1, dtype=torch.int32, device=dev),
             logits_idx=torch.arange(B, dtype=torch.long, device=dev),
             max_q=1, max_kv=self.kv.max_ctx)
-        entry = {"bufs": bufs, "meta": meta, "graph": None, "logits": None}
+        entry = {"bufs": bufs, "meta": meta, "graph": None, "logits": None,
+                 "out": None, "ring_i": 0,
+                 "pinned": [torch.empty(B, dtype=torch.int64,
+                                        pin_memory=True) for _ in range(2)],
+                 "events": [torch.cuda.Event() for _ in range(2)]}
         self._graphs[B] = entry
         return entry
 
-    def _decode_forward_graphed(self, seqs, token_list, pos_list):
-        B = len(seqs)
-        entry = self._graph_entry(B)
+    def _fill_bufs(self, entry, seqs, token_list, pos_list):
         bufs = entry["bufs"]
         host = torch.tensor(
             [token_list, pos_list, [s.slot for s in seqs],
@@ -357,7 +372,9 @@ class LlamaEngine:
         bufs["pos"].copy_(staged[1])
         bufs["slot"].copy_(staged[2])
         bufs["lens"].copy_(staged[3])
-        meta = entry["meta"]
+
+    def _graph_replay(self, entry):
+        bufs, meta = entry["bufs"], entry["meta"]
         if entry["graph"] is None:
             # warmup twice on a side stream, then capture
             s = torch.cuda.Stream()
@@ -371,12 +388,34 @@ class LlamaEngine:
             with torch.cuda.graph(g):
                 entry["logits"] = self.model.forward(
                     bufs["tok"], bufs["pos"], self.kv, bufs["slot"], meta)
+                # self-advancing tail: greedy-sample in-graph and stage the
+                # NEXT step's inputs on-device, so a steady decode batch
+                # replays back-to-back with no host round-trip.  The
+                # non-pipelined/stochastic path simply overwrites the bufs
+                # next step and samples logits itself.
+                toks = ops.sample(entry["logits"], 0.0, 0, 1.0, None)
+                entry["out"] = toks
+                bufs["tok"].copy_(toks)
+                bufs["pos"] += 1
+                bufs["lens"] += 1
             entry["graph"] = g
         entry["graph"].replay()
         return entry["logits"]
 
+    def _decode_forward_graphed(self, seqs, token_list, pos_list):
+        entry = self._graph_entry(len(seqs))
+        self._fill_bufs(entry, seqs, token_list, pos_list)
+        return self._graph_replay(entry)
+
     def _decode_step(self):
         seqs = self.running
+        if (self.use_pipeline
+                and all(s.params.temperature <= 0 for s in seqs)
+                and all(self.kv.seq_lens[s.slot] + 1 <= self.kv.max_ctx
+                        for s in seqs)):
+            self._decode_step_pipelined(seqs)
+            return
+        self._drain_pipe()
         token_list = [s.generated[-1] for s in seqs]
         pos_list = [s.total_len - 1 for s in seqs]
         q_lens = [1] * len(seqs)
@@ -390,6 +429,61 @@ class LlamaEngine:
         toks = self._sample(seqs, logits)
         now = time.monotonic()
         for s, tok in zip(seqs, toks):
+            s.generated.append(int(tok))
+            self.tokens_out += 1
+            if s.first_token_at is None:
+                s.first_token_at = now
+            if s.on_token:
+                s.on_token(int(tok), False)
+
+    # -- pipelined decode --------------------------------------------------
+    def _decode_step_pipelined(self, seqs):
+        """One decode step with the host one token behind the device.
+
+        The graph's tail already advanced tok/pos/lens on-device, so a
+        steady batch needs no input upload; after replay the sampled
+        tokens stream back asynchronously (pinned ring + event) and the
+        PREVIOUS step's tokens are applied while this step runs.
+        """
+        key = tuple(s.seq_id for s in seqs)
+        if self._pipe is not None and self._pipe["key"] != key:
+            self._drain_pipe()
+        # host/device page bookkeeping for the token this replay appends
+        for s in seqs:
+            self.kv.ensure(s.slot, self.kv.seq_lens[s.slot] + 1)
+        entry = self._graph_entry(len(seqs))
+        if self._pipe is None:
+            # (re)sync device input buffers from host truth
+            self._fill_bufs(entry, seqs,
+                            [s.generated[-1] for s in seqs],
+                            [s.total_len - 1 for s in seqs])
+            self._pipe = {"key": key, "inflight": None}
+        self._graph_replay(entry)
+        i = entry["ring_i"]
+        entry["ring_i"] = i ^ 1
+        pinned, ev = entry["pinned"][i], entry["events"][i]
+        pinned.copy_(entry["out"], non_blocking=True)
+        ev.record()
+        prev = self._pipe["inflight"]
+        self._pipe["inflight"] = (ev, pinned, list(seqs))
+        if prev is not None:
+            self._apply_tokens(*prev)
+
+    def _drain_pipe(self):
+        if self._pipe is None:
+            return
+        inflight = self._pipe["inflight"]
+        self._pipe = None
+        if inflight is not None:
+            self._apply_tokens(*inflight)
+
+    def _apply_tokens(self, ev, pinned, seqs):
+        ev.synchronize()
+        toks = pinned.tolist()
+        now = time.monotonic()
+        for s, tok in zip(seqs, toks):
+            if s.state != "running":   # finished/cancelled after issue
+                continue
             s.generated.append(int(tok))
             self.tokens_out += 1
             if s.first_token_at is None:
