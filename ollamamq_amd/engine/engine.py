@@ -409,7 +409,7 @@ class LlamaEngine:
 
     def _decode_step(self):
         seqs = self.running
-        if (self.use_pipeline
+        if (self.use_pipeline and self.use_graphs
                 and all(s.params.temperature <= 0 for s in seqs)
                 and all(self.kv.seq_lens[s.slot] + 1 <= self.kv.max_ctx
                         for s in seqs)):

@@ -191,6 +191,10 @@ def test_forward_e2e_vs_cpu():
         kv = PagedKVCache.for_model(cfg, n_pages=64, max_slots=4,
                                     max_ctx=256, device=device, dtype=dtype)
         eng = LlamaEngine(model, kv, max_batch=4)
+        # pipelined decode samples in-graph and never calls _sample; this
+        # test captures logits through _sample, so use the plain path
+        # (pipelined-vs-plain equivalence is covered by test_pipeline_gpu)
+        eng.use_pipeline = False
         eng.submit(list(range(1, 33)), GenParams(max_tokens=4))
         eng.submit(list(range(40, 52)), GenParams(max_tokens=4))
         logits_log = []
