@@ -47,10 +47,16 @@ DEV float wave_sum(float x) {
 #define DHEAD 128
 #define DKPAD 4
 
-// grid: (n_seqs * split, KVH); block: G*64 threads.
+// grid: (n_seqs * split, KVH); block: G*64 threads (G templated: the
+// next-chunk register-prefetch array must be compile-time sized, guide
+// rule 20 — runtime-indexed arrays land in scratch).
 // Partials: o_part [S][Hq][split][128] f32, ml_part [S][Hq][split][2] f32.
 // When split == 1, writes normalized bf16 straight to out.
-__global__ __launch_bounds__(512) void k_decode_attn(
+// launch_bounds min-waves/EU = 1: LDS (~37 KB/block) already caps
+// residency at 4 blocks/CU, so a high compiler occupancy target only
+// forces the prefetch registers to spill (measured 144 B/lane scratch)
+template <int GT>
+__global__ __launch_bounds__(GT * 64, 1) void k_decode_attn(
     bf16* __restrict__ out,               // [S, Hq, D] (split==1)
     float* __restrict__ o_part,           // split>1 partials
     float* __restrict__ ml_part,
@@ -66,18 +72,24 @@ __global__ __launch_bounds__(512) void k_decode_attn(
     const int S_idx = blockIdx.x / split;
     const int seg = blockIdx.x % split;
     const int kvh = blockIdx.y;
-    const int G = Hq / KVH;
+    constexpr int G = GT;
     const int wid = threadIdx.x >> 6;
     const int lane = threadIdx.x & 63;
     const int qh = kvh * G + wid;
 
+    constexpr int NTHR = GT * 64;
+    constexpr int NSLOT = DCHUNK * (DHEAD / 8);   // uint4 slots per tile
+    constexpr int NPF = NSLOT / NTHR;             // prefetch regs / thread
+    constexpr bool PF = (NPF <= 4);               // G>=4: T14 double-buffer
+
     const int slot = slot_ids[S_idx];
     const int kv_len = seq_lens[S_idx];
-    // segment bounds in whole chunks so pages don't straddle segments
+    // balanced segment bounds in whole chunks (pages don't straddle
+    // chunks): ceil-div per_seg gave e.g. 9 chunks @ split 4 -> 3,3,3,0
+    // with a whole workgroup idle; floor-interpolated bounds give 2,2,2,3
     const int n_chunks = (kv_len + DCHUNK - 1) / DCHUNK;
-    const int per_seg = (n_chunks + split - 1) / split;
-    const int c0 = seg * per_seg;
-    const int c1 = min(n_chunks, c0 + per_seg);
+    const int c0 = (int)(((int64_t)n_chunks * seg) / split);
+    const int c1 = (int)(((int64_t)n_chunks * (seg + 1)) / split);
 
     extern __shared__ __attribute__((aligned(16))) char smem[];
     const int KROW = DHEAD + DKPAD;
@@ -96,30 +108,59 @@ __global__ __launch_bounds__(512) void k_decode_attn(
 
     float m = -3.0e38f, l = 0.f, o0 = 0.f, o1 = 0.f;
 
+    // clamp instead of guarding: a conditional load would de-pipeline the
+    // whole stage (guide §5 traps (c)); the duplicated tail rows are
+    // masked to p=0 by the softmax
+    auto src_of = [&](int base, int u) -> int64_t {
+        const int tok = u / (DHEAD / 8), dv = u % (DHEAD / 8);
+        const int tk = base + tok < kv_len ? base + tok : kv_len - 1;
+        const int gp = page_table[(int64_t)slot * max_pages + tk / page];
+        return (((int64_t)gp * KVH + kvh) * page + tk % page) * DHEAD
+               + dv * 8;
+    };
+    auto stage_direct = [&](int ch2) {
+        const int base2 = ch2 * DCHUNK;
+        for (int u = threadIdx.x; u < NSLOT; u += NTHR) {
+            const int tok = u / (DHEAD / 8), dv = u % (DHEAD / 8);
+            const int64_t src = src_of(base2, u);
+            *reinterpret_cast<uint4*>(k_tile + tok * KROW + dv * 8) =
+                *reinterpret_cast<const uint4*>(kpool + src);
+            *reinterpret_cast<uint4*>(v_tile + tok * KROW + dv * 8) =
+                *reinterpret_cast<const uint4*>(vpool + src);
+        }
+    };
+
+    if (c0 < c1) stage_direct(c0);
+    __syncthreads();
+
     for (int ch = c0; ch < c1; ch++) {
         const int base = ch * DCHUNK;
         const int n_here = min(DCHUNK, kv_len - base);
-        // ---- cooperative K/V chunk stage (4 pages @ page=16) ----
-        {
-            const int tid = threadIdx.x, nthr = blockDim.x;
-            for (int u = tid; u < DCHUNK * (DHEAD / 8); u += nthr) {
-                const int tok = u / (DHEAD / 8), dv = u % (DHEAD / 8);
-                // clamp instead of guarding: a conditional load would
-                // de-pipeline the whole stage (guide §5 traps (c)); the
-                // duplicated tail rows are masked to p=0 by the softmax
-                const int tk = base + tok < kv_len ? base + tok
-                                                   : kv_len - 1;
-                const int gp = page_table[(int64_t)slot * max_pages
-                                          + tk / page];
-                const int64_t src = (((int64_t)gp * KVH + kvh) * page
-                                     + tk % page) * DHEAD + dv * 8;
-                *reinterpret_cast<uint4*>(k_tile + tok * KROW + dv * 8) =
-                    *reinterpret_cast<const uint4*>(kpool + src);
-                *reinterpret_cast<uint4*>(v_tile + tok * KROW + dv * 8) =
-                    *reinterpret_cast<const uint4*>(vpool + src);
+        // ---- issue next chunk's HBM loads NOW, write to LDS after the
+        // compute phase (T14 issue-early/write-late): the fetch rides
+        // under the score/PV work instead of serializing with it.
+        // Named registers, not an array — hipcc demotes even
+        // constant-indexed uint4 arrays here to scratch (24 scratch ops,
+        // 144 B/lane), and a spilled prefetch is HBM traffic, not a win.
+        uint4 k0, k1, k2, k3, v0, v1, v2, v3;
+        const bool have_next = PF && (ch + 1 < c1);
+        if (have_next) {
+            const int nbase = (ch + 1) * DCHUNK;
+            const int64_t s0 = src_of(nbase, threadIdx.x);
+            const int64_t s1 = src_of(nbase, threadIdx.x + NTHR);
+            k0 = *reinterpret_cast<const uint4*>(kpool + s0);
+            v0 = *reinterpret_cast<const uint4*>(vpool + s0);
+            k1 = *reinterpret_cast<const uint4*>(kpool + s1);
+            v1 = *reinterpret_cast<const uint4*>(vpool + s1);
+            if constexpr (NPF >= 4) {
+                const int64_t s2 = src_of(nbase, threadIdx.x + 2 * NTHR);
+                const int64_t s3 = src_of(nbase, threadIdx.x + 3 * NTHR);
+                k2 = *reinterpret_cast<const uint4*>(kpool + s2);
+                v2 = *reinterpret_cast<const uint4*>(vpool + s2);
+                k3 = *reinterpret_cast<const uint4*>(kpool + s3);
+                v3 = *reinterpret_cast<const uint4*>(vpool + s3);
             }
         }
-        __syncthreads();
 
         // ---- score for key j = lane ----
         float s = 0.f;
@@ -165,7 +206,26 @@ __global__ __launch_bounds__(512) void k_decode_attn(
             o0 = fmaf(pj, bf2f(v2.x), o0);
             o1 = fmaf(pj, bf2f(v2.y), o1);
         }
-        __syncthreads();  // protect k/v tiles before next stage
+        __syncthreads();  // every wave done reading the k/v tiles
+        if (have_next) {
+            auto put = [&](int u, uint4 kx, uint4 vx) {
+                const int tok = u / (DHEAD / 8), dv = u % (DHEAD / 8);
+                *reinterpret_cast<uint4*>(k_tile + tok * KROW + dv * 8)
+                    = kx;
+                *reinterpret_cast<uint4*>(v_tile + tok * KROW + dv * 8)
+                    = vx;
+            };
+            put(threadIdx.x, k0, v0);
+            put(threadIdx.x + NTHR, k1, v1);
+            if constexpr (NPF >= 4) {
+                put(threadIdx.x + 2 * NTHR, k2, v2);
+                put(threadIdx.x + 3 * NTHR, k3, v3);
+            }
+            __syncthreads();  // tiles ready for next iteration
+        } else if (!PF && ch + 1 < c1) {
+            stage_direct(ch + 1);
+            __syncthreads();
+        }
     }
 
     // ---- emit ----
@@ -223,11 +283,20 @@ extern "C" int decode_attn_bf16(
     const int lds = 2 * DCHUNK * (DHEAD + DKPAD) * 2 + G * DHEAD * 4
                   + G * (DCHUNK + 1) * 4;
     dim3 grid(S * split, KVH);
-    k_decode_attn<<<grid, G * 64, lds, stream>>>(
-        (bf16*)out, (float*)o_part, (float*)ml_part, (const bf16*)q,
-        (const bf16*)kpool, (const bf16*)vpool, (const int*)page_table,
-        (const int*)slot_ids, (const int*)seq_lens, Hq, KVH, page,
-        max_pages, scale, q_stride, split);
+#define DA_LAUNCH(GT)                                                     \
+    k_decode_attn<GT><<<grid, GT * 64, lds, stream>>>(                    \
+        (bf16*)out, (float*)o_part, (float*)ml_part, (const bf16*)q,      \
+        (const bf16*)kpool, (const bf16*)vpool, (const int*)page_table,   \
+        (const int*)slot_ids, (const int*)seq_lens, Hq, KVH, page,        \
+        max_pages, scale, q_stride, split)
+    switch (G) {
+        case 1: DA_LAUNCH(1); break;
+        case 2: DA_LAUNCH(2); break;
+        case 4: DA_LAUNCH(4); break;
+        case 8: DA_LAUNCH(8); break;
+        default: return (int)hipErrorInvalidValue;
+    }
+#undef DA_LAUNCH
     if (split > 1) {
         const int waves = S * Hq;
         k_decode_combine<<<(waves + 3) / 4, 256, 0, stream>>>(

@@ -239,6 +239,9 @@ def attention_decode(q, cache, layer, meta):
     # max_ctx, so an unconditional max-split would freeze in dozens of
     # empty segments per sequence (measured: 5.55k -> 4.3k tok/s)
     want = max(1, 1024 // max(1, S * kvh))
+    env = os.environ.get("OLLAMAMQ_DECODE_SPLIT")
+    if env:                       # tuning override (tools/perf_decode.py)
+        want = int(env)
     max_seg = max(1, (meta.max_kv + 63) // 64)
     split = int(min(want, max_seg, 32))
     o_part, ml_part = (None, None)

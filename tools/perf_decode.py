@@ -19,7 +19,10 @@ def main():
     hip.require()
     ctx = int(sys.argv[1]) if len(sys.argv) > 1 else 1024
     B = int(sys.argv[2]) if len(sys.argv) > 2 else 32
-    Hq, KVH, D, L = 32, 8, 128, 1
+    # 8 layers cycled per call: the working set (8 × B×KVH×ctx×512B) blows
+    # past the 256 MiB L3 like the real 32-layer model does — timing one
+    # resident layer would measure L3, not HBM (the perf_gemm.py lesson)
+    Hq, KVH, D, L = 32, 8, 128, 8
     dev = "cuda"
     n_pages = B * (ctx // 16 + 2)
     cache = PagedKVCache(L, KVH, D, page_size=16, n_pages=n_pages,
@@ -42,19 +45,20 @@ def main():
 
     def bench(fn, label):
         for _ in range(5):
-            out = fn()
+            out = fn(0)
         torch.cuda.synchronize()
         t0 = time.perf_counter()
-        n = 50
-        for _ in range(n):
-            out = fn()
+        n = 64
+        for i in range(n):
+            out = fn(i % L)     # cycle layers: cold KV every call
         torch.cuda.synchronize()
         dt = (time.perf_counter() - t0) / n
         print(f"{label}: {dt*1e6:8.1f} us  {kv_bytes/dt/1e12:6.2f} TB/s")
         return out
 
-    o_new = bench(lambda: hip.attention_decode(q, cache, 0, meta), "split ")
-    o_old = bench(lambda: hip._attention(q, cache, 0, meta, 1), "legacy")
+    o_new = bench(lambda l: hip.attention_decode(q, cache, l, meta),
+                  "split ")
+    o_old = bench(lambda l: hip._attention(q, cache, l, meta, 1), "legacy")
     diff = (o_new.float() - o_old.float()).abs().max().item()
     print("max |new-old| =", diff)
     assert diff < 3e-2, "split path disagrees with legacy"
