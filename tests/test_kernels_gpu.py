@@ -140,6 +140,26 @@ def test_attention_decode():
                                atol=3e-2, rtol=3e-2)
 
 
+@pytest.mark.parametrize("Hq,KVH", [(2, 2), (4, 2), (8, 2), (16, 2)])
+def test_attention_decode_gqa_groups(Hq, KVH):
+    """All decode-kernel template instantiations (GQA group G=Hq/KVH in
+    {1,2,4,8}: G>=4 takes the register-prefetch pipeline, G<4 the direct
+    staging path) against the fp32 reference, ragged lengths."""
+    hip = _hip()
+    D = 128
+    lens = [3, 64, 200, 516]
+    gc, cc = make_caches(KVH=KVH, n_pages=256, ctx=1024)
+    fill_caches(gc, cc, lens, KVH=KVH)
+    S = len(lens)
+    q = rnd(S, Hq, D, seed=100 + Hq)
+    meta_g = _meta(dev(), list(range(S)), lens, [1] * S)
+    meta_c = _meta("cpu", list(range(S)), lens, [1] * S)
+    out = hip.attention_decode(q, gc, 1, meta_g)
+    out_ref = ref.attention(q.float().cpu(), cc, 1, meta_c)
+    torch.testing.assert_close(out.float().cpu(), out_ref,
+                               atol=3e-2, rtol=3e-2)
+
+
 def test_attention_prefill_varlen():
     hip = _hip()
     Hq, KVH, D = 4, 2, 128
