@@ -67,7 +67,7 @@ __global__ __launch_bounds__(GT * 64, 1) void k_decode_attn(
     const int* __restrict__ slot_ids,     // [S]
     const int* __restrict__ seq_lens,     // [S] (kv length incl. this tok)
     int Hq, int KVH, int page, int max_pages, float scale,
-    int64_t qs, int split)
+    int64_t qs, int split, int window)    // window 0 = full causal
 {
     const int S_idx = blockIdx.x / split;
     const int seg = blockIdx.x % split;
@@ -88,8 +88,13 @@ __global__ __launch_bounds__(GT * 64, 1) void k_decode_attn(
     // chunks): ceil-div per_seg gave e.g. 9 chunks @ split 4 -> 3,3,3,0
     // with a whole workgroup idle; floor-interpolated bounds give 2,2,2,3
     const int n_chunks = (kv_len + DCHUNK - 1) / DCHUNK;
-    const int c0 = (int)(((int64_t)n_chunks * seg) / split);
-    const int c1 = (int)(((int64_t)n_chunks * (seg + 1)) / split);
+    // sliding window: the (single) query sits at kv_len-1, so only keys
+    // >= lo_tok participate; split the ACTIVE chunk range
+    const int lo_tok = (window > 0 && kv_len > window) ? kv_len - window : 0;
+    const int ch_base = lo_tok / DCHUNK;
+    const int n_act = n_chunks - ch_base;
+    const int c0 = ch_base + (int)(((int64_t)n_act * seg) / split);
+    const int c1 = ch_base + (int)(((int64_t)n_act * (seg + 1)) / split);
 
     extern __shared__ __attribute__((aligned(16))) char smem[];
     const int KROW = DHEAD + DKPAD;
@@ -164,7 +169,7 @@ __global__ __launch_bounds__(GT * 64, 1) void k_decode_attn(
 
         // ---- score for key j = lane ----
         float s = 0.f;
-        const bool live = lane < n_here;
+        const bool live = lane < n_here && base + lane >= lo_tok;
         {
             const bf16* krow = k_tile + lane * KROW;
             const float* qrow = q_lds + wid * DHEAD;
@@ -277,7 +282,7 @@ extern "C" int decode_attn_bf16(
     const void* kpool, const void* vpool, const void* page_table,
     const void* slot_ids, const void* seq_lens, int S, int Hq, int KVH,
     int page, int max_pages, float scale, int64_t q_stride, int split,
-    hipStream_t stream)
+    int window, hipStream_t stream)
 {
     const int G = Hq / KVH;
     const int lds = 2 * DCHUNK * (DHEAD + DKPAD) * 2 + G * DHEAD * 4
@@ -288,7 +293,7 @@ extern "C" int decode_attn_bf16(
         (bf16*)out, (float*)o_part, (float*)ml_part, (const bf16*)q,      \
         (const bf16*)kpool, (const bf16*)vpool, (const int*)page_table,   \
         (const int*)slot_ids, (const int*)seq_lens, Hq, KVH, page,        \
-        max_pages, scale, q_stride, split)
+        max_pages, scale, q_stride, split, window)
     switch (G) {
         case 1: DA_LAUNCH(1); break;
         case 2: DA_LAUNCH(2); break;

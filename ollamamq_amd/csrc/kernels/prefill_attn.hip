@@ -56,7 +56,8 @@ __global__ __launch_bounds__(512) void k_prefill_attn(
     const int* __restrict__ tile_q0,
     const int* __restrict__ tile_pos0,
     const int* __restrict__ tile_rows,
-    int Hq, int KVH, int page, int max_pages, float scale, int64_t qs)
+    int Hq, int KVH, int page, int max_pages, float scale, int64_t qs,
+    int window)                           // 0 = full causal
 {
     const int tile = blockIdx.x;
     const int kvh = blockIdx.y;
@@ -143,10 +144,14 @@ __global__ __launch_bounds__(512) void k_prefill_attn(
     };
 
     const int n_chunks = (kv_len + PF_CHUNK - 1) / PF_CHUNK;
-    issue_chunk(0);
-    write_chunk(0);
+    // sliding window: the earliest key any q row of this tile can see is
+    // pos0 - window + 1 (row 0); skip chunks entirely below it
+    const int ch_lo = (window > 0 && pos0 >= window)
+                          ? (pos0 - window + 1) / PF_CHUNK : 0;
+    issue_chunk(ch_lo * PF_CHUNK);
+    write_chunk(ch_lo & 1);
     __syncthreads();
-    for (int ch = 0; ch < n_chunks; ch++) {
+    for (int ch = ch_lo; ch < n_chunks; ch++) {
         const int base = ch * PF_CHUNK;
         const int n_here = min(PF_CHUNK, kv_len - base);
         (void)n_here;
@@ -177,7 +182,9 @@ __global__ __launch_bounds__(512) void k_prefill_attn(
         // non-MFMA instructions (uniform branch: fine)
         const bool interior = (base + PF_CHUNK <= pos0 + 1)
                               && (base + PF_CHUNK <= kv_len)
-                              && (rows == PF_ROWS);
+                              && (rows == PF_ROWS)
+                              && (window <= 0
+                                  || base >= pos0 + rows - window);
         if (interior) {
             #pragma unroll
             for (int t = 0; t < 2; t++)
@@ -193,7 +200,8 @@ __global__ __launch_bounds__(512) void k_prefill_attn(
                 for (int r = 0; r < 16; r++) {
                     const int kv = base + t * 32 + (r & 3) + 8 * (r >> 2)
                                    + 4 * (lane >> 5);
-                    const bool ok = kv < kv_len && kv <= qpos && col < rows;
+                    const bool ok = kv < kv_len && kv <= qpos && col < rows
+                                    && (window <= 0 || kv > qpos - window);
                     s[t][r] = ok ? s[t][r] * scale : -3.0e38f;
                     pmax = fmaxf(pmax, s[t][r]);
                 }
@@ -284,7 +292,7 @@ extern "C" int prefill_attn_bf16(
     const void* page_table, const void* tile_slot, const void* tile_q0,
     const void* tile_pos0, const void* tile_rows, int n_tiles,
     int Hq, int KVH, int page, int max_pages, float scale, int64_t q_stride,
-    hipStream_t stream)
+    int window, hipStream_t stream)
 {
     const int G = Hq / KVH;
     const int lds = 2 * PF_CHUNK * KTROW * 2 + 2 * DH * VTROW * 2
@@ -305,7 +313,7 @@ extern "C" int prefill_attn_bf16(
                 (const bf16*)vpool, (const int*)page_table,                \
                 (const int*)tile_slot, (const int*)tile_q0,                \
                 (const int*)tile_pos0, (const int*)tile_rows,              \
-                Hq, KVH, page, max_pages, scale, q_stride);                \
+                Hq, KVH, page, max_pages, scale, q_stride, window);        \
         } while (0)
     switch (G) {
         case 1: PF_LAUNCH(1); break;

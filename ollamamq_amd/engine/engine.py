@@ -173,7 +173,8 @@ class LlamaEngine:
                                   device=dev),
                 logits_idx=torch.tensor([len(prompt) - 1], dtype=torch.long,
                                         device=dev),
-                max_q=len(prompt), max_kv=len(prompt))
+                max_q=len(prompt), max_kv=len(prompt),
+                window=getattr(self.model.cfg, "sliding_window", 0))
             h = self.model.forward(tokens, positions, self.kv, slot_t, meta,
                                    return_hidden=True)
             return h[0].float().tolist()
@@ -278,6 +279,7 @@ class LlamaEngine:
             logits_idx=logits_idx, max_q=max(q_lens),
             max_kv=int(max(self.kv.seq_lens[s.slot] for s in seqs)),
             n_decode=n_decode,
+            window=getattr(self.model.cfg, "sliding_window", 0),
         )
 
     def _forward(self, seqs, token_list, pos_list, q_lens, mode,
@@ -353,7 +355,8 @@ class LlamaEngine:
             mode="decode", slot_ids=bufs["slot"], seq_lens=bufs["lens"],
             cu_q=torch.arange(B + 1, dtype=torch.int32, device=dev),
             logits_idx=torch.arange(B, dtype=torch.long, device=dev),
-            max_q=1, max_kv=self.kv.max_ctx)
+            max_q=1, max_kv=self.kv.max_ctx,
+            window=getattr(self.model.cfg, "sliding_window", 0))
         entry = {"bufs": bufs, "meta": meta, "graph": None, "logits": None,
                  "out": None, "ring_i": 0,
                  "pinned": [torch.empty(B, dtype=torch.int64,

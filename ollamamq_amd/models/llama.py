@@ -38,6 +38,10 @@ class LlamaConfig:
     norm_eps: float = 1e-5
     rope_theta: float = 500000.0
     max_ctx: int = 8192
+    # family variants sharing the Llama block: Qwen2 adds a bias on the
+    # QKV projection; Mistral restricts attention to a sliding window
+    qkv_bias: bool = False
+    sliding_window: int = 0       # 0 = full causal
 
     @property
     def q_dim(self) -> int:
@@ -56,6 +60,17 @@ PRESETS = {
         name="llama3-70b", n_layers=80, hidden=8192, n_heads=64, n_kv_heads=8,
         ffn=28672,
     ),
+    # Qwen2-7B: QKV bias, GQA 28Q/4KV, rope theta 1e6, vocab 152064
+    "qwen2-7b": LlamaConfig(
+        name="qwen2-7b", n_layers=28, hidden=3584, n_heads=28, n_kv_heads=4,
+        ffn=18944, vocab=152064, rope_theta=1_000_000.0, qkv_bias=True,
+    ),
+    # Mistral-7B: sliding-window attention (W=4096), vocab 32000
+    "mistral-7b": LlamaConfig(
+        name="mistral-7b", n_layers=32, hidden=4096, n_heads=32,
+        n_kv_heads=8, ffn=14336, vocab=32000, rope_theta=10000.0,
+        sliding_window=4096,
+    ),
     # Tiny config for CPU tests and fast GPU smoke: same head_dim=128 the
     # kernels are specialized for.
     "tiny": LlamaConfig(
@@ -66,6 +81,18 @@ PRESETS = {
     "tiny-cpu": LlamaConfig(
         name="tiny-cpu", n_layers=2, hidden=256, n_heads=2, n_kv_heads=1,
         head_dim=128, ffn=512, vocab=256, max_ctx=256, rope_theta=10000.0,
+    ),
+    # Tiny family-variant configs exercising the Qwen2 bias path and the
+    # Mistral sliding-window path end to end (CPU tests + GPU smoke).
+    "tiny-qwen": LlamaConfig(
+        name="tiny-qwen", n_layers=2, hidden=512, n_heads=4, n_kv_heads=2,
+        head_dim=128, ffn=1024, vocab=512, max_ctx=512, rope_theta=10000.0,
+        qkv_bias=True,
+    ),
+    "tiny-swa": LlamaConfig(
+        name="tiny-swa", n_layers=2, hidden=512, n_heads=4, n_kv_heads=2,
+        head_dim=128, ffn=1024, vocab=512, max_ctx=512, rope_theta=10000.0,
+        sliding_window=96,
     ),
 }
 
@@ -87,7 +114,7 @@ class LlamaLayer:
     CPU tests assert logits parity across degrees)."""
 
     __slots__ = (
-        "wqkv", "wo", "wgate_up", "wdown", "attn_norm", "mlp_norm",
+        "wqkv", "bqkv", "wo", "wgate_up", "wdown", "attn_norm", "mlp_norm",
     )
 
     def __init__(self, cfg: LlamaConfig, dev, dtype, gen, tp: int, rank: int):
@@ -105,6 +132,16 @@ class LlamaLayer:
                    (Hq + KVH) * d + (rank + 1) * nkv * d]
         self.wqkv = torch.cat([q, k, v], dim=0).contiguous()
         del wqkv_g
+        # Qwen2-style QKV bias, sharded with the same row split
+        self.bqkv = None
+        if cfg.qkv_bias:
+            b_g = _randn(((Hq + 2 * KVH) * d,), dev, dtype, gen, scale)
+            bq = b_g[rank * nh * d:(rank + 1) * nh * d]
+            bk = b_g[Hq * d + rank * nkv * d: Hq * d + (rank + 1) * nkv * d]
+            bv = b_g[(Hq + KVH) * d + rank * nkv * d:
+                     (Hq + KVH) * d + (rank + 1) * nkv * d]
+            self.bqkv = torch.cat([bq, bk, bv], dim=0).contiguous()
+            del b_g
         # row-parallel output projection: columns of the global [h, Hq*d]
         wo_g = _randn((h, Hq * d), dev, dtype, gen, scale)
         self.wo = wo_g[:, rank * nh * d:(rank + 1) * nh * d].contiguous()
@@ -193,6 +230,7 @@ class LlamaModel:
             n += (
                 l.wqkv.numel() + l.wo.numel() + l.wgate_up.numel()
                 + l.wdown.numel() + l.attn_norm.numel() + l.mlp_norm.numel()
+                + (l.bqkv.numel() if l.bqkv is not None else 0)
             )
         return n * self.embed.element_size()
 
@@ -215,7 +253,7 @@ class LlamaModel:
             normed, residual = ops.rmsnorm_residual(
                 x, residual, layer.attn_norm, cfg.norm_eps
             )
-            qkv = ops.linear(normed, layer.wqkv)
+            qkv = ops.linear(normed, layer.wqkv, layer.bqkv)
             nl, nkl, d = self.n_local_heads, self.n_local_kv_heads, cfg.head_dim
             q, k, v = qkv.split([nl * d, nkl * d, nkl * d], dim=-1)
             q = q.view(-1, nl, d)

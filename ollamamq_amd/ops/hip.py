@@ -45,13 +45,13 @@ def _try_load():
     lib.swiglu_bf16.argtypes = [vp, vp, i, i, vp]
     lib.argmax_bf16.argtypes = [vp, vp, i, i, vp]
     lib.decode_attn_bf16.argtypes = [vp, vp, vp, vp, vp, vp, vp, vp, vp,
-                                     i, i, i, i, i, f, i64, i, vp]
+                                     i, i, i, i, i, f, i64, i, i, vp]
     lib.skinny_gemm_bf16.argtypes = [vp, vp, vp, vp, i, i, i, i64, i, vp]
     lib.skinny_direct_bf16.argtypes = [vp, vp, vp, vp, i, i, i, i64, i, vp]
     lib.rope_append_bf16.argtypes = [vp, vp, vp, vp, vp, vp, vp, vp, vp, vp,
                                      i, i, i, i, i, i, i64, i64, i64, vp]
     lib.prefill_attn_bf16.argtypes = [vp, vp, vp, vp, vp, vp, vp, vp, vp,
-                                      i, i, i, i, i, f, i64, vp]
+                                      i, i, i, i, i, f, i64, i, vp]
     for fn in ("rmsnorm_residual_bf16", "rope_bf16", "kv_append_bf16",
                "paged_attn_bf16", "swiglu_bf16", "argmax_bf16",
                "decode_attn_bf16", "skinny_gemm_bf16",
@@ -257,7 +257,7 @@ def attention_decode(q, cache, layer, meta):
         _p(out), op, mp, _p(q), kp, vp, _p(cache.page_table),
         _p(slot32), _p(len32), S, Hq, kvh, cache.page_size,
         cache.page_table.shape[1], 1.0 / (D ** 0.5), _row_stride(q, D),
-        split, _stream()), "decode_attn")
+        split, meta.window, _stream()), "decode_attn")
     return out
 
 
@@ -275,7 +275,7 @@ def attention_mixed(q, cache, layer, meta):
             mode="decode", slot_ids=meta.slot_ids[:nd],
             seq_lens=meta.seq_lens[:nd],
             cu_q=meta.cu_q[:nd + 1], logits_idx=None, max_q=1,
-            max_kv=meta.max_kv)
+            max_kv=meta.max_kv, window=meta.window)
         meta._mixed_dec = sub
     if nd:
         out[:nd] = attention_decode(q[:nd], cache, layer, sub)
@@ -309,14 +309,15 @@ def attention_mixed(q, cache, layer, meta):
             _p(tile_slot), _p(tile_q0), _p(tile_pos0), _p(tile_rows),
             n_tiles, Hq, cache.n_kv_heads, cache.page_size,
             cache.page_table.shape[1], 1.0 / (D ** 0.5),
-            _row_stride(q, D), _stream()), "prefill_attn")
+            _row_stride(q, D), meta.window, _stream()), "prefill_attn")
     return out
 
 
 def attention_prefill(q, cache, layer, meta):
     """MFMA flash prefill (prefill_attn.hip); set OLLAMAMQ_VALU_PREFILL=1
     to fall back to the VALU paged_attn path (A/B + debugging)."""
-    if os.environ.get("OLLAMAMQ_VALU_PREFILL") == "1":
+    if os.environ.get("OLLAMAMQ_VALU_PREFILL") == "1" \
+            and not meta.window:   # VALU fallback predates sliding window
         return _attention(q, cache, layer, meta, 16)
     T, Hq, D = q.shape
     assert D == 128
@@ -332,7 +333,7 @@ def attention_prefill(q, cache, layer, meta):
         _p(tile_slot), _p(tile_q0), _p(tile_pos0), _p(tile_rows),
         n_tiles, Hq, cache.n_kv_heads, cache.page_size,
         cache.page_table.shape[1], 1.0 / (D ** 0.5), _row_stride(q, D),
-        _stream()), "prefill_attn")
+        meta.window, _stream()), "prefill_attn")
     return out
 
 
@@ -358,12 +359,12 @@ def _skinny_ksplit(N, K):
     return ks
 
 
-def linear(x, weight):
+def linear(x, weight, bias=None):
     M, K = x.shape
     N = weight.shape[0]
     # gated to the shapes where the hand-written kernel beats hipBLASLt
     # (measured tools/perf_gemm.py); widen via env as the kernel improves
-    if M <= 32 and K % 256 == 0 and N % 32 == 0 \
+    if bias is None and M <= 32 and K % 256 == 0 and N % 32 == 0 \
             and N <= SKINNY_MAX_N and K <= SKINNY_MAX_K \
             and x.dtype == torch.bfloat16 and weight.stride(1) == 1:
         y = torch.empty((M, N), dtype=x.dtype, device=x.device)
@@ -392,7 +393,7 @@ def linear(x, weight):
         _check(fn(_p(y), part, _p(x), _p(weight),
                   M, N, K, x.stride(0), ks, _stream()), "skinny_gemm")
         return y
-    return torch.nn.functional.linear(x, weight)
+    return torch.nn.functional.linear(x, weight, bias)
 
 
 def swiglu(gate_up):

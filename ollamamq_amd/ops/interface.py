@@ -47,6 +47,9 @@ class AttnMeta:
     max_q: int
     max_kv: int
     n_decode: int = 0
+    # sliding-window attention (Mistral-style): query at position p attends
+    # keys [max(0, p - window + 1), p]; 0 = full causal
+    window: int = 0
 
 
 def embedding(tokens: torch.Tensor, table: torch.Tensor) -> torch.Tensor:
@@ -98,14 +101,14 @@ def attention(q, cache, layer, meta: AttnMeta):
     return ref.attention(q, cache, layer, meta)
 
 
-def linear(x, weight):
-    """y = x @ weight.T — library GEMM normally; hand-written MFMA skinny
-    GEMM on the decode path (M<=32), where hipBLASLt is far off the
+def linear(x, weight, bias=None):
+    """y = x @ weight.T (+ bias) — library GEMM normally; hand-written MFMA
+    skinny GEMM on the decode path (M<=32), where hipBLASLt is far off the
     weights-streaming roofline."""
     if _use_hip(x):
         from . import hip
-        return hip.linear(x, weight)
-    return torch.nn.functional.linear(x, weight)
+        return hip.linear(x, weight, bias)
+    return torch.nn.functional.linear(x, weight, bias)
 
 
 def swiglu(gate_up):

@@ -93,6 +93,9 @@ def attention(q, cache, layer, meta):
         qpos = torch.arange(kv_len - qlen, kv_len, device=q.device)
         kpos = torch.arange(kv_len, device=q.device)
         mask = kpos[None, :] > qpos[:, None]
+        if meta.window and meta.window > 0:
+            # sliding window: query p attends [max(0, p-window+1), p]
+            mask |= kpos[None, :] < qpos[:, None] - meta.window + 1
         scores.masked_fill_(mask.unsqueeze(0), float("-inf"))
         p = torch.softmax(scores, dim=-1)
         o = torch.einsum("hqk,khd->qhd", p, v)
