@@ -305,7 +305,23 @@ class LlamaEngine:
         top_k = max((s.params.top_k or 0) for s in seqs)
         top_p = min((s.params.top_p if s.params.top_p else 1.0) for s in seqs)
         toks = ops.sample(logits, temps, top_k, top_p, self._gen)
-        return toks.tolist()
+        out = toks.tolist()
+        # per-request `seed` (Ollama options.seed): stochastic rows with a
+        # seed re-sample alone under a generator keyed (seed, step-index)
+        # so the same request reproduces regardless of batch composition
+        for i, s in enumerate(seqs):
+            p = s.params
+            if p.seed is not None and p.temperature > 0:
+                g = torch.Generator(device=logits.device)
+                g.manual_seed((int(p.seed) << 20)
+                              ^ (len(s.generated) + len(s.prompt)))
+                t = ops.sample(logits[i:i + 1],
+                               torch.tensor([p.temperature],
+                                            dtype=torch.float32,
+                                            device=logits.device),
+                               p.top_k or 0, p.top_p or 1.0, g)
+                out[i] = int(t[0])
+        return out
 
     def _prefill_step(self, batch: List[Sequence]):
         self._drain_pipe()

@@ -114,3 +114,29 @@ def test_stop_token():
     run_all(eng)
     assert seq2.finish_reason == "stop"
     assert seq2.generated[-1] == stop
+
+
+def test_per_request_seed_reproducible():
+    """options.seed (Ollama parity): a seeded stochastic request yields
+    the same tokens regardless of what else shares the batch."""
+    from ollamamq_amd.engine import GenParams as GP
+
+    def run(extra):
+        eng = make_engine()
+        if extra:
+            eng.submit([9, 9, 9], GP(max_tokens=6, temperature=1.0))
+        sid = eng.submit([1, 2, 3], GP(max_tokens=6, temperature=0.9,
+                                       seed=1234))
+        seq = eng.seqs[sid]
+        run_all(eng)
+        return seq.generated
+
+    a = run(extra=False)
+    b = run(extra=True)
+    assert a == b and len(a) == 6
+
+    eng = make_engine()
+    sid = eng.seqs[eng.submit([1, 2, 3], GP(max_tokens=6, temperature=0.9,
+                                            seed=77))]
+    run_all(eng)
+    assert sid.generated != a  # different seed, different trajectory
