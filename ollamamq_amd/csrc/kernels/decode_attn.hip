@@ -67,7 +67,8 @@ __global__ __launch_bounds__(GT * 64, 1) void k_decode_attn(
     const int* __restrict__ slot_ids,     // [S]
     const int* __restrict__ seq_lens,     // [S] (kv length incl. this tok)
     int Hq, int KVH, int page, int max_pages, float scale,
-    int64_t qs, int split, int window)    // window 0 = full causal
+    int64_t qs, int split, int window,    // window 0 = full causal
+    unsigned* __restrict__ sem)           // [S*KVH] tickets (fused combine)
 {
     const int S_idx = blockIdx.x / split;
     const int seg = blockIdx.x % split;
@@ -239,13 +240,63 @@ __global__ __launch_bounds__(GT * 64, 1) void k_decode_attn(
         bf16* orow = out + ((int64_t)S_idx * Hq + qh) * DHEAD;
         reinterpret_cast<bf162*>(orow)[lane] =
             __float22bfloat162_rn(make_float2(o0 * linv, o1 * linv));
-    } else {
+        return;
+    }
+    {
         const int64_t pi = (((int64_t)S_idx * Hq + qh) * split + seg);
         float* op = o_part + pi * DHEAD;
         op[2 * lane] = o0;
         op[2 * lane + 1] = o1;
         ml_part[pi * 2] = m;
         ml_part[pi * 2 + 1] = l;
+    }
+    if (sem == nullptr) return;   // separate k_decode_combine pass (A/B)
+
+    // ---- in-launch combine (guide §6 G16 split-K recipe): publish the
+    // partials with an agent-scope release BEFORE the ticket fetch_add;
+    // the block that draws split-1 acquires and reduces its (seq, kvh)
+    // group — saves the combine kernel's launch boundary + HBM round trip
+    asm volatile("s_waitcnt vmcnt(0)" ::: "memory");
+    __syncthreads();              // all waves' partials issued
+    int* flag = reinterpret_cast<int*>(smem);   // tiles are dead; reuse
+    if (threadIdx.x == 0) {
+        __builtin_amdgcn_fence(__ATOMIC_RELEASE, "agent");
+        // restate the post-wbl2 wait the compiler may drop (pitfall 12)
+        asm volatile("s_waitcnt vmcnt(0)" ::: "memory");
+        const unsigned t = __hip_atomic_fetch_add(
+            &sem[(int64_t)S_idx * gridDim.y + kvh], 1u,
+            __ATOMIC_RELAXED, __HIP_MEMORY_SCOPE_AGENT);
+        *flag = (t == (unsigned)(split - 1)) ? 1 : 0;
+    }
+    __syncthreads();
+    if (*flag == 0) return;
+    if (threadIdx.x == 0) {
+        __builtin_amdgcn_fence(__ATOMIC_ACQUIRE, "agent");
+        // reset for the next launch (stream-ordered; all arrivals done)
+        __hip_atomic_store(&sem[(int64_t)S_idx * gridDim.y + kvh], 0u,
+                           __ATOMIC_RELAXED, __HIP_MEMORY_SCOPE_AGENT);
+    }
+    __syncthreads();              // acquire visible to the whole block
+    {
+        // wave `wid` reduces its head, lane covers dims (2*lane, 2*lane+1)
+        const int64_t sh = (int64_t)S_idx * Hq + qh;
+        float mm = -3.0e38f;
+        for (int i = 0; i < split; i++)
+            mm = fmaxf(mm, ml_part[(sh * split + i) * 2]);
+        float ll = 0.f, a0 = 0.f, a1 = 0.f;
+        for (int i = 0; i < split; i++) {
+            const int64_t pi = sh * split + i;
+            const float mi = ml_part[pi * 2];
+            if (mi <= -3.0e38f) continue;
+            const float w = __expf(mi - mm);
+            ll += w * ml_part[pi * 2 + 1];
+            const float* op = o_part + pi * DHEAD;
+            a0 = fmaf(w, op[2 * lane], a0);
+            a1 = fmaf(w, op[2 * lane + 1], a1);
+        }
+        const float linv = ll > 0.f ? 1.f / ll : 0.f;
+        reinterpret_cast<bf162*>(out + sh * DHEAD)[lane] =
+            __float22bfloat162_rn(make_float2(a0 * linv, a1 * linv));
     }
 }
 
@@ -282,7 +333,7 @@ extern "C" int decode_attn_bf16(
     const void* kpool, const void* vpool, const void* page_table,
     const void* slot_ids, const void* seq_lens, int S, int Hq, int KVH,
     int page, int max_pages, float scale, int64_t q_stride, int split,
-    int window, hipStream_t stream)
+    int window, void* sem, hipStream_t stream)
 {
     const int G = Hq / KVH;
     const int lds = 2 * DCHUNK * (DHEAD + DKPAD) * 2 + G * DHEAD * 4
@@ -293,7 +344,7 @@ extern "C" int decode_attn_bf16(
         (bf16*)out, (float*)o_part, (float*)ml_part, (const bf16*)q,      \
         (const bf16*)kpool, (const bf16*)vpool, (const int*)page_table,   \
         (const int*)slot_ids, (const int*)seq_lens, Hq, KVH, page,        \
-        max_pages, scale, q_stride, split, window)
+        max_pages, scale, q_stride, split, window, (unsigned*)sem)
     switch (G) {
         case 1: DA_LAUNCH(1); break;
         case 2: DA_LAUNCH(2); break;
@@ -302,7 +353,7 @@ extern "C" int decode_attn_bf16(
         default: return (int)hipErrorInvalidValue;
     }
 #undef DA_LAUNCH
-    if (split > 1) {
+    if (split > 1 && sem == nullptr) {
         const int waves = S * Hq;
         k_decode_combine<<<(waves + 3) / 4, 256, 0, stream>>>(
             (bf16*)out, (const float*)o_part, (const float*)ml_part, waves,

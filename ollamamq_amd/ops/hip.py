@@ -45,7 +45,7 @@ def _try_load():
     lib.swiglu_bf16.argtypes = [vp, vp, i, i, vp]
     lib.argmax_bf16.argtypes = [vp, vp, i, i, vp]
     lib.decode_attn_bf16.argtypes = [vp, vp, vp, vp, vp, vp, vp, vp, vp,
-                                     i, i, i, i, i, f, i64, i, i, vp]
+                                     i, i, i, i, i, f, i64, i, i, vp, vp]
     lib.skinny_gemm_bf16.argtypes = [vp, vp, vp, vp, i, i, i, i64, i, vp]
     lib.skinny_direct_bf16.argtypes = [vp, vp, vp, vp, i, i, i, i64, i, vp]
     lib.rope_append_bf16.argtypes = [vp, vp, vp, vp, vp, vp, vp, vp, vp, vp,
@@ -224,6 +224,18 @@ def _decode_scratch(S, Hq, split, dev):
     return t
 
 
+def _combine_sem(S, KVH, dev):
+    """Ticket counters for the in-launch split-K combine: zero-initialized
+    once; the last-arriving block resets its counter, so the buffer is
+    always all-zero between launches (guide G16 counter recipe)."""
+    key = ("sem", S, KVH, str(dev))
+    t = _scratch.get(key)
+    if t is None:
+        t = torch.zeros(S * KVH, dtype=torch.int32, device=dev)
+        _scratch[key] = t
+    return t
+
+
 def attention_decode(q, cache, layer, meta):
     """Flash-decoding: KV-split partials + exact online-softmax combine.
 
@@ -246,9 +258,12 @@ def attention_decode(q, cache, layer, meta):
     split = int(min(want, max_seg, 32))
     o_part, ml_part = (None, None)
     op = mp = ctypes.c_void_p(0)
+    sem = ctypes.c_void_p(0)
     if split > 1:
         o_part, ml_part = _decode_scratch(S, Hq, split, q.device)
         op, mp = _p(o_part), _p(ml_part)
+        if os.environ.get("OLLAMAMQ_NO_FUSED_COMBINE") != "1":
+            sem = _p(_combine_sem(S, kvh, q.device))
     slot32 = meta.slot_ids.int() if meta.slot_ids.dtype != torch.int32 \
         else meta.slot_ids
     len32 = meta.seq_lens.int() if meta.seq_lens.dtype != torch.int32 \
@@ -257,7 +272,7 @@ def attention_decode(q, cache, layer, meta):
         _p(out), op, mp, _p(q), kp, vp, _p(cache.page_table),
         _p(slot32), _p(len32), S, Hq, kvh, cache.page_size,
         cache.page_table.shape[1], 1.0 / (D ** 0.5), _row_stride(q, D),
-        split, meta.window, _stream()), "decode_attn")
+        split, meta.window, sem, _stream()), "decode_attn")
     return out
 
 
