@@ -262,7 +262,12 @@ def attention_decode(q, cache, layer, meta):
     if split > 1:
         o_part, ml_part = _decode_scratch(S, Hq, split, q.device)
         op, mp = _p(o_part), _p(ml_part)
-        if os.environ.get("OLLAMAMQ_NO_FUSED_COMBINE") != "1":
+        # in-launch combine measured WORSE (41.8 vs 29.3 us @ctx512 B32;
+        # bench 5131 vs 5580): all S*split*KVH blocks pay the agent-scope
+        # release fence (buffer_wbl2, ~1.7-6.5 us at 4 blocks/CU) to save
+        # one 5 us combine launch. Kept for re-evaluation on shapes with
+        # fewer, longer blocks.
+        if os.environ.get("OLLAMAMQ_FUSED_COMBINE") == "1":
             sem = _p(_combine_sem(S, kvh, q.device))
     slot32 = meta.slot_ids.int() if meta.slot_ids.dtype != torch.int32 \
         else meta.slot_ids
