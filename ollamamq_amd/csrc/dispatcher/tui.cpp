@@ -12,6 +12,7 @@
 
 #include <algorithm>
 #include <cstdio>
+#include <set>
 #include <sstream>
 #include <vector>
 
@@ -151,6 +152,7 @@ void run_tui(Server& server) {
     int sel_backend = 0;
     int focus = 1;            // 0 = Backends panel, 1 = Users panel (Tab)
     bool show_all = false;    // 'a': show available models per backend
+    std::set<int> expanded;   // Space/Enter: expand one backend's models
     std::string input;        // typed model name for L/U
     char input_mode = 0;      // 'L' or 'U' when typing
     bool help = false;
@@ -203,7 +205,7 @@ void run_tui(Server& server) {
                 << "act:" << b.active << " done:" << b.processed << " "
                 << pad(b.loaded, 24)
                 << (b.op.empty() ? "" : " [" + b.op + "]") << "\r\n";
-            if (show_all && !b.avail.empty())
+            if ((show_all || expanded.count(i)) && !b.avail.empty())
                 out << "    available: " << pad(b.avail, 80) << "\r\n";
         }
         out << "\x1b[7m"
@@ -277,8 +279,20 @@ void run_tui(Server& server) {
                     case 'q':
                     case 27:
                         return;
+                    case 'h':
+                    case 'l':
                     case '\t':
                         focus ^= 1;
+                        break;
+                    case ' ':
+                    case '\r':
+                    case '\n':
+                        if (focus == 0) {
+                            if (expanded.count(sel_backend))
+                                expanded.erase(sel_backend);
+                            else
+                                expanded.insert(sel_backend);
+                        }
                         break;
                     case 'a':
                         show_all = !show_all;
