@@ -43,7 +43,7 @@ def _try_load():
     lib.paged_attn_bf16.argtypes = [vp, vp, vp, vp, vp, vp, vp, vp, vp,
                                     i, i, i, i, i, i, f, i64, vp]
     lib.swiglu_bf16.argtypes = [vp, vp, i, i, vp]
-    lib.argmax_bf16.argtypes = [vp, vp, i, i, vp]
+    lib.argmax_bf16.argtypes = [vp, vp, i, i, vp, vp, i, vp]
     lib.decode_attn_bf16.argtypes = [vp, vp, vp, vp, vp, vp, vp, vp, vp,
                                      i, i, i, i, i, f, i64, i, i, vp, vp]
     lib.skinny_gemm_bf16.argtypes = [vp, vp, vp, vp, i, i, i, i64, i, vp]
@@ -434,7 +434,21 @@ def sample(logits, temperature, top_k, top_p, generator=None):
         B, V = logits.shape
         out = torch.empty(B, dtype=torch.int32, device=logits.device)
         l = logits if logits.dtype == torch.bfloat16 else logits.bfloat16()
-        _check(_lib.argmax_bf16(_p(out), _p(l), B, V, _stream()), "argmax")
+        # split-V: B blocks alone underfill the chip at decode batch sizes
+        sp = min(32, max(1, 768 // max(1, B)))
+        pb = pi = ctypes.c_void_p(0)
+        if sp > 1:
+            key = ("amax", B, sp, str(logits.device))
+            t = _scratch.get(key)
+            if t is None:
+                t = (torch.empty(B * sp, dtype=torch.float32,
+                                 device=logits.device),
+                     torch.empty(B * sp, dtype=torch.int32,
+                                 device=logits.device))
+                _scratch[key] = t
+            pb, pi = _p(t[0]), _p(t[1])
+        _check(_lib.argmax_bf16(_p(out), _p(l), B, V, pb, pi, sp,
+                                _stream()), "argmax")
         return out.long()
     # stochastic paths compose on-GPU torch ops (sort/softmax/multinomial)
     return ref.sample(logits, temperature, top_k, top_p, generator)
