@@ -81,7 +81,9 @@ __global__ __launch_bounds__(GT * 64, 1) void k_decode_attn(
     constexpr int NTHR = GT * 64;
     constexpr int NSLOT = DCHUNK * (DHEAD / 8);   // uint4 slots per tile
     constexpr int NPF = NSLOT / NTHR;             // prefetch regs / thread
-    constexpr bool PF = (NPF <= 4);               // G>=4: T14 double-buffer
+    // register prefetch needs every thread to own exactly NPF slots; odd
+    // GQA groups (Qwen2: G=7) take the strided direct-stage path
+    constexpr bool PF = (NSLOT % NTHR == 0) && (NPF <= 4);
 
     const int slot = slot_ids[S_idx];
     const int kv_len = seq_lens[S_idx];
@@ -348,7 +350,11 @@ extern "C" int decode_attn_bf16(
     switch (G) {
         case 1: DA_LAUNCH(1); break;
         case 2: DA_LAUNCH(2); break;
+        case 3: DA_LAUNCH(3); break;
         case 4: DA_LAUNCH(4); break;
+        case 5: DA_LAUNCH(5); break;
+        case 6: DA_LAUNCH(6); break;
+        case 7: DA_LAUNCH(7); break;
         case 8: DA_LAUNCH(8); break;
         default: return (int)hipErrorInvalidValue;
     }

@@ -140,7 +140,7 @@ def test_attention_decode():
                                atol=3e-2, rtol=3e-2)
 
 
-@pytest.mark.parametrize("Hq,KVH", [(2, 2), (4, 2), (8, 2), (16, 2)])
+@pytest.mark.parametrize("Hq,KVH", [(2, 2), (4, 2), (8, 2), (16, 2), (14, 2)])
 def test_attention_decode_gqa_groups(Hq, KVH):
     """All decode-kernel template instantiations (GQA group G=Hq/KVH in
     {1,2,4,8}: G>=4 takes the register-prefetch pipeline, G<4 the direct
@@ -230,6 +230,25 @@ def test_family_forward_gpu_vs_cpu(preset):
     assert len(gpu) == len(cpu) == 8
     # greedy ties can flip late under bf16; the first tokens must agree
     assert gpu[:2] == cpu[:2], f"{gpu} vs {cpu}"
+
+
+def test_attention_prefill_odd_gqa_group():
+    """G=7 (Qwen2-7B's 28Q/4KV) routes prefill through the VALU paged
+    path — numerics must still match the fp32 reference."""
+    hip = _hip()
+    Hq, KVH, D = 14, 2, 128
+    kv_lens = [64, 130]
+    q_lens = [64, 130]
+    gc, cc = make_caches(KVH=KVH, n_pages=128, ctx=512)
+    fill_caches(gc, cc, kv_lens, KVH=KVH)
+    q = rnd(sum(q_lens), Hq, D, seed=21)
+    slots = list(range(len(kv_lens)))
+    meta_g = _meta(dev(), slots, kv_lens, q_lens)
+    meta_c = _meta("cpu", slots, kv_lens, q_lens)
+    out = hip.attention_prefill(q, gc, 0, meta_g)
+    out_ref = ref.attention(q.float().cpu(), cc, 0, meta_c)
+    torch.testing.assert_close(out.float().cpu(), out_ref,
+                               atol=3e-2, rtol=3e-2)
 
 
 def test_attention_prefill_varlen():
