@@ -106,14 +106,18 @@ __global__ __launch_bounds__(512) void k_prefill_attn(
     // keeps the piece arrays in registers (runtime indexing would go to
     // scratch — guide §5.4 rule 20).
     constexpr int NTHR = G * 64;
-    constexpr int N_PIECES = PF_CHUNK * (DH / 8) / NTHR;
+    constexpr int NSLOT = PF_CHUNK * (DH / 8);
+    constexpr int N_PIECES = (NSLOT + NTHR - 1) / NTHR;
+    // odd GQA groups (G=7, Qwen2): NSLOT % NTHR != 0 — clamp the last
+    // piece's slot; the duplicate threads re-load/re-write identical data
+    // (a guarded load/write would de-pipeline the stage, traps (c))
     uint4 stK[N_PIECES], stV[N_PIECES];
     const int tid = threadIdx.x;
 
     auto issue_chunk = [&](int base) {
         #pragma unroll
         for (int pi = 0; pi < N_PIECES; pi++) {
-            const int u = tid + pi * NTHR;
+            const int u = min(tid + pi * NTHR, NSLOT - 1);
             const int tok = u / (DH / 8);
             const int dv = u % (DH / 8);
             const int tk = base + tok < kv_len ? base + tok : kv_len - 1;
@@ -130,7 +134,7 @@ __global__ __launch_bounds__(512) void k_prefill_attn(
         bf16* vt = v_t0 + buf * DH * VTROW;
         #pragma unroll
         for (int pi = 0; pi < N_PIECES; pi++) {
-            const int u = tid + pi * NTHR;
+            const int u = min(tid + pi * NTHR, NSLOT - 1);
             const int tok = u / (DH / 8);
             const int dv = u % (DH / 8);
             *reinterpret_cast<uint4*>(kt + tok * KTROW + dv * 8) = stK[pi];
@@ -316,6 +320,10 @@ extern "C" int prefill_attn_bf16(
                 Hq, KVH, page, max_pages, scale, q_stride, window);        \
         } while (0)
     switch (G) {
+        case 3: PF_LAUNCH(3); break;
+        case 5: PF_LAUNCH(5); break;
+        case 6: PF_LAUNCH(6); break;
+        case 7: PF_LAUNCH(7); break;
         case 1: PF_LAUNCH(1); break;
         case 2: PF_LAUNCH(2); break;
         case 4: PF_LAUNCH(4); break;

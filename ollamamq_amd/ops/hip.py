@@ -322,18 +322,7 @@ def attention_mixed(q, cache, layer, meta):
         meta._mixed_plan = plan
     tile_slot, tile_q0, tile_pos0, tile_rows = plan
     n_tiles = tile_slot.shape[0]
-    if n_tiles and (Hq // cache.n_kv_heads) not in (1, 2, 4, 8):
-        sub = getattr(meta, "_mixed_pref", None)
-        if sub is None:
-            sub = AttnMeta(
-                mode="prefill", slot_ids=meta.slot_ids[nd:],
-                seq_lens=meta.seq_lens[nd:],
-                cu_q=meta.cu_q[nd:] - meta.cu_q[nd],
-                logits_idx=None, max_q=meta.max_q, max_kv=meta.max_kv,
-                window=meta.window)
-            meta._mixed_pref = sub
-        out[nd:] = attention_prefill(q[nd:], cache, layer, sub)
-    elif n_tiles:
+    if n_tiles:
         kp, vp = _layer_ptrs(cache, layer)
         _check(_lib.prefill_attn_bf16(
             _p(out), _p(q), kp, vp, _p(cache.page_table),
@@ -351,11 +340,6 @@ def attention_prefill(q, cache, layer, meta):
             and not meta.window:   # VALU fallback predates sliding window
         return _attention(q, cache, layer, meta, 16)
     T, Hq, D = q.shape
-    if (Hq // cache.n_kv_heads) not in (1, 2, 4, 8):
-        # MFMA prefill stages tiles with per-thread-exact piece counts;
-        # odd GQA groups (Qwen2-7B: G=7) take the VALU paged path
-        assert not meta.window, "windowed prefill needs MFMA-supported GQA"
-        return _attention(q, cache, layer, meta, 16)
     assert D == 128
     out = torch.empty((T, Hq, D), dtype=q.dtype, device=q.device)
     kp, vp = _layer_ptrs(cache, layer)

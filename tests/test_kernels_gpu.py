@@ -233,8 +233,8 @@ def test_family_forward_gpu_vs_cpu(preset):
 
 
 def test_attention_prefill_odd_gqa_group():
-    """G=7 (Qwen2-7B's 28Q/4KV) routes prefill through the VALU paged
-    path — numerics must still match the fp32 reference."""
+    """G=7 (Qwen2-7B's 28Q/4KV): MFMA prefill with clamped staging
+    slots (NSLOT % NTHR != 0) must match the fp32 reference."""
     hip = _hip()
     Hq, KVH, D = 14, 2, 128
     kv_lens = [64, 130]
