@@ -135,7 +135,7 @@ def main():
             "metric": "agg_tokens_per_sec",
             "value": round(value, 2),
             "unit": "tokens/s",
-            "n_gpus": world if has_gpu else 0,
+            "n_gpus": world,
             "steps": args.steps,
             "warmup": args.warmup,
             "ms_per_step": round(elapsed / args.steps * 1e3, 3),
