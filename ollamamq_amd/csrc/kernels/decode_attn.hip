@@ -55,7 +55,7 @@ DEV float wave_sum(float x) {
 // launch_bounds min-waves/EU = 1: LDS (~37 KB/block) already caps
 // residency at 4 blocks/CU, so a high compiler occupancy target only
 // forces the prefetch registers to spill (measured 144 B/lane scratch)
-template <int GT>
+template <int GT, int CH>
 __global__ __launch_bounds__(GT * 64, 1) void k_decode_attn(
     bf16* __restrict__ out,               // [S, Hq, D] (split==1)
     float* __restrict__ o_part,           // split>1 partials
@@ -79,7 +79,7 @@ __global__ __launch_bounds__(GT * 64, 1) void k_decode_attn(
     const int qh = kvh * G + wid;
 
     constexpr int NTHR = GT * 64;
-    constexpr int NSLOT = DCHUNK * (DHEAD / 8);   // uint4 slots per tile
+    constexpr int NSLOT = CH * (DHEAD / 8);       // uint4 slots per tile
     constexpr int NPF = NSLOT / NTHR;             // prefetch regs / thread
     // register prefetch needs every thread to own exactly NPF slots; odd
     // GQA groups (Qwen2: G=7) take the strided direct-stage path
@@ -90,11 +90,11 @@ __global__ __launch_bounds__(GT * 64, 1) void k_decode_attn(
     // balanced segment bounds in whole chunks (pages don't straddle
     // chunks): ceil-div per_seg gave e.g. 9 chunks @ split 4 -> 3,3,3,0
     // with a whole workgroup idle; floor-interpolated bounds give 2,2,2,3
-    const int n_chunks = (kv_len + DCHUNK - 1) / DCHUNK;
+    const int n_chunks = (kv_len + CH - 1) / CH;
     // sliding window: the (single) query sits at kv_len-1, so only keys
     // >= lo_tok participate; split the ACTIVE chunk range
     const int lo_tok = (window > 0 && kv_len > window) ? kv_len - window : 0;
-    const int ch_base = lo_tok / DCHUNK;
+    const int ch_base = lo_tok / CH;
     const int n_act = n_chunks - ch_base;
     const int c0 = ch_base + (int)(((int64_t)n_act * seg) / split);
     const int c1 = ch_base + (int)(((int64_t)n_act * (seg + 1)) / split);
@@ -102,8 +102,8 @@ __global__ __launch_bounds__(GT * 64, 1) void k_decode_attn(
     extern __shared__ __attribute__((aligned(16))) char smem[];
     const int KROW = DHEAD + DKPAD;
     bf16* k_tile = reinterpret_cast<bf16*>(smem);
-    bf16* v_tile = k_tile + DCHUNK * KROW;
-    float* q_lds = reinterpret_cast<float*>(v_tile + DCHUNK * KROW); // [G][128]
+    bf16* v_tile = k_tile + CH * KROW;
+    float* q_lds = reinterpret_cast<float*>(v_tile + CH * KROW); // [G][128]
     float* p_lds = q_lds + (int64_t)G * DHEAD;                       // [G][65]
 
     // stage q (fp32 in LDS: repeated broadcast reads in the score loop)
@@ -127,7 +127,7 @@ __global__ __launch_bounds__(GT * 64, 1) void k_decode_attn(
                + dv * 8;
     };
     auto stage_direct = [&](int ch2) {
-        const int base2 = ch2 * DCHUNK;
+        const int base2 = ch2 * CH;
         for (int u = threadIdx.x; u < NSLOT; u += NTHR) {
             const int tok = u / (DHEAD / 8), dv = u % (DHEAD / 8);
             const int64_t src = src_of(base2, u);
@@ -142,8 +142,8 @@ __global__ __launch_bounds__(GT * 64, 1) void k_decode_attn(
     __syncthreads();
 
     for (int ch = c0; ch < c1; ch++) {
-        const int base = ch * DCHUNK;
-        const int n_here = min(DCHUNK, kv_len - base);
+        const int base = ch * CH;
+        const int n_here = min(CH, kv_len - base);
         // ---- issue next chunk's HBM loads NOW, write to LDS after the
         // compute phase (T14 issue-early/write-late): the fetch rides
         // under the score/PV work instead of serializing with it.
@@ -153,13 +153,16 @@ __global__ __launch_bounds__(GT * 64, 1) void k_decode_attn(
         uint4 k0, k1, k2, k3, v0, v1, v2, v3;
         const bool have_next = PF && (ch + 1 < c1);
         if (have_next) {
-            const int nbase = (ch + 1) * DCHUNK;
+            const int nbase = (ch + 1) * CH;
             const int64_t s0 = src_of(nbase, threadIdx.x);
-            const int64_t s1 = src_of(nbase, threadIdx.x + NTHR);
+            const int64_t s1 = NPF >= 2 ? src_of(nbase, threadIdx.x + NTHR)
+                                        : s0;
             k0 = *reinterpret_cast<const uint4*>(kpool + s0);
             v0 = *reinterpret_cast<const uint4*>(vpool + s0);
-            k1 = *reinterpret_cast<const uint4*>(kpool + s1);
-            v1 = *reinterpret_cast<const uint4*>(vpool + s1);
+            if constexpr (NPF >= 2) {
+                k1 = *reinterpret_cast<const uint4*>(kpool + s1);
+                v1 = *reinterpret_cast<const uint4*>(vpool + s1);
+            }
             if constexpr (NPF >= 4) {
                 const int64_t s2 = src_of(nbase, threadIdx.x + 2 * NTHR);
                 const int64_t s3 = src_of(nbase, threadIdx.x + 3 * NTHR);
@@ -199,15 +202,15 @@ __global__ __launch_bounds__(GT * 64, 1) void k_decode_attn(
             o0 *= corr; o1 *= corr;
             m = mn;
         }
-        p_lds[wid * (DCHUNK + 1) + lane] = p;
+        p_lds[wid * (CH + 1) + lane] = p;
 
         // ---- PV: lane accumulates dims (2*lane, 2*lane+1) ----
         // fixed bound + full unroll: a runtime bound left this loop a
         // serial ds_read latency chain (PMC: 25% SQ_WAIT_INST_ANY);
-        // dead keys contribute p=0 so looping to DCHUNK is exact
-        const float* prow = p_lds + wid * (DCHUNK + 1);
+        // dead keys contribute p=0 so looping to CH is exact
+        const float* prow = p_lds + wid * (CH + 1);
         #pragma unroll 8
-        for (int j = 0; j < DCHUNK; j++) {
+        for (int j = 0; j < CH; j++) {
             const bf162 v2 = *reinterpret_cast<const bf162*>(
                 v_tile + j * KROW + 2 * lane);
             const float pj = prow[j];
@@ -224,7 +227,7 @@ __global__ __launch_bounds__(GT * 64, 1) void k_decode_attn(
                     = vx;
             };
             put(threadIdx.x, k0, v0);
-            put(threadIdx.x + NTHR, k1, v1);
+            if constexpr (NPF >= 2) put(threadIdx.x + NTHR, k1, v1);
             if constexpr (NPF >= 4) {
                 put(threadIdx.x + 2 * NTHR, k2, v2);
                 put(threadIdx.x + 3 * NTHR, k3, v3);
@@ -335,18 +338,24 @@ extern "C" int decode_attn_bf16(
     const void* kpool, const void* vpool, const void* page_table,
     const void* slot_ids, const void* seq_lens, int S, int Hq, int KVH,
     int page, int max_pages, float scale, int64_t q_stride, int split,
-    int window, void* sem, hipStream_t stream)
+    int window, int chunk, void* sem, hipStream_t stream)
 {
     const int G = Hq / KVH;
-    const int lds = 2 * DCHUNK * (DHEAD + DKPAD) * 2 + G * DHEAD * 4
-                  + G * (DCHUNK + 1) * 4;
+    const int lds = 2 * chunk * (DHEAD + DKPAD) * 2 + G * DHEAD * 4
+                  + G * (chunk + 1) * 4;
     dim3 grid(S * split, KVH);
-#define DA_LAUNCH(GT)                                                     \
-    k_decode_attn<GT><<<grid, GT * 64, lds, stream>>>(                    \
+#define DA_LAUNCH1(GT, CH)                                                \
+    k_decode_attn<GT, CH><<<grid, GT * 64, lds, stream>>>(                \
         (bf16*)out, (float*)o_part, (float*)ml_part, (const bf16*)q,      \
         (const bf16*)kpool, (const bf16*)vpool, (const int*)page_table,   \
         (const int*)slot_ids, (const int*)seq_lens, Hq, KVH, page,        \
         max_pages, scale, q_stride, split, window, (unsigned*)sem)
+#define DA_LAUNCH(GT)                                                     \
+    switch (chunk) {                                                      \
+        case 32: DA_LAUNCH1(GT, 32); break;                               \
+        case 128: DA_LAUNCH1(GT, 128); break;                             \
+        default: DA_LAUNCH1(GT, 64); break;                               \
+    }
     switch (G) {
         case 1: DA_LAUNCH(1); break;
         case 2: DA_LAUNCH(2); break;
@@ -359,6 +368,7 @@ extern "C" int decode_attn_bf16(
         default: return (int)hipErrorInvalidValue;
     }
 #undef DA_LAUNCH
+#undef DA_LAUNCH1
     if (split > 1 && sem == nullptr) {
         const int waves = S * Hq;
         k_decode_combine<<<(waves + 3) / 4, 256, 0, stream>>>(

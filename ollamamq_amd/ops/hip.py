@@ -45,7 +45,8 @@ def _try_load():
     lib.swiglu_bf16.argtypes = [vp, vp, i, i, vp]
     lib.argmax_bf16.argtypes = [vp, vp, i, i, vp, vp, i, vp]
     lib.decode_attn_bf16.argtypes = [vp, vp, vp, vp, vp, vp, vp, vp, vp,
-                                     i, i, i, i, i, f, i64, i, i, vp, vp]
+                                     i, i, i, i, i, f, i64, i, i, i, vp,
+                                     vp]
     lib.skinny_gemm_bf16.argtypes = [vp, vp, vp, vp, i, i, i, i64, i, vp]
     lib.skinny_direct_bf16.argtypes = [vp, vp, vp, vp, i, i, i, i64, i, vp]
     lib.rope_append_bf16.argtypes = [vp, vp, vp, vp, vp, vp, vp, vp, vp, vp,
@@ -236,6 +237,10 @@ def _combine_sem(S, KVH, dev):
     return t
 
 
+# staged KV chunk length (tokens) per pipeline stage; 32/64/128 compiled
+DECODE_CHUNK = int(os.environ.get("OLLAMAMQ_DECODE_CHUNK", "64"))
+
+
 def attention_decode(q, cache, layer, meta):
     """Flash-decoding: KV-split partials + exact online-softmax combine.
 
@@ -277,7 +282,7 @@ def attention_decode(q, cache, layer, meta):
         _p(out), op, mp, _p(q), kp, vp, _p(cache.page_table),
         _p(slot32), _p(len32), S, Hq, kvh, cache.page_size,
         cache.page_table.shape[1], 1.0 / (D ** 0.5), _row_stride(q, D),
-        split, meta.window, sem, _stream()), "decode_attn")
+        split, meta.window, DECODE_CHUNK, sem, _stream()), "decode_attn")
     return out
 
 
