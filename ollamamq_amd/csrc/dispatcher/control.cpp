@@ -1,5 +1,7 @@
 #include "control.h"
 
+#include <sstream>
+
 #include <thread>
 
 #include "matching.h"
@@ -456,6 +458,69 @@ void apply_model_config(AppState& st) {
             }
         }).detach();
     }
+}
+
+}  // namespace omq
+
+namespace omq {
+
+// ------------------------------------------------------- /metrics (Prom)
+// Prometheus text exposition of the same counters admin_stats reports.
+std::string metrics_text(AppState& st) {
+    std::ostringstream o;
+    o << "# HELP ollamamq_uptime_seconds Dispatcher uptime.\n"
+         "# TYPE ollamamq_uptime_seconds gauge\n"
+      << "ollamamq_uptime_seconds "
+      << (st.started_ms ? (now_ms() - st.started_ms) / 1000.0 : 0) << "\n";
+    int64_t processed = 0, dropped = 0, queued = 0, processing = 0;
+    {
+        std::lock_guard<std::mutex> g(st.queues_mu);
+        for (const auto& [name, us] : st.users) {
+            processed += us.processed;
+            dropped += us.dropped;
+            queued += (int64_t)us.queue.size();
+            processing += us.processing;
+        }
+    }
+    o << "# TYPE ollamamq_requests_processed_total counter\n"
+      << "ollamamq_requests_processed_total " << processed << "\n"
+      << "# TYPE ollamamq_requests_dropped_total counter\n"
+      << "ollamamq_requests_dropped_total " << dropped << "\n"
+      << "# TYPE ollamamq_requests_queued gauge\n"
+      << "ollamamq_requests_queued " << queued << "\n"
+      << "# TYPE ollamamq_requests_processing gauge\n"
+      << "ollamamq_requests_processing " << processing << "\n";
+    {
+        std::lock_guard<std::mutex> g(st.backends_mu);
+        o << "# TYPE ollamamq_backend_online gauge\n"
+             "# TYPE ollamamq_backend_active_requests gauge\n"
+             "# TYPE ollamamq_backend_processed_total counter\n";
+        for (const auto& b : st.backends) {
+            o << "ollamamq_backend_online{url=\"" << b.url << "\"} "
+              << (b.is_online ? 1 : 0) << "\n"
+              << "ollamamq_backend_active_requests{url=\"" << b.url
+              << "\"} " << b.active_requests << "\n"
+              << "ollamamq_backend_processed_total{url=\"" << b.url
+              << "\"} " << b.processed_count << "\n";
+        }
+    }
+    {
+        std::lock_guard<std::mutex> g(st.waits_mu);
+        std::vector<int64_t> w(st.wait_samples_ms.begin(),
+                               st.wait_samples_ms.end());
+        std::sort(w.begin(), w.end());
+        auto pct = [&](double p) -> double {
+            if (w.empty()) return 0;
+            return (double)w[(size_t)(p * (w.size() - 1))];
+        };
+        o << "# TYPE ollamamq_queue_wait_ms summary\n"
+          << "ollamamq_queue_wait_ms{quantile=\"0.5\"} " << pct(0.50) << "\n"
+          << "ollamamq_queue_wait_ms{quantile=\"0.9\"} " << pct(0.90) << "\n"
+          << "ollamamq_queue_wait_ms{quantile=\"0.99\"} " << pct(0.99)
+          << "\n"
+          << "ollamamq_queue_wait_ms_count " << w.size() << "\n";
+    }
+    return o.str();
 }
 
 }  // namespace omq

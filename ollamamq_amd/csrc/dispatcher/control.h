@@ -32,6 +32,7 @@ ControlOutcome start_model_control(AppState& st, const ControlRequest& req);
 // Admin handlers (return status + JSON body)
 ControlOutcome admin_models_state(AppState& st);
 ControlOutcome admin_stats(AppState& st);
+std::string metrics_text(AppState& st);
 ControlOutcome admin_model_load(AppState& st, const std::string& body);
 ControlOutcome admin_model_unload(AppState& st, const std::string& body);
 

@@ -120,6 +120,15 @@ void Server::handle(const HttpRequest& req, HttpConn& conn) {
         conn.send(out.http_status, {}, out.body.dump());
         return;
     }
+    if (req.path == "/metrics" && req.method == "GET") {
+        // Prometheus exposition (beyond the reference: SURVEY §5 calls for
+        // a first-class metrics surface for the tokens/queue-wait metric)
+        conn.send(200,
+                  {{"Content-Type",
+                    "text/plain; version=0.0.4; charset=utf-8"}},
+                  metrics_text(st_));
+        return;
+    }
     if (req.path == "/admin/models" && req.method == "GET") {
         auto out = admin_models_state(st_);
         conn.send(out.http_status, {}, out.body.dump());

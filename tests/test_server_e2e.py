@@ -294,3 +294,26 @@ def test_admin_stats_surface(proxy):
     assert "queue_wait" in st and "p50_ms" in st["queue_wait"]
     assert any(u["processed"] > 0 for u in st["users"])
     assert len(st["backends"]) == 2
+
+
+def test_metrics_prometheus(fleet, tmp_path_factory):
+    p = Proxy([fleet.ollama_url], tmp_path_factory.mktemp("metrics"))
+    try:
+        # generate one request so counters move
+        httpx.post(p.base + "/api/generate",
+                   json={"model": "llama3:8b", "prompt": "x",
+                         "stream": False},
+                   headers={"X-User-ID": "m1"}, timeout=30.0)
+        r = httpx.get(p.base + "/metrics", timeout=10.0)
+        assert r.status_code == 200
+        assert "text/plain" in r.headers["content-type"]
+        body = r.text
+        assert "ollamamq_requests_processed_total" in body
+        assert "ollamamq_backend_online{url=" in body
+        assert 'ollamamq_queue_wait_ms{quantile="0.5"}' in body
+        # processed counter reflects the request above
+        line = [l for l in body.splitlines()
+                if l.startswith("ollamamq_requests_processed_total")][0]
+        assert float(line.split()[-1]) >= 1
+    finally:
+        p.stop()
