@@ -140,3 +140,56 @@ def test_per_request_seed_reproducible():
                                             seed=77))]
     run_all(eng)
     assert sid.generated != a  # different seed, different trajectory
+
+
+def test_step_failure_recovery_worker_sequence():
+    """The worker loop's recovery after a failed step (engine/worker.py
+    run loop): cancel all in-flight, reap with done markers, engine stays
+    serviceable for new requests."""
+    eng = make_engine()
+    calls = []
+    orig = eng.model.forward
+
+    def flaky(*a, **k):
+        calls.append(1)
+        if len(calls) == 3:
+            raise RuntimeError("injected forward failure")
+        return orig(*a, **k)
+
+    eng.model.forward = flaky
+    done_markers = []
+    sid = eng.submit(list(range(1, 9)),
+                     GenParams(max_tokens=50))
+    eng.seqs[sid].on_token = lambda t, d: done_markers.append(d)
+    with pytest.raises(RuntimeError):
+        for _ in range(100):
+            eng.step()
+    # worker-style recovery
+    for seq in list(eng.waiting) + list(eng.running):
+        seq.cancelled = True
+    eng.step()
+    assert done_markers and done_markers[-1] is True
+    assert not eng.has_work() and eng.seqs == {}
+    assert len(eng.kv._free_slots) == eng.kv.max_slots
+
+    # still serves new work afterwards
+    eng.model.forward = orig
+    sid2 = eng.submit([4, 5, 6], GenParams(max_tokens=3))
+    seq2 = eng.seqs[sid2]
+    run_all(eng)
+    assert len(seq2.generated) == 3
+
+
+def test_long_run_resource_stability():
+    """Waves of sequences through a small engine: every slot and KV page
+    must return to the pool and the registry must empty each wave."""
+    eng = make_engine(max_slots=4)
+    for wave in range(5):
+        sids = [eng.submit([wave + 1, i + 1, 3],
+                           GenParams(max_tokens=4 + (i % 3)))
+                for i in range(8)]
+        run_all(eng)
+        assert eng.seqs == {}
+        assert len(eng.kv._free_slots) == eng.kv.max_slots
+        assert eng.kv.free_page_count() == eng.kv.n_pages
+        assert all(s not in eng.seqs for s in sids)
