@@ -93,6 +93,26 @@ def build_tsan_stress(force=False):
     return out
 
 
+def build_asan_stress(force=False):
+    """ASan+UBSan variant of the same hammer (heap errors, UB)."""
+    import glob
+    srcs = sorted(glob.glob(os.path.join(DDIR, "*.cpp")))
+    core = [s for s in srcs if not s.endswith("main.cpp")
+            and not s.endswith("bindings.cpp")
+            and not s.endswith("tsan_stress.cpp")]
+    out = os.path.join(DDIR, "asan_stress")
+    stress = os.path.join(DDIR, "tsan_stress.cpp")
+    if force or not _newer(out, stress, *core):
+        cxx = "/opt/rocm/lib/llvm/bin/clang++"
+        if not os.path.exists(cxx):
+            cxx = "g++"
+        _run([cxx, "-O1", "-g", "-std=c++17",
+              "-fsanitize=address,undefined",
+              "-fno-sanitize-recover=all",
+              "-I" + DDIR, stress, *core, "-o", out, "-lpthread"])
+    return out
+
+
 def build_all(force=False):
     build_kernels(force)
     build_dispatcher(force)
