@@ -1,0 +1,42 @@
+#!/bin/bash
+# Serving soak on a GPU box: bring up the full node (dispatcher + GPU
+# worker, Llama-3-8B), then drive repeated randomized stress waves for
+# $SOAK_SECS while watching VRAM and dispatcher RSS for drift.
+set -x
+cd /root/repo
+PORT=18222
+SOAK_SECS=${SOAK_SECS:-300}
+python -m ollamamq_amd.launch --gpus 1 --model llama3-8b --max-ctx 2048 \
+    --max-batch 32 --no-tui --port $PORT --sock-dir /tmp \
+    -c /tmp/absent.yaml &
+LAUNCH=$!
+trap "kill $LAUNCH 2>/dev/null" EXIT
+
+for i in $(seq 1 180); do
+  curl -sf http://127.0.0.1:$PORT/health >/dev/null && break; sleep 1
+done
+for i in $(seq 1 180); do
+  curl -s http://127.0.0.1:$PORT/admin/models | grep -q '"llama3-8b"' && break
+  sleep 1
+done
+
+vram() { rocm-smi --showmeminfo vram --json 2>/dev/null \
+         | python3 -c "import json,sys;d=json.load(sys.stdin);print(next(iter(d.values()))['VRAM Total Used Memory (B)'])"; }
+rss() { ps -o rss= -p $(pgrep -P $LAUNCH -f ollamamq-server | head -1) 2>/dev/null || echo 0; }
+
+V0=$(vram); R0=$(rss)
+echo "baseline vram=$V0 rss=$R0"
+END=$(( $(date +%s) + SOAK_SECS ))
+WAVE=0
+while [ "$(date +%s)" -lt "$END" ]; do
+  WAVE=$((WAVE+1))
+  timeout 200 python tools/stress.py --base http://127.0.0.1:$PORT \
+      --users 24 --models llama3-8b --max-tokens 24 2>&1 | tail -2
+done
+V1=$(vram); R1=$(rss)
+echo "after $WAVE waves: vram=$V1 (delta $((V1-V0))) rss=$R1 (delta $((R1-R0)))"
+curl -s http://127.0.0.1:$PORT/admin/stats | head -c 300; echo
+curl -s http://127.0.0.1:$PORT/metrics | grep -E "processed_total|dropped"
+# VRAM must not grow across waves (KV pool is preallocated); allow 256 MB
+# slack for allocator pools, 200 MB RSS slack for the dispatcher
+python3 -c "assert $V1 - $V0 < 256*2**20, 'VRAM drift'; assert $R1 - $R0 < 200*1024, 'RSS drift'; print('SOAK OK')"
