@@ -24,8 +24,12 @@ vram() { rocm-smi --showmeminfo vram --json 2>/dev/null \
          | python3 -c "import json,sys;d=json.load(sys.stdin);print(next(iter(d.values()))['VRAM Total Used Memory (B)'])"; }
 rss() { ps -o rss= -p $(pgrep -P $LAUNCH -f ollamamq-server | head -1) 2>/dev/null || echo 0; }
 
+# one warmup wave first: decode graphs are captured lazily per batch
+# size (bounded by --max-batch, ~75 MB each); baseline AFTER they exist
+timeout 200 python tools/stress.py --base http://127.0.0.1:$PORT \
+    --users 24 --models llama3-8b --max-tokens 24 2>&1 | tail -1
 V0=$(vram); R0=$(rss)
-echo "baseline vram=$V0 rss=$R0"
+echo "post-warmup baseline vram=$V0 rss=$R0"
 END=$(( $(date +%s) + SOAK_SECS ))
 WAVE=0
 while [ "$(date +%s)" -lt "$END" ]; do
