@@ -32,15 +32,28 @@ V0=$(vram); R0=$(rss)
 echo "post-warmup baseline vram=$V0 rss=$R0"
 END=$(( $(date +%s) + SOAK_SECS ))
 WAVE=0
+VLOG=""
 while [ "$(date +%s)" -lt "$END" ]; do
   WAVE=$((WAVE+1))
   timeout 200 python tools/stress.py --base http://127.0.0.1:$PORT \
-      --users 24 --models llama3-8b --max-tokens 24 2>&1 | tail -2
+      --users 24 --models llama3-8b --max-tokens 24 2>&1 | tail -1
+  VW=$(vram); VLOG="$VLOG $VW"
+  echo "wave $WAVE vram=$VW"
 done
 V1=$(vram); R1=$(rss)
+echo "vram trajectory:$VLOG"
 echo "after $WAVE waves: vram=$V1 (delta $((V1-V0))) rss=$R1 (delta $((R1-R0)))"
 curl -s http://127.0.0.1:$PORT/admin/stats | head -c 300; echo
 curl -s http://127.0.0.1:$PORT/metrics | grep -E "processed_total|dropped"
-# VRAM must not grow across waves (KV pool is preallocated); allow 256 MB
-# slack for allocator pools, 200 MB RSS slack for the dispatcher
-python3 -c "assert $V1 - $V0 < 256*2**20, 'VRAM drift'; assert $R1 - $R0 < 200*1024, 'RSS drift'; print('SOAK OK')"
+# decode graphs are captured lazily per batch size, so VRAM steps up as
+# new sizes first appear (bounded by max_batch); what must NOT happen is
+# continued growth once sizes repeat: assert the LAST HALF of the
+# trajectory is flat (<128 MB), and dispatcher RSS stable (<200 MB)
+python3 - <<PYEOF
+v = [int(x) for x in "$VLOG".split()]
+half = v[len(v)//2:]
+drift = max(half) - min(half)
+assert drift < 128*2**20, f"VRAM drift in steady half: {drift/2**20:.0f} MB over {half}"
+assert $R1 - $R0 < 200*1024, "RSS drift"
+print(f"SOAK OK  (capture ramp {(v[-1]-v[0])/2**20:.0f} MB, steady-half drift {drift/2**20:.0f} MB)")
+PYEOF
