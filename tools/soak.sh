@@ -51,9 +51,11 @@ curl -s http://127.0.0.1:$PORT/metrics | grep -E "processed_total|dropped"
 # trajectory is flat (<128 MB), and dispatcher RSS stable (<200 MB)
 python3 - <<PYEOF
 v = [int(x) for x in "$VLOG".split()]
-half = v[len(v)//2:]
-drift = max(half) - min(half)
-assert drift < 128*2**20, f"VRAM drift in steady half: {drift/2**20:.0f} MB over {half}"
+# measured trajectory (profiles/r01_soak.log): capture ramp flattens by
+# ~wave 15 of 24; judge the final 8 waves
+tail = v[-8:] if len(v) >= 8 else v
+drift = max(tail) - min(tail)
+assert drift < 64*2**20, f"VRAM drift in steady tail: {drift/2**20:.0f} MB over {tail}"
 assert $R1 - $R0 < 200*1024, "RSS drift"
 print(f"SOAK OK  (capture ramp {(v[-1]-v[0])/2**20:.0f} MB, steady-half drift {drift/2**20:.0f} MB)")
 PYEOF
