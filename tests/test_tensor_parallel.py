@@ -18,8 +18,8 @@ PROMPT = [5, 9, 2, 7, 11, 3]
 N_TOKENS = 6
 
 
-def run_engine(tp_rank=0, tp_size=1, group=None):
-    cfg = PRESETS["tiny"]
+def run_engine(tp_rank=0, tp_size=1, group=None, preset="tiny"):
+    cfg = PRESETS[preset]
     model = LlamaModel(cfg, device="cpu", dtype=torch.float32, seed=31,
                        tp_rank=tp_rank, tp_size=tp_size,
                        process_group=group)
@@ -36,13 +36,13 @@ def run_engine(tp_rank=0, tp_size=1, group=None):
     return seq.generated
 
 
-def _tp_worker(rank, world, port, out_q):
+def _tp_worker(rank, world, port, out_q, preset="tiny"):
     os.environ["MASTER_ADDR"] = "127.0.0.1"
     os.environ["MASTER_PORT"] = str(port)
     torch.distributed.init_process_group("gloo", rank=rank,
                                          world_size=world)
     try:
-        toks = run_engine(tp_rank=rank, tp_size=world)
+        toks = run_engine(tp_rank=rank, tp_size=world, preset=preset)
         out_q.put((rank, toks))
     finally:
         torch.distributed.destroy_process_group()
@@ -69,3 +69,28 @@ def test_tp2_matches_tp1():
     assert results[0] == results[1], "ranks diverged"
     assert results[0] == single, (
         f"TP=2 tokens {results[0]} != TP=1 tokens {single}")
+
+
+@pytest.mark.parametrize("preset,port", [("tiny-qwen", 29573),
+                                         ("tiny-swa", 29574)])
+def test_tp2_matches_tp1_family_variants(preset, port):
+    """Qwen2 bias sharding and Mistral sliding-window attention must be
+    TP-invariant too (bias rows split with the column-parallel QKV; the
+    window mask is per-head and head-partitioned)."""
+    single = run_engine(preset=preset)
+    assert len(single) == N_TOKENS
+
+    ctx = mp.get_context("spawn")
+    q = ctx.Queue()
+    procs = [ctx.Process(target=_tp_worker, args=(r, 2, port, q, preset))
+             for r in range(2)]
+    for p in procs:
+        p.start()
+    results = {}
+    for _ in range(2):
+        rank, toks = q.get(timeout=300)
+        results[rank] = toks
+    for p in procs:
+        p.join(timeout=60)
+        assert p.exitcode == 0
+    assert results[0] == results[1] == single
