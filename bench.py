@@ -121,7 +121,9 @@ def main():
         torch.cuda.synchronize()
     elapsed = time.monotonic() - t0
     if dist:
-        t = torch.tensor([elapsed], dtype=torch.float64)
+        # nccl (=RCCL) requires device tensors; gloo (CPU tests) takes CPU
+        t = torch.tensor([elapsed], dtype=torch.float64,
+                         device=f"cuda:{local_rank}" if has_gpu else "cpu")
         torch.distributed.all_reduce(t, op=torch.distributed.ReduceOp.MAX)
         elapsed = float(t.item())
         torch.distributed.barrier()

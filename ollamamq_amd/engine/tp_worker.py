@@ -203,6 +203,9 @@ def main():
     rank = int(os.environ.get("RANK", "0"))
     world = int(os.environ.get("WORLD_SIZE", "1"))
     backend = "nccl" if torch.cuda.is_available() else "gloo"
+    if torch.cuda.is_available():
+        # bind the device BEFORE the RCCL communicator is created
+        torch.cuda.set_device(args.gpu_base + rank)
     if world > 1:
         dist.init_process_group(backend)
 
