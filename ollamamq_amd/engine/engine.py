@@ -428,6 +428,36 @@ class LlamaEngine:
         self._fill_bufs(entry, seqs, token_list, pos_list)
         return self._graph_replay(entry)
 
+    def warm_graphs(self, sizes=None):
+        """Pre-capture decode graphs for common batch sizes at load time
+        (capture costs ~150 ms each; lazily hitting them mid-serving adds
+        first-request jitter).  Uses throwaway KV slots with one token."""
+        if not self.use_graphs:
+            return
+        sizes = [b for b in (sizes or (1, 2, 4, 8, 16, self.max_batch))
+                 if 0 < b <= self.max_batch]
+        for b in sorted(set(sizes)):
+            slots = []
+            try:
+                for _ in range(b):
+                    s = self.kv.alloc_slot()
+                    self.kv.ensure(s, 1)
+                    slots.append(s)
+            except RuntimeError:      # not enough slots/pages: skip size
+                for s in slots:
+                    self.kv.free_slot(s)
+                continue
+            entry = self._graph_entry(b)
+            host = torch.tensor(
+                [[0] * b, [0] * b, slots, [1] * b], dtype=torch.int32)
+            staged = host.to(self.dev, non_blocking=True)
+            for i, k in enumerate(("tok", "pos", "slot", "lens")):
+                entry["bufs"][k].copy_(staged[i])
+            self._graph_replay(entry)
+            torch.cuda.synchronize()
+            for s in slots:
+                self.kv.free_slot(s)
+
     def _decode_step(self):
         seqs = self.running
         if (self.use_pipeline and self.use_graphs

@@ -87,8 +87,9 @@ class Worker:
                 kv = PagedKVCache.for_model(
                     cfg, n_pages=pages, max_slots=self.max_batch + 2,
                     max_ctx=ctx, device=self.device, dtype=self.dtype)
-                self.engines[model] = LlamaEngine(m, kv,
-                                                 max_batch=self.max_batch)
+                eng = LlamaEngine(m, kv, max_batch=self.max_batch)
+                eng.warm_graphs()   # pre-capture: no first-request jitter
+                self.engines[model] = eng
                 self.tokenizers[model] = ByteTokenizer(cfg.vocab)
                 self.loaded_ctx[model] = ctx
                 return None

@@ -104,3 +104,17 @@ def test_pipelined_kv_pages_all_freed():
     _workload(eng)
     run_all(eng)
     assert eng.kv.free_page_count() == eng.kv.n_pages
+
+
+def test_warm_graphs_precapture():
+    eng = make_engine(pipelined=True)
+    eng.warm_graphs()
+    assert set(eng._graphs) >= {1, 2, 4, 8}
+    assert all(e["graph"] is not None for e in eng._graphs.values())
+    # throwaway slots and pages all returned
+    assert len(eng.kv._free_slots) == eng.kv.max_slots
+    assert eng.kv.free_page_count() == eng.kv.n_pages
+    # serving still correct after pre-capture
+    sids = _workload(eng)
+    run_all(eng)
+    assert all(s.finish_reason for s in sids)
