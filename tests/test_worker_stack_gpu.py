@@ -93,3 +93,28 @@ def test_gpu_worker_concurrent(gpu_stack):
 
     with cf.ThreadPoolExecutor(6) as ex:
         assert list(ex.map(one, range(6))) == [200] * 6
+
+
+def test_gpu_worker_embeddings(gpu_stack):
+    r = httpx.post(gpu_stack + "/api/embed",
+                   json={"model": "tiny", "input": "hello world"},
+                   headers={"X-User-ID": "ge"}, timeout=120.0)
+    assert r.status_code == 200, r.text
+    emb = r.json()["embeddings"]
+    assert len(emb) == 1 and len(emb[0]) == 512  # tiny hidden size
+    assert any(abs(x) > 1e-6 for x in emb[0])
+
+
+def test_gpu_worker_openai_sse(gpu_stack):
+    with httpx.stream(
+            "POST", gpu_stack + "/v1/chat/completions",
+            json={"model": "tiny", "stream": True,
+                  "messages": [{"role": "user", "content": "hi"}],
+                  "max_tokens": 6},
+            headers={"X-User-ID": "gs"}, timeout=120.0) as r:
+        assert r.status_code == 200
+        lines = [l for l in r.iter_lines() if l.startswith("data: ")]
+    assert lines[-1].strip() == "data: [DONE]"
+    import json as j
+    chunks = [j.loads(l[6:]) for l in lines[:-1]]
+    assert any(c["choices"][0]["delta"].get("content") for c in chunks)
