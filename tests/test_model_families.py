@@ -163,3 +163,31 @@ def test_family_presets_shapes():
     m = PRESETS["mistral-7b"]
     assert (m.hidden, m.n_heads, m.n_kv_heads, m.vocab,
             m.sliding_window) == (4096, 32, 8, 32000, 4096)
+
+
+def test_swa_mixed_step_late_admission():
+    """A prompt admitted while another sequence decodes (mixed
+    prefill+decode step) must produce the same tokens as the same
+    admission pattern run with full-causal masking disabled only where
+    lengths stay inside the window — i.e. consistency of the windowed
+    mixed path with the windowed sequential path."""
+    prompt_a = list(range(10, 130))     # 120 > window 96
+    prompt_b = [7, 8, 9]
+
+    def run(stagger):
+        eng = make_engine("tiny-swa")
+        sa = eng.submit(prompt_a, GenParams(max_tokens=12))
+        seq_a = eng.seqs[sa]
+        if stagger:
+            for _ in range(4):
+                eng.step()
+        sb = eng.submit(prompt_b, GenParams(max_tokens=12))
+        seq_b = eng.seqs[sb]
+        run_all(eng)
+        return seq_a.generated, seq_b.generated
+
+    a1, b1 = run(stagger=True)    # b admitted mid-decode -> mixed step
+    a2, b2 = run(stagger=False)   # both admitted together
+    assert len(a1) == len(a2) == 12
+    assert a1 == a2, "windowed decode depends on admission timing"
+    assert b1 == b2
