@@ -29,7 +29,7 @@ typedef __attribute__((ext_vector_type(8))) __bf16 bf16x8;
 typedef __attribute__((ext_vector_type(16))) float f32x16;
 
 #define DEV static __device__ __forceinline__
-DEV float bf2f(bf16 x) { return __bfloat162float(x); }
+[[maybe_unused]] DEV float bf2f(bf16 x) { return __bfloat162float(x); }
 
 #define PF_CHUNK 64          // kv tokens per staged chunk
 #define PF_ROWS 32           // q rows per tile
