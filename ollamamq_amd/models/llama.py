@@ -42,6 +42,9 @@ class LlamaConfig:
     # QKV projection; Mistral restricts attention to a sliding window
     qkv_bias: bool = False
     sliding_window: int = 0       # 0 = full causal
+    # Llama-3.1-style NTK rope scaling (factor, low_freq_factor,
+    # high_freq_factor, original_max_ctx); None = plain RoPE
+    rope_scaling: Optional[tuple] = None
 
     @property
     def q_dim(self) -> int:
@@ -59,6 +62,22 @@ PRESETS = {
     "llama3-70b": LlamaConfig(
         name="llama3-70b", n_layers=80, hidden=8192, n_heads=64, n_kv_heads=8,
         ffn=28672,
+    ),
+    # Llama-3.1 8B: same arch as 8B + NTK-scaled RoPE for 32k+ contexts
+    "llama3.1-8b": LlamaConfig(
+        name="llama3.1-8b", max_ctx=32768,
+        rope_scaling=(8.0, 1.0, 4.0, 8192),
+    ),
+    # Llama-2 7B/13B: MHA (no GQA), rope theta 1e4, vocab 32000
+    "llama2-7b": LlamaConfig(
+        name="llama2-7b", n_layers=32, hidden=4096, n_heads=32,
+        n_kv_heads=32, ffn=11008, vocab=32000, rope_theta=10000.0,
+        max_ctx=4096,
+    ),
+    "llama2-13b": LlamaConfig(
+        name="llama2-13b", n_layers=40, hidden=5120, n_heads=40,
+        n_kv_heads=40, ffn=13824, vocab=32000, rope_theta=10000.0,
+        max_ctx=4096,
     ),
     # Qwen2-7B: QKV bias, GQA 28Q/4KV, rope theta 1e6, vocab 152064
     "qwen2-7b": LlamaConfig(
@@ -214,6 +233,17 @@ class LlamaModel:
             cfg.rope_theta
             ** (torch.arange(0, cfg.head_dim, 2, dtype=torch.float32) / cfg.head_dim)
         )
+        if cfg.rope_scaling is not None:
+            # Llama-3.1 NTK scaling: long wavelengths divided by `factor`,
+            # short kept, smooth ramp between (host-side table only — the
+            # RoPE kernel is scaling-agnostic)
+            factor, lo_f, hi_f, orig = cfg.rope_scaling
+            wavelen = 2 * math.pi / inv
+            lo_wl, hi_wl = orig / lo_f, orig / hi_f
+            smooth = ((orig / wavelen - lo_f) / (hi_f - lo_f)).clamp(0, 1)
+            scaled = (1 - smooth) * inv / factor + smooth * inv
+            inv = torch.where(wavelen > lo_wl, inv / factor,
+                              torch.where(wavelen < hi_wl, inv, scaled))
         ang = torch.outer(pos, inv)  # [max_ctx, head_dim/2]
         self.rope_cos = ang.cos().to(dev)
         self.rope_sin = ang.sin().to(dev)

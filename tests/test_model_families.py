@@ -191,3 +191,50 @@ def test_swa_mixed_step_late_admission():
     assert len(a1) == len(a2) == 12
     assert a1 == a2, "windowed decode depends on admission timing"
     assert b1 == b2
+
+
+def test_llama2_mha_presets_run():
+    """Llama-2 shape class: MHA (GQA group 1), non-power-of-2 hidden for
+    13B; exercised via a tiny MHA config variant."""
+    import dataclasses
+    cfg = dataclasses.replace(PRESETS["tiny"], name="tiny-mha",
+                              n_kv_heads=PRESETS["tiny"].n_heads)
+    model = LlamaModel(cfg, device="cpu", dtype=torch.float32, seed=3)
+    kv = PagedKVCache.for_model(cfg, n_pages=64, max_slots=2,
+                                max_ctx=cfg.max_ctx)
+    eng = LlamaEngine(model, kv, max_batch=2)
+    s = eng.submit([1, 2, 3], GenParams(max_tokens=5))
+    seq = eng.seqs[s]
+    run_all(eng)
+    assert len(seq.generated) == 5
+    p2, p13 = PRESETS["llama2-7b"], PRESETS["llama2-13b"]
+    assert p2.n_heads == p2.n_kv_heads == 32 and p2.ffn == 11008
+    assert p13.hidden == 5120 and p13.n_kv_heads == 40
+
+
+def test_llama31_rope_scaling_table():
+    """Llama-3.1 NTK scaling: long-wavelength frequencies divided by the
+    factor, short ones untouched, monotone ramp between.  (Checked on a
+    tiny config carrying the 3.1 scaling tuple — the 8B preset only
+    differs in size.)"""
+    import math
+    import dataclasses
+    assert PRESETS["llama3.1-8b"].rope_scaling == (8.0, 1.0, 4.0, 8192)
+    cfg = dataclasses.replace(PRESETS["tiny"], rope_theta=500000.0,
+                              max_ctx=256, rope_scaling=(8.0, 1.0, 4.0, 8192))
+    scaled = LlamaModel(cfg, device="cpu", dtype=torch.float32, seed=1)
+    plain_cfg = dataclasses.replace(cfg, rope_scaling=None)
+    plain = LlamaModel(plain_cfg, device="cpu", dtype=torch.float32, seed=1)
+    # reconstruct inv-freqs from the angle tables at position 1
+    inv_s = torch.atan2(scaled.rope_sin[1], scaled.rope_cos[1])
+    inv_p = torch.atan2(plain.rope_sin[1], plain.rope_cos[1])
+    wavelen = 2 * math.pi / inv_p
+    long_wl = wavelen > 8192          # scaled down by factor 8
+    short_wl = wavelen < 8192 / 4     # untouched
+    torch.testing.assert_close(inv_s[long_wl], inv_p[long_wl] / 8.0,
+                               atol=1e-6, rtol=1e-5)
+    torch.testing.assert_close(inv_s[short_wl], inv_p[short_wl],
+                               atol=1e-6, rtol=1e-6)
+    mid = ~(long_wl | short_wl)
+    assert ((inv_s[mid] <= inv_p[mid] + 1e-6)
+            & (inv_s[mid] >= inv_p[mid] / 8.0 - 1e-6)).all()
