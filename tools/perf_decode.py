@@ -19,10 +19,12 @@ def main():
     hip.require()
     ctx = int(sys.argv[1]) if len(sys.argv) > 1 else 1024
     B = int(sys.argv[2]) if len(sys.argv) > 2 else 32
+    Hq_arg = int(sys.argv[3]) if len(sys.argv) > 3 else 32
+    KVH_arg = int(sys.argv[4]) if len(sys.argv) > 4 else 8
     # 8 layers cycled per call: the working set (8 × B×KVH×ctx×512B) blows
     # past the 256 MiB L3 like the real 32-layer model does — timing one
     # resident layer would measure L3, not HBM (the perf_gemm.py lesson)
-    Hq, KVH, D, L = 32, 8, 128, 8
+    Hq, KVH, D, L = Hq_arg, KVH_arg, 128, 8
     dev = "cuda"
     n_pages = B * (ctx // 16 + 2)
     cache = PagedKVCache(L, KVH, D, page_size=16, n_pages=n_pages,
