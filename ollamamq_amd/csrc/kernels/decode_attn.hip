@@ -82,8 +82,10 @@ __global__ __launch_bounds__(GT * 64, 1) void k_decode_attn(
     constexpr int NSLOT = CH * (DHEAD / 8);       // uint4 slots per tile
     constexpr int NPF = NSLOT / NTHR;             // prefetch regs / thread
     // register prefetch needs every thread to own exactly NPF slots; odd
-    // GQA groups (Qwen2: G=7) take the strided direct-stage path
-    constexpr bool PF = (NSLOT % NTHR == 0) && (NPF <= 4);
+    // GQA groups (Qwen2: G=7) take the strided direct-stage path.  NPF=8
+    // (G=1 at chunk 32, G=2 at chunk 64) costs 64 VGPRs — free at the
+    // 1-2 wave/SIMD occupancy LDS already imposes there
+    constexpr bool PF = (NSLOT % NTHR == 0) && (NPF <= 8);
 
     const int slot = slot_ids[S_idx];
     const int kv_len = seq_lens[S_idx];
@@ -151,6 +153,7 @@ __global__ __launch_bounds__(GT * 64, 1) void k_decode_attn(
         // constant-indexed uint4 arrays here to scratch (24 scratch ops,
         // 144 B/lane), and a spilled prefetch is HBM traffic, not a win.
         uint4 k0, k1, k2, k3, v0, v1, v2, v3;
+        uint4 k4, k5, k6, k7, v4, v5, v6, v7;
         const bool have_next = PF && (ch + 1 < c1);
         if (have_next) {
             const int nbase = (ch + 1) * CH;
@@ -170,6 +173,20 @@ __global__ __launch_bounds__(GT * 64, 1) void k_decode_attn(
                 v2 = *reinterpret_cast<const uint4*>(vpool + s2);
                 k3 = *reinterpret_cast<const uint4*>(kpool + s3);
                 v3 = *reinterpret_cast<const uint4*>(vpool + s3);
+            }
+            if constexpr (NPF >= 8) {
+                const int64_t s4 = src_of(nbase, threadIdx.x + 4 * NTHR);
+                const int64_t s5 = src_of(nbase, threadIdx.x + 5 * NTHR);
+                const int64_t s6 = src_of(nbase, threadIdx.x + 6 * NTHR);
+                const int64_t s7 = src_of(nbase, threadIdx.x + 7 * NTHR);
+                k4 = *reinterpret_cast<const uint4*>(kpool + s4);
+                v4 = *reinterpret_cast<const uint4*>(vpool + s4);
+                k5 = *reinterpret_cast<const uint4*>(kpool + s5);
+                v5 = *reinterpret_cast<const uint4*>(vpool + s5);
+                k6 = *reinterpret_cast<const uint4*>(kpool + s6);
+                v6 = *reinterpret_cast<const uint4*>(vpool + s6);
+                k7 = *reinterpret_cast<const uint4*>(kpool + s7);
+                v7 = *reinterpret_cast<const uint4*>(vpool + s7);
             }
         }
 
@@ -231,6 +248,12 @@ __global__ __launch_bounds__(GT * 64, 1) void k_decode_attn(
             if constexpr (NPF >= 4) {
                 put(threadIdx.x + 2 * NTHR, k2, v2);
                 put(threadIdx.x + 3 * NTHR, k3, v3);
+            }
+            if constexpr (NPF >= 8) {
+                put(threadIdx.x + 4 * NTHR, k4, v4);
+                put(threadIdx.x + 5 * NTHR, k5, v5);
+                put(threadIdx.x + 6 * NTHR, k6, v6);
+                put(threadIdx.x + 7 * NTHR, k7, v7);
             }
             __syncthreads();  // tiles ready for next iteration
         } else if (!PF && ch + 1 < c1) {

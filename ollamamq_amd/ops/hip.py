@@ -237,8 +237,17 @@ def _combine_sem(S, KVH, dev):
     return t
 
 
-# staged KV chunk length (tokens) per pipeline stage; 32/64/128 compiled
-DECODE_CHUNK = int(os.environ.get("OLLAMAMQ_DECODE_CHUNK", "64"))
+# staged KV chunk length (tokens) per pipeline stage; 32/64/128 compiled.
+# 64 wins for GQA groups >= 2 (sweep in NOTES.md); G=1 (MHA, Llama-2)
+# takes 32 so the register-prefetch pipeline fits (NPF=8) — measured
+# 142.9 -> 101.4 us at ctx 512 (see NOTES.md)
+_CHUNK_ENV = os.environ.get("OLLAMAMQ_DECODE_CHUNK")
+
+
+def _decode_chunk(G):
+    if _CHUNK_ENV:
+        return int(_CHUNK_ENV)
+    return 32 if G == 1 else 64
 
 
 def attention_decode(q, cache, layer, meta):
@@ -282,7 +291,8 @@ def attention_decode(q, cache, layer, meta):
         _p(out), op, mp, _p(q), kp, vp, _p(cache.page_table),
         _p(slot32), _p(len32), S, Hq, kvh, cache.page_size,
         cache.page_table.shape[1], 1.0 / (D ** 0.5), _row_stride(q, D),
-        split, meta.window, DECODE_CHUNK, sem, _stream()), "decode_attn")
+        split, meta.window, _decode_chunk(Hq // kvh), sem, _stream()),
+        "decode_attn")
     return out
 
 
