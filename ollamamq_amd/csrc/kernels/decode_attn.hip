@@ -373,10 +373,12 @@ extern "C" int decode_attn_bf16(
         (const bf16*)kpool, (const bf16*)vpool, (const int*)page_table,   \
         (const int*)slot_ids, (const int*)seq_lens, Hq, KVH, page,        \
         max_pages, scale, q_stride, split, window, (unsigned*)sem)
+// chunk is capped at 64: the score phase assigns ONE key per lane of a
+// wave64, so a larger staged chunk would silently drop keys 64+ (a 128
+// arm measured wrong before this guard)
 #define DA_LAUNCH(GT)                                                     \
     switch (chunk) {                                                      \
         case 32: DA_LAUNCH1(GT, 32); break;                               \
-        case 128: DA_LAUNCH1(GT, 128); break;                             \
         default: DA_LAUNCH1(GT, 64); break;                               \
     }
     switch (G) {

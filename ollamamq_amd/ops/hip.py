@@ -237,7 +237,8 @@ def _combine_sem(S, KVH, dev):
     return t
 
 
-# staged KV chunk length (tokens) per pipeline stage; 32/64/128 compiled.
+# staged KV chunk length (tokens) per pipeline stage; 32/64 compiled
+# (the kernel scores one key per wave64 lane, so 64 is the ceiling).
 # 64 wins for GQA groups >= 2 (sweep in NOTES.md); G=1 (MHA, Llama-2)
 # takes 32 so the register-prefetch pipeline fits (NPF=8) — measured
 # 142.9 -> 101.4 us at ctx 512 (see NOTES.md)
@@ -246,7 +247,7 @@ _CHUNK_ENV = os.environ.get("OLLAMAMQ_DECODE_CHUNK")
 
 def _decode_chunk(G):
     if _CHUNK_ENV:
-        return int(_CHUNK_ENV)
+        return min(64, int(_CHUNK_ENV))
     return 32 if G == 1 else 64
 
 
