@@ -80,12 +80,16 @@ __global__ __launch_bounds__(GT * 64, 1) void k_decode_attn(
 
     constexpr int NTHR = GT * 64;
     constexpr int NSLOT = CH * (DHEAD / 8);       // uint4 slots per tile
-    constexpr int NPF = NSLOT / NTHR;             // prefetch regs / thread
-    // register prefetch needs every thread to own exactly NPF slots; odd
-    // GQA groups (Qwen2: G=7) take the strided direct-stage path.  NPF=8
-    // (G=1 at chunk 32, G=2 at chunk 64) costs 64 VGPRs — free at the
-    // 1-2 wave/SIMD occupancy LDS already imposes there
-    constexpr bool PF = (NSLOT % NTHR == 0) && (NPF <= 8);
+    constexpr int NPF = (NSLOT + NTHR - 1) / NTHR;  // prefetch regs/thread
+    // Register-prefetch pipeline for every GQA group a preset uses.
+    // Non-dividing groups (G=7, Qwen2: 1024 slots over 448 threads)
+    // clamp the tail slot exactly like the prefill stager — duplicate
+    // threads re-load/re-write identical bytes, which is benign, while a
+    // guarded load/write would de-pipeline the stage (traps (c)).
+    // NPF=8 (G=1 at chunk 32, G=2 at chunk 64) costs 64 VGPRs — free at
+    // the 1-2 wave/SIMD occupancy LDS already imposes there.
+    constexpr bool PF =
+        (NPF <= 8) && (NSLOT % NTHR == 0 || NPF == 3);
 
     const int slot = slot_ids[S_idx];
     const int kv_len = seq_lens[S_idx];
@@ -166,11 +170,14 @@ __global__ __launch_bounds__(GT * 64, 1) void k_decode_attn(
                 k1 = *reinterpret_cast<const uint4*>(kpool + s1);
                 v1 = *reinterpret_cast<const uint4*>(vpool + s1);
             }
-            if constexpr (NPF >= 4) {
-                const int64_t s2 = src_of(nbase, threadIdx.x + 2 * NTHR);
-                const int64_t s3 = src_of(nbase, threadIdx.x + 3 * NTHR);
+            if constexpr (NPF >= 3) {
+                const int64_t s2 = src_of(
+                    nbase, min((int)threadIdx.x + 2 * NTHR, NSLOT - 1));
                 k2 = *reinterpret_cast<const uint4*>(kpool + s2);
                 v2 = *reinterpret_cast<const uint4*>(vpool + s2);
+            }
+            if constexpr (NPF >= 4) {
+                const int64_t s3 = src_of(nbase, threadIdx.x + 3 * NTHR);
                 k3 = *reinterpret_cast<const uint4*>(kpool + s3);
                 v3 = *reinterpret_cast<const uint4*>(vpool + s3);
             }
@@ -245,10 +252,9 @@ __global__ __launch_bounds__(GT * 64, 1) void k_decode_attn(
             };
             put(threadIdx.x, k0, v0);
             if constexpr (NPF >= 2) put(threadIdx.x + NTHR, k1, v1);
-            if constexpr (NPF >= 4) {
-                put(threadIdx.x + 2 * NTHR, k2, v2);
-                put(threadIdx.x + 3 * NTHR, k3, v3);
-            }
+            if constexpr (NPF >= 3)
+                put(min((int)threadIdx.x + 2 * NTHR, NSLOT - 1), k2, v2);
+            if constexpr (NPF >= 4) put(threadIdx.x + 3 * NTHR, k3, v3);
             if constexpr (NPF >= 8) {
                 put(threadIdx.x + 4 * NTHR, k4, v4);
                 put(threadIdx.x + 5 * NTHR, k5, v5);
