@@ -231,24 +231,62 @@ void HttpServer::handle_conn(int fd, std::string peer_ip) {
             }
         }
         if (req.method.empty() || req.path.empty()) break;
-        // body
-        size_t clen = 0;
-        const std::string cl = req.header("Content-Length");
-        if (!cl.empty()) clen = strtoull(cl.c_str(), nullptr, 10);
-        if (clen > BODY_CAP) {
-            HttpConn c(fd);
-            c.send(400, {}, "{\"error\":\"body too large\"}");
-            break;
-        }
         buf.erase(0, hend);
-        while (buf.size() < clen) {
-            char tmp[65536];
-            ssize_t n = ::recv(fd, tmp, sizeof tmp, 0);
-            if (n <= 0) { ::close(fd); return; }
-            buf.append(tmp, n);
+        // body: Content-Length, or chunked transfer coding (axum parity —
+        // reference clients may stream request bodies)
+        auto fill = [&](size_t need) -> bool {   // grow buf to >= need
+            while (buf.size() < need) {
+                char tmp[65536];
+                ssize_t n = ::recv(fd, tmp, sizeof tmp, 0);
+                if (n <= 0) return false;
+                buf.append(tmp, n);
+            }
+            return true;
+        };
+        if (ieq(req.header("Transfer-Encoding"), "chunked")) {
+            std::string body;
+            bool ok = true;
+            for (;;) {
+                size_t eol;
+                while ((eol = buf.find("\r\n")) == std::string::npos) {
+                    if (!fill(buf.size() + 1)) { ok = false; break; }
+                }
+                if (!ok) break;
+                const size_t csz = strtoull(buf.c_str(), nullptr, 16);
+                buf.erase(0, eol + 2);
+                if (body.size() + csz > BODY_CAP) { ok = false; break; }
+                if (csz == 0) {
+                    // consume optional trailers up to the blank line
+                    size_t tend;
+                    while ((tend = buf.find("\r\n")) == std::string::npos) {
+                        if (!fill(buf.size() + 1)) { ok = false; break; }
+                    }
+                    if (ok) buf.erase(0, tend + 2);
+                    break;
+                }
+                if (!fill(csz + 2)) { ok = false; break; }
+                body.append(buf, 0, csz);
+                buf.erase(0, csz + 2);   // chunk + CRLF
+            }
+            if (!ok) {
+                HttpConn c(fd);
+                c.send(400, {}, "{\"error\":\"bad chunked body\"}");
+                break;
+            }
+            req.body = std::move(body);
+        } else {
+            size_t clen = 0;
+            const std::string cl = req.header("Content-Length");
+            if (!cl.empty()) clen = strtoull(cl.c_str(), nullptr, 10);
+            if (clen > BODY_CAP) {
+                HttpConn c(fd);
+                c.send(400, {}, "{\"error\":\"body too large\"}");
+                break;
+            }
+            if (!fill(clen)) { ::close(fd); return; }
+            req.body = buf.substr(0, clen);
+            buf.erase(0, clen);
         }
-        req.body = buf.substr(0, clen);
-        buf.erase(0, clen);
 
         HttpConn conn(fd);
         handler_(req, conn);
