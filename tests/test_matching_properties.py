@@ -1,0 +1,52 @@
+"""Property-based fuzz of the native model matcher/resolver (C10/C12):
+resolution must never guess — any returned name is exactly available, a
+smart (tag/case) match, or a UNIQUE case-insensitive substring; ambiguity
+returns None.  Inputs include junk unicode, colons, empty strings."""
+import hypothesis.strategies as st
+from hypothesis import given, settings
+
+from ollamamq_amd.dispatch import load
+
+d = load()
+
+# ASCII names: the matcher's case-folding contract is ASCII (like the
+# reference's eq_ignore_ascii_case); unicode goes through byte-equal only
+name_st = st.text(
+    alphabet="abcdefghijklmnopqrstuvwxyzABCDEFGHIJKLMNOPQRSTUVWXYZ"
+             "0123456789:.-_/@",
+    min_size=0, max_size=24)
+
+
+@settings(max_examples=150, deadline=None)
+@given(name_st, st.lists(name_st, min_size=0, max_size=6))
+def test_resolve_never_guesses(req, avail):
+    r = d.resolve_model_name(req, avail, {})
+    if r is None:
+        return
+    assert r in avail, "resolved to a name not in the inventory"
+    low = req.lower()
+    exact = [a for a in avail if a.lower() == low]
+    smart = [a for a in avail if d.smart_model_match_one(req, a)]
+    subs = [a for a in avail if low and low in a.lower()]
+    assert (r in exact) or (r in smart) or (subs == [r]), (
+        f"guessy resolve: {req!r} -> {r!r} from {avail}")
+
+
+@settings(max_examples=150, deadline=None)
+@given(name_st, name_st)
+def test_smart_match_symmetric_on_case(a, b):
+    # case-insensitivity: matching is invariant under case of the request
+    assert d.smart_model_match_one(a, b) == \
+        d.smart_model_match_one(a.upper(), b)
+
+
+@settings(max_examples=100, deadline=None)
+@given(name_st, st.lists(name_st, max_size=5))
+def test_routable_iff_some_match(req, models):
+    routable = d.model_routable(req, models)
+    if not req:
+        assert routable   # no model constraint routes anywhere (C10)
+        return
+    any_match = any(d.smart_model_match_one(req, m) for m in models) \
+        or d.fuzzy_model_match(req, models)
+    assert routable == any_match
