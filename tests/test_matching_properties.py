@@ -3,7 +3,7 @@ resolution must never guess — any returned name is exactly available, a
 smart (tag/case) match, or a UNIQUE case-insensitive substring; ambiguity
 returns None.  Inputs include junk unicode, colons, empty strings."""
 import hypothesis.strategies as st
-from hypothesis import given, settings
+from hypothesis import assume, given, settings
 
 from ollamamq_amd.dispatch import load
 
@@ -35,7 +35,9 @@ def test_resolve_never_guesses(req, avail):
 @settings(max_examples=150, deadline=None)
 @given(name_st, name_st)
 def test_smart_match_symmetric_on_case(a, b):
-    # case-insensitivity: matching is invariant under case of the request
+    # case-insensitivity of the BASE-name path: names with an empty base
+    # (":tag") only match byte-exact (exact-first semantics), so exclude
+    assume(not a.startswith(":") and not b.startswith(":"))
     assert d.smart_model_match_one(a, b) == \
         d.smart_model_match_one(a.upper(), b)
 
