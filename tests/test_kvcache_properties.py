@@ -7,7 +7,7 @@ from hypothesis import given, settings
 from ollamamq_amd.engine.kvcache import PagedKVCache
 
 
-@settings(max_examples=60, deadline=None)
+@settings(derandomize=True, max_examples=60, deadline=None)
 @given(st.lists(st.tuples(st.sampled_from(["alloc", "grow", "free"]),
                           st.integers(0, 7), st.integers(1, 140)),
                 min_size=1, max_size=60))

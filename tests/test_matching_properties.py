@@ -17,7 +17,7 @@ name_st = st.text(
     min_size=0, max_size=24)
 
 
-@settings(max_examples=150, deadline=None)
+@settings(derandomize=True, max_examples=150, deadline=None)
 @given(name_st, st.lists(name_st, min_size=0, max_size=6))
 def test_resolve_never_guesses(req, avail):
     r = d.resolve_model_name(req, avail, {})
@@ -32,7 +32,7 @@ def test_resolve_never_guesses(req, avail):
         f"guessy resolve: {req!r} -> {r!r} from {avail}")
 
 
-@settings(max_examples=150, deadline=None)
+@settings(derandomize=True, max_examples=150, deadline=None)
 @given(name_st, name_st)
 def test_smart_match_symmetric_on_case(a, b):
     # case-insensitivity of the BASE-name path: names with an empty base
@@ -42,7 +42,7 @@ def test_smart_match_symmetric_on_case(a, b):
         d.smart_model_match_one(a.upper(), b)
 
 
-@settings(max_examples=100, deadline=None)
+@settings(derandomize=True, max_examples=100, deadline=None)
 @given(name_st, st.lists(name_st, max_size=5))
 def test_routable_iff_some_match(req, models):
     routable = d.model_routable(req, models)

@@ -8,7 +8,7 @@ from hypothesis import given, settings
 from ollamamq_amd.ops import reference as ref
 
 
-@settings(max_examples=40, deadline=None)
+@settings(derandomize=True, max_examples=40, deadline=None)
 @given(st.integers(1, 8), st.integers(2, 200), st.integers(0, 10 ** 6),
        st.floats(0.05, 4.0), st.integers(0, 8),
        st.floats(0.05, 1.0))
@@ -30,7 +30,7 @@ def test_sampler_invariants(B, V, seed, temp, top_k, top_p):
     assert torch.equal(out, out2)
 
 
-@settings(max_examples=25, deadline=None)
+@settings(derandomize=True, max_examples=25, deadline=None)
 @given(st.integers(1, 6), st.integers(2, 100), st.integers(0, 10 ** 6))
 def test_greedy_always_argmax(B, V, seed):
     g = torch.Generator().manual_seed(seed)
