@@ -110,6 +110,9 @@ __global__ __launch_bounds__(GT * 64, 1) void k_decode_attn(
     bf16* k_tile = reinterpret_cast<bf16*>(smem);
     bf16* v_tile = k_tile + CH * KROW;
     float* q_lds = reinterpret_cast<float*>(v_tile + CH * KROW); // [G][128]
+    // p row is per-LANE (65 = 64+pad), NOT per-chunk-key: every lane
+    // stores its p (dead lanes 0.0), so a CH=32 row of 33 floats would
+    // send lanes 32-63 past the row (OOB at G=1, cross-wave race at G>1)
     float* p_lds = q_lds + (int64_t)G * DHEAD;                       // [G][65]
 
     // stage q (fp32 in LDS: repeated broadcast reads in the score loop)
@@ -226,13 +229,13 @@ __global__ __launch_bounds__(GT * 64, 1) void k_decode_attn(
             o0 *= corr; o1 *= corr;
             m = mn;
         }
-        p_lds[wid * (CH + 1) + lane] = p;
+        p_lds[wid * 65 + lane] = p;
 
         // ---- PV: lane accumulates dims (2*lane, 2*lane+1) ----
         // fixed bound + full unroll: a runtime bound left this loop a
         // serial ds_read latency chain (PMC: 25% SQ_WAIT_INST_ANY);
         // dead keys contribute p=0 so looping to CH is exact
-        const float* prow = p_lds + wid * (CH + 1);
+        const float* prow = p_lds + wid * 65;
         #pragma unroll 8
         for (int j = 0; j < CH; j++) {
             const bf162 v2 = *reinterpret_cast<const bf162*>(
@@ -371,7 +374,7 @@ extern "C" int decode_attn_bf16(
 {
     const int G = Hq / KVH;
     const int lds = 2 * chunk * (DHEAD + DKPAD) * 2 + G * DHEAD * 4
-                  + G * (chunk + 1) * 4;
+                  + G * 65 * 4;
     dim3 grid(S * split, KVH);
 #define DA_LAUNCH1(GT, CH)                                                \
     k_decode_attn<GT, CH><<<grid, GT * 64, lds, stream>>>(                \
