@@ -156,10 +156,13 @@ class Worker:
                 self.work_ev.clear()
 
     def generate(self, model: str, prompt_tokens, params: GenParams,
-                 on_token) -> int:
+                 on_token, num_ctx: int = 0) -> int:
         with self.lock:
             if model not in self.engines:
-                err = self.load(model)
+                # Ollama parity: a request's options.num_ctx shapes the
+                # context of an on-demand load (already-resident models
+                # keep their context; /admin reload changes it)
+                err = self.load(model, num_ctx)
                 if err:
                     raise RuntimeError(err)
             sid = self.engines[model].submit(prompt_tokens, params, on_token)
@@ -333,7 +336,9 @@ class Conn:
         q: "queue.Queue" = queue.Queue()
         t0 = time.time()
         sid = w.generate(model, prompt, params,
-                         lambda t, done: q.put((t, done)))
+                         lambda t, done: q.put((t, done)),
+                         num_ctx=int((body.get("options") or {})
+                                     .get("num_ctx", 0) or 0))
 
         if openai:
             ct = "text/event-stream" if stream else "application/json"

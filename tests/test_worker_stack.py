@@ -219,3 +219,29 @@ def test_multimodal_fields_tolerated(stack):
     assert r.status_code == 200, r.text
     obj = json.loads(r.text.strip())
     assert obj["done"] is True and "created_at" in obj
+
+
+def test_request_num_ctx_shapes_on_demand_load(stack):
+    """options.num_ctx on a generate request sets the context of an
+    on-demand model load (Ollama parity; resident models keep their
+    context); /api/ps reports it."""
+    # ensure not resident (earlier tests may have loaded it)
+    httpx.post(stack + "/admin/models/unload",
+               json={"model": "tiny-cpu", "backend": "any"}, timeout=30.0)
+    deadline = time.time() + 20
+    while time.time() < deadline:
+        ps0 = httpx.get(stack + "/api/ps", timeout=10.0).json()["models"]
+        if not any(m["name"] == "tiny-cpu" for m in ps0):
+            break
+        time.sleep(0.3)
+    r = httpx.post(stack + "/api/generate",
+                   json={"model": "tiny-cpu", "prompt": "x",
+                         "stream": False,
+                         "options": {"num_predict": 2, "num_ctx": 128}},
+                   headers={"X-User-ID": "nctx"}, timeout=120.0)
+    assert r.status_code == 200, r.text
+    ps = httpx.post(stack + "/api/ps", json={}, timeout=30.0)
+    if ps.status_code != 200:
+        ps = httpx.get(stack + "/api/ps", timeout=30.0)
+    models = {m["name"]: m for m in ps.json()["models"]}
+    assert models["tiny-cpu"]["context_length"] == 128
