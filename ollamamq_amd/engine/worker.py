@@ -98,11 +98,15 @@ class Worker:
             except Exception as e:  # pragma: no cover
                 return f"load failed: {e}"
 
-    def unload(self, model: str) -> Optional[str]:
+    def unload(self, model: str, only_if_idle: bool = False) \
+            -> Optional[str]:
         with self.lock:
-            eng = self.engines.pop(model, None)
+            eng = self.engines.get(model)
             if eng is None:
                 return f"model not loaded: {model}"
+            if only_if_idle and eng.has_work():
+                return "model busy"
+            self.engines.pop(model, None)
             self.tokenizers.pop(model, None)
             self.loaded_ctx.pop(model, None)
             del eng
@@ -368,6 +372,12 @@ class Conn:
                     self._stream_piece(path, model, piece, openai)
             self._final(path, model, pieces, n_out, t0, openai, stream,
                         len(prompt))
+            # Ollama parity: "keep_alive": 0 on the request frees the
+            # model after the response (only if the engine is idle —
+            # other users' in-flight sequences always win)
+            if body.get("keep_alive") == 0 or \
+                    (body.get("options") or {}).get("keep_alive") == 0:
+                w.unload(model, only_if_idle=True)
         except (BrokenPipeError, ConnectionResetError, OSError):
             w.cancel(model, sid)
         except queue.Empty:

@@ -245,3 +245,21 @@ def test_request_num_ctx_shapes_on_demand_load(stack):
         ps = httpx.get(stack + "/api/ps", timeout=30.0)
     models = {m["name"]: m for m in ps.json()["models"]}
     assert models["tiny-cpu"]["context_length"] == 128
+
+
+def test_request_keep_alive_zero_unloads(stack):
+    """Ollama parity: "keep_alive": 0 on a generate request frees the
+    model after the response completes (if idle)."""
+    r = httpx.post(stack + "/api/generate",
+                   json={"model": "tiny-cpu", "prompt": "y",
+                         "stream": False, "keep_alive": 0,
+                         "options": {"num_predict": 2}},
+                   headers={"X-User-ID": "ka0"}, timeout=120.0)
+    assert r.status_code == 200, r.text
+    deadline = time.time() + 20
+    while time.time() < deadline:
+        ps = httpx.get(stack + "/api/ps", timeout=10.0).json()["models"]
+        if not any(m["name"] == "tiny-cpu" for m in ps):
+            return
+        time.sleep(0.3)
+    raise AssertionError(f"tiny-cpu still resident: {ps}")
