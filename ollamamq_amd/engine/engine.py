@@ -397,7 +397,13 @@ class LlamaEngine:
     def _graph_replay(self, entry):
         bufs, meta = entry["bufs"], entry["meta"]
         if entry["graph"] is None:
-            # warmup twice on a side stream, then capture
+            # warmup twice on a side stream, then capture.  The Python GC
+            # must not run DURING capture: collecting a dead CUDA tensor
+            # issues a hipFree, which is illegal while a stream is
+            # capturing and aborts the process (observed under the full
+            # GPU suite) — flush pending frees first, then hold GC off.
+            import gc
+            gc.collect()
             s = torch.cuda.Stream()
             s.wait_stream(torch.cuda.current_stream())
             with torch.cuda.stream(s):
@@ -405,20 +411,27 @@ class LlamaEngine:
                     self.model.forward(bufs["tok"], bufs["pos"], self.kv,
                                        bufs["slot"], meta)
             torch.cuda.current_stream().wait_stream(s)
-            g = torch.cuda.CUDAGraph()
-            with torch.cuda.graph(g):
-                entry["logits"] = self.model.forward(
-                    bufs["tok"], bufs["pos"], self.kv, bufs["slot"], meta)
-                # self-advancing tail: greedy-sample in-graph and stage the
-                # NEXT step's inputs on-device, so a steady decode batch
-                # replays back-to-back with no host round-trip.  The
-                # non-pipelined/stochastic path simply overwrites the bufs
-                # next step and samples logits itself.
-                toks = ops.sample(entry["logits"], 0.0, 0, 1.0, None)
-                entry["out"] = toks
-                bufs["tok"].copy_(toks)
-                bufs["pos"] += 1
-                bufs["lens"] += 1
+            torch.cuda.synchronize()
+            gc.disable()
+            try:
+                g = torch.cuda.CUDAGraph()
+                with torch.cuda.graph(g):
+                    entry["logits"] = self.model.forward(
+                        bufs["tok"], bufs["pos"], self.kv, bufs["slot"],
+                        meta)
+                    # self-advancing tail: greedy-sample in-graph and
+                    # stage the NEXT step's inputs on-device, so a steady
+                    # decode batch replays back-to-back with no host
+                    # round-trip.  The non-pipelined/stochastic path
+                    # simply overwrites the bufs next step and samples
+                    # logits itself.
+                    toks = ops.sample(entry["logits"], 0.0, 0, 1.0, None)
+                    entry["out"] = toks
+                    bufs["tok"].copy_(toks)
+                    bufs["pos"] += 1
+                    bufs["lens"] += 1
+            finally:
+                gc.enable()
             entry["graph"] = g
         entry["graph"].replay()
         return entry["logits"]
