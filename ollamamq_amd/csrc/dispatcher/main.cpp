@@ -34,6 +34,7 @@ static void usage() {
         "  -t, --timeout <S>         request timeout seconds (default 300)\n"
         "      --load-keep-alive <S> control-load keep_alive (default 86400)\n"
         "      --stuck-timeout <S>   queue stuck timeout -> 503 (default 60)\n"
+        "      --probe-interval-ms <M> health probe cadence (default 10000)\n"
         "      --allow-all-routes    proxy unknown routes too\n"
         "      --no-tui              headless (logs to stderr)\n"
         "  -c, --model-config <F>    config file (default appconf.yaml)\n";

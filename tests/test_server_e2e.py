@@ -317,3 +317,16 @@ def test_metrics_prometheus(fleet, tmp_path_factory):
         assert float(line.split()[-1]) >= 1
     finally:
         p.stop()
+
+
+def test_cli_help_and_unknown_flag():
+    r = subprocess.run([BIN, "--help"], capture_output=True, text=True,
+                       timeout=15)
+    assert r.returncode == 0
+    for flag in ("-p, --port", "--workers", "--stuck-timeout",
+                 "--probe-interval-ms", "--no-tui"):
+        assert flag in r.stdout
+    r2 = subprocess.run([BIN, "--bogus-flag"], capture_output=True,
+                        text=True, timeout=15)
+    assert r2.returncode == 2
+    assert "unknown flag" in r2.stderr
