@@ -30,8 +30,14 @@ def one_request(base, user, model, rng, cancel_pct, max_tokens):
     openai = ep.startswith("/v1/")
     prompt = "".join(rng.choice("abcdefghij ") for _ in range(rng.randint(8, 200)))
     if "chat" in ep:
-        body = {"model": model,
-                "messages": [{"role": "user", "content": prompt}]}
+        msg = {"role": "user", "content": prompt}
+        # reference-stress parity: ~5% multimodal requests carry a
+        # base64 image; text models accept and ignore the bytes
+        if rng.random() < 0.05:
+            import base64
+            msg["images"] = [base64.b64encode(
+                rng.randbytes(256)).decode()]
+        body = {"model": model, "messages": [msg]}
     else:
         body = {"model": model, "prompt": prompt}
     if openai:
