@@ -18,7 +18,7 @@ write failure and cancels the sequence (KV pages freed) — the reference
 only dropped bytes, we stop the compute (SURVEY.md §7 hard-part 4).
 
 Run:  python -m ollamamq_amd.engine.worker --socket /tmp/omq0.sock \
-          --gpu 0 [--model llama3-8b --max-ctx 4096 --preload]
+          --gpu 0 [--model llama3-8b --max-ctx 4096 --max-batch 64]
 """
 from __future__ import annotations
 
