@@ -263,3 +263,19 @@ def test_request_keep_alive_zero_unloads(stack):
             return
         time.sleep(0.3)
     raise AssertionError(f"tiny-cpu still resident: {ps}")
+
+
+def test_multimodal_images_accepted_and_ignored(stack):
+    """Ollama multimodal shape: base64 `images` on a chat message must
+    not break a text-only model (bytes accepted, ignored)."""
+    import base64
+    r = httpx.post(stack + "/api/chat",
+                   json={"model": "tiny-cpu",
+                         "messages": [{"role": "user", "content": "hi",
+                                       "images": [base64.b64encode(
+                                           b"\x89PNG fake").decode()]}],
+                         "stream": False,
+                         "options": {"num_predict": 3}},
+                   headers={"X-User-ID": "img"}, timeout=120.0)
+    assert r.status_code == 200, r.text
+    assert r.json()["done"] is True
