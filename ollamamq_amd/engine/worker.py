@@ -495,13 +495,19 @@ class Conn:
             obj = {"version": VERSION}
         elif path == "/api/show":
             m = body.get("model", "")
-            cfg = PRESETS.get(m)
-            obj = ({"error": "model not found"} if cfg is None else
-                   {"details": {"family": "llama",
-                                "parameter_size": m},
-                    "model_info": {"n_layers": cfg.n_layers,
-                                   "hidden": cfg.hidden,
-                                   "context_length": cfg.max_ctx}})
+            cfg = PRESETS.get(m) or PRESETS.get(w.resolve(m) or "")
+            if cfg is None:
+                # Ollama returns 404 for unknown models on /api/show
+                self._line({"status": 404,
+                            "content_type": "application/json"})
+                self.sock.sendall(json.dumps(
+                    {"error": f"model not found: {m}"}).encode())
+                return
+            obj = {"details": {"family": "llama",
+                               "parameter_size": cfg.name},
+                   "model_info": {"n_layers": cfg.n_layers,
+                                  "hidden": cfg.hidden,
+                                  "context_length": cfg.max_ctx}}
         else:
             obj = {"status": "ollamamq-amd worker",
                    "device": w.device, "loaded": loaded}

@@ -279,3 +279,13 @@ def test_multimodal_images_accepted_and_ignored(stack):
                    headers={"X-User-ID": "img"}, timeout=120.0)
     assert r.status_code == 200, r.text
     assert r.json()["done"] is True
+
+
+def test_api_show_known_and_unknown(stack):
+    r = httpx.post(stack + "/api/show", json={"model": "tiny-cpu"},
+                   timeout=30.0)
+    assert r.status_code == 200
+    assert r.json()["model_info"]["hidden"] == 256
+    r = httpx.post(stack + "/api/show", json={"model": "nope-xyz"},
+                   timeout=30.0)
+    assert r.status_code == 404
