@@ -286,6 +286,10 @@ def test_api_show_known_and_unknown(stack):
                    timeout=30.0)
     assert r.status_code == 200
     assert r.json()["model_info"]["hidden"] == 256
-    r = httpx.post(stack + "/api/show", json={"model": "nope-xyz"},
+    # an UNROUTABLE model parks in the dispatcher queue (reference
+    # semantics) — the worker-level 404 needs a name that routes (fuzzy
+    # substring) but is ambiguous at the worker resolver ("t" matches
+    # both tiny and tiny-cpu, and never-guess resolution returns none)
+    r = httpx.post(stack + "/api/show", json={"model": "t"},
                    timeout=30.0)
     assert r.status_code == 404
