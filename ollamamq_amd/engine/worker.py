@@ -302,7 +302,7 @@ class Conn:
         w = self.worker
 
         if path in ("/api/tags", "/v1/models", "/api/ps", "/api/version",
-                    "/", "/api/show"):
+                    "/", "/api/show") or path.startswith("/v1/models/"):
             return self._meta(path, body)
         if path in ("/api/embed", "/api/embeddings", "/v1/embeddings"):
             return self._embed(path, body)
@@ -493,6 +493,16 @@ class Conn:
                               for m in loaded]}
         elif path == "/api/version":
             obj = {"version": VERSION}
+        elif path.startswith("/v1/models/"):
+            m = path[len("/v1/models/"):]
+            r = w.resolve(m)
+            if r is None:
+                self._line({"status": 404,
+                            "content_type": "application/json"})
+                self.sock.sendall(json.dumps(
+                    {"error": f"model not found: {m}"}).encode())
+                return
+            obj = {"id": r, "object": "model"}
         elif path == "/api/show":
             m = body.get("model", "")
             cfg = PRESETS.get(m) or PRESETS.get(w.resolve(m) or "")

@@ -293,3 +293,9 @@ def test_api_show_known_and_unknown(stack):
     r = httpx.post(stack + "/api/show", json={"model": "t"},
                    timeout=30.0)
     assert r.status_code == 404
+
+
+def test_v1_models_item_route(stack):
+    r = httpx.get(stack + "/v1/models/tiny-cpu", timeout=30.0)
+    assert r.status_code == 200
+    assert r.json() == {"id": "tiny-cpu", "object": "model"}
