@@ -110,3 +110,15 @@ def test_chunked_request_body_garbage_size(srv):
     raw(srv, b"POST /api/chat HTTP/1.1\r\n"
              b"Transfer-Encoding: chunked\r\n\r\nZZZ\r\n")
     assert alive(srv)
+
+
+def test_deeply_nested_json_body(srv):
+    """A 100k-deep JSON nesting bomb must be rejected by the parser's
+    depth guard, not blow the stack."""
+    bomb = b"[" * 100000 + b"]" * 100000
+    data = (b"POST /admin/models/load HTTP/1.1\r\nHost: x\r\n"
+            b"Content-Length: " + str(len(bomb)).encode() + b"\r\n\r\n"
+            + bomb)
+    r = raw(srv, data, expect_reply=True)
+    assert r.startswith(b"HTTP/1.1 4"), r[:60]   # clean 4xx, no crash
+    assert alive(srv)
