@@ -201,6 +201,11 @@ static bool read_headers(int fd, std::string& buf, size_t& header_end) {
 void HttpServer::handle_conn(int fd, std::string peer_ip) {
     int one = 1;
     setsockopt(fd, IPPROTO_TCP, TCP_NODELAY, &one, sizeof one);
+    // read timeout: a client trickling headers/body (slowloris) must not
+    // pin a connection thread forever; writes stay unbounded so slow
+    // CONSUMERS of long token streams are unaffected
+    timeval rto{120, 0};
+    setsockopt(fd, SOL_SOCKET, SO_RCVTIMEO, &rto, sizeof rto);
     std::string buf;
     while (!stopping_) {
         size_t hend = 0;
