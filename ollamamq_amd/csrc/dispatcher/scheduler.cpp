@@ -36,11 +36,15 @@ std::vector<std::string> candidate_order(
     const std::string& vip, const std::string& boost, uint64_t counter) {
     // stable sort by total processed ascending (least-served first),
     // reference src/dispatcher.rs:518-534
+    // boost is pulled out only on the ticks it is actually boosted (every
+    // 2nd); on other ticks it competes at its least-served sorted position
+    // like any other user (src/dispatcher.rs:538-549)
+    const bool boost_tick = counter % 2 == 0;
     std::vector<std::pair<std::string, int64_t>> rest;
     bool has_vip = false, has_boost = false;
     for (const auto& [u, n] : active_users) {
         if (u == vip) { has_vip = true; continue; }
-        if (u == boost) { has_boost = true; continue; }
+        if (u == boost && boost_tick) { has_boost = true; continue; }
         rest.emplace_back(u, n);
     }
     std::stable_sort(rest.begin(), rest.end(),
@@ -55,10 +59,7 @@ std::vector<std::string> candidate_order(
             order.push_back(rest[(rot + i) % rest.size()].first);
     }
     // boost: prepended on every 2nd tick (src/dispatcher.rs:538-549)
-    if (has_boost && counter % 2 == 0)
-        order.insert(order.begin(), boost);
-    else if (has_boost)
-        order.push_back(boost);
+    if (has_boost) order.insert(order.begin(), boost);
     // VIP: absolute head (src/dispatcher.rs:507,538)
     if (has_vip) order.insert(order.begin(), vip);
     return order;
@@ -66,10 +67,11 @@ std::vector<std::string> candidate_order(
 
 // ------------------------------------------------------------- eligibility
 static bool family_supported(ApiType t, const std::string& path) {
-    if (t == ApiType::Both) return true;
+    // Unknown-type backends are allowed for BOTH families: the health loop
+    // will classify them (reference src/dispatcher.rs:609-612)
+    if (t == ApiType::Both || t == ApiType::Unknown) return true;
     const bool openai = path.rfind("/v1/", 0) == 0;
-    if (openai) return t == ApiType::OpenAi;
-    return t == ApiType::Ollama || t == ApiType::Unknown;
+    return openai ? t == ApiType::OpenAi : t == ApiType::Ollama;
 }
 
 bool backend_eligible(const BackendStatus& b, bool has_control_op,
@@ -78,11 +80,13 @@ bool backend_eligible(const BackendStatus& b, bool has_control_op,
     if (!b.is_online) return false;
     if (b.active_requests >= b.max_concurrency) return false;
     if (has_control_op) return false;
-    if (!family_supported(b.api_type, path)) return false;
-    if (!requested_model.empty() &&
-        !model_routable(requested_model, b.available_models))
-        return false;
-    return true;
+    // A specific model is the HARD gate when present; the API-family check
+    // applies only to model-less requests (reference src/dispatcher.rs:599-616:
+    // "If a specific model is requested, backend MUST have it. If no model is
+    // requested, fall back to API family check.")
+    if (!requested_model.empty())
+        return model_routable(requested_model, b.available_models);
+    return family_supported(b.api_type, path);
 }
 
 size_t pick_backend(const std::vector<BackendStatus>& backends,

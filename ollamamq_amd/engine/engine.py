@@ -138,7 +138,12 @@ class LlamaEngine:
             # the new prompts prefill a chunk in the SAME forward — one
             # weight pass serves both, so admission never stalls decode
             self._mixed_step(admitted)
-            finished = self._postprocess(self.running + admitted)
+            # admitted seqs that finished prefill were moved into
+            # self.running by _mixed_step — don't postprocess them twice
+            # (a double _finish would emit the done marker twice)
+            finished = self._postprocess(
+                self.running + [s for s in admitted
+                                if s not in self.running])
         elif admitted:
             self._prefill_step(admitted)
             finished = self._postprocess(admitted)
@@ -571,6 +576,8 @@ class LlamaEngine:
         return finished
 
     def _finish(self, seq: Sequence, emit: bool = True):
+        if seq.state == "done":    # already finished: never double-emit
+            return
         if seq in self.running:
             self.running.remove(seq)
         if seq in self.waiting:

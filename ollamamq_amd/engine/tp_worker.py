@@ -2,12 +2,29 @@
 
 Launched by ollamamq_amd.launch via torch.distributed.run (one process per
 GPU over RCCL/xGMI).  Rank 0 owns the dispatcher socket and the request
-surface; every rank runs an identical engine replica in lockstep: rank 0
-broadcasts the op-stream (submissions / cancels / loads) each loop tick
-and all ranks call engine.step() together, so the 2 all-reduces per layer
-and the vocab-parallel all-gather line up by construction (the engine is
-deterministic given the same op order; sequence ids and sampling
-generators are seeded identically).
+surface; every rank runs an identical engine replica in lockstep, so the
+2 all-reduces per layer and the vocab-parallel all-gather line up by
+construction (the engine is deterministic given the same op order;
+sequence ids and sampling generators are seeded identically).
+
+Steady-state protocol (the part the reference has no analog for — its
+"collective" is a tokio Notify inside one process, reference
+src/dispatcher.rs:170-171): the op stream (submit/cancel/load/unload) is
+synchronized with ONE 8-byte tensor broadcast per sync point, and sync
+points happen every SYNC_EVERY engine steps — NOT every step.  Between
+sync points all ranks free-run SYNC_EVERY lockstep decode steps with no
+host-side exchange at all: because every rank is a replica, has_work()
+and sequence finishes agree without communication.  The pickled
+broadcast_object_list is only paid when ops actually arrived (rare — at
+request cadence, not token cadence).
+
+Decode steps run inside hipGraphs with the RCCL collectives captured
+in-graph (engine.use_graphs stays ON at TP>1): each replay issues the
+~7·n_layers kernels AND the 2·n_layers all-reduces from the graph,
+eliminating per-step launch + collective-issue overhead.  Graph capture
+happens in lockstep on all ranks (same op order ⇒ same capture step),
+and the pre-capture eager warmups initialize the RCCL communicator
+before capture begins (a capture-time communicator init is illegal).
 
 CPU-testable with gloo (tests/test_tp_worker.py drives a 2-rank group).
 """
@@ -27,6 +44,12 @@ from .kvcache import PagedKVCache
 from .tokenizer import ByteTokenizer
 from . import worker as worker_mod
 
+# Engine steps between op-stream sync points.  Ops arriving mid-window
+# wait at most SYNC_EVERY steps (~45 ms at 5.6 ms/step) — request-level
+# latency, invisible next to prefill.  Smaller values add one tiny
+# broadcast per window; at 1 the protocol degenerates to per-step sync.
+SYNC_EVERY = 8
+
 
 class TPWorker(worker_mod.Worker):
     """Rank-0 worker whose engine loop drives the whole TP group."""
@@ -38,6 +61,11 @@ class TPWorker(worker_mod.Worker):
         self.gpu_base = gpu_base
         self._pending_ops = []
         self._ops_mu = threading.Lock()
+        # op-count sync word: device-resident under RCCL so the idle-tick
+        # broadcast never touches host memory paths
+        flag_dev = (f"cuda:{gpu_base + rank}"
+                    if torch.cuda.is_available() else "cpu")
+        self._flag = torch.zeros(1, dtype=torch.int64, device=flag_dev)
         super().__init__(gpu_base + rank, max_batch, default_ctx)
 
     # --- model lifecycle (applied on every rank via the op stream) -----
@@ -61,8 +89,9 @@ class TPWorker(worker_mod.Worker):
                 max_slots=self.max_batch + 2, max_ctx=ctx,
                 device=self.device, dtype=self.dtype)
             eng = LlamaEngine(m, kv, max_batch=self.max_batch)
-            if self.world > 1:
-                eng.use_graphs = False  # graphs+RCCL: enable once validated
+            # lockstep capture: every rank reaches warm_graphs at the same
+            # op, so the in-graph RCCL collectives pair up across ranks
+            eng.warm_graphs()
             self.engines[model] = eng
             self.tokenizers[model] = ByteTokenizer(cfg.vocab)
             self.loaded_ctx[model] = ctx
@@ -70,15 +99,18 @@ class TPWorker(worker_mod.Worker):
         except torch.cuda.OutOfMemoryError:
             return "out of HBM: model + KV pool do not fit"
 
-    # --- op-stream plumbing -------------------------------------------
+    # --- op-stream plumbing (rank-0 request surface) -------------------
+    # Signatures MUST match worker.Worker: Conn._request calls
+    # generate(..., num_ctx=...) and unload(..., only_if_idle=True)
+    # (worker.py:342-345,380).
     def load(self, model, num_ctx=0):
         # queued into the op stream so every rank allocates together
         return self._rpc(("load", model, num_ctx))
 
-    def unload(self, model):
-        return self._rpc(("unload", model))
+    def unload(self, model, only_if_idle=False):
+        return self._rpc(("unload", model, only_if_idle))
 
-    def generate(self, model, prompt_tokens, params, on_token):
+    def generate(self, model, prompt_tokens, params, on_token, num_ctx=0):
         done = threading.Event()
         box = {}
 
@@ -88,7 +120,8 @@ class TPWorker(worker_mod.Worker):
 
         with self._ops_mu:
             self._pending_ops.append(
-                (("submit", model, prompt_tokens, params), record, on_token))
+                (("submit", model, prompt_tokens, params, num_ctx),
+                 record, on_token))
         self.work_ev.set()
         done.wait(timeout=900)
         r = box.get("r")
@@ -121,19 +154,23 @@ class TPWorker(worker_mod.Worker):
         if kind == "load":
             return self._do_load(op[1], op[2])
         if kind == "unload":
-            eng = self.engines.pop(op[1], None)
-            self.tokenizers.pop(op[1], None)
-            self.loaded_ctx.pop(op[1], None)
+            model, only_if_idle = op[1], op[2]
+            eng = self.engines.get(model)
             if eng is None:
-                return f"model not loaded: {op[1]}"
+                return f"model not loaded: {model}"
+            if only_if_idle and eng.has_work():
+                return "model busy"
+            self.engines.pop(model, None)
+            self.tokenizers.pop(model, None)
+            self.loaded_ctx.pop(model, None)
             del eng
             if torch.cuda.is_available():
                 torch.cuda.empty_cache()
             return None
         if kind == "submit":
-            _, model, prompt, params = op
+            _, model, prompt, params, num_ctx = op
             if model not in self.engines:
-                err = self._do_load(model, 0)
+                err = self._do_load(model, num_ctx)
                 if err:
                     return err
             return self.engines[model].submit(prompt, params, on_token)
@@ -144,6 +181,23 @@ class TPWorker(worker_mod.Worker):
             return None
         return f"bad op {kind}"
 
+    def _run_steps(self, budget: int) -> bool:
+        """Free-run up to `budget` lockstep engine steps (no collectives
+        beyond the ones inside forward).  All ranks take the same number
+        of steps because the engines are replicas — has_work() and
+        finishes agree by determinism, not by communication."""
+        did = False
+        for _ in range(budget):
+            stepped = False
+            for eng in list(self.engines.values()):
+                if eng.has_work():
+                    eng.step()
+                    stepped = True
+            if not stepped:
+                break
+            did = True
+        return did
+
     def _engine_loop(self):
         while True:
             if self.rank == 0:
@@ -151,38 +205,35 @@ class TPWorker(worker_mod.Worker):
                     batch = self._pending_ops
                     self._pending_ops = []
                 ops = [b[0] for b in batch]
-                with self.lock:
-                    work = any(e.has_work() for e in self.engines.values())
-                do_step = work or bool(ops)
-                payload = [ops, do_step]
                 if self.world > 1:
-                    dist.broadcast_object_list(payload, src=0)
+                    self._flag[0] = len(ops)
+                    dist.broadcast(self._flag, src=0)
+                    if ops:
+                        dist.broadcast_object_list([ops], src=0)
                 results = []
                 with self.lock:
                     for op, record, on_token in batch:
                         results.append((record,
                                         self._apply_op(op, on_token)))
-                    if do_step:
-                        for eng in list(self.engines.values()):
-                            if eng.has_work():
-                                eng.step()
+                    did = self._run_steps(SYNC_EVERY)
                 for record, r in results:
                     if record:
                         record(r)
-                if not do_step:
+                if not did and not ops:
                     self.work_ev.wait(timeout=0.02)
                     self.work_ev.clear()
             else:
-                payload = [None, None]
-                dist.broadcast_object_list(payload, src=0)
-                ops, do_step = payload
+                dist.broadcast(self._flag, src=0)
+                n = int(self._flag.item())
+                ops = []
+                if n:
+                    payload = [None]
+                    dist.broadcast_object_list(payload, src=0)
+                    ops = payload[0]
                 with self.lock:
                     for op in ops:
                         self._apply_op(op, None)
-                    if do_step:
-                        for eng in list(self.engines.values()):
-                            if eng.has_work():
-                                eng.step()
+                    self._run_steps(SYNC_EVERY)
 
 
 def follower_loop(w: TPWorker):

@@ -250,7 +250,12 @@ class LlamaModel:
 
     # -- helpers -----------------------------------------------------------
     def _allreduce(self, x: torch.Tensor) -> torch.Tensor:
-        if self.tp_size > 1:
+        # tp_size > 1: the 2-per-layer TP all-reduces.  An explicitly
+        # passed group at tp_size == 1 ALSO reduces (numerically a no-op):
+        # that is how the graph+RCCL capture mechanics are validated on a
+        # single GPU (tests/test_tp_graph_gpu.py) before an 8-GPU node
+        # ever runs them.
+        if self.tp_size > 1 or self.group is not None:
             torch.distributed.all_reduce(x, group=self.group)
         return x
 

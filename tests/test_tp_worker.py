@@ -71,3 +71,80 @@ def test_tp_worker_lockstep_matches_single():
         p.terminate()
         p.join(timeout=30)
     assert toks == expected, f"{toks} != {expected}"
+
+
+def _rank_uds(rank, world, port, sock_path, q):
+    """Rank 0 serves the REAL UDS wire (worker.serve + Conn), exercising
+    Conn._request -> TPWorker.generate(..., num_ctx=...) and the
+    keep_alive:0 -> unload(only_if_idle=True) path that ADVICE r01 found
+    untested (TPWorker signatures must match worker.Worker's)."""
+    os.environ["MASTER_ADDR"] = "127.0.0.1"
+    os.environ["MASTER_PORT"] = str(port)
+    import json
+    import socket
+    import threading
+    import time
+    import torch.distributed as dist
+    from ollamamq_amd.engine.tp_worker import TPWorker
+    from ollamamq_amd.engine import worker as worker_mod
+    dist.init_process_group("gloo", rank=rank, world_size=world)
+    w = TPWorker(rank, world, 0, max_batch=4, default_ctx=128)
+    if rank != 0:
+        time.sleep(120)
+        return
+    threading.Thread(target=worker_mod.serve, args=(sock_path, w),
+                     daemon=True).start()
+    for _ in range(100):
+        if os.path.exists(sock_path):
+            break
+        time.sleep(0.05)
+
+    def req(body):
+        c = socket.socket(socket.AF_UNIX, socket.SOCK_STREAM)
+        c.connect(sock_path)
+        c.sendall((json.dumps({"cmd": "request", "method": "POST",
+                               "path": "/api/generate",
+                               "body": json.dumps(body)}) + "\n").encode())
+        data = b""
+        while True:
+            chunk = c.recv(65536)
+            if not chunk:
+                break
+            data += chunk
+        c.close()
+        return data.decode()
+
+    # num_ctx flows through generate(..., num_ctx=96) into the on-demand
+    # load; keep_alive:0 triggers unload(only_if_idle=True) after the
+    # response — both raise TypeError if the TP signatures drift again
+    out = req({"model": "tiny", "prompt": "hi",
+               "options": {"num_ctx": 96, "num_predict": 3},
+               "keep_alive": 0, "stream": False})
+    lines = [json.loads(l) for l in out.splitlines() if l.strip()]
+    assert lines[0]["status"] == 200, out
+    assert lines[-1].get("done") is True, out
+    # keep_alive:0 + idle engine => model was unloaded on all ranks
+    deadline = time.time() + 10
+    while time.time() < deadline and "tiny" in w.engines:
+        time.sleep(0.05)
+    q.put(("ok", "tiny" not in w.engines))
+
+
+def test_tp_worker_uds_wire_and_keepalive_unload():
+    ctx = mp.get_context("spawn")
+    q = ctx.Queue()
+    sock = f"/tmp/omq_tp_test_{os.getpid()}.sock"
+    procs = [ctx.Process(target=_rank_uds, args=(r, 2, 29591, sock, q),
+                         daemon=True) for r in range(2)]
+    for p in procs:
+        p.start()
+    tag, unloaded = q.get(timeout=300)
+    for p in procs:
+        p.terminate()
+        p.join(timeout=30)
+    try:
+        os.unlink(sock)
+    except FileNotFoundError:
+        pass
+    assert tag == "ok"
+    assert unloaded, "keep_alive:0 did not unload the idle model"
