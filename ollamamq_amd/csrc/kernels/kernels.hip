@@ -691,3 +691,38 @@ extern "C" int rope_append_bf16(
         T, Hq, KVH, D, page, max_pages, qs, ks, vs);
     return (int)hipGetLastError();
 }
+
+// ---------------------------------------------------------------------------
+// Embedding gather: out[t] = table[tokens[t]]  (bf16 -> bf16, no casts).
+// The torch path (index_select on a .long() cast) cost two kernels and 5%
+// of the decode trace (profiles/r01_final_kernel_stats.md); this is one
+// memory-bound pass of 16 B vector copies with the row indirection read
+// through L1 (tokens[] is tiny).  Grid-stride so prefill T=2048 fills the
+// chip and decode T=32 stays a single small launch.
+// ---------------------------------------------------------------------------
+__global__ __launch_bounds__(256) void k_embed_gather(
+    uint4* __restrict__ out, const uint4* __restrict__ table,
+    const int* __restrict__ tokens, int64_t chunks_per_row, int64_t total)
+{
+    int64_t g = (int64_t)blockIdx.x * blockDim.x + threadIdx.x;
+    const int64_t stride = (int64_t)gridDim.x * blockDim.x;
+    for (; g < total; g += stride) {
+        const int64_t row = g / chunks_per_row;
+        const int64_t col = g - row * chunks_per_row;
+        out[g] = table[(int64_t)tokens[row] * chunks_per_row + col];
+    }
+}
+
+extern "C" int embed_gather_bf16(void* out, const void* table,
+                                 const void* tokens, int T, int H,
+                                 hipStream_t stream)
+{
+    const int64_t cpr = H / 8;           // 16 B chunks per row (H % 8 == 0)
+    const int64_t total = (int64_t)T * cpr;
+    if (total == 0) return 0;
+    int64_t blocks = (total + 255) / 256;
+    if (blocks > 8192) blocks = 8192;    // grid-stride covers the rest
+    k_embed_gather<<<(int)blocks, 256, 0, stream>>>(
+        (uint4*)out, (const uint4*)table, (const int*)tokens, cpr, total);
+    return (int)hipGetLastError();
+}

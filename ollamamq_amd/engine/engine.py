@@ -305,12 +305,16 @@ class LlamaEngine:
         return logits
 
     def _sample(self, seqs, logits):
+        dev = logits.device
         temps = torch.tensor([s.params.temperature for s in seqs],
-                             dtype=torch.float32, device=logits.device)
-        # engine-wide top_k/top_p: use per-seq params of the first seq's
-        # class; mixed params fall back to the max (conservative).
-        top_k = max((s.params.top_k or 0) for s in seqs)
-        top_p = min((s.params.top_p if s.params.top_p else 1.0) for s in seqs)
+                             dtype=torch.float32, device=dev)
+        # per-request top_k/top_p: each row is filtered by its own params
+        # (vectorized in ops.sample — mixed batches never bleed)
+        top_k = torch.tensor([s.params.top_k or 0 for s in seqs],
+                             dtype=torch.long, device=dev)
+        top_p = torch.tensor(
+            [s.params.top_p if s.params.top_p else 1.0 for s in seqs],
+            dtype=torch.float32, device=dev)
         toks = ops.sample(logits, temps, top_k, top_p, self._gen)
         out = toks.tolist()
         # per-request `seed` (Ollama options.seed): stochastic rows with a

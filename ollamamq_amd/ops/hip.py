@@ -53,11 +53,12 @@ def _try_load():
                                      i, i, i, i, i, i, i64, i64, i64, vp]
     lib.prefill_attn_bf16.argtypes = [vp, vp, vp, vp, vp, vp, vp, vp, vp,
                                       i, i, i, i, i, f, i64, i, vp]
+    lib.embed_gather_bf16.argtypes = [vp, vp, vp, i, i, vp]
     for fn in ("rmsnorm_residual_bf16", "rope_bf16", "kv_append_bf16",
                "paged_attn_bf16", "swiglu_bf16", "argmax_bf16",
                "decode_attn_bf16", "skinny_gemm_bf16",
                "skinny_direct_bf16", "prefill_attn_bf16",
-               "rope_append_bf16"):
+               "rope_append_bf16", "embed_gather_bf16"):
         getattr(lib, fn).restype = ctypes.c_int
     _lib = lib
 
@@ -91,6 +92,19 @@ def _p(t: Optional[torch.Tensor]):
 
 
 # ---------------------------------------------------------------------------
+
+def embedding(tokens, table):
+    """out[t] = table[tokens[t]]: one gather kernel, int32 ids, no casts."""
+    T = tokens.shape[0]
+    V, H = table.shape
+    if H % 8 or table.dtype != torch.bfloat16:
+        return table.index_select(0, tokens.long())
+    tok32 = tokens if tokens.dtype == torch.int32 else tokens.int()
+    out = torch.empty((T, H), dtype=table.dtype, device=table.device)
+    _check(_lib.embed_gather_bf16(_p(out), _p(table), _p(tok32), T, H,
+                                  _stream()), "embed_gather")
+    return out
+
 
 def rmsnorm_residual(x, residual, weight, eps):
     T, H = x.shape

@@ -53,7 +53,9 @@ class AttnMeta:
 
 
 def embedding(tokens: torch.Tensor, table: torch.Tensor) -> torch.Tensor:
-    # index_select is already a single gather kernel on ROCm; no custom op.
+    if _use_hip(table):
+        from . import hip
+        return hip.embedding(tokens, table)
     return table.index_select(0, tokens.long())
 
 

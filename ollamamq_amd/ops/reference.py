@@ -114,33 +114,50 @@ def swiglu(gate_up):
 def sample(logits, temperature, top_k, top_p, generator=None):
     """Per-row sampling: greedy when temperature<=0, else temp/top-k/top-p.
 
-    logits: [B, V]; temperature/top_k/top_p: python floats/ints (uniform
-    across the batch rows that share a sampling config) or 1-D tensors.
-    Returns [B] int64 token ids.
+    logits: [B, V]; temperature/top_k/top_p are python scalars (uniform
+    across the batch) OR 1-D [B] tensors — each row is filtered by ITS OWN
+    top_k/top_p (per-request semantics: mixed sampling params in one batch
+    never bleed into each other).  top_k 0 and top_p >= 1 disable the
+    respective filter for that row.  Returns [B] int64 token ids.
     """
     B, V = logits.shape
+    dev = logits.device
     logits = logits.float()
     if not torch.is_tensor(temperature):
-        temperature = torch.full((B,), float(temperature), device=logits.device)
+        temperature = torch.full((B,), float(temperature), device=dev)
+    if not torch.is_tensor(top_k):
+        top_k = torch.full((B,), int(top_k or 0), dtype=torch.long,
+                           device=dev)
+    if not torch.is_tensor(top_p):
+        top_p = torch.full((B,), float(top_p if top_p else 1.0), device=dev)
+    top_k = top_k.long()
     greedy = temperature <= 0
-    out = torch.empty(B, dtype=torch.long, device=logits.device)
+    out = torch.empty(B, dtype=torch.long, device=dev)
     if greedy.any():
         out[greedy] = logits[greedy].argmax(dim=-1)
     rest = ~greedy
     if rest.any():
         l = logits[rest] / temperature[rest].unsqueeze(1)
-        if top_k and top_k > 0 and top_k < V:
-            kth = l.topk(int(top_k), dim=-1).values[:, -1:]
-            l = l.masked_fill(l < kth, float("-inf"))
-        probs = torch.softmax(l, dim=-1)
-        if top_p and 0.0 < top_p < 1.0:
-            sp, si = probs.sort(dim=-1, descending=True)
+        tk = top_k[rest]
+        tp = top_p[rest]
+        if bool(((tk > 0) & (tk < V)).any()) or bool((tp < 1.0).any()):
+            # one descending sort serves both filters; applying top-k as a
+            # post-softmax mask + renormalize is identical to the -inf
+            # pre-softmax mask (softmax restricted to a subset)
+            sv, si = l.sort(dim=-1, descending=True)
+            sp = torch.softmax(sv, dim=-1)
+            ar = torch.arange(V, device=dev)
+            kk = torch.where(tk > 0, tk.clamp(max=V), torch.full_like(tk, V))
+            keep = ar.unsqueeze(0) < kk.unsqueeze(1)          # per-row top-k
             cum = sp.cumsum(dim=-1)
-            keep = (cum - sp) < top_p   # keep first token crossing the mass
+            keep &= (cum - sp) < tp.unsqueeze(1)   # keep token crossing mass
+            keep[:, 0] = True                      # never empty a row
             sp = sp * keep
             sp = sp / sp.sum(dim=-1, keepdim=True)
             pick = torch.multinomial(sp, 1, generator=generator).squeeze(1)
             out[rest] = si.gather(1, pick.unsqueeze(1)).squeeze(1)
         else:
-            out[rest] = torch.multinomial(probs, 1, generator=generator).squeeze(1)
+            probs = torch.softmax(l, dim=-1)
+            out[rest] = torch.multinomial(
+                probs, 1, generator=generator).squeeze(1)
     return out

@@ -410,3 +410,17 @@ def test_mixed_batch_continuous_batching():
     assert not eng.has_work()
     assert seq1.generated == alone["s1"], "mixed step changed s1 tokens"
     assert seq2.generated == alone["s2"], "late joiner diverged"
+
+
+def test_embed_gather():
+    """Embedding gather kernel vs index_select (exact: it is a copy)."""
+    hip = _hip()
+    for T, V, H in [(1, 512, 4096), (32, 128256, 4096), (2048, 512, 512),
+                    (7, 1000, 3584)]:
+        g = torch.Generator().manual_seed(T)
+        table = rnd(V, H, seed=V + T)
+        toks = torch.randint(0, V, (T,), generator=g,
+                             dtype=torch.int32).to(dev())
+        out = hip.embedding(toks, table)
+        expect = table.index_select(0, toks.long())
+        assert torch.equal(out, expect)
