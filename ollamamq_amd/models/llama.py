@@ -134,6 +134,7 @@ class LlamaLayer:
 
     __slots__ = (
         "wqkv", "bqkv", "wo", "wgate_up", "wdown", "attn_norm", "mlp_norm",
+        "wqkv_pk", "wo_pk", "wgate_up_pk", "wdown_pk",
     )
 
     def __init__(self, cfg: LlamaConfig, dev, dtype, gen, tp: int, rank: int):
@@ -178,6 +179,8 @@ class LlamaLayer:
         del wd_g
         self.attn_norm = torch.ones(h, device=dev, dtype=dtype)
         self.mlp_norm = torch.ones(h, device=dev, dtype=dtype)
+        # packed decode copies filled by LlamaModel._pack_weights()
+        self.wqkv_pk = self.wo_pk = self.wgate_up_pk = self.wdown_pk = None
 
 
 class LlamaModel:
@@ -247,6 +250,34 @@ class LlamaModel:
         ang = torch.outer(pos, inv)  # [max_ctx, head_dim/2]
         self.rope_cos = ang.cos().to(dev)
         self.rope_sin = ang.sin().to(dev)
+        self.lm_head_pk = None
+        self._pack_weights()
+
+    def _pack_weights(self):
+        """Build MFMA-fragment-order packed copies of every projection for
+        the weight-streaming decode GEMM.  Doubles weight HBM (8B: 16->32
+        of 288 GB — prefill keeps the standard layout for hipBLASLt); on
+        models where the copy would not fit (70B on one GPU) packing is
+        skipped wholesale and decode stays on the library GEMMs."""
+        import os
+        if self.device.type != "cuda" or self.dtype != torch.bfloat16 \
+                or os.environ.get("OLLAMAMQ_NO_PACK") == "1":
+            return
+        free, _ = torch.cuda.mem_get_info(self.device)
+        if self.weight_bytes() > free * 0.45:
+            return
+        try:
+            for l in self.layers:
+                l.wqkv_pk = ops.pack_weight(l.wqkv)
+                l.wo_pk = ops.pack_weight(l.wo)
+                l.wgate_up_pk = ops.pack_weight(l.wgate_up)
+                l.wdown_pk = ops.pack_weight(l.wdown)
+            self.lm_head_pk = ops.pack_weight(self.lm_head)
+        except torch.cuda.OutOfMemoryError:
+            for l in self.layers:
+                l.wqkv_pk = l.wo_pk = l.wgate_up_pk = l.wdown_pk = None
+            self.lm_head_pk = None
+            torch.cuda.empty_cache()
 
     # -- helpers -----------------------------------------------------------
     def _allreduce(self, x: torch.Tensor) -> torch.Tensor:
@@ -288,7 +319,8 @@ class LlamaModel:
             normed, residual = ops.rmsnorm_residual(
                 x, residual, layer.attn_norm, cfg.norm_eps
             )
-            qkv = ops.linear(normed, layer.wqkv, layer.bqkv)
+            qkv = ops.linear(normed, layer.wqkv, layer.bqkv,
+                             packed=layer.wqkv_pk)
             nl, nkl, d = self.n_local_heads, self.n_local_kv_heads, cfg.head_dim
             q, k, v = qkv.split([nl * d, nkl * d, nkl * d], dim=-1)
             q = q.view(-1, nl, d)
@@ -297,14 +329,16 @@ class LlamaModel:
             ops.rope_append(kv_cache, li, q, k, v, positions, slot_ids,
                             self.rope_cos, self.rope_sin)
             attn = ops.attention(q, kv_cache, li, attn_meta)
-            x = ops.linear(attn.view(-1, nl * d), layer.wo)
+            x = ops.linear(attn.view(-1, nl * d), layer.wo,
+                           packed=layer.wo_pk)
             self._allreduce(x)  # RCCL all-reduce #1 (TP)
             normed, residual = ops.rmsnorm_residual(
                 x, residual, layer.mlp_norm, cfg.norm_eps
             )
-            gate_up = ops.linear(normed, layer.wgate_up)
+            gate_up = ops.linear(normed, layer.wgate_up,
+                                 packed=layer.wgate_up_pk)
             act = ops.swiglu(gate_up)
-            x = ops.linear(act, layer.wdown)
+            x = ops.linear(act, layer.wdown, packed=layer.wdown_pk)
             self._allreduce(x)  # RCCL all-reduce #2 (TP)
 
         # Only the last token of each sequence needs logits.
@@ -316,7 +350,7 @@ class LlamaModel:
         h, _ = ops.rmsnorm_residual(h, None, self.final_norm, cfg.norm_eps)
         if return_hidden:
             return h
-        logits = ops.linear(h, self.lm_head)
+        logits = ops.linear(h, self.lm_head, packed=self.lm_head_pk)
         if self.tp_size > 1:
             # vocab-parallel logits: all-gather shards on the last dim
             shards = [torch.empty_like(logits) for _ in range(self.tp_size)]

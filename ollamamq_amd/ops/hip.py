@@ -54,11 +54,14 @@ def _try_load():
     lib.prefill_attn_bf16.argtypes = [vp, vp, vp, vp, vp, vp, vp, vp, vp,
                                       i, i, i, i, i, f, i64, i, vp]
     lib.embed_gather_bf16.argtypes = [vp, vp, vp, i, i, vp]
+    lib.wstream_gemm_bf16.argtypes = [vp, vp, vp, vp, vp, i, i, i, i64, i,
+                                      vp]
     for fn in ("rmsnorm_residual_bf16", "rope_bf16", "kv_append_bf16",
                "paged_attn_bf16", "swiglu_bf16", "argmax_bf16",
                "decode_attn_bf16", "skinny_gemm_bf16",
                "skinny_direct_bf16", "prefill_attn_bf16",
-               "rope_append_bf16", "embed_gather_bf16"):
+               "rope_append_bf16", "embed_gather_bf16",
+               "wstream_gemm_bf16"):
         getattr(lib, fn).restype = ctypes.c_int
     _lib = lib
 
@@ -409,9 +412,63 @@ def _skinny_ksplit(N, K):
     return ks
 
 
-def linear(x, weight, bias=None):
+def pack_weight(w):
+    """Pre-pack a [N, K] bf16 weight into MFMA fragment order for the
+    weight-streaming decode GEMM (wstream_gemm.hip): element W[n][k] with
+    n = t*32+r, k = b*64 + j*16 + h*8 + e goes to 16-byte unit
+    ((t*NB + b)*4 + j)*64 + (h*32 + r).  The kernel's loads then walk the
+    buffer LINEARLY (full 128 B line per instruction, nt-tagged) — the fix
+    for the 4x cold-line over-fetch of fragment-direct loads (NOTES r01).
+    Returns None when the shape doesn't qualify."""
+    N, K = w.shape
+    if N % 32 or K % 64 or w.dtype != torch.bfloat16 \
+            or not w.is_contiguous():
+        return None
+    p = w.reshape(N // 32, 32, K // 64, 4, 2, 8) \
+         .permute(0, 2, 3, 4, 1, 5).contiguous()
+    return p.view(-1)
+
+
+USE_WSTREAM = os.environ.get("OLLAMAMQ_NO_WSTREAM") != "1"
+
+
+def _wstream_ksplit(N, K):
+    """Fill the 256-CU chip: target >= 768 blocks; each of a block's 8
+    waves should keep >= 2 k-blocks so the register prefetch has a loop."""
+    tiles = N // 32
+    ks = max(1, min(8, (768 + tiles - 1) // tiles))
+    nb = K // 64
+    while ks > 1 and nb // (ks * 8) < 2:
+        ks -= 1
+    return ks
+
+
+def linear_packed(x, packed, bias, N):
+    """y = x @ W^T via the weight-streaming kernel over pre-packed W."""
+    M, K = x.shape
+    y = torch.empty((M, N), dtype=x.dtype, device=x.device)
+    ks = _wstream_ksplit(N, K)
+    part = ctypes.c_void_p(0)
+    if ks > 1:
+        key = ("ws", M, N, ks, str(x.device))
+        t = _gemm_scratch.get(key)
+        if t is None:
+            t = torch.empty(ks * M * N, dtype=torch.float32,
+                            device=x.device)
+            _gemm_scratch[key] = t
+        part = _p(t)
+    _check(_lib.wstream_gemm_bf16(
+        _p(y), part, _p(x), _p(packed), _p(bias), M, N, K, x.stride(0),
+        ks, _stream()), "wstream_gemm")
+    return y
+
+
+def linear(x, weight, bias=None, packed=None):
     M, K = x.shape
     N = weight.shape[0]
+    if packed is not None and USE_WSTREAM and M <= 64 \
+            and x.dtype == torch.bfloat16 and x.stride(1) == 1:
+        return linear_packed(x, packed, bias, N)
     # gated to the shapes where the hand-written kernel beats hipBLASLt
     # (measured tools/perf_gemm.py); widen via env as the kernel improves
     if bias is None and M <= 32 and K % 256 == 0 and N % 32 == 0 \
