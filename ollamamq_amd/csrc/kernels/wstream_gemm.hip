@@ -38,7 +38,56 @@ typedef __attribute__((ext_vector_type(4))) unsigned int u32x4;
 
 union u4bf8 { u32x4 u; bf16x8 v; };
 
-template <int MT>   // m-tiles of 32 rows (1: M<=32, 2: M<=64)
+// Pure-stream diagnostic: same grid/geometry/addressing as the GEMM but
+// only the nt weight loads (no x, no MFMA) — measures this geometry's
+// load-path ceiling so kernel iterations know what they are chasing.
+__global__ __launch_bounds__(512) void k_wstream_pure(
+    float* __restrict__ sink, const u32x4* __restrict__ wp,
+    int N, int K, int ksplit)
+{
+    const int t = blockIdx.x;
+    const int ks = blockIdx.y;
+    const int wid = threadIdx.x >> 6;
+    const int lane = threadIdx.x & 63;
+    const int NB = K >> 6;
+    const int GW = ksplit * 8;
+    const int gw = ks * 8 + wid;
+    const int b_lo = (int)(((int64_t)gw * NB) / GW);
+    const int b_hi = (int)(((int64_t)(gw + 1) * NB) / GW);
+    unsigned acc = 0;
+    const u32x4* wq = wp + (((int64_t)t * NB + b_lo) * 4) * 64 + lane;
+    for (int b = b_lo; b < b_hi; ++b, wq += 256) {
+        u32x4 v0 = __builtin_nontemporal_load(wq);
+        u32x4 v1 = __builtin_nontemporal_load(wq + 64);
+        u32x4 v2 = __builtin_nontemporal_load(wq + 128);
+        u32x4 v3 = __builtin_nontemporal_load(wq + 192);
+        acc ^= v0.x ^ v0.w ^ v1.x ^ v1.w ^ v2.x ^ v2.w ^ v3.x ^ v3.w;
+    }
+    if (acc == 0xDEADBEEFu) sink[0] = 1.f;   // never: keeps loads alive
+}
+
+extern "C" int wstream_pure_bf16(void* sink, const void* wp, int N, int K,
+                                 int ksplit, hipStream_t stream)
+{
+    dim3 grid(N / 32, ksplit);
+    k_wstream_pure<<<grid, 512, 0, stream>>>(
+        (float*)sink, (const u32x4*)wp, N, K, ksplit);
+    return (int)hipGetLastError();
+}
+
+// MT: m-tiles of 32 rows (1: M<=32, 2: M<=64).
+// DEPTH: weight prefetch depth in 4 KiB iterations (1 or 2); 2 doubles
+// the b-register ring so a whole iteration's loads stay in flight across
+// one full compute iteration (covers ~2x the HBM latency per wave).
+// XLDS (requires MT==1): stage the x fragments through a wave-PRIVATE
+// LDS tile instead of loading them in MFMA layout.  The direct x
+// fragment load is 32 rows x 16 B at an 8 KB stride = 32 cache-line
+// requests per instruction x 4 per iteration; staged, the global x read
+// is 4 coalesced instructions (8 lines each), a dense ds_write and a
+// minimum-phase swizzled ds_read (u = r*8 + (c ^ (r&7))).  Wave-private
+// double-buffered tiles -> NO barrier anywhere in the stream (a
+// __syncthreads would drain vmcnt and kill the weight prefetch).
+template <int MT, int DEPTH = 1, int XLDS = 0>
 __global__ __launch_bounds__(512) void k_wstream_gemm(
     bf16* __restrict__ y,            // [M, N] (ksplit == 1)
     float* __restrict__ part,        // [ksplit, M, N] (ksplit > 1)
@@ -72,79 +121,146 @@ __global__ __launch_bounds__(512) void k_wstream_gemm(
         xr1 = x + (int64_t)ar1 * xs + khalf;
     }
 
+    // dynamic LDS: 32 KiB (wave-partial reduce) or 64 KiB (XLDS adds 8
+    // double-buffered 4 KiB wave-private x tiles, reduce space aliased —
+    // a __syncthreads separates the stream epoch from the reduce epoch)
+    extern __shared__ __align__(16) char smem[];
+
     f32x16 acc0 = {}, acc1 = {};
     if (b_lo < b_hi) {
-        const u32x4* wq = wp + (((int64_t)t * NB + b_lo) * 4) * 64 + lane;
-        u4bf8 b0, b1, b2, b3;
+        const u32x4* wbase = wp + ((int64_t)t * NB * 4) * 64 + lane;
+        u4bf8 br[DEPTH][4];
         bf16x8 a0[2], a1[2], a2[2], a3[2];
         auto lda16 = [&](bf16x8* d, int b, int off) {
             const int64_t k = ((int64_t)b << 6) + off;
             d[0] = *reinterpret_cast<const bf16x8*>(xr0 + k);
             if (MT == 2) d[1] = *reinterpret_cast<const bf16x8*>(xr1 + k);
         };
-        // prologue: first iteration's 4 weight vectors + activations
-        b0.u = __builtin_nontemporal_load(wq);
-        b1.u = __builtin_nontemporal_load(wq + 64);
-        b2.u = __builtin_nontemporal_load(wq + 128);
-        b3.u = __builtin_nontemporal_load(wq + 192);
-        lda16(a0, b_lo, 0);
-        lda16(a1, b_lo, 16);
-        lda16(a2, b_lo, 32);
-        lda16(a3, b_lo, 48);
-        const u32x4* wnext = wq + 256;
-        for (int b = b_lo; b < b_hi - 1; ++b, wnext += 256) {
-            bf16x8 vb0 = b0.v, vb1 = b1.v, vb2 = b2.v, vb3 = b3.v;
+        auto lda = [&](int b) {
+            lda16(a0, b, 0);
+            lda16(a1, b, 16);
+            lda16(a2, b, 32);
+            lda16(a3, b, 48);
+        };
+        // clamped prefetch: an over-the-end refill re-reads a valid block
+        // instead of branching per lane (the tail never computes it)
+        auto ldb = [&](int s, int b) {
+            const u32x4* q = wbase + (int64_t)(b < b_hi ? b : b_hi - 1)
+                                     * 256;
+            br[s][0].u = __builtin_nontemporal_load(q);
+            br[s][1].u = __builtin_nontemporal_load(q + 64);
+            br[s][2].u = __builtin_nontemporal_load(q + 128);
+            br[s][3].u = __builtin_nontemporal_load(q + 192);
+        };
+        auto compute = [&](int s) {
+            bf16x8 vb0 = br[s][0].v, vb1 = br[s][1].v;
+            bf16x8 vb2 = br[s][2].v, vb3 = br[s][3].v;
             acc0 = __builtin_amdgcn_mfma_f32_32x32x16_bf16(
                 a0[0], vb0, acc0, 0, 0, 0);
-            if (MT == 2)
-                acc1 = __builtin_amdgcn_mfma_f32_32x32x16_bf16(
-                    a0[1], vb0, acc1, 0, 0, 0);
-            b0.u = __builtin_nontemporal_load(wnext);
-            lda16(a0, b + 1, 0);
             acc0 = __builtin_amdgcn_mfma_f32_32x32x16_bf16(
                 a1[0], vb1, acc0, 0, 0, 0);
-            if (MT == 2)
-                acc1 = __builtin_amdgcn_mfma_f32_32x32x16_bf16(
-                    a1[1], vb1, acc1, 0, 0, 0);
-            b1.u = __builtin_nontemporal_load(wnext + 64);
-            lda16(a1, b + 1, 16);
             acc0 = __builtin_amdgcn_mfma_f32_32x32x16_bf16(
                 a2[0], vb2, acc0, 0, 0, 0);
-            if (MT == 2)
-                acc1 = __builtin_amdgcn_mfma_f32_32x32x16_bf16(
-                    a2[1], vb2, acc1, 0, 0, 0);
-            b2.u = __builtin_nontemporal_load(wnext + 128);
-            lda16(a2, b + 1, 32);
             acc0 = __builtin_amdgcn_mfma_f32_32x32x16_bf16(
                 a3[0], vb3, acc0, 0, 0, 0);
-            if (MT == 2)
+            if (MT == 2) {
+                acc1 = __builtin_amdgcn_mfma_f32_32x32x16_bf16(
+                    a0[1], vb0, acc1, 0, 0, 0);
+                acc1 = __builtin_amdgcn_mfma_f32_32x32x16_bf16(
+                    a1[1], vb1, acc1, 0, 0, 0);
+                acc1 = __builtin_amdgcn_mfma_f32_32x32x16_bf16(
+                    a2[1], vb2, acc1, 0, 0, 0);
                 acc1 = __builtin_amdgcn_mfma_f32_32x32x16_bf16(
                     a3[1], vb3, acc1, 0, 0, 0);
-            b3.u = __builtin_nontemporal_load(wnext + 192);
-            lda16(a3, b + 1, 48);
-        }
-        acc0 = __builtin_amdgcn_mfma_f32_32x32x16_bf16(a0[0], b0.v, acc0,
-                                                       0, 0, 0);
-        acc0 = __builtin_amdgcn_mfma_f32_32x32x16_bf16(a1[0], b1.v, acc0,
-                                                       0, 0, 0);
-        acc0 = __builtin_amdgcn_mfma_f32_32x32x16_bf16(a2[0], b2.v, acc0,
-                                                       0, 0, 0);
-        acc0 = __builtin_amdgcn_mfma_f32_32x32x16_bf16(a3[0], b3.v, acc0,
-                                                       0, 0, 0);
-        if (MT == 2) {
-            acc1 = __builtin_amdgcn_mfma_f32_32x32x16_bf16(a0[1], b0.v,
-                                                           acc1, 0, 0, 0);
-            acc1 = __builtin_amdgcn_mfma_f32_32x32x16_bf16(a1[1], b1.v,
-                                                           acc1, 0, 0, 0);
-            acc1 = __builtin_amdgcn_mfma_f32_32x32x16_bf16(a2[1], b2.v,
-                                                           acc1, 0, 0, 0);
-            acc1 = __builtin_amdgcn_mfma_f32_32x32x16_bf16(a3[1], b3.v,
-                                                           acc1, 0, 0, 0);
+            }
+        };
+        if (XLDS) {
+            // --- LDS-staged x fragments (MT==1 only) ---
+            bf16x8 xs4[4];
+            auto glbx = [&](int b) {       // 4 coalesced reads: 8 rows
+                const int64_t kk = (int64_t)b << 6;   // x 128 B each
+                #pragma unroll
+                for (int i2 = 0; i2 < 4; i2++) {
+                    const int r = 8 * i2 + (lane >> 3);
+                    const int ar = r < M ? r : (M - 1);
+                    xs4[i2] = *reinterpret_cast<const bf16x8*>(
+                        x + (int64_t)ar * xs + kk + (lane & 7) * 8);
+                }
+            };
+            auto dswx = [&](int s) {       // dense swizzled tile write
+                bf16* tile = reinterpret_cast<bf16*>(smem)
+                             + (wid * 2 + s) * 2048;
+                #pragma unroll
+                for (int i2 = 0; i2 < 4; i2++) {
+                    const int u = (8 * i2 + (lane >> 3)) * 8
+                                  + ((lane & 7) ^ (lane >> 3));
+                    *reinterpret_cast<bf16x8*>(tile + u * 8) = xs4[i2];
+                }
+            };
+            auto dsrx = [&](int s) {       // min-phase fragment reads
+                const bf16* tile = reinterpret_cast<const bf16*>(smem)
+                                   + (wid * 2 + s) * 2048;
+                const int r = lane & 31;
+                #pragma unroll
+                for (int j = 0; j < 4; j++) {
+                    const int c = j * 2 + (lane >> 5);
+                    const int u = r * 8 + (c ^ (r & 7));
+                    bf16x8 v = *reinterpret_cast<const bf16x8*>(
+                        tile + u * 8);
+                    if (j == 0) a0[0] = v;
+                    else if (j == 1) a1[0] = v;
+                    else if (j == 2) a2[0] = v;
+                    else a3[0] = v;
+                }
+            };
+            ldb(0, b_lo);
+            glbx(b_lo);
+            dswx(0);
+            int b = b_lo;
+            for (; b < b_hi - 1; ++b) {
+                const int s = (b - b_lo) & 1;
+                dsrx(s);
+                compute(0);
+                ldb(0, b + 1);
+                glbx(b + 1);
+                dswx(s ^ 1);
+            }
+            dsrx((b - b_lo) & 1);
+            compute(0);
+        } else {
+            ldb(0, b_lo);
+            if (DEPTH == 2) ldb(1, b_lo + 1);
+            lda(b_lo);
+            int b = b_lo;
+            if (DEPTH == 1) {
+                for (; b < b_hi - 1; ++b) {
+                    compute(0);
+                    ldb(0, b + 1);
+                    lda(b + 1);
+                }
+                compute(0);
+            } else {
+                for (; b + 2 < b_hi; b += 2) {
+                    compute(0);
+                    ldb(0, b + 2);
+                    lda(b + 1);
+                    compute(1);
+                    ldb(1, b + 3);
+                    lda(b + 2);
+                }
+                compute(0);
+                if (b + 1 < b_hi) {
+                    lda(b + 1);
+                    compute(1);
+                }
+            }
         }
     }
 
-    // ---- the one barrier: reduce 8 wave partials through LDS ----
-    __shared__ float red8[8][32][32];
+    // ---- reduce 8 wave partials through LDS (aliases the x tiles: the
+    // barrier closes the streaming epoch before anyone writes) ----
+    float (*red8)[32][32] = reinterpret_cast<float (*)[32][32]>(smem);
+    __syncthreads();
     const int n0 = t * 32;
     for (int mt = 0; mt < MT; ++mt) {
         const f32x16& acc = mt ? acc1 : acc0;
@@ -190,17 +306,35 @@ __global__ __launch_bounds__(256) void k_wstream_combine(
 
 extern "C" int wstream_gemm_bf16(
     void* y, void* part, const void* x, const void* wp, const void* bias,
-    int M, int N, int K, int64_t xs, int ksplit, hipStream_t stream)
+    int M, int N, int K, int64_t xs, int ksplit, int depth, int xlds,
+    hipStream_t stream)
 {
     dim3 grid(N / 32, ksplit);
-    if (M <= 32)
-        k_wstream_gemm<1><<<grid, 512, 0, stream>>>(
-            (bf16*)y, (float*)part, (const bf16*)x, (const u32x4*)wp,
-            (const bf16*)bias, M, N, K, xs, ksplit);
-    else
-        k_wstream_gemm<2><<<grid, 512, 0, stream>>>(
-            (bf16*)y, (float*)part, (const bf16*)x, (const u32x4*)wp,
-            (const bf16*)bias, M, N, K, xs, ksplit);
+    const int lds_red = 8 * 32 * 32 * 4;           // 32 KiB reduce
+    const int lds_x = 16 * 4096;                   // + 8x2 x tiles
+    if (M <= 32) {
+        if (xlds)
+            k_wstream_gemm<1, 1, 1><<<grid, 512, lds_x, stream>>>(
+                (bf16*)y, (float*)part, (const bf16*)x, (const u32x4*)wp,
+                (const bf16*)bias, M, N, K, xs, ksplit);
+        else if (depth == 2)
+            k_wstream_gemm<1, 2><<<grid, 512, lds_red, stream>>>(
+                (bf16*)y, (float*)part, (const bf16*)x, (const u32x4*)wp,
+                (const bf16*)bias, M, N, K, xs, ksplit);
+        else
+            k_wstream_gemm<1, 1><<<grid, 512, lds_red, stream>>>(
+                (bf16*)y, (float*)part, (const bf16*)x, (const u32x4*)wp,
+                (const bf16*)bias, M, N, K, xs, ksplit);
+    } else {
+        if (depth == 2)
+            k_wstream_gemm<2, 2><<<grid, 512, lds_red, stream>>>(
+                (bf16*)y, (float*)part, (const bf16*)x, (const u32x4*)wp,
+                (const bf16*)bias, M, N, K, xs, ksplit);
+        else
+            k_wstream_gemm<2, 1><<<grid, 512, lds_red, stream>>>(
+                (bf16*)y, (float*)part, (const bf16*)x, (const u32x4*)wp,
+                (const bf16*)bias, M, N, K, xs, ksplit);
+    }
     if (ksplit > 1) {
         const int64_t mn = (int64_t)M * N;
         const int64_t want = (mn + 255) / 256;

@@ -266,13 +266,17 @@ class LlamaModel:
         free, _ = torch.cuda.mem_get_info(self.device)
         if self.weight_bytes() > free * 0.45:
             return
+        # a packed copy nothing will read is pure HBM waste: only pack
+        # tensors the decode dispatch will actually route to the
+        # streaming kernel (N <= the measured lib-wins threshold)
+        pk = ops.pack_weight
         try:
             for l in self.layers:
-                l.wqkv_pk = ops.pack_weight(l.wqkv)
-                l.wo_pk = ops.pack_weight(l.wo)
-                l.wgate_up_pk = ops.pack_weight(l.wgate_up)
-                l.wdown_pk = ops.pack_weight(l.wdown)
-            self.lm_head_pk = ops.pack_weight(self.lm_head)
+                l.wqkv_pk = pk(l.wqkv)
+                l.wo_pk = pk(l.wo)
+                l.wgate_up_pk = pk(l.wgate_up)
+                l.wdown_pk = pk(l.wdown)
+            self.lm_head_pk = pk(self.lm_head)
         except torch.cuda.OutOfMemoryError:
             for l in self.layers:
                 l.wqkv_pk = l.wo_pk = l.wgate_up_pk = l.wdown_pk = None

@@ -55,7 +55,8 @@ def _try_load():
                                       i, i, i, i, i, f, i64, i, vp]
     lib.embed_gather_bf16.argtypes = [vp, vp, vp, i, i, vp]
     lib.wstream_gemm_bf16.argtypes = [vp, vp, vp, vp, vp, i, i, i, i64, i,
-                                      vp]
+                                      i, i, vp]
+    lib.wstream_pure_bf16.argtypes = [vp, vp, i, i, i, vp]
     for fn in ("rmsnorm_residual_bf16", "rope_bf16", "kv_append_bf16",
                "paged_attn_bf16", "swiglu_bf16", "argmax_bf16",
                "decode_attn_bf16", "skinny_gemm_bf16",
@@ -433,21 +434,43 @@ USE_WSTREAM = os.environ.get("OLLAMAMQ_NO_WSTREAM") != "1"
 
 
 def _wstream_ksplit(N, K):
-    """Fill the 256-CU chip: target >= 768 blocks; each of a block's 8
-    waves should keep >= 2 k-blocks so the register prefetch has a loop."""
+    """r02 sweep (profiles/r02_gemm_sweep.md): the sweet spot is ~8
+    k-iterations per wave — deep enough that the register prefetch
+    pipelines, while blocks stay >= 128 so the grid isn't starved.  Pick
+    the ks whose iters/wave is nearest 8 (ties -> smaller ks: fewer
+    partials)."""
     tiles = N // 32
-    ks = max(1, min(8, (768 + tiles - 1) // tiles))
     nb = K // 64
-    while ks > 1 and nb // (ks * 8) < 2:
-        ks -= 1
-    return ks
+    best, best_cost = 1, 1 << 30
+    for ks in range(1, 9):
+        iters = nb // (ks * 8)
+        if iters < 1:
+            break
+        cost = abs(iters - 8) * 4 + ks
+        if tiles * ks < 128:
+            cost += 64          # starved grid: strong penalty
+        if cost < best_cost:
+            best, best_cost = ks, cost
+    return best
 
 
-def linear_packed(x, packed, bias, N):
+_WS_DEPTH = int(os.environ.get("OLLAMAMQ_WS_DEPTH", "1"))
+_WS_XLDS = int(os.environ.get("OLLAMAMQ_WS_XLDS", "1"))
+_WS_KS = os.environ.get("OLLAMAMQ_WS_KS")   # sweep override
+# N above which the library GEMM wins (measured r02 sweep,
+# profiles/r02_gemm_sweep.md: wstream beats hipBLASLt on the
+# ramp-dominated small-N decode shapes qkv/o/down; on gate_up/logits the
+# lib's big-grid kernels are closer to the stream ceiling than our
+# x-load overhead allows — revisit after the x staging rework)
+_WS_MAX_N = int(os.environ.get("OLLAMAMQ_WS_MAX_N", "16384"))
+
+
+def linear_packed(x, packed, bias, N, ks=None, depth=None, xlds=None):
     """y = x @ W^T via the weight-streaming kernel over pre-packed W."""
     M, K = x.shape
     y = torch.empty((M, N), dtype=x.dtype, device=x.device)
-    ks = _wstream_ksplit(N, K)
+    if ks is None:
+        ks = int(_WS_KS) if _WS_KS else _wstream_ksplit(N, K)
     part = ctypes.c_void_p(0)
     if ks > 1:
         key = ("ws", M, N, ks, str(x.device))
@@ -457,16 +480,27 @@ def linear_packed(x, packed, bias, N):
                             device=x.device)
             _gemm_scratch[key] = t
         part = _p(t)
+    if xlds is None:
+        xlds = _WS_XLDS if M <= 32 else 0
     _check(_lib.wstream_gemm_bf16(
         _p(y), part, _p(x), _p(packed), _p(bias), M, N, K, x.stride(0),
-        ks, _stream()), "wstream_gemm")
+        ks, depth if depth is not None else _WS_DEPTH, xlds, _stream()),
+        "wstream_gemm")
     return y
+
+
+def wstream_pure(packed, N, K, ks):
+    """Bandwidth diagnostic: the GEMM's weight stream alone."""
+    sink = torch.zeros(1, dtype=torch.float32, device=packed.device)
+    _check(_lib.wstream_pure_bf16(_p(sink), _p(packed), N, K, ks,
+                                  _stream()), "wstream_pure")
+    return sink
 
 
 def linear(x, weight, bias=None, packed=None):
     M, K = x.shape
     N = weight.shape[0]
-    if packed is not None and USE_WSTREAM and M <= 64 \
+    if packed is not None and USE_WSTREAM and M <= 64 and N <= _WS_MAX_N \
             and x.dtype == torch.bfloat16 and x.stride(1) == 1:
         return linear_packed(x, packed, bias, N)
     # gated to the shapes where the hand-written kernel beats hipBLASLt

@@ -115,10 +115,13 @@ def linear(x, weight, bias=None, packed=None):
 
 
 def pack_weight(w):
-    """MFMA-fragment-order packed copy for the decode GEMM (GPU only)."""
+    """MFMA-fragment-order packed copy for the decode GEMM (GPU only).
+    Returns None for shapes the decode dispatch would never route to the
+    streaming kernel (N above the measured lib-wins threshold) so no HBM
+    is spent on copies nothing reads."""
     if w.is_cuda:
         from . import hip
-        if hip.available():
+        if hip.available() and w.shape[0] <= hip._WS_MAX_N:
             return hip.pack_weight(w)
     return None
 
