@@ -481,7 +481,10 @@ def linear_packed(x, packed, bias, N, ks=None, depth=None, xlds=None):
             _gemm_scratch[key] = t
         part = _p(t)
     if xlds is None:
-        xlds = _WS_XLDS if M <= 32 else 0
+        # measured (profiles/r02_gemm_sweep.md): LDS x-staging pays on
+        # long streams (down/gate_up/logits), the direct fragment load
+        # wins on the short ramp-bound ones (qkv/o)
+        xlds = (_WS_XLDS if M <= 32 and N * K * 2 > (64 << 20) else 0)
     _check(_lib.wstream_gemm_bf16(
         _p(y), part, _p(x), _p(packed), _p(bias), M, N, K, x.stride(0),
         ks, depth if depth is not None else _WS_DEPTH, xlds, _stream()),
@@ -500,7 +503,8 @@ def wstream_pure(packed, N, K, ks):
 def linear(x, weight, bias=None, packed=None):
     M, K = x.shape
     N = weight.shape[0]
-    if packed is not None and USE_WSTREAM and M <= 64 and N <= _WS_MAX_N \
+    if packed is not None and USE_WSTREAM and M <= 64 \
+            and (N <= _WS_MAX_N or N >= 65536) \
             and x.dtype == torch.bfloat16 and x.stride(1) == 1:
         return linear_packed(x, packed, bias, N)
     # gated to the shapes where the hand-written kernel beats hipBLASLt

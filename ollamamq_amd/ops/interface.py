@@ -121,7 +121,8 @@ def pack_weight(w):
     is spent on copies nothing reads."""
     if w.is_cuda:
         from . import hip
-        if hip.available() and w.shape[0] <= hip._WS_MAX_N:
+        if hip.available() and (w.shape[0] <= hip._WS_MAX_N
+                                or w.shape[0] >= 65536):
             return hip.pack_weight(w)
     return None
 
