@@ -87,7 +87,14 @@ extern "C" int wstream_pure_bf16(void* sink, const void* wp, int N, int K,
 // minimum-phase swizzled ds_read (u = r*8 + (c ^ (r&7))).  Wave-private
 // double-buffered tiles -> NO barrier anywhere in the stream (a
 // __syncthreads would drain vmcnt and kill the weight prefetch).
-template <int MT, int DEPTH = 1, int XLDS = 0>
+// GU (requires MT==1, ksplit==1): fused gate_up@SwiGLU.  The weights are
+// packed with gate and up INTERLEAVED (tile t = 16 gate rows f=t*16..+16
+// then the 16 matching up rows — ops/hip.py pack_weight_gu), so each
+// block's 32-column output tile holds the (gate, up) pairs of 16 ffn
+// columns; the epilogue computes silu(g)*u and stores [M, 16] of the
+// activation directly.  Kills the separate swiglu kernel (9.7 us x 32
+// layers/step in the r02 trace) and the gate_up intermediate round-trip.
+template <int MT, int DEPTH = 1, int XLDS = 0, int GU = 0>
 __global__ __launch_bounds__(512) void k_wstream_gemm(
     bf16* __restrict__ y,            // [M, N] (ksplit == 1)
     float* __restrict__ part,        // [ksplit, M, N] (ksplit > 1)
@@ -260,6 +267,28 @@ __global__ __launch_bounds__(512) void k_wstream_gemm(
     // ---- reduce 8 wave partials through LDS (aliases the x tiles: the
     // barrier closes the streaming epoch before anyone writes) ----
     float (*red8)[32][32] = reinterpret_cast<float (*)[32][32]>(smem);
+    if (GU) {
+        __syncthreads();
+        #pragma unroll
+        for (int r = 0; r < 16; r++) {
+            const int crow = (r & 3) + 8 * (r >> 2) + 4 * (lane >> 5);
+            red8[wid][crow][lane & 31] = acc0[r];
+        }
+        __syncthreads();
+        const int F = N >> 1;                  // ffn width
+        const int m = tid >> 4, c = tid & 15;  // 512 threads = 32 x 16
+        if (m < M) {
+            float sg = 0.f, su = 0.f;
+            #pragma unroll
+            for (int wv = 0; wv < 8; wv++) {
+                sg += red8[wv][m][c];
+                su += red8[wv][m][c + 16];
+            }
+            const float act = (sg / (1.f + __expf(-sg))) * su;
+            y[(int64_t)m * F + t * 16 + c] = __float2bfloat16(act);
+        }
+        return;
+    }
     __syncthreads();
     const int n0 = t * 32;
     for (int mt = 0; mt < MT; ++mt) {
@@ -288,6 +317,25 @@ __global__ __launch_bounds__(512) void k_wstream_gemm(
             }
         }
     }
+}
+
+// Fused gate_up @ SwiGLU over GU-interleaved packed weights.
+// N = 2F total weight rows; output is the activation [M, F].
+extern "C" int wstream_gu_bf16(
+    void* act, const void* x, const void* wp, int M, int N, int K,
+    int64_t xs, int xlds, hipStream_t stream)
+{
+    dim3 grid(N / 32, 1);
+    const int lds = xlds ? 16 * 4096 : 8 * 32 * 32 * 4;
+    if (xlds)
+        k_wstream_gemm<1, 1, 1, 1><<<grid, 512, lds, stream>>>(
+            (bf16*)act, nullptr, (const bf16*)x, (const u32x4*)wp,
+            nullptr, M, N, K, xs, 1);
+    else
+        k_wstream_gemm<1, 1, 0, 1><<<grid, 512, lds, stream>>>(
+            (bf16*)act, nullptr, (const bf16*)x, (const u32x4*)wp,
+            nullptr, M, N, K, xs, 1);
+    return (int)hipGetLastError();
 }
 
 __global__ __launch_bounds__(256) void k_wstream_combine(

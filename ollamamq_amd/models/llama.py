@@ -134,7 +134,7 @@ class LlamaLayer:
 
     __slots__ = (
         "wqkv", "bqkv", "wo", "wgate_up", "wdown", "attn_norm", "mlp_norm",
-        "wqkv_pk", "wo_pk", "wgate_up_pk", "wdown_pk",
+        "wqkv_pk", "wo_pk", "wgu_pk", "wdown_pk",
     )
 
     def __init__(self, cfg: LlamaConfig, dev, dtype, gen, tp: int, rank: int):
@@ -180,7 +180,7 @@ class LlamaLayer:
         self.attn_norm = torch.ones(h, device=dev, dtype=dtype)
         self.mlp_norm = torch.ones(h, device=dev, dtype=dtype)
         # packed decode copies filled by LlamaModel._pack_weights()
-        self.wqkv_pk = self.wo_pk = self.wgate_up_pk = self.wdown_pk = None
+        self.wqkv_pk = self.wo_pk = self.wgu_pk = self.wdown_pk = None
 
 
 class LlamaModel:
@@ -274,12 +274,12 @@ class LlamaModel:
             for l in self.layers:
                 l.wqkv_pk = pk(l.wqkv)
                 l.wo_pk = pk(l.wo)
-                l.wgate_up_pk = pk(l.wgate_up)
+                l.wgu_pk = ops.pack_weight_gu(l.wgate_up)
                 l.wdown_pk = pk(l.wdown)
             self.lm_head_pk = pk(self.lm_head)
         except torch.cuda.OutOfMemoryError:
             for l in self.layers:
-                l.wqkv_pk = l.wo_pk = l.wgate_up_pk = l.wdown_pk = None
+                l.wqkv_pk = l.wo_pk = l.wgu_pk = l.wdown_pk = None
             self.lm_head_pk = None
             torch.cuda.empty_cache()
 
@@ -339,9 +339,8 @@ class LlamaModel:
             normed, residual = ops.rmsnorm_residual(
                 x, residual, layer.mlp_norm, cfg.norm_eps
             )
-            gate_up = ops.linear(normed, layer.wgate_up,
-                                 packed=layer.wgate_up_pk)
-            act = ops.swiglu(gate_up)
+            act = ops.gateup_swiglu(normed, layer.wgate_up,
+                                    layer.wgu_pk)
             x = ops.linear(act, layer.wdown, packed=layer.wdown_pk)
             self._allreduce(x)  # RCCL all-reduce #2 (TP)
 

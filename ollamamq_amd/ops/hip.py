@@ -57,6 +57,7 @@ def _try_load():
     lib.wstream_gemm_bf16.argtypes = [vp, vp, vp, vp, vp, i, i, i, i64, i,
                                       i, i, vp]
     lib.wstream_pure_bf16.argtypes = [vp, vp, i, i, i, vp]
+    lib.wstream_gu_bf16.argtypes = [vp, vp, vp, i, i, i, i64, i, vp]
     for fn in ("rmsnorm_residual_bf16", "rope_bf16", "kv_append_bf16",
                "paged_attn_bf16", "swiglu_bf16", "argmax_bf16",
                "decode_attn_bf16", "skinny_gemm_bf16",
@@ -428,6 +429,33 @@ def pack_weight(w):
     p = w.reshape(N // 32, 32, K // 64, 4, 2, 8) \
          .permute(0, 2, 3, 4, 1, 5).contiguous()
     return p.view(-1)
+
+
+def pack_weight_gu(w):
+    """GU-interleaved pack of a fused [2F, K] gate_up weight: tile t holds
+    gate rows f=t*16..t*16+16 then the matching 16 up rows, so the fused
+    kernel's 32-column output tile pairs each gate with its up and the
+    epilogue can apply SwiGLU in-register (wstream_gemm.hip GU=1)."""
+    N, K = w.shape
+    F = N // 2
+    if N % 32 or F % 16 or K % 64 or w.dtype != torch.bfloat16 \
+            or not w.is_contiguous():
+        return None
+    p = w.reshape(2, F // 16, 16, K // 64, 4, 2, 8) \
+         .permute(1, 3, 4, 5, 0, 2, 6).contiguous()
+    return p.view(-1)
+
+
+def linear_gu(x, packed, N):
+    """act = swiglu(x @ Wgu^T) fused; N = 2F total weight rows."""
+    M, K = x.shape
+    F = N // 2
+    act = torch.empty((M, F), dtype=x.dtype, device=x.device)
+    xlds = 1 if N * K * 2 > (64 << 20) else 0
+    _check(_lib.wstream_gu_bf16(
+        _p(act), _p(x), _p(packed), M, N, K, x.stride(0), xlds,
+        _stream()), "wstream_gu")
+    return act
 
 
 USE_WSTREAM = os.environ.get("OLLAMAMQ_NO_WSTREAM") != "1"

@@ -134,6 +134,27 @@ def swiglu(gate_up):
     return ref.swiglu(gate_up)
 
 
+def gateup_swiglu(x, weight, packed_gu=None):
+    """act = swiglu(x @ Wgu^T).  On the decode path (M<=32) with a
+    GU-packed copy this is ONE fused weight-streaming kernel; otherwise
+    the library GEMM + the swiglu kernel."""
+    if packed_gu is not None and x.shape[0] <= 32 and _use_hip(x):
+        from . import hip
+        if hip.USE_WSTREAM and x.dtype == torch.bfloat16 \
+                and x.stride(1) == 1:
+            return hip.linear_gu(x, packed_gu, weight.shape[0])
+    return swiglu(linear(x, weight))
+
+
+def pack_weight_gu(w):
+    """GU-interleaved packed copy for the fused gate_up+SwiGLU kernel."""
+    if w.is_cuda:
+        from . import hip
+        if hip.available():
+            return hip.pack_weight_gu(w)
+    return None
+
+
 def sample(logits, temperature, top_k, top_p, generator=None):
     # Sampler: torch ops compose on-GPU; custom fused kernel in ops/hip.py
     # handles the greedy + temperature paths.

@@ -471,3 +471,20 @@ def test_wstream_pack_roundtrip_cpu_check():
     back = pk.view(N // 32, K // 64, 4, 2, 32, 8) \
              .permute(0, 4, 1, 2, 3, 5).reshape(N, K)
     assert torch.equal(back, w)
+
+
+def test_wstream_gu_fused():
+    """Fused gate_up+SwiGLU kernel vs F.linear + fp32-reference swiglu."""
+    hip = _hip()
+    for M, F, K in [(32, 14336, 4096), (1, 14336, 4096), (7, 512, 512),
+                    (32, 9472, 3584)]:
+        x = rnd(M, K, seed=M + F)
+        w = rnd(2 * F, K, seed=K + F)
+        pk = hip.pack_weight_gu(w)
+        assert pk is not None
+        act = hip.linear_gu(x, pk, 2 * F)
+        gu = torch.nn.functional.linear(x.float(), w.float())
+        g, u = gu[:, :F], gu[:, F:]
+        expect = torch.nn.functional.silu(g) * u
+        torch.testing.assert_close(act.float(), expect,
+                                   atol=8e-2, rtol=8e-2)
