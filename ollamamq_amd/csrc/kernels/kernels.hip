@@ -726,3 +726,35 @@ extern "C" int embed_gather_bf16(void* out, const void* table,
         (uint4*)out, (const uint4*)table, (const int*)tokens, cpr, total);
     return (int)hipGetLastError();
 }
+
+// ---------------------------------------------------------------------------
+// Row sum-of-squares: sq[m] = sum_k x[m][k]^2 (fp32).  Seeds the fused
+// rmsnorm-in-GEMM decode chain after the embedding gather (wstream_gemm
+// epilogues produce the per-tile partials for every later layer).
+// ---------------------------------------------------------------------------
+__global__ __launch_bounds__(256) void k_row_sumsq(
+    const bf16* __restrict__ x, float* __restrict__ sq, int H)
+{
+    const int m = blockIdx.x;
+    const bf16* row = x + (int64_t)m * H;
+    float s = 0.f;
+    for (int i = threadIdx.x * 8; i < H; i += 256 * 8) {
+        float v[8];
+        load8f(row + i, v);
+        #pragma unroll
+        for (int j = 0; j < 8; j++) s += v[j] * v[j];
+    }
+    __shared__ float red[4];
+    s = wave_reduce_sum(s);
+    if ((threadIdx.x & 63) == 0) red[threadIdx.x >> 6] = s;
+    __syncthreads();
+    if (threadIdx.x == 0)
+        sq[m] = red[0] + red[1] + red[2] + red[3];
+}
+
+extern "C" int row_sumsq_bf16(void* sq, const void* x, int M, int H,
+                              hipStream_t stream)
+{
+    k_row_sumsq<<<M, 256, 0, stream>>>((const bf16*)x, (float*)sq, H);
+    return (int)hipGetLastError();
+}

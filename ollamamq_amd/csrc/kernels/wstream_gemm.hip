@@ -94,6 +94,15 @@ extern "C" int wstream_pure_bf16(void* sink, const void* wp, int N, int K,
 // columns; the epilogue computes silu(g)*u and stores [M, 16] of the
 // activation directly.  Kills the separate swiglu kernel (9.7 us x 32
 // layers/step in the r02 trace) and the gate_up intermediate round-trip.
+// Fused-rmsnorm chain extensions (all nullable; tp=1 decode path):
+//  * rstd_parts/rstd_nt: per-tile sum-of-squares partials of the RAW
+//    residual state x (written by the producing GEMM's epilogue).  The
+//    epilogue scales row m by rstd[m] = rsqrt(sum*inv_h + eps) — the
+//    norm WEIGHT is folded into the packed W at pack time, so
+//    y = rmsnorm(x)@W' without any rmsnorm kernel in the step.
+//  * res_in/sq_parts: epilogue adds the residual stream (in-place safe:
+//    each (m,n) is read+written by exactly one block) and emits this
+//    block's sum-of-squares partials for the NEXT GEMM's rstd.
 template <int MT, int DEPTH = 1, int XLDS = 0, int GU = 0>
 __global__ __launch_bounds__(512) void k_wstream_gemm(
     bf16* __restrict__ y,            // [M, N] (ksplit == 1)
@@ -101,7 +110,10 @@ __global__ __launch_bounds__(512) void k_wstream_gemm(
     const bf16* __restrict__ x,      // [M, K], row stride xs
     const u32x4* __restrict__ wp,    // packed W (see header)
     const bf16* __restrict__ bias,   // [N] or null (ksplit==1 path)
-    int M, int N, int K, int64_t xs, int ksplit)
+    int M, int N, int K, int64_t xs, int ksplit,
+    const float* __restrict__ rstd_parts, int rstd_nt, float inv_h,
+    float eps, const bf16* __restrict__ res_in,
+    float* __restrict__ sq_parts)
 {
     const int t = blockIdx.x;              // n-tile (32 cols of y)
     const int ks = blockIdx.y;
@@ -267,6 +279,17 @@ __global__ __launch_bounds__(512) void k_wstream_gemm(
     // ---- reduce 8 wave partials through LDS (aliases the x tiles: the
     // barrier closes the streaming epoch before anyone writes) ----
     float (*red8)[32][32] = reinterpret_cast<float (*)[32][32]>(smem);
+    // rmsnorm scale of the raw-residual input: sum this row's sq
+    // partials from the producing GEMM (L2-hot, overlapped with other
+    // waves' stream drain), publish rstd through the smem tail
+    float* rstd_sh = reinterpret_cast<float*>(
+        smem + (XLDS ? 16 * 4096 : 8 * 32 * 32 * 4));
+    if (rstd_parts && tid < 32) {
+        float s = 0.f;
+        for (int i = 0; i < rstd_nt; ++i)
+            s += rstd_parts[i * 32 + tid];
+        rstd_sh[tid] = rsqrtf(s * inv_h + eps);
+    }
     if (GU) {
         __syncthreads();
         #pragma unroll
@@ -284,6 +307,11 @@ __global__ __launch_bounds__(512) void k_wstream_gemm(
                 sg += red8[wv][m][c];
                 su += red8[wv][m][c + 16];
             }
+            if (rstd_parts) {
+                const float rs = rstd_sh[m];   // pre-silu: nonlinear
+                sg *= rs;
+                su *= rs;
+            }
             const float act = (sg / (1.f + __expf(-sg))) * su;
             y[(int64_t)m * F + t * 16 + c] = __float2bfloat16(act);
         }
@@ -300,8 +328,10 @@ __global__ __launch_bounds__(512) void k_wstream_gemm(
             red8[wid][crow][lane & 31] = acc[r];
         }
         __syncthreads();
+        float rsq[2] = {0.f, 0.f};
         #pragma unroll
-        for (int e = tid; e < 1024; e += 512) {
+        for (int ee = 0; ee < 2; ee++) {
+            const int e = tid + ee * 512;
             const int m = (e >> 5) + mt * 32, n = e & 31;
             if (m < M) {
                 float s = 0.f;
@@ -309,11 +339,31 @@ __global__ __launch_bounds__(512) void k_wstream_gemm(
                 for (int wv = 0; wv < 8; wv++)
                     s += red8[wv][m - mt * 32][n];
                 if (ksplit == 1) {
+                    if (rstd_parts) s *= rstd_sh[m];
                     if (bias) s += __bfloat162float(bias[n0 + n]);
+                    if (res_in) {
+                        s += __bfloat162float(res_in[(int64_t)m * N
+                                                     + n0 + n]);
+                        rsq[ee] = s * s;
+                    }
                     y[(int64_t)m * N + n0 + n] = __float2bfloat16(s);
                 } else {
                     part[((int64_t)ks * M + m) * N + n0 + n] = s;
                 }
+            }
+        }
+        // emit this block's sum-of-squares partials (M<=32, MT==1 only)
+        if (MT == 1 && sq_parts && ksplit == 1) {
+            __syncthreads();
+            float (*sqt)[32] = reinterpret_cast<float (*)[32]>(smem);
+            sqt[tid >> 5][tid & 31] = rsq[0];          // m 0..15
+            sqt[(tid >> 5) + 16][tid & 31] = rsq[1];   // m 16..31
+            __syncthreads();
+            if (tid < 32) {
+                float s = 0.f;
+                #pragma unroll
+                for (int n2 = 0; n2 < 32; n2++) s += sqt[tid][n2];
+                sq_parts[t * 32 + tid] = s;
             }
         }
     }
@@ -321,20 +371,24 @@ __global__ __launch_bounds__(512) void k_wstream_gemm(
 
 // Fused gate_up @ SwiGLU over GU-interleaved packed weights.
 // N = 2F total weight rows; output is the activation [M, F].
+// rstd_parts: optional fused-rmsnorm scale of the raw-residual input.
 extern "C" int wstream_gu_bf16(
     void* act, const void* x, const void* wp, int M, int N, int K,
-    int64_t xs, int xlds, hipStream_t stream)
+    int64_t xs, int xlds, const void* rstd_parts, int rstd_nt,
+    float inv_h, float eps, hipStream_t stream)
 {
     dim3 grid(N / 32, 1);
-    const int lds = xlds ? 16 * 4096 : 8 * 32 * 32 * 4;
+    const int lds = (xlds ? 16 * 4096 : 8 * 32 * 32 * 4) + 128;
     if (xlds)
         k_wstream_gemm<1, 1, 1, 1><<<grid, 512, lds, stream>>>(
             (bf16*)act, nullptr, (const bf16*)x, (const u32x4*)wp,
-            nullptr, M, N, K, xs, 1);
+            nullptr, M, N, K, xs, 1, (const float*)rstd_parts, rstd_nt,
+            inv_h, eps, nullptr, nullptr);
     else
         k_wstream_gemm<1, 1, 0, 1><<<grid, 512, lds, stream>>>(
             (bf16*)act, nullptr, (const bf16*)x, (const u32x4*)wp,
-            nullptr, M, N, K, xs, 1);
+            nullptr, M, N, K, xs, 1, (const float*)rstd_parts, rstd_nt,
+            inv_h, eps, nullptr, nullptr);
     return (int)hipGetLastError();
 }
 
@@ -352,44 +406,101 @@ __global__ __launch_bounds__(256) void k_wstream_combine(
     }
 }
 
+// Tile-structured combine for the fused chain (M <= 32): folds the
+// k-split partials, adds the residual stream in place and emits per-tile
+// sum-of-squares partials for the next GEMM's rstd.
+__global__ __launch_bounds__(256) void k_wstream_combine_tiles(
+    bf16* __restrict__ y, const float* __restrict__ part,
+    const bf16* __restrict__ res_in, float* __restrict__ sq_parts,
+    int M, int N, int ksplit)
+{
+    const int t = blockIdx.x;
+    const int tid = threadIdx.x;
+    __shared__ float sqt[32][32];
+    const int n0 = t * 32;
+    #pragma unroll
+    for (int ee = 0; ee < 4; ee++) {
+        const int e = tid + ee * 256;
+        const int m = e >> 5, n = e & 31;
+        float rsq = 0.f;
+        if (m < M) {
+            float s = 0.f;
+            for (int k = 0; k < ksplit; k++)
+                s += part[((int64_t)k * M + m) * N + n0 + n];
+            if (res_in)
+                s += __bfloat162float(res_in[(int64_t)m * N + n0 + n]);
+            y[(int64_t)m * N + n0 + n] = __float2bfloat16(s);
+            rsq = s * s;
+        }
+        sqt[m][n] = rsq;
+    }
+    if (sq_parts) {
+        __syncthreads();
+        if (tid < 32) {
+            float s = 0.f;
+            #pragma unroll
+            for (int n2 = 0; n2 < 32; n2++) s += sqt[tid][n2];
+            sq_parts[t * 32 + tid] = s;
+        }
+    }
+}
+
 extern "C" int wstream_gemm_bf16(
     void* y, void* part, const void* x, const void* wp, const void* bias,
     int M, int N, int K, int64_t xs, int ksplit, int depth, int xlds,
-    hipStream_t stream)
+    const void* rstd_parts, int rstd_nt, float inv_h, float eps,
+    const void* res_in, void* sq_parts, hipStream_t stream)
 {
     dim3 grid(N / 32, ksplit);
-    const int lds_red = 8 * 32 * 32 * 4;           // 32 KiB reduce
-    const int lds_x = 16 * 4096;                   // + 8x2 x tiles
+    const int lds_red = 8 * 32 * 32 * 4 + 128;     // reduce + rstd tail
+    const int lds_x = 16 * 4096 + 128;             // + 8x2 x tiles
+    const float* rp = (const float*)rstd_parts;
+    const bf16* ri = ksplit == 1 ? (const bf16*)res_in : nullptr;
+    float* sq = ksplit == 1 ? (float*)sq_parts : nullptr;
+    if (ksplit > 1 && rstd_parts) return -100;     // fused rstd needs ks==1
     if (M <= 32) {
         if (xlds)
             k_wstream_gemm<1, 1, 1><<<grid, 512, lds_x, stream>>>(
                 (bf16*)y, (float*)part, (const bf16*)x, (const u32x4*)wp,
-                (const bf16*)bias, M, N, K, xs, ksplit);
+                (const bf16*)bias, M, N, K, xs, ksplit, rp, rstd_nt,
+                inv_h, eps, ri, sq);
         else if (depth == 2)
             k_wstream_gemm<1, 2><<<grid, 512, lds_red, stream>>>(
                 (bf16*)y, (float*)part, (const bf16*)x, (const u32x4*)wp,
-                (const bf16*)bias, M, N, K, xs, ksplit);
+                (const bf16*)bias, M, N, K, xs, ksplit, rp, rstd_nt,
+                inv_h, eps, ri, sq);
         else
             k_wstream_gemm<1, 1><<<grid, 512, lds_red, stream>>>(
                 (bf16*)y, (float*)part, (const bf16*)x, (const u32x4*)wp,
-                (const bf16*)bias, M, N, K, xs, ksplit);
+                (const bf16*)bias, M, N, K, xs, ksplit, rp, rstd_nt,
+                inv_h, eps, ri, sq);
     } else {
+        if (rstd_parts || res_in || sq_parts) return -101;  // M<=32 only
         if (depth == 2)
             k_wstream_gemm<2, 2><<<grid, 512, lds_red, stream>>>(
                 (bf16*)y, (float*)part, (const bf16*)x, (const u32x4*)wp,
-                (const bf16*)bias, M, N, K, xs, ksplit);
+                (const bf16*)bias, M, N, K, xs, ksplit, nullptr, 0,
+                0.f, 0.f, nullptr, nullptr);
         else
             k_wstream_gemm<2, 1><<<grid, 512, lds_red, stream>>>(
                 (bf16*)y, (float*)part, (const bf16*)x, (const u32x4*)wp,
-                (const bf16*)bias, M, N, K, xs, ksplit);
+                (const bf16*)bias, M, N, K, xs, ksplit, nullptr, 0,
+                0.f, 0.f, nullptr, nullptr);
     }
     if (ksplit > 1) {
-        const int64_t mn = (int64_t)M * N;
-        const int64_t want = (mn + 255) / 256;
-        const int blocks = (int)(want < 1024 ? want : 1024);
-        k_wstream_combine<<<blocks, 256, 0, stream>>>(
-            (bf16*)y, (const float*)part, (const bf16*)bias, mn, mn, N,
-            ksplit);
+        if (res_in || sq_parts) {
+            k_wstream_combine_tiles<<<N / 32, 256, 0, stream>>>(
+                (bf16*)y, (const float*)part, (const bf16*)res_in,
+                (float*)sq_parts, M, N, ksplit);
+        } else {
+            const int64_t mn = (int64_t)M * N;
+            const int64_t want = (mn + 255) / 256;
+            const int blocks = (int)(want < 1024 ? want : 1024);
+            k_wstream_combine<<<blocks, 256, 0, stream>>>(
+                (bf16*)y, (const float*)part, (const bf16*)bias, mn, mn,
+                N, ksplit);
+        }
     }
     return (int)hipGetLastError();
 }
+

@@ -251,6 +251,7 @@ class LlamaModel:
         self.rope_cos = ang.cos().to(dev)
         self.rope_sin = ang.sin().to(dev)
         self.lm_head_pk = None
+        self.fused_chain = False
         self._pack_weights()
 
     def _pack_weights(self):
@@ -266,22 +267,39 @@ class LlamaModel:
         free, _ = torch.cuda.mem_get_info(self.device)
         if self.weight_bytes() > free * 0.45:
             return
-        # a packed copy nothing will read is pure HBM waste: only pack
-        # tensors the decode dispatch will actually route to the
-        # streaming kernel (N <= the measured lib-wins threshold)
+        # The rmsnorm WEIGHT is folded into the qkv / gate_up / lm_head
+        # packs (rmsnorm(x)*w @ W^T == rmsnorm_unit(x) @ (W*diag(w))^T),
+        # so the fused decode chain needs no rmsnorm kernel at all — the
+        # GEMM epilogues carry rstd scaling + residual + stats.  Folded
+        # packs are ONLY read by the fused chain; the generic path keeps
+        # explicit rmsnorm + unfolded weights.
         pk = ops.pack_weight
         try:
             for l in self.layers:
-                l.wqkv_pk = pk(l.wqkv)
+                l.wqkv_pk = pk((l.wqkv * l.attn_norm).contiguous())
                 l.wo_pk = pk(l.wo)
-                l.wgu_pk = ops.pack_weight_gu(l.wgate_up)
+                l.wgu_pk = ops.pack_weight_gu(
+                    (l.wgate_up * l.mlp_norm).contiguous())
                 l.wdown_pk = pk(l.wdown)
-            self.lm_head_pk = pk(self.lm_head)
+            self.lm_head_pk = pk((self.lm_head * self.final_norm)
+                                 .contiguous())
         except torch.cuda.OutOfMemoryError:
             for l in self.layers:
                 l.wqkv_pk = l.wo_pk = l.wgu_pk = l.wdown_pk = None
             self.lm_head_pk = None
             torch.cuda.empty_cache()
+            return
+        self.fused_chain = (
+            self.tp_size == 1 and self.group is None
+            and self.lm_head_pk is not None
+            and all(l.wqkv_pk is not None and l.wo_pk is not None
+                    and l.wgu_pk is not None and l.wdown_pk is not None
+                    for l in self.layers))
+        if self.fused_chain:
+            th = self.cfg.hidden // 32
+            self._sq_a = torch.zeros(max(th, 1) * 32, dtype=torch.float32,
+                                     device=self.device)
+            self._sq_b = torch.zeros_like(self._sq_a)
 
     # -- helpers -----------------------------------------------------------
     def _allreduce(self, x: torch.Tensor) -> torch.Tensor:
@@ -317,14 +335,19 @@ class LlamaModel:
         """Returns logits [T_last, vocab_full] for the tokens attn_meta
         selects as "last" (decode: all; prefill: final token per seq)."""
         cfg = self.cfg
+        if (self.fused_chain and attn_meta.mode == "decode"
+                and tokens.shape[0] <= 32 and not return_hidden):
+            return self._forward_decode_fused(tokens, positions, kv_cache,
+                                              slot_ids, attn_meta)
         x = ops.embedding(tokens, self.embed)
         residual = None
         for li, layer in enumerate(self.layers):
             normed, residual = ops.rmsnorm_residual(
                 x, residual, layer.attn_norm, cfg.norm_eps
             )
-            qkv = ops.linear(normed, layer.wqkv, layer.bqkv,
-                             packed=layer.wqkv_pk)
+            # NB: no packed= here — wqkv_pk has the norm weight folded in
+            # and is readable only by the fused chain
+            qkv = ops.linear(normed, layer.wqkv, layer.bqkv)
             nl, nkl, d = self.n_local_heads, self.n_local_kv_heads, cfg.head_dim
             q, k, v = qkv.split([nl * d, nkl * d, nkl * d], dim=-1)
             q = q.view(-1, nl, d)
@@ -339,8 +362,7 @@ class LlamaModel:
             normed, residual = ops.rmsnorm_residual(
                 x, residual, layer.mlp_norm, cfg.norm_eps
             )
-            act = ops.gateup_swiglu(normed, layer.wgate_up,
-                                    layer.wgu_pk)
+            act = ops.gateup_swiglu(normed, layer.wgate_up)
             x = ops.linear(act, layer.wdown, packed=layer.wdown_pk)
             self._allreduce(x)  # RCCL all-reduce #2 (TP)
 
@@ -353,10 +375,55 @@ class LlamaModel:
         h, _ = ops.rmsnorm_residual(h, None, self.final_norm, cfg.norm_eps)
         if return_hidden:
             return h
-        logits = ops.linear(h, self.lm_head, packed=self.lm_head_pk)
+        logits = ops.linear(h, self.lm_head)
         if self.tp_size > 1:
             # vocab-parallel logits: all-gather shards on the last dim
             shards = [torch.empty_like(logits) for _ in range(self.tp_size)]
             torch.distributed.all_gather(shards, logits, group=self.group)
             logits = torch.cat(shards, dim=-1)
         return logits
+
+    def _forward_decode_fused(self, tokens, positions, kv_cache, slot_ids,
+                              attn_meta):
+        """Decode step as a fused weight-streaming chain (tp=1, B<=32):
+        4 GEMM-class kernels per layer, zero rmsnorm/swiglu kernels.
+        rmsnorm weights are folded into the packs; the rstd scale rides
+        each consumer GEMM's epilogue from sum-of-squares partials the
+        producer GEMM emitted; residual adds happen in place inside the
+        o/down epilogues (SURVEY.md §2 kernel-table GEMM rows — this is
+        the MFMA-native decode path)."""
+        cfg = self.cfg
+        B = tokens.shape[0]
+        nl, nkl, d = self.n_local_heads, self.n_local_kv_heads, cfg.head_dim
+        res = ops.embedding(tokens, self.embed)
+        if not res.is_contiguous():
+            res = res.contiguous()
+        sq_a, sq_b = self._sq_a, self._sq_b
+        ops.row_sumsq(res, out=sq_a[:B])
+        nt = 1                      # embed stats = one partial tile
+        inv_h = 1.0 / cfg.hidden
+        eps = cfg.norm_eps
+        tiles_h = cfg.hidden // 32
+        for li, layer in enumerate(self.layers):
+            qkv = ops.linear_fused(res, layer.wqkv_pk,
+                                   layer.wqkv.shape[0], bias=layer.bqkv,
+                                   rstd=sq_a, rstd_nt=nt, inv_h=inv_h,
+                                   eps=eps)
+            q, k, v = qkv.split([nl * d, nkl * d, nkl * d], dim=-1)
+            q = q.view(-1, nl, d)
+            k = k.view(-1, nkl, d)
+            v = v.view(-1, nkl, d)
+            ops.rope_append(kv_cache, li, q, k, v, positions, slot_ids,
+                            self.rope_cos, self.rope_sin)
+            attn = ops.attention(q, kv_cache, li, attn_meta)
+            ops.linear_fused(attn.view(-1, nl * d), layer.wo_pk,
+                             cfg.hidden, res=res, sq_out=sq_b, y=res)
+            act = ops.gu_fused(res, layer.wgu_pk,
+                               layer.wgate_up.shape[0], rstd=sq_b,
+                               rstd_nt=tiles_h, inv_h=inv_h, eps=eps)
+            ops.linear_fused(act, layer.wdown_pk, cfg.hidden,
+                             res=res, sq_out=sq_a, y=res)
+            nt = tiles_h
+        return ops.linear_fused(res, self.lm_head_pk,
+                                self.lm_head.shape[0], rstd=sq_a,
+                                rstd_nt=nt, inv_h=inv_h, eps=eps)

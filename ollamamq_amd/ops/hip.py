@@ -55,9 +55,11 @@ def _try_load():
                                       i, i, i, i, i, f, i64, i, vp]
     lib.embed_gather_bf16.argtypes = [vp, vp, vp, i, i, vp]
     lib.wstream_gemm_bf16.argtypes = [vp, vp, vp, vp, vp, i, i, i, i64, i,
-                                      i, i, vp]
+                                      i, i, vp, i, f, f, vp, vp, vp]
     lib.wstream_pure_bf16.argtypes = [vp, vp, i, i, i, vp]
-    lib.wstream_gu_bf16.argtypes = [vp, vp, vp, i, i, i, i64, i, vp]
+    lib.wstream_gu_bf16.argtypes = [vp, vp, vp, i, i, i, i64, i,
+                                    vp, i, f, f, vp]
+    lib.row_sumsq_bf16.argtypes = [vp, vp, i, i, vp]
     for fn in ("rmsnorm_residual_bf16", "rope_bf16", "kv_append_bf16",
                "paged_attn_bf16", "swiglu_bf16", "argmax_bf16",
                "decode_attn_bf16", "skinny_gemm_bf16",
@@ -446,14 +448,17 @@ def pack_weight_gu(w):
     return p.view(-1)
 
 
-def linear_gu(x, packed, N):
-    """act = swiglu(x @ Wgu^T) fused; N = 2F total weight rows."""
+def linear_gu(x, packed, N, rstd=None, rstd_nt=0, inv_h=0.0, eps=0.0):
+    """act = swiglu(x @ Wgu^T) fused; N = 2F total weight rows.
+    With rstd: x is the raw residual and the epilogue applies the
+    rmsnorm scale before SwiGLU (norm weight folded into the pack)."""
     M, K = x.shape
     F = N // 2
     act = torch.empty((M, F), dtype=x.dtype, device=x.device)
     xlds = 1 if N * K * 2 > (64 << 20) else 0
     _check(_lib.wstream_gu_bf16(
         _p(act), _p(x), _p(packed), M, N, K, x.stride(0), xlds,
+        _p(rstd), rstd_nt, float(inv_h), float(eps),
         _stream()), "wstream_gu")
     return act
 
@@ -493,12 +498,24 @@ _WS_KS = os.environ.get("OLLAMAMQ_WS_KS")   # sweep override
 _WS_MAX_N = int(os.environ.get("OLLAMAMQ_WS_MAX_N", "16384"))
 
 
-def linear_packed(x, packed, bias, N, ks=None, depth=None, xlds=None):
-    """y = x @ W^T via the weight-streaming kernel over pre-packed W."""
+def linear_packed(x, packed, bias, N, ks=None, depth=None, xlds=None,
+                  rstd=None, rstd_nt=0, inv_h=0.0, eps=0.0,
+                  res=None, sq_out=None, y=None):
+    """y = x @ W^T via the weight-streaming kernel over pre-packed W.
+
+    Fused-chain extras (decode, M<=32): `rstd`/`rstd_nt`/`inv_h`/`eps`
+    apply rmsnorm scaling of the raw-residual input x (norm weight must
+    be folded into the pack); `res` adds the residual stream into the
+    output (pass y=res for the in-place residual update) and `sq_out`
+    emits per-tile sum-of-squares partials for the next GEMM."""
     M, K = x.shape
-    y = torch.empty((M, N), dtype=x.dtype, device=x.device)
+    if y is None:
+        y = torch.empty((M, N), dtype=x.dtype, device=x.device)
     if ks is None:
         ks = int(_WS_KS) if _WS_KS else _wstream_ksplit(N, K)
+    if rstd is not None:
+        ks_ok = ks        # rstd fusion requires ks == 1
+        ks = 1
     part = ctypes.c_void_p(0)
     if ks > 1:
         key = ("ws", M, N, ks, str(x.device))
@@ -515,9 +532,20 @@ def linear_packed(x, packed, bias, N, ks=None, depth=None, xlds=None):
         xlds = (_WS_XLDS if M <= 32 and N * K * 2 > (64 << 20) else 0)
     _check(_lib.wstream_gemm_bf16(
         _p(y), part, _p(x), _p(packed), _p(bias), M, N, K, x.stride(0),
-        ks, depth if depth is not None else _WS_DEPTH, xlds, _stream()),
-        "wstream_gemm")
+        ks, depth if depth is not None else _WS_DEPTH, xlds,
+        _p(rstd), rstd_nt, float(inv_h), float(eps), _p(res), _p(sq_out),
+        _stream()), "wstream_gemm")
     return y
+
+
+def row_sumsq(x, out=None):
+    """sq[m] = sum(x[m]**2) fp32 — seeds the fused-rmsnorm chain."""
+    M, H = x.shape
+    if out is None:
+        out = torch.empty(M, dtype=torch.float32, device=x.device)
+    _check(_lib.row_sumsq_bf16(_p(out), _p(x), M, H, _stream()),
+           "row_sumsq")
+    return out
 
 
 def wstream_pure(packed, N, K, ks):

@@ -116,13 +116,12 @@ def linear(x, weight, bias=None, packed=None):
 
 def pack_weight(w):
     """MFMA-fragment-order packed copy for the decode GEMM (GPU only).
-    Returns None for shapes the decode dispatch would never route to the
-    streaming kernel (N above the measured lib-wins threshold) so no HBM
-    is spent on copies nothing reads."""
+    Every projection is packed: the fused decode chain streams all of
+    them (the per-shape lib-vs-stream policy only applies to the generic
+    non-chain dispatch in hip.linear)."""
     if w.is_cuda:
         from . import hip
-        if hip.available() and (w.shape[0] <= hip._WS_MAX_N
-                                or w.shape[0] >= 65536):
+        if hip.available():
             return hip.pack_weight(w)
     return None
 
@@ -153,6 +152,40 @@ def pack_weight_gu(w):
         if hip.available():
             return hip.pack_weight_gu(w)
     return None
+
+
+# ---- fused-rmsnorm decode chain (GPU-only; the model gates on
+# fused_chain availability — tp=1, decode, batch<=32) -----------------
+
+def row_sumsq(x, out=None):
+    """sq[m] = sum(x[m]**2) in fp32 (seeds the chain after embedding)."""
+    if _use_hip(x):
+        from . import hip
+        return hip.row_sumsq(x, out)
+    s = (x.float() ** 2).sum(-1)
+    if out is not None:
+        out.copy_(s)
+        return out
+    return s
+
+
+def linear_fused(x, packed, N, bias=None, rstd=None, rstd_nt=0,
+                 inv_h=0.0, eps=0.0, res=None, sq_out=None, y=None):
+    """Weight-streaming GEMM with the fused-chain epilogue/prologue:
+    optional rmsnorm scaling of the raw-residual input (norm weight
+    folded into the pack), in-place residual add (pass y=res) and
+    sum-of-squares partial emission for the next GEMM's rstd."""
+    from . import hip
+    return hip.linear_packed(x, packed, bias, N, rstd=rstd,
+                             rstd_nt=rstd_nt, inv_h=inv_h, eps=eps,
+                             res=res, sq_out=sq_out, y=y)
+
+
+def gu_fused(x, packed, N, rstd, rstd_nt, inv_h, eps):
+    """rmsnorm -> gate_up GEMM -> SwiGLU, one kernel."""
+    from . import hip
+    return hip.linear_gu(x, packed, N, rstd=rstd, rstd_nt=rstd_nt,
+                         inv_h=inv_h, eps=eps)
 
 
 def sample(logits, temperature, top_k, top_p, generator=None):
