@@ -284,11 +284,20 @@ __global__ __launch_bounds__(512) void k_wstream_gemm(
     // waves' stream drain), publish rstd through the smem tail
     float* rstd_sh = reinterpret_cast<float*>(
         smem + (XLDS ? 16 * 4096 : 8 * 32 * 32 * 4));
-    if (rstd_parts && tid < 32) {
+    if (rstd_parts) {
+        // parallel partial fold: 16 lanes per row (each <= nt/16 loads,
+        // independent), group-reduced with wave shuffles — a serial
+        // 32-thread loop here measurably stalled short-stream blocks
+        const int m2 = tid >> 4, c2 = tid & 15;
         float s = 0.f;
-        for (int i = 0; i < rstd_nt; ++i)
-            s += rstd_parts[i * 32 + tid];
-        rstd_sh[tid] = rsqrtf(s * inv_h + eps);
+        if (m2 < 32)
+            for (int i = c2; i < rstd_nt; i += 16)
+                s += rstd_parts[i * 32 + m2];
+        #pragma unroll
+        for (int off = 8; off; off >>= 1)
+            s += __shfl_down(s, off, 64);
+        if (m2 < 32 && c2 == 0)
+            rstd_sh[m2] = rsqrtf(s * inv_h + eps);
     }
     if (GU) {
         __syncthreads();

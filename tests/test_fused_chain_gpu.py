@@ -100,11 +100,10 @@ def test_fused_chain_logits_close():
                                             device=dev),
                     max_q=1, max_kv=max(lens), window=0)
     # generic first (it appends KV; rewind lens between runs)
-    import copy
-    saved = dict(kv.seq_lens)
+    saved = list(kv.seq_lens)
     model.fused_chain = False
     lg = model.forward(toks, pos, kv, st, meta)
-    kv.seq_lens.update(saved)
+    kv.seq_lens[:] = saved
     meta2 = AttnMeta(mode="decode", slot_ids=st, seq_lens=meta.seq_lens,
                      cu_q=meta.cu_q, logits_idx=meta.logits_idx,
                      max_q=1, max_kv=max(lens), window=0)
