@@ -292,7 +292,7 @@ __global__ __launch_bounds__(512) void k_wstream_gemm(
         float s = 0.f;
         if (m2 < 32)
             for (int i = c2; i < rstd_nt; i += 16)
-                s += rstd_parts[i * 32 + m2];
+                s += rstd_parts[(int64_t)m2 * rstd_nt + i];
         #pragma unroll
         for (int off = 8; off; off >>= 1)
             s += __shfl_down(s, off, 64);
@@ -372,7 +372,10 @@ __global__ __launch_bounds__(512) void k_wstream_gemm(
                 float s = 0.f;
                 #pragma unroll
                 for (int n2 = 0; n2 < 32; n2++) s += sqt[tid][n2];
-                sq_parts[t * 32 + tid] = s;
+                // m-major [32][n_tiles]: the consumer's fold walks tiles
+                // CONTIGUOUSLY (a tile-major layout cost 1 float per
+                // 128 B line and ~3 us per consumer block)
+                sq_parts[(int64_t)tid * gridDim.x + t] = s;
             }
         }
     }
@@ -449,7 +452,7 @@ __global__ __launch_bounds__(256) void k_wstream_combine_tiles(
             float s = 0.f;
             #pragma unroll
             for (int n2 = 0; n2 < 32; n2++) s += sqt[tid][n2];
-            sq_parts[t * 32 + tid] = s;
+            sq_parts[(int64_t)tid * gridDim.x + t] = s;   // m-major
         }
     }
 }
