@@ -291,6 +291,7 @@ class LlamaModel:
             return
         self.fused_chain = (
             self.tp_size == 1 and self.group is None
+            and os.environ.get("OLLAMAMQ_NO_CHAIN") != "1"
             and self.lm_head_pk is not None
             and all(l.wqkv_pk is not None and l.wo_pk is not None
                     and l.wgu_pk is not None and l.wdown_pk is not None

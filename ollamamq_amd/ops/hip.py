@@ -512,7 +512,11 @@ def linear_packed(x, packed, bias, N, ks=None, depth=None, xlds=None,
     if y is None:
         y = torch.empty((M, N), dtype=x.dtype, device=x.device)
     if ks is None:
-        ks = int(_WS_KS) if _WS_KS else _wstream_ksplit(N, K)
+        per_k = os.environ.get(f"OLLAMAMQ_WS_KS_{K}")   # per-shape tuning
+        if per_k:
+            ks = int(per_k)
+        else:
+            ks = int(_WS_KS) if _WS_KS else _wstream_ksplit(N, K)
     if rstd is not None:
         ks_ok = ks        # rstd fusion requires ks == 1
         ks = 1
