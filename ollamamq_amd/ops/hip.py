@@ -467,21 +467,31 @@ USE_WSTREAM = os.environ.get("OLLAMAMQ_NO_WSTREAM") != "1"
 
 
 def _wstream_ksplit(N, K):
-    """r02 sweep (profiles/r02_gemm_sweep.md): the sweet spot is ~8
-    k-iterations per wave — deep enough that the register prefetch
-    pipelines, while blocks stay >= 128 so the grid isn't starved.  Pick
-    the ks whose iters/wave is nearest 8 (ties -> smaller ks: fewer
-    partials)."""
+    """Measured rules (profiles/r02_gemm_sweep.md + in-graph A/B):
+    * small weights (< 64 MB: qkv/o) are ramp/latency-floor-bound —
+      fewer blocks with longer streams win (ks=1 beat every split);
+    * big weights need >= 256 blocks to reach the full-chip stream rate
+      (down at 128 blocks ran at half the chip: 41 vs 26 us);
+    * uneven wave ranges (NB % 8ks != 0) cost ~10% tail imbalance
+      (down ks3 measured worse than both ks2 and ks4);
+    * very long (>16 iters) or very short (<4) per-wave streams lose."""
     tiles = N // 32
     nb = K // 64
+    big = N * K * 2 > (64 << 20)
     best, best_cost = 1, 1 << 30
     for ks in range(1, 9):
         iters = nb // (ks * 8)
         if iters < 1:
             break
-        cost = abs(iters - 8) * 4 + ks
-        if tiles * ks < 128:
-            cost += 64          # starved grid: strong penalty
+        cost = ks
+        if nb % (ks * 8):
+            cost += 16
+        if big and tiles * ks < 256:
+            cost += 24
+        if iters < 4:
+            cost += (4 - iters) * 8
+        if iters > 16:
+            cost += (iters - 16) * 2
         if cost < best_cost:
             best, best_cost = ks, cost
     return best
