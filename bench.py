@@ -24,9 +24,134 @@ import json
 import os
 import statistics
 import sys
+import threading
 import time
 
 import torch
+
+
+def run_stack(args):
+    """Full-stack serving benchmark: native dispatcher + UDS engine
+    workers + HTTP streaming clients.  The timed
+    region crosses the whole reference-parity path — HTTP ingress,
+    per-user queues, scheduler, UDS hop, engine, token stream-out — and
+    p50 queue-wait is the DISPATCHER's own /admin/stats definition
+    (enqueue -> dispatch), not engine TTFT."""
+    import socket
+    import subprocess
+    import urllib.request
+
+    has_gpu = torch.cuda.is_available()
+    model = args.model or ("llama3-8b" if has_gpu else "tiny-cpu")
+    users = args.users if has_gpu else min(args.users, 4)
+    port = 11640 + (os.getpid() % 199)
+    max_tokens = 48 if has_gpu else 12
+    prompt_len = min(args.prompt_len, 256) if has_gpu else 16
+    warm_s = max(4, args.warmup)
+    meas_s = max(8, args.steps // 3) if has_gpu else max(6, args.steps // 6)
+
+    cmd = [sys.executable, "-m", "ollamamq_amd.launch",
+           "--gpus", str(args.gpus), "--model", model,
+           "--max-ctx", str(args.max_ctx),
+           "--max-batch", str(max(32, users)),
+           "--port", str(port), "--no-tui",
+           "-c", "/nonexistent/appconf.yaml"]
+    node = subprocess.Popen(cmd, stdout=subprocess.DEVNULL,
+                            stderr=subprocess.DEVNULL,
+                            start_new_session=True)
+    base = f"http://127.0.0.1:{port}"
+    try:
+        deadline = time.time() + 600
+        while time.time() < deadline:
+            try:
+                with urllib.request.urlopen(base + "/health",
+                                            timeout=2) as r:
+                    if r.read() == b"OK":
+                        break
+            except OSError:
+                time.sleep(0.5)
+        else:
+            raise SystemExit("stack did not come up")
+
+        stop = threading.Event()
+        token_ts = []          # (time, 1) per streamed token
+        ts_lock = threading.Lock()
+
+        def user_loop(uid):
+            body = json.dumps({
+                "model": model, "prompt": "x" * prompt_len,
+                "options": {"num_predict": max_tokens}, "stream": True,
+            }).encode()
+            while not stop.is_set():
+                try:
+                    req = urllib.request.Request(
+                        base + "/api/generate", data=body,
+                        headers={"Content-Type": "application/json",
+                                 "X-User-ID": f"bench-user-{uid}"})
+                    local = []
+                    with urllib.request.urlopen(req, timeout=120) as r:
+                        for line in r:
+                            if b'"done": false' in line or \
+                                    b'"done":false' in line:
+                                local.append(time.monotonic())
+                    with ts_lock:
+                        token_ts.extend(local)
+                except OSError:
+                    time.sleep(0.2)
+
+        threads = [threading.Thread(target=user_loop, args=(u,),
+                                    daemon=True) for u in range(users)]
+        for t in threads:
+            t.start()
+        time.sleep(warm_s)
+        t0 = time.monotonic()
+        time.sleep(meas_s)
+        t1 = time.monotonic()
+        stop.set()
+        with urllib.request.urlopen(base + "/admin/stats", timeout=10) \
+                as r:
+            stats = json.loads(r.read())
+        time.sleep(0.5)
+        with ts_lock:
+            n_tok = sum(1 for ts in token_ts if t0 <= ts <= t1)
+        value = n_tok / (t1 - t0)
+        out = {
+            "metric": "agg_tokens_per_sec",
+            "value": round(value, 2),
+            "unit": "tokens/s",
+            "n_gpus": args.gpus,
+            "steps": args.steps,
+            "warmup": args.warmup,
+            "ms_per_step": round((t1 - t0) * 1e3 / max(1, n_tok // users),
+                                 3),
+            "higher_is_better": True,
+            "scaling": "weak",
+            "vs_baseline": None,
+            "dtype": "bf16" if has_gpu else "fp32",
+            "data": "synthetic prompts, random-init weights",
+            "config": {
+                "model": model,
+                "mode": "full-stack (HTTP -> dispatcher -> UDS workers)",
+                "global_batch": users,
+                "users_per_gpu": users // max(1, args.gpus),
+                "seq_len": prompt_len,
+                "max_ctx": args.max_ctx,
+                "parallelism": f"dp{args.gpus} via native dispatcher",
+                "p50_queue_wait_ms":
+                    stats.get("queue_wait", {}).get("p50_ms"),
+                "p99_queue_wait_ms":
+                    stats.get("queue_wait", {}).get("p99_ms"),
+                "requests_processed": stats.get("processed"),
+                "measure_s": round(t1 - t0, 1),
+            },
+        }
+        print(json.dumps(out), flush=True)
+    finally:
+        try:
+            os.killpg(node.pid, 15)
+        except (ProcessLookupError, PermissionError):
+            node.terminate()
+        node.wait(timeout=30)
 
 
 def main():
@@ -40,7 +165,15 @@ def main():
     ap.add_argument("--model", type=str, default=None)
     ap.add_argument("--max-ctx", type=int, default=4096)
     ap.add_argument("--prefill-chunk", type=int, default=4096)
+    ap.add_argument("--stack", action="store_true",
+                    help="benchmark through the full dispatcher stack "
+                         "(HTTP + scheduler + UDS workers) instead of "
+                         "the engine-only kernel metric")
     args = ap.parse_args()
+
+    if args.stack:
+        run_stack(args)
+        return
 
     from ollamamq_amd.models import LlamaModel, PRESETS
     from ollamamq_amd.engine import LlamaEngine, PagedKVCache, GenParams

@@ -57,3 +57,21 @@ def test_bench_torchrun_two_ranks():
     n_json = sum(1 for l in r.stdout.splitlines()
                  if l.strip().startswith("{") and '"metric"' in l)
     assert n_json == 1, r.stdout
+
+
+def test_bench_stack_mode():
+    """--stack measures through the full dispatcher + UDS worker path and
+    reports the dispatcher's own queue-wait percentiles (VERDICT r01 #3)."""
+    r = subprocess.run(
+        [sys.executable, BENCH, "--stack", "--gpus", "1",
+         "--steps", "36", "--warmup", "3"],
+        cwd=REPO, capture_output=True, text=True, timeout=600)
+    assert r.returncode == 0, (r.stdout[-1000:], r.stderr[-2000:])
+    out = last_json_line(r.stdout)
+    for k in REQUIRED:
+        assert k in out, f"missing field {k}"
+    assert out["value"] > 0
+    cfg = out["config"]
+    assert "full-stack" in cfg["mode"]
+    assert cfg["p50_queue_wait_ms"] is not None
+    assert cfg["requests_processed"] > 0
