@@ -59,12 +59,58 @@ bool load_config(const std::string& path, AppConfig* out, std::string* err) {
         }
     };
 
+    // The parser accepts exactly the documented 3-section subset
+    // (appconf.yaml.example).  Anything it cannot represent is a HARD
+    // error with a line number — a config that silently parses to
+    // something else is worse than one that refuses to load (ADVICE r01:
+    // "parses the shipped examples; any real-world appconf fails
+    // silently or oddly").
+    auto unsupported = [&](int ln, const std::string& what,
+                           std::string* e) {
+        if (e)
+            *e = "line " + std::to_string(ln) + ": unsupported YAML (" +
+                 what + ") — appconf.yaml supports plain 2-level " +
+                 "mappings, '- ' lists and quoted scalars only " +
+                 "(see appconf.yaml.example)";
+        return false;
+    };
+
     int lineno = 0;
     while (std::getline(f, line)) {
         lineno++;
         line = decomment(line);
         const std::string t = strip(line);
         if (t.empty()) continue;
+        if (t == "---" || t == "...") continue;       // document markers
+        // constructs the subset cannot represent -> refuse loudly
+        if (t[0] == '&' || t[0] == '*')
+            return unsupported(lineno, "anchor/alias", err);
+        if (t[0] == '?')
+            return unsupported(lineno, "complex mapping key", err);
+        if (t.back() == '|' || t.back() == '>')
+            return unsupported(lineno, "block scalar", err);
+        {   // flow collections / anchors as values — except the empty
+            // flow collections, which mean "no entries" and are harmless
+            std::string item = t;
+            if (item.rfind("- ", 0) == 0) item = strip(item.substr(2));
+            if (!item.empty() && (item[0] == '&' || item[0] == '*' ||
+                                  item[0] == '{' ||
+                                  (item[0] == '[' && item != "[]")))
+                return unsupported(lineno,
+                                   "anchor/alias/flow list item", err);
+            auto c = item.find(':');
+            std::string v = c == std::string::npos
+                                ? "" : strip(item.substr(c + 1));
+            if ((!v.empty() && (v[0] == '{' || v[0] == '[')) &&
+                v != "[]" && v != "{}")
+                return unsupported(lineno, "flow collection", err);
+            if (!v.empty() && (v[0] == '&' || v[0] == '*'))
+                return unsupported(lineno, "anchor/alias value", err);
+        }
+        if (t.rfind("<<", 0) == 0)
+            return unsupported(lineno, "merge key", err);
+        if (line[0] == '\t')
+            return unsupported(lineno, "tab indentation", err);
         const size_t indent = line.find_first_not_of(" \t");
 
         if (indent == 0 && t.back() == ':') {

@@ -97,3 +97,42 @@ def test_url_normalization():
         "http://10.0.0.1:11434"
     assert d.normalize_backend_url("http://x/") == "http://x"
     assert d.normalize_backend_url(" https://y ") == "https://y"
+
+
+def test_adversarial_yaml_refused(tmp_path):
+    """Constructs outside the documented subset must refuse to load with
+    a line-numbered error, never silently misparse."""
+    bad = [
+        ("anchors", "backends:\n  - &b http://x\n"),
+        ("alias value", "models:\n  - name: *b\n"),
+        ("block scalar", "settings:\n  host: |\n    multi\n    line\n"),
+        ("flow mapping", "models:\n  - {name: x, max_ctx: 1}\n"),
+        ("flow list", "backends: [http://a, http://b]\n"),
+        ("tabs", "settings:\n\tport: 1\n"),
+        ("merge key", "settings:\n  <<: *base\n"),
+    ]
+    for label, text in bad:
+        p = write(tmp_path, text)
+        with pytest.raises(RuntimeError) as ei:
+            d.load_config(p)
+        msg = str(ei.value)
+        assert "line " in msg and "unsupported" in msg, (label, msg)
+
+
+def test_benign_yaml_still_parses(tmp_path):
+    """Doc markers, comments, quotes, empty flow lists and unknown keys
+    stay accepted (serde_yaml ignores unknown fields; reference
+    config.rs:137-238)."""
+    p = write(tmp_path,
+              "---\n"
+              "# comment\n"
+              "backends:\n"
+              "  - \"http://a:1\"   # inline comment\n"
+              "settings:\n"
+              "  port: 1234\n"
+              "  future_knob: whatever\n"
+              "models: []\n"
+              "...\n")
+    cfg = d.load_config(p)
+    assert cfg["backends"] == ["http://a:1"]
+    assert cfg["settings"]["port"] == 1234
