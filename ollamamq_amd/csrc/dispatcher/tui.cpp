@@ -36,6 +36,7 @@ struct Snapshot {
         int64_t processed;
         std::string loaded;
         std::string avail;
+        std::vector<std::string> models;   // sorted available (cursor)
         std::string op;
     };
     struct U {
@@ -70,7 +71,9 @@ Snapshot capture(AppState& st) {
             for (const auto& m : b.available_models) {
                 if (!e.avail.empty()) e.avail += ",";
                 e.avail += m;
+                e.models.push_back(m);
             }
+            std::sort(e.models.begin(), e.models.end());
             auto op = st.control_ops.find(i);
             if (op != st.control_ops.end())
                 e.op = (op->second.action == ControlAction::Load ? "load "
@@ -153,6 +156,12 @@ void run_tui(Server& server) {
     int focus = 1;            // 0 = Backends panel, 1 = Users panel (Tab)
     bool show_all = false;    // 'a': show available models per backend
     std::set<int> expanded;   // Space/Enter: expand one backend's models
+    std::map<int, int> mcur;  // per-backend model cursor (reference
+                              // tui.rs:607-636): Tab walks the expanded
+                              // backend's sorted models; L/U act on the
+                              // highlighted model directly
+    std::set<int> unfolded;   // expanded lists fold to 5 until the
+                              // cursor walks past (tui.rs:631-634)
     std::string input;        // typed model name for L/U
     char input_mode = 0;      // 'L' or 'U' when typing
     bool help = false;
@@ -205,8 +214,27 @@ void run_tui(Server& server) {
                 << "act:" << b.active << " done:" << b.processed << " "
                 << pad(b.loaded, 24)
                 << (b.op.empty() ? "" : " [" + b.op + "]") << "\r\n";
-            if ((show_all || expanded.count(i)) && !b.avail.empty())
+            if (expanded.count(i) && !b.models.empty()) {
+                // per-model rows with the cursor marker; folded to 5
+                // unless 'a' or the cursor forced the full list
+                const bool full = show_all || unfolded.count(i);
+                const size_t lim = full ? b.models.size()
+                                        : std::min<size_t>(5,
+                                                           b.models.size());
+                auto cit = mcur.find(i);
+                for (size_t mi = 0; mi < lim; mi++) {
+                    const bool cur = cit != mcur.end() &&
+                                     (int)mi == cit->second;
+                    out << (cur ? "    \x1b[7m> " : "      ")
+                        << pad(b.models[mi], 40)
+                        << (cur ? "\x1b[0m" : "") << "\r\n";
+                }
+                if (lim < b.models.size())
+                    out << "      … +"
+                        << (b.models.size() - lim) << " more (Tab)\r\n";
+            } else if (show_all && !b.avail.empty()) {
                 out << "    available: " << pad(b.avail, 80) << "\r\n";
+            }
         }
         out << "\x1b[7m"
             << pad(focus == 1 ? " Users [focused] (j/k, p VIP, b Boost, "
@@ -279,9 +307,27 @@ void run_tui(Server& server) {
                     case 'q':
                     case 27:
                         return;
+                    case '\t':
+                        // Tab advances the model cursor of the selected
+                        // expanded backend; falls through to panel cycle
+                        // when there is none (reference tui.rs:607-636)
+                        if (focus == 0 && expanded.count(sel_backend) &&
+                            sel_backend < (int)s.backends.size() &&
+                            !s.backends[sel_backend].models.empty()) {
+                            const int len =
+                                (int)s.backends[sel_backend].models.size();
+                            auto it = mcur.find(sel_backend);
+                            const int next =
+                                it == mcur.end() ? 0
+                                                 : (it->second + 1) % len;
+                            mcur[sel_backend] = next;
+                            if (next >= 5) unfolded.insert(sel_backend);
+                            break;
+                        }
+                        focus ^= 1;
+                        break;
                     case 'h':
                     case 'l':
-                    case '\t':
                         focus ^= 1;
                         break;
                     case ' ':
@@ -313,10 +359,37 @@ void run_tui(Server& server) {
                             sel_backend = std::max(sel_backend - 1, 0);
                         break;
                     case 'L':
-                    case 'U':
+                    case 'U': {
+                        // cursor-model direct control when the selected
+                        // backend is expanded with a highlighted model
+                        // (reference tui.rs:345-387); typed-name input
+                        // mode otherwise
+                        auto it = mcur.find(sel_backend);
+                        if (focus == 0 && expanded.count(sel_backend) &&
+                            it != mcur.end() &&
+                            sel_backend < (int)s.backends.size() &&
+                            it->second <
+                                (int)s.backends[sel_backend].models
+                                    .size()) {
+                            ControlRequest req;
+                            req.action = c == 'L' ? ControlAction::Load
+                                                  : ControlAction::Unload;
+                            req.model =
+                                s.backends[sel_backend].models[it->second];
+                            req.backend_idx = (size_t)sel_backend;
+                            auto r = start_model_control(st, req);
+                            status_msg =
+                                r.http_status == 202
+                                    ? (std::string(c == 'L' ? "load "
+                                                            : "unload ") +
+                                       req.model + " accepted")
+                                    : r.body.get_str("error");
+                            break;
+                        }
                         input_mode = c;
                         input.clear();
                         break;
+                    }
                     case 'r':
                         reload_model_config(st, server.config_path);
                         status_msg = "config reloaded";

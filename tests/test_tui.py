@@ -53,3 +53,66 @@ def test_tui_renders_and_quits(tmp_path):
     text = out.decode(errors="replace")
     assert "Backends" in text and "Users" in text and "Logs" in text
     assert "Keys" in text, "help overlay never rendered"
+
+
+def test_tui_model_cursor_direct_load(tmp_path):
+    """Reference tui.rs:607-636,345-387: Tab walks the expanded backend's
+    sorted model list and L loads the highlighted model directly (no
+    typed-name input mode)."""
+    sys.path.insert(0, os.path.join(REPO, "tests"))
+    from mocks import MockFleet
+    fleet = MockFleet()
+    try:
+        master, slave = pty.openpty()
+        p = subprocess.Popen(
+            [BIN, "-p", "0", "-o", fleet.ollama_url,
+             "-c", os.path.join(str(tmp_path), "absent.yaml")],
+            stdin=slave, stdout=slave, stderr=subprocess.DEVNULL,
+            cwd=str(tmp_path))
+        os.close(slave)
+        out = b""
+        deadline = time.time() + 40
+        stage = 0
+        try:
+            while time.time() < deadline:
+                try:
+                    chunk = os.read(master, 65536)
+                except OSError:
+                    break
+                out += chunk
+                # wait for the probe to mark the mock backend online
+                if stage == 0 and b"\xe2\x97\x8f" in out:
+                    # backend online; expand + cursor + L
+                    os.write(master, b"\t")    # focus Backends
+                    time.sleep(0.3)
+                    os.write(master, b" ")     # expand selected backend
+                    time.sleep(0.5)
+                    os.write(master, b"\t")    # model cursor -> first
+                    time.sleep(0.3)
+                    os.write(master, b"L")     # direct load
+                    stage = 1
+                    t_fire = time.time()
+                if stage == 1 and (b"accepted" in out
+                                   or time.time() - t_fire > 8):
+                    os.write(master, b"q")
+                    stage = 2
+                if stage == 2 and p.poll() is not None:
+                    break
+            assert stage >= 2, f"never fired cursor-load: {out[-600:]!r}"
+            p.wait(timeout=10)
+        finally:
+            os.close(master)
+            if p.poll() is None:
+                p.terminate()
+        text = out.decode(errors="replace")
+        assert "accepted" in text, text[-800:]
+        # the direct load reached the mock backend's generate endpoint
+        deadline = time.time() + 10
+        while time.time() < deadline:
+            if fleet.recorder.of(path="/api/generate"):
+                break
+            time.sleep(0.3)
+        assert fleet.recorder.of(path="/api/generate"), \
+            "cursor-load never hit the backend"
+    finally:
+        fleet.stop()
