@@ -44,10 +44,16 @@ VERSION = "0.1.0-ollamamq-amd"
 
 class Worker:
     def __init__(self, gpu: int, max_batch: int = 32,
-                 default_ctx: int = 4096):
+                 default_ctx: int = 4096, models: Optional[list] = None):
         self.gpu = gpu
         self.max_batch = max_batch
         self.default_ctx = default_ctx
+        # optional fleet-sharding restriction: this worker only advertises
+        # (and loads) these models, so the dispatcher's model routing
+        # (smart/fuzzy match against available_models) steers each request
+        # to the workers that carry its model (reference
+        # src/dispatcher.rs:599-620; BASELINE.json config 5 mixed fleet)
+        self.model_filter = list(models) if models else None
         self.device = f"cuda:{gpu}" if torch.cuda.is_available() else "cpu"
         self.dtype = (torch.bfloat16 if torch.cuda.is_available()
                       else torch.float32)
@@ -61,6 +67,8 @@ class Worker:
 
     # ---------------------------------------------------------- control
     def available_models(self):
+        if self.model_filter is not None:
+            return list(self.model_filter)
         if self.device == "cpu":
             return ["tiny", "tiny-cpu"]
         return [n for n in PRESETS if n != "tiny-cpu"]
@@ -121,12 +129,17 @@ class Worker:
             for name in self.engines:
                 if requested.split(":")[0].lower() == name.lower():
                     return name
-            # auto-load when a known preset is requested
-            if requested in PRESETS:
+            # auto-load when a known preset is requested (within this
+            # worker's fleet-shard restriction, if any)
+            allowed = (set(self.model_filter)
+                       if self.model_filter is not None else None)
+            if requested in PRESETS and \
+                    (allowed is None or requested in allowed):
                 return requested
             base = requested.split(":")[0].lower()
             for name in PRESETS:
-                if name.lower() == base:
+                if name.lower() == base and \
+                        (allowed is None or name in allowed):
                     return name
         return None
 
@@ -548,9 +561,13 @@ def main():
                     help="preload this model")
     ap.add_argument("--max-ctx", type=int, default=4096)
     ap.add_argument("--max-batch", type=int, default=32)
+    ap.add_argument("--models", type=str, default=None,
+                    help="comma list restricting this worker's advertised "
+                         "models (fleet sharding for multi-model routing)")
     args = ap.parse_args()
 
-    w = Worker(args.gpu, max_batch=args.max_batch, default_ctx=args.max_ctx)
+    w = Worker(args.gpu, max_batch=args.max_batch, default_ctx=args.max_ctx,
+               models=args.models.split(",") if args.models else None)
     if args.model:
         err = w.load(args.model, args.max_ctx)
         if err:

@@ -54,6 +54,15 @@ def main():
     ap.add_argument("-c", "--model-config", type=str, default="appconf.yaml")
     ap.add_argument("--extra-backends", type=str, default="",
                     help="comma list of external HTTP backends to add")
+    ap.add_argument("--workers", type=str, default="",
+                    help="heterogeneous fleet spec overriding "
+                         "--gpus/--tp/--model: comma list of "
+                         "MODEL[*COUNT][/tpN], e.g. "
+                         "'llama3-8b*4,llama3-70b/tp4' = four 1-GPU 8B "
+                         "backends + one TP4 70B backend (BASELINE.json "
+                         "config 5).  Each worker advertises ONLY its "
+                         "model, so /api/tags routing steers requests "
+                         "(reference src/dispatcher.rs:599-620).")
     args = ap.parse_args()
 
     if not os.path.exists(SERVER_BIN):
@@ -65,32 +74,54 @@ def main():
     env = dict(os.environ)
     env.setdefault("MASTER_ADDR", "127.0.0.1")
 
-    assert args.gpus % args.tp == 0, "--gpus must be a multiple of --tp"
-    n_backends = args.gpus // args.tp
-    for b in range(n_backends):
+    # Build the backend plan: homogeneous (--gpus/--tp/--model) or a
+    # heterogeneous --workers spec; each entry = (model, tp_degree).
+    plan = []
+    if args.workers:
+        for item in args.workers.split(","):
+            item = item.strip()
+            if not item:
+                continue
+            count, tp = 1, 1
+            if "/tp" in item:
+                item, tp_s = item.rsplit("/tp", 1)
+                tp = int(tp_s)
+            if "*" in item:
+                item, cnt_s = item.rsplit("*", 1)
+                count = int(cnt_s)
+            plan.extend([(item.strip(), tp)] * count)
+    else:
+        assert args.gpus % args.tp == 0, "--gpus must be a multiple of --tp"
+        plan = [(args.model, args.tp)] * (args.gpus // args.tp)
+
+    gpu_base = 0
+    for b, (model, tp) in enumerate(plan):
         sock = os.path.join(args.sock_dir, f"omq_worker{b}.sock")
         try:
             os.unlink(sock)
         except FileNotFoundError:
             pass
         socks.append(sock)
-        if args.tp == 1:
+        if tp == 1:
             cmd = [sys.executable, "-m", "ollamamq_amd.engine.worker",
-                   "--socket", sock, "--gpu", str(b),
-                   "--model", args.model, "--max-ctx", str(args.max_ctx),
+                   "--socket", sock, "--gpu", str(gpu_base),
+                   "--model", model, "--max-ctx", str(args.max_ctx),
                    "--max-batch", str(args.max_batch)]
+            if args.workers:
+                cmd += ["--models", model]
             procs.append(subprocess.Popen(cmd, env=env))
         else:
-            gpus = range(b * args.tp, (b + 1) * args.tp)
             cmd = [sys.executable, "-m", "torch.distributed.run",
-                   "--nnodes=1", f"--nproc-per-node={args.tp}",
+                   "--nnodes=1", f"--nproc-per-node={tp}",
                    "--master-addr", "127.0.0.1",
                    "--master-port", str(29600 + b),
                    "-m", "ollamamq_amd.engine.tp_worker",
-                   "--socket", sock, "--gpu-base", str(min(gpus)),
-                   "--model", args.model, "--max-ctx", str(args.max_ctx),
+                   "--socket", sock, "--gpu-base", str(gpu_base),
+                   "--model", model, "--max-ctx", str(args.max_ctx),
                    "--max-batch", str(args.max_batch)]
             procs.append(subprocess.Popen(cmd, env=env))
+        gpu_base += tp
+    n_backends = len(plan)
 
     print(f"waiting for {n_backends} worker socket(s)...", flush=True)
     if not wait_sockets(socks):

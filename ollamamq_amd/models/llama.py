@@ -113,6 +113,14 @@ PRESETS = {
         head_dim=128, ffn=1024, vocab=512, max_ctx=512, rope_theta=10000.0,
         sliding_window=96,
     ),
+    # Disjoint NAME from every other preset: multi-model routing tests
+    # need a model that neither smart- nor fuzzy-matches "tiny*"
+    # (fuzzy matching is substring-both-ways, reference
+    # src/dispatcher.rs:381-393, so "tiny" routes to "tiny-qwen" too).
+    "nano": LlamaConfig(
+        name="nano", n_layers=2, hidden=512, n_heads=4, n_kv_heads=2,
+        head_dim=128, ffn=1024, vocab=512, max_ctx=512, rope_theta=10000.0,
+    ),
 }
 
 

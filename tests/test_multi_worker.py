@@ -97,3 +97,97 @@ def test_load_spreads_across_workers(node):
     assert sum(counts) >= 24
     # least-connections must have used BOTH workers substantially
     assert min(counts) >= 5, counts
+
+
+# --------------------------------------------------------------- config 5
+@pytest.fixture(scope="module")
+def hetero_node(tmp_path_factory):
+    """Mixed-fleet rehearsal (BASELINE.json config 5 on CPU): two workers
+    carrying ONLY tiny + one carrying ONLY tiny-qwen; requests must route
+    by model (reference src/dispatcher.rs:599-620) and fair-share across
+    users."""
+    tmp = tmp_path_factory.mktemp("hetero")
+    spec = [("tiny",), ("tiny",), ("nano",)]
+    socks, workers = [], []
+    for i, (model,) in enumerate(spec):
+        sock = os.path.join(str(tmp), f"h{i}.sock")
+        socks.append(sock)
+        workers.append(subprocess.Popen(
+            [sys.executable, "-m", "ollamamq_amd.engine.worker",
+             "--socket", sock, "--model", model, "--models", model,
+             "--max-ctx", "256", "--max-batch", "4"],
+            cwd=REPO, stdout=subprocess.DEVNULL,
+            stderr=subprocess.STDOUT))
+    for s in socks:
+        assert _wait_socket(s)
+    server = subprocess.Popen(
+        [BIN, "--no-tui", "-p", "0", "-w", ",".join(socks),
+         "--probe-interval-ms", "500",
+         "-c", os.path.join(str(tmp), "absent.yaml")],
+        stderr=subprocess.PIPE, cwd=str(tmp), text=True)
+    line = server.stderr.readline()
+    port = int(line.rsplit(":", 1)[1].split()[0])
+    base = f"http://127.0.0.1:{port}"
+    deadline = time.time() + 30
+    while time.time() < deadline:
+        try:
+            r = httpx.get(base + "/admin/models", timeout=2.0).json()
+            if all(b["online"] for b in r["backends"]) and \
+                    len(r["backends"]) == 3:
+                break
+        except Exception:
+            pass
+        time.sleep(0.2)
+    yield base
+    server.terminate()
+    for w in workers:
+        w.terminate()
+
+
+def test_hetero_advertises_per_worker_models(hetero_node):
+    r = httpx.get(hetero_node + "/admin/models").json()
+    avail = [set(b["available_models"]) for b in r["backends"]]
+    assert avail[0] == {"tiny"} and avail[1] == {"tiny"}
+    assert avail[2] == {"nano"}
+
+
+def test_hetero_routes_by_model(hetero_node):
+    """nano requests must ONLY land on worker 2; tiny requests only
+    on workers 0/1, fair-shared across users."""
+    def gen(model, user):
+        return httpx.post(
+            hetero_node + "/api/generate",
+            json={"model": model, "prompt": "hi",
+                  "options": {"num_predict": 2}, "stream": False},
+            headers={"X-User-ID": user}, timeout=120)
+
+    before = {b["url"]: b["processed_count"]
+              for b in httpx.get(hetero_node + "/admin/stats")
+              .json()["backends"]}
+    import concurrent.futures as cf
+    with cf.ThreadPoolExecutor(12) as ex:
+        futs = []
+        for i in range(8):
+            futs.append(ex.submit(gen, "tiny", f"mixu{i % 4}"))
+        for i in range(4):
+            futs.append(ex.submit(gen, "nano", f"mixq{i % 2}"))
+        for f in futs:
+            r = f.result()
+            assert r.status_code == 200, r.text
+            assert json.loads(r.text.splitlines()[-1])["done"] is True
+    after = httpx.get(hetero_node + "/admin/stats").json()["backends"]
+    delta = {b["url"]: b["processed_count"] - before[b["url"]]
+             for b in after}
+    d = list(delta.values())
+    # workers 0/1 carried the 8 tiny requests; worker 2 the 4 qwen ones
+    assert d[2] == 4, delta
+    assert d[0] + d[1] == 8, delta
+    assert d[0] >= 2 and d[1] >= 2, f"least-conn spread broken: {delta}"
+
+
+def test_hetero_unknown_model_parks_then_503(hetero_node):
+    """A model nobody carries must never dispatch (stuck-timeout 503 is
+    covered in dispatch tests; here assert immediate non-routing)."""
+    r = httpx.get(hetero_node + "/admin/models").json()
+    names = {m for b in r["backends"] for m in b["available_models"]}
+    assert "gemma" not in names
