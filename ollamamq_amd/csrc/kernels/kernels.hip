@@ -758,3 +758,135 @@ extern "C" int row_sumsq_bf16(void* sq, const void* x, int M, int H,
     k_row_sumsq<<<M, 256, 0, stream>>>((const bf16*)x, (float*)sq, H);
     return (int)hipGetLastError();
 }
+
+// ---------------------------------------------------------------------------
+// Stochastic sampler: exact temperature sampling via the Gumbel-max
+// trick — sampling token v with prob softmax(l/T)[v] is EXACTLY
+// argmax_v(l[v]/T + g_v) with iid Gumbel noise g = -log(-log(u)).  The
+// noise is counter-based (splitmix64 of (seed[row], ctr[row], v)), so a
+// hipGraph replay draws fresh randomness with no host RNG round-trip
+// and a request with a fixed seed reproduces its tokens regardless of
+// batch composition (ctr = the sequence's own length that step).
+// Rows with temps[row] <= 0 take the pure argmax path (greedy), so ONE
+// captured graph serves mixed greedy/sampled batches.
+// Split-V two-stage exactly like argmax (the noise is recomputed from
+// (row, v) so partials stay exact).  noise_override (tests): u = that
+// buffer instead of the hash.
+// ---------------------------------------------------------------------------
+DEV float u01_hash(uint64_t key) {
+    key ^= key >> 33; key *= 0xff51afd7ed558ccdULL;
+    key ^= key >> 33; key *= 0xc4ceb9fe1a85ec53ULL;
+    key ^= key >> 33;
+    return ((float)(key >> 40) + 0.5f) * (1.0f / 16777216.0f);
+}
+
+DEV float gumbel_of(uint64_t seed, uint64_t ctr, int v,
+                    const float* noise, int64_t noff) {
+    const float u = noise ? noise[noff + v]
+                          : u01_hash((seed * 0x9E3779B97F4A7C15ULL)
+                                     ^ (ctr * 0xD1B54A32D192ED03ULL)
+                                     ^ (uint64_t)v);
+    return -__logf(-__logf(u));
+}
+
+template <bool PART>
+__global__ __launch_bounds__(256) void k_gumbel_part(
+    int* __restrict__ out, float* __restrict__ pb, int* __restrict__ pi,
+    const bf16* __restrict__ logits, const float* __restrict__ temps,
+    const unsigned long long* __restrict__ seeds,
+    const int* __restrict__ ctrs, const float* __restrict__ noise,
+    int V, int seg)
+{
+    const int row = blockIdx.x, slice = PART ? blockIdx.y : 0;
+    const int lo = slice * seg, hi = min(V, lo + seg);
+    const bf16* lr = logits + (int64_t)row * V;
+    const float T = temps ? temps[row] : 0.f;
+    const bool greedy = T <= 0.f;
+    const float invT = greedy ? 1.f : 1.f / T;
+    const uint64_t sd = seeds ? seeds[row] : 1234567ULL;
+    const uint64_t ct = ctrs ? (uint64_t)(unsigned)ctrs[row] : 0ULL;
+    const int64_t noff = (int64_t)row * V;
+    float best = -3.0e38f;
+    int bidx = 0;
+    for (int i = lo + threadIdx.x * 8; i < hi; i += blockDim.x * 8) {
+        const int n = min(8, hi - i);
+        float v[8];
+        if (n == 8) load8f(lr + i, v);
+        else for (int e = 0; e < n; e++) v[e] = bf2f(lr[i + e]);
+        for (int e = 0; e < n; e++) {
+            float s = v[e] * invT;
+            if (!greedy)
+                s += gumbel_of(sd, ct, i + e, noise, noff);
+            if (s > best || (s == best && i + e < bidx)) {
+                best = s; bidx = i + e;
+            }
+        }
+    }
+    #pragma unroll
+    for (int off = 32; off > 0; off >>= 1) {
+        const float ob = __shfl_down(best, off, 64);
+        const int oi = __shfl_down(bidx, off, 64);
+        if (ob > best || (ob == best && oi < bidx)) { best = ob; bidx = oi; }
+    }
+    __shared__ float wb[4];
+    __shared__ int wi[4];
+    const int wid = threadIdx.x >> 6, lane = threadIdx.x & 63;
+    if (lane == 0) { wb[wid] = best; wi[wid] = bidx; }
+    __syncthreads();
+    if (threadIdx.x == 0) {
+        for (int wv = 1; wv < 4; wv++)
+            if (wb[wv] > best || (wb[wv] == best && wi[wv] < bidx)) {
+                best = wb[wv]; bidx = wi[wv];
+            }
+        if (PART) {
+            pb[row * gridDim.y + slice] = best;
+            pi[row * gridDim.y + slice] = bidx;
+        } else {
+            out[row] = bidx;
+        }
+    }
+}
+
+__global__ __launch_bounds__(64) void k_gumbel_comb(
+    int* __restrict__ out, const float* __restrict__ pb,
+    const int* __restrict__ pi, int sp)
+{
+    const int row = blockIdx.x;
+    float best = -3.0e38f;
+    int bidx = 0;
+    for (int s = threadIdx.x; s < sp; s += 64) {
+        const float v = pb[row * sp + s];
+        const int ix = pi[row * sp + s];
+        if (v > best || (v == best && ix < bidx)) { best = v; bidx = ix; }
+    }
+    #pragma unroll
+    for (int off = 32; off > 0; off >>= 1) {
+        const float ob = __shfl_down(best, off, 64);
+        const int oi = __shfl_down(bidx, off, 64);
+        if (ob > best || (ob == best && oi < bidx)) { best = ob; bidx = oi; }
+    }
+    if (threadIdx.x == 0) out[row] = bidx;
+}
+
+extern "C" int sample_gumbel_bf16(
+    void* out, const void* logits, const void* temps, const void* seeds,
+    const void* ctrs, const void* noise, int B, int V,
+    void* pb, void* pi, int sp, hipStream_t stream)
+{
+    if (sp > 1) {
+        const int seg = (V + sp - 1) / sp;
+        dim3 grid(B, sp);
+        k_gumbel_part<true><<<grid, 256, 0, stream>>>(
+            nullptr, (float*)pb, (int*)pi, (const bf16*)logits,
+            (const float*)temps, (const unsigned long long*)seeds,
+            (const int*)ctrs, (const float*)noise, V, seg);
+        k_gumbel_comb<<<B, 64, 0, stream>>>(
+            (int*)out, (const float*)pb, (const int*)pi, sp);
+    } else {
+        k_gumbel_part<false><<<B, 256, 0, stream>>>(
+            (int*)out, nullptr, nullptr, (const bf16*)logits,
+            (const float*)temps, (const unsigned long long*)seeds,
+            (const int*)ctrs, (const float*)noise, V, V);
+    }
+    return (int)hipGetLastError();
+}

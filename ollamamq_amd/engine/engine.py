@@ -108,6 +108,13 @@ class LlamaEngine:
             prompt = [0]
         sid = next(self._ids)
         seq = Sequence(sid, list(prompt), params, on_token)
+        # per-sequence noise seed for the in-graph Gumbel sampler:
+        # request seed (reproducible regardless of batch composition,
+        # the counter is the sequence's own length) or a per-sequence
+        # engine default (deterministic across TP ranks: sid matches)
+        seq.noise_seed = (int(params.seed) if params.seed is not None
+                          else (0x5EED0000 ^ (sid * 0x9E3779B1)) &
+                               0x7FFFFFFFFFFFFFFF)
         self.seqs[sid] = seq
         self.waiting.append(seq)
         return sid
@@ -365,10 +372,25 @@ class LlamaEngine:
                 if s.on_token:
                     s.on_token(int(tok), False)
 
-    def _graph_entry(self, B: int):
-        """Capture (once per batch size) a full decode forward as a hipGraph
-        reading its inputs from static device buffers."""
-        entry = self._graphs.get(B)
+    def _samp_class(self, seqs) -> int:
+        """0 = all greedy (argmax tail), 1 = temperature-only mix (Gumbel
+        tail — exact, still one captured graph), 2 = top-k/top-p present
+        (host torch sampling; pipelined replay not captured yet)."""
+        k = 0
+        for s in seqs:
+            p = s.params
+            if p.temperature > 0:
+                if (p.top_k or 0) > 0 or (p.top_p if p.top_p else 1.0) \
+                        < 1.0:
+                    return 2
+                k = 1
+        return k
+
+    def _graph_entry(self, B: int, klass: int = 0):
+        """Capture (once per (batch size, sampling class)) a full decode
+        forward as a hipGraph reading its inputs from static device
+        buffers."""
+        entry = self._graphs.get((B, klass))
         if entry is not None:
             return entry
         dev = self.dev
@@ -377,6 +399,8 @@ class LlamaEngine:
             "pos": torch.zeros(B, dtype=torch.int32, device=dev),
             "slot": torch.zeros(B, dtype=torch.int32, device=dev),
             "lens": torch.zeros(B, dtype=torch.int32, device=dev),
+            "temps": torch.zeros(B, dtype=torch.float32, device=dev),
+            "seeds": torch.zeros(B, dtype=torch.int64, device=dev),
         }
         meta = AttnMeta(
             mode="decode", slot_ids=bufs["slot"], seq_lens=bufs["lens"],
@@ -385,11 +409,11 @@ class LlamaEngine:
             max_q=1, max_kv=self.kv.max_ctx,
             window=getattr(self.model.cfg, "sliding_window", 0))
         entry = {"bufs": bufs, "meta": meta, "graph": None, "logits": None,
-                 "out": None, "ring_i": 0,
+                 "out": None, "ring_i": 0, "klass": klass,
                  "pinned": [torch.empty(B, dtype=torch.int64,
                                         pin_memory=True) for _ in range(2)],
                  "events": [torch.cuda.Event() for _ in range(2)]}
-        self._graphs[B] = entry
+        self._graphs[(B, klass)] = entry
         return entry
 
     def _fill_bufs(self, entry, seqs, token_list, pos_list):
@@ -402,6 +426,13 @@ class LlamaEngine:
         bufs["pos"].copy_(staged[1])
         bufs["slot"].copy_(staged[2])
         bufs["lens"].copy_(staged[3])
+        if entry["klass"] == 1:
+            bufs["temps"].copy_(torch.tensor(
+                [s.params.temperature for s in seqs],
+                dtype=torch.float32).to(self.dev, non_blocking=True))
+            bufs["seeds"].copy_(torch.tensor(
+                [getattr(s, "noise_seed", 1234) for s in seqs],
+                dtype=torch.int64).to(self.dev, non_blocking=True))
 
     def _graph_replay(self, entry):
         bufs, meta = entry["bufs"], entry["meta"]
@@ -428,15 +459,26 @@ class LlamaEngine:
                     entry["logits"] = self.model.forward(
                         bufs["tok"], bufs["pos"], self.kv, bufs["slot"],
                         meta)
-                    # self-advancing tail: greedy-sample in-graph and
-                    # stage the NEXT step's inputs on-device, so a steady
-                    # decode batch replays back-to-back with no host
-                    # round-trip.  The non-pipelined/stochastic path
-                    # simply overwrites the bufs next step and samples
-                    # logits itself.
-                    toks = ops.sample(entry["logits"], 0.0, 0, 1.0, None)
+                    # self-advancing tail: sample in-graph and stage the
+                    # NEXT step's inputs on-device, so a steady decode
+                    # batch replays back-to-back with no host round-trip.
+                    # Class 0 = greedy argmax; class 1 = exact Gumbel-max
+                    # temperature sampling (kernels.hip sample_gumbel:
+                    # greedy rows ride the same kernel with T<=0), noise
+                    # keyed by (per-seq seed, kv length) so replays draw
+                    # fresh randomness with no host RNG.
+                    if entry["klass"] == 1:
+                        from ..ops import hip as _hip
+                        toks32 = _hip.sample_gumbel(
+                            entry["logits"], bufs["temps"], bufs["seeds"],
+                            bufs["lens"])
+                        toks = toks32.long()
+                        bufs["tok"].copy_(toks32)
+                    else:
+                        toks = ops.sample(entry["logits"], 0.0, 0, 1.0,
+                                          None)
+                        bufs["tok"].copy_(toks)
                     entry["out"] = toks
-                    bufs["tok"].copy_(toks)
                     bufs["pos"] += 1
                     bufs["lens"] += 1
             finally:
@@ -446,8 +488,8 @@ class LlamaEngine:
         return entry["logits"]
 
     def _decode_forward_graphed(self, seqs, token_list, pos_list):
-        entry = self._graph_entry(len(seqs))
-        self._fill_bufs(entry, seqs, token_list, pos_list)
+        entry = self._graph_entry(len(seqs))   # class-0 graph: the host
+        self._fill_bufs(entry, seqs, token_list, pos_list)   # samples
         return self._graph_replay(entry)
 
     def warm_graphs(self, sizes=None):
@@ -483,7 +525,7 @@ class LlamaEngine:
     def _decode_step(self):
         seqs = self.running
         if (self.use_pipeline and self.use_graphs
-                and all(s.params.temperature <= 0 for s in seqs)
+                and self._samp_class(seqs) <= 1
                 and all(self.kv.seq_lens[s.slot] + 1 <= self.kv.max_ctx
                         for s in seqs)):
             self._decode_step_pipelined(seqs)
@@ -518,13 +560,13 @@ class LlamaEngine:
         tokens stream back asynchronously (pinned ring + event) and the
         PREVIOUS step's tokens are applied while this step runs.
         """
-        key = tuple(s.seq_id for s in seqs)
+        key = (tuple(s.seq_id for s in seqs), self._samp_class(seqs))
         if self._pipe is not None and self._pipe["key"] != key:
             self._drain_pipe()
         # host/device page bookkeeping for the token this replay appends
         for s in seqs:
             self.kv.ensure(s.slot, self.kv.seq_lens[s.slot] + 1)
-        entry = self._graph_entry(len(seqs))
+        entry = self._graph_entry(len(seqs), key[1])
         if self._pipe is None:
             # (re)sync device input buffers from host truth
             self._fill_bufs(entry, seqs,

@@ -60,6 +60,8 @@ def _try_load():
     lib.wstream_gu_bf16.argtypes = [vp, vp, vp, i, i, i, i64, i,
                                     vp, i, f, f, vp]
     lib.row_sumsq_bf16.argtypes = [vp, vp, i, i, vp]
+    lib.sample_gumbel_bf16.argtypes = [vp, vp, vp, vp, vp, vp, i, i,
+                                       vp, vp, i, vp]
     for fn in ("rmsnorm_residual_bf16", "rope_bf16", "kv_append_bf16",
                "paged_attn_bf16", "swiglu_bf16", "argmax_bf16",
                "decode_attn_bf16", "skinny_gemm_bf16",
@@ -617,6 +619,33 @@ def swiglu(gate_up):
     out = torch.empty((T, F), dtype=gate_up.dtype, device=gate_up.device)
     _check(_lib.swiglu_bf16(_p(out), _p(gate_up), T, F, _stream()),
            "swiglu")
+    return out
+
+
+def sample_gumbel(logits, temps, seeds, ctrs, noise=None, out=None):
+    """Exact temperature sampling (Gumbel-max): token ~ softmax(l/T) per
+    row; rows with temps<=0 are greedy argmax.  Counter-based noise from
+    (seeds[row], ctrs[row], v) — graph-replay-safe and reproducible per
+    request.  noise: optional [B,V] uniform(0,1) override (tests)."""
+    B, V = logits.shape
+    l = logits if logits.dtype == torch.bfloat16 else logits.bfloat16()
+    if out is None:
+        out = torch.empty(B, dtype=torch.int32, device=logits.device)
+    sp = min(32, max(1, 768 // max(1, B)))
+    pb = pi = ctypes.c_void_p(0)
+    if sp > 1:
+        key = ("gmb", B, sp, str(logits.device))
+        t = _scratch.get(key)
+        if t is None:
+            t = (torch.empty(B * sp, dtype=torch.float32,
+                             device=logits.device),
+                 torch.empty(B * sp, dtype=torch.int32,
+                             device=logits.device))
+            _scratch[key] = t
+        pb, pi = _p(t[0]), _p(t[1])
+    _check(_lib.sample_gumbel_bf16(
+        _p(out), _p(l), _p(temps), _p(seeds), _p(ctrs), _p(noise),
+        B, V, pb, pi, sp, _stream()), "sample_gumbel")
     return out
 
 

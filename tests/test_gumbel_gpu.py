@@ -1,0 +1,111 @@
+"""Gumbel-max sampler kernel + in-graph sampled decode (1 GPU).
+
+VERDICT r01 item 5: the stochastic sampler is a hand-written kernel and
+sampled workloads replay hipGraphs.  Exactness: with a supplied uniform
+noise buffer the kernel must equal the fp32 reference
+argmax(l/T - log(-log(u))) bit-for-bit in its argmax decision; without
+it, the counter-based noise must reproduce per (seed, position) and
+match the softmax distribution.
+"""
+import pytest
+import torch
+
+pytestmark = pytest.mark.gpu
+
+
+def _hip():
+    from ollamamq_amd.ops import hip
+    hip.require()
+    return hip
+
+
+def test_gumbel_matches_fp32_reference_with_noise():
+    hip = _hip()
+    g = torch.Generator().manual_seed(11)
+    for B, V in [(4, 128256), (32, 128256), (1, 512), (7, 32000)]:
+        logits = (torch.randn(B, V, generator=g) * 3).bfloat16().cuda()
+        temps = (torch.rand(B, generator=g) * 1.5 + 0.05).cuda()
+        noise = torch.rand(B, V, generator=g).clamp(1e-7, 1 - 1e-7).cuda()
+        seeds = torch.zeros(B, dtype=torch.int64, device="cuda")
+        ctrs = torch.zeros(B, dtype=torch.int32, device="cuda")
+        out = hip.sample_gumbel(logits, temps, seeds, ctrs, noise=noise)
+        ref = (logits.float() / temps[:, None]
+               - torch.log(-torch.log(noise.float()))).argmax(-1)
+        assert torch.equal(out.long().cpu(), ref.cpu())
+        # greedy rows (T<=0) ride the same kernel
+        temps0 = torch.zeros(B, device="cuda")
+        out0 = hip.sample_gumbel(logits, temps0, seeds, ctrs, noise=noise)
+        assert torch.equal(out0.long().cpu(),
+                           logits.float().argmax(-1).cpu())
+
+
+def test_gumbel_counter_noise_reproducible_and_distributed():
+    hip = _hip()
+    B, V = 8, 512
+    g = torch.Generator().manual_seed(5)
+    logits = (torch.randn(1, V, generator=g) * 2).bfloat16().cuda() \
+        .expand(B, V).contiguous()
+    temps = torch.full((B,), 1.0, device="cuda")
+    seeds = torch.arange(B, dtype=torch.int64, device="cuda") + 99
+    ctrs = torch.full((B,), 17, dtype=torch.int32, device="cuda")
+    a = hip.sample_gumbel(logits, temps, seeds, ctrs)
+    b = hip.sample_gumbel(logits, temps, seeds, ctrs)
+    assert torch.equal(a, b), "same (seed, ctr) must reproduce"
+    c = hip.sample_gumbel(logits, temps, seeds, ctrs + 1)
+    assert not torch.equal(a, c), "ctr change must redraw"
+    # distribution: many draws over a small vocab ~ softmax
+    n = 4000
+    l1 = (torch.randn(1, 64, generator=g) * 1.5).bfloat16().cuda()
+    big = l1.expand(n, 64).contiguous()
+    t1 = torch.ones(n, device="cuda")
+    sd = torch.arange(n, dtype=torch.int64, device="cuda")
+    ct = torch.zeros(n, dtype=torch.int32, device="cuda")
+    draws = hip.sample_gumbel(big, t1, sd, ct).cpu()
+    emp = torch.bincount(draws.long(), minlength=64).float() / n
+    expect = torch.softmax(l1[0].float().cpu(), -1)
+    # loose L1 bound: 4000 draws over 64 bins
+    assert (emp - expect).abs().sum() < 0.25, \
+        (emp - expect).abs().sum()
+
+
+def test_sampled_decode_replays_graphs():
+    """Temperature-only sampled sequences keep the pipelined graph path
+    (engine._samp_class==1): graphs capture once and replay; tokens are
+    diverse and seeded requests reproduce across batch compositions."""
+    from ollamamq_amd.models import LlamaModel, PRESETS
+    from ollamamq_amd.engine import GenParams, LlamaEngine, PagedKVCache
+    cfg = PRESETS["tiny"]
+    model = LlamaModel(cfg, device="cuda:0", dtype=torch.bfloat16,
+                       seed=1234)
+    kv = PagedKVCache.for_model(cfg, n_pages=128, max_slots=6,
+                                max_ctx=256, device="cuda:0",
+                                dtype=torch.bfloat16)
+    eng = LlamaEngine(model, kv, max_batch=4)
+    prompts = [[1, 2, 3, 4, 5], [9, 8, 7, 6], [3, 3, 3]]
+    sids = [eng.submit(p, GenParams(max_tokens=12, temperature=0.9,
+                                    seed=100 + i))
+            for i, p in enumerate(prompts)]
+    seqs = [eng.seqs[s] for s in sids]
+    for _ in range(64):
+        eng.step()
+        if not eng.has_work():
+            break
+    torch.cuda.synchronize()
+    toks_a = [list(s.generated) for s in seqs]
+    assert all(len(t) == 12 for t in toks_a)
+    # the pipelined sampled path captured a class-1 graph
+    assert any(k[1] == 1 for k in eng._graphs), eng._graphs.keys()
+    # diversity: stochastic rows should not all be a constant token
+    assert any(len(set(t)) > 1 for t in toks_a), toks_a
+    # reproducibility: seeded request rerun in a DIFFERENT batch
+    # composition produces the same tokens
+    sid = eng.submit(prompts[0], GenParams(max_tokens=12, temperature=0.9,
+                                           seed=100))
+    solo = eng.seqs[sid]
+    for _ in range(64):
+        eng.step()
+        if not eng.has_work():
+            break
+    torch.cuda.synchronize()
+    assert list(solo.generated) == toks_a[0], \
+        (solo.generated, toks_a[0])
