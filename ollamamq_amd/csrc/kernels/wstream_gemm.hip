@@ -103,7 +103,25 @@ extern "C" int wstream_pure_bf16(void* sink, const void* wp, int N, int K,
 //  * res_in/sq_parts: epilogue adds the residual stream (in-place safe:
 //    each (m,n) is read+written by exactly one block) and emits this
 //    block's sum-of-squares partials for the NEXT GEMM's rstd.
-template <int MT, int DEPTH = 1, int XLDS = 0, int GU = 0>
+// RoPE/KV-append epilogue bundle (RP=1 qkv GEMM): the packed qkv
+// weight is PAIR-ORDERED per head (16 lo-dims then their 16 hi-dims,
+// ops/hip.py pack_weight_qkv_rope), so each 32-col output tile holds
+// complete rotation pairs; the epilogue rotates q/k with the host
+// cos/sin tables and writes k/v straight into the paged KV pool —
+// the separate rope_append kernel (8.6 us x layers/step in the r02
+// trace) disappears from the decode step.
+struct RopeEpi {
+    const float* cos_t;   // [max_ctx, 64]
+    const float* sin_t;
+    const int* pos;       // [M] position of this step's token
+    const int* slot;      // [M] kv slot per row
+    const int* ptab;      // page table [slots, maxp]
+    bf16* kp;             // this layer's K pool base
+    bf16* vp;
+    int nl, nkl, psz, maxp;
+};
+
+template <int MT, int DEPTH = 1, int XLDS = 0, int GU = 0, int RP = 0>
 __global__ __launch_bounds__(512) void k_wstream_gemm(
     bf16* __restrict__ y,            // [M, N] (ksplit == 1)
     float* __restrict__ part,        // [ksplit, M, N] (ksplit > 1)
@@ -113,7 +131,7 @@ __global__ __launch_bounds__(512) void k_wstream_gemm(
     int M, int N, int K, int64_t xs, int ksplit,
     const float* __restrict__ rstd_parts, int rstd_nt, float inv_h,
     float eps, const bf16* __restrict__ res_in,
-    float* __restrict__ sq_parts)
+    float* __restrict__ sq_parts, RopeEpi rp = {})
 {
     const int t = blockIdx.x;              // n-tile (32 cols of y)
     const int ks = blockIdx.y;
@@ -326,6 +344,85 @@ __global__ __launch_bounds__(512) void k_wstream_gemm(
         }
         return;
     }
+    if (RP) {
+        __syncthreads();
+        #pragma unroll
+        for (int r = 0; r < 16; r++) {
+            const int crow = (r & 3) + 8 * (r >> 2) + 4 * (lane >> 5);
+            red8[wid][crow][lane & 31] = acc0[r];
+        }
+        __syncthreads();
+        const int qt = rp.nl * 4, kt = rp.nkl * 4;   // tiles per section
+        if (t < qt + kt) {
+            // pair-packed q/k tile: cols {d0+c, d0+64+c}, c<16
+            const int m = tid >> 4, c = tid & 15;
+            if (m < M) {
+                const int sec_t = t < qt ? t : t - qt;
+                const int head = sec_t >> 2;
+                const int d0 = (sec_t & 3) * 16;
+                float lo = 0.f, hi = 0.f;
+                #pragma unroll
+                for (int wv = 0; wv < 8; wv++) {
+                    lo += red8[wv][m][c];
+                    hi += red8[wv][m][c + 16];
+                }
+                if (rstd_parts) {
+                    const float rs = rstd_sh[m];
+                    lo *= rs;
+                    hi *= rs;
+                }
+                if (bias) {   // bias is pair-reordered like the pack
+                    lo += __bfloat162float(bias[t * 32 + c]);
+                    hi += __bfloat162float(bias[t * 32 + 16 + c]);
+                }
+                const int p = rp.pos[m];
+                const float co = rp.cos_t[p * 64 + d0 + c];
+                const float si = rp.sin_t[p * 64 + d0 + c];
+                const float rlo = lo * co - hi * si;
+                const float rhi = hi * co + lo * si;
+                if (t < qt) {          // q: standard layout in y
+                    bf16* qr = y + (int64_t)m * N + head * 128;
+                    qr[d0 + c] = __float2bfloat16(rlo);
+                    qr[64 + d0 + c] = __float2bfloat16(rhi);
+                } else {               // k: rotated, straight to pool
+                    const int pg = rp.ptab[rp.slot[m] * rp.maxp
+                                           + p / rp.psz];
+                    bf16* dst = rp.kp
+                        + (((int64_t)pg * rp.nkl + head) * rp.psz
+                           + p % rp.psz) * 128;
+                    dst[d0 + c] = __float2bfloat16(rlo);
+                    dst[64 + d0 + c] = __float2bfloat16(rhi);
+                }
+            }
+        } else {
+            // v tile (plain 32-col order): straight to pool, no rope
+            const int sec_t = t - qt - kt;
+            const int head = sec_t >> 2;
+            const int d0 = (sec_t & 3) * 32;
+            #pragma unroll
+            for (int ee = 0; ee < 2; ee++) {
+                const int e = tid + ee * 512;
+                const int m = e >> 5, n = e & 31;
+                if (m < M) {
+                    float sv = 0.f;
+                    #pragma unroll
+                    for (int wv = 0; wv < 8; wv++)
+                        sv += red8[wv][m][n];
+                    if (rstd_parts) sv *= rstd_sh[m];
+                    if (bias)
+                        sv += __bfloat162float(bias[t * 32 + n]);
+                    const int p = rp.pos[m];
+                    const int pg = rp.ptab[rp.slot[m] * rp.maxp
+                                           + p / rp.psz];
+                    bf16* dst = rp.vp
+                        + (((int64_t)pg * rp.nkl + head) * rp.psz
+                           + p % rp.psz) * 128;
+                    dst[d0 + n] = __float2bfloat16(sv);
+                }
+            }
+        }
+        return;
+    }
     __syncthreads();
     const int n0 = t * 32;
     for (int mt = 0; mt < MT; ++mt) {
@@ -516,3 +613,25 @@ extern "C" int wstream_gemm_bf16(
     return (int)hipGetLastError();
 }
 
+
+// qkv GEMM with fused RoPE + paged KV append (fused decode chain):
+// q lands rotated in y's standard layout; k/v land in the paged pool.
+extern "C" int wstream_qkv_rope_bf16(
+    void* y, const void* x, const void* wp, const void* bias,
+    int M, int N, int K, int64_t xs,
+    const void* rstd_parts, int rstd_nt, float inv_h, float eps,
+    const void* rcos, const void* rsin, const void* pos,
+    const void* slot, const void* ptab, void* kp, void* vp,
+    int nl, int nkl, int psz, int maxp, hipStream_t stream)
+{
+    dim3 grid(N / 32, 1);
+    const int lds = 8 * 32 * 32 * 4 + 128;
+    RopeEpi rp{(const float*)rcos, (const float*)rsin, (const int*)pos,
+               (const int*)slot, (const int*)ptab, (bf16*)kp, (bf16*)vp,
+               nl, nkl, psz, maxp};
+    k_wstream_gemm<1, 1, 0, 0, 1><<<grid, 512, lds, stream>>>(
+        (bf16*)y, nullptr, (const bf16*)x, (const u32x4*)wp,
+        (const bf16*)bias, M, N, K, xs, 1, (const float*)rstd_parts,
+        rstd_nt, inv_h, eps, nullptr, nullptr, rp);
+    return (int)hipGetLastError();
+}

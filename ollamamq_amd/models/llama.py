@@ -142,7 +142,7 @@ class LlamaLayer:
 
     __slots__ = (
         "wqkv", "bqkv", "wo", "wgate_up", "wdown", "attn_norm", "mlp_norm",
-        "wqkv_pk", "wo_pk", "wgu_pk", "wdown_pk",
+        "wqkv_pk", "wo_pk", "wgu_pk", "wdown_pk", "bqkv_rp",
     )
 
     def __init__(self, cfg: LlamaConfig, dev, dtype, gen, tp: int, rank: int):
@@ -189,6 +189,7 @@ class LlamaLayer:
         self.mlp_norm = torch.ones(h, device=dev, dtype=dtype)
         # packed decode copies filled by LlamaModel._pack_weights()
         self.wqkv_pk = self.wo_pk = self.wgu_pk = self.wdown_pk = None
+        self.bqkv_rp = None
 
 
 class LlamaModel:
@@ -284,7 +285,18 @@ class LlamaModel:
         pk = ops.pack_weight
         try:
             for l in self.layers:
-                l.wqkv_pk = pk((l.wqkv * l.attn_norm).contiguous())
+                # qkv pack is PAIR-ORDERED for the fused RoPE/KV-append
+                # epilogue (head_dim==128 models; the chain requires it)
+                if self.cfg.head_dim == 128:
+                    l.wqkv_pk = ops.pack_weight_qkv_rope(
+                        (l.wqkv * l.attn_norm).contiguous(),
+                        self.n_local_heads, self.n_local_kv_heads)
+                    if l.bqkv is not None:
+                        l.bqkv_rp = ops.qkv_rope_bias_order(
+                            l.bqkv, self.n_local_heads,
+                            self.n_local_kv_heads)
+                else:
+                    l.wqkv_pk = None
                 l.wo_pk = pk(l.wo)
                 l.wgu_pk = ops.pack_weight_gu(
                     (l.wgate_up * l.mlp_norm).contiguous())
@@ -414,16 +426,14 @@ class LlamaModel:
         eps = cfg.norm_eps
         tiles_h = cfg.hidden // 32
         for li, layer in enumerate(self.layers):
-            qkv = ops.linear_fused(res, layer.wqkv_pk,
-                                   layer.wqkv.shape[0], bias=layer.bqkv,
-                                   rstd=sq_a, rstd_nt=nt, inv_h=inv_h,
-                                   eps=eps)
-            q, k, v = qkv.split([nl * d, nkl * d, nkl * d], dim=-1)
-            q = q.view(-1, nl, d)
-            k = k.view(-1, nkl, d)
-            v = v.view(-1, nkl, d)
-            ops.rope_append(kv_cache, li, q, k, v, positions, slot_ids,
-                            self.rope_cos, self.rope_sin)
+            # fused rmsnorm -> qkv -> RoPE -> paged KV append: q comes
+            # back rotated in standard layout, k/v land in the pool
+            qkv = ops.qkv_rope_fused(
+                res, layer.wqkv_pk, layer.wqkv.shape[0], layer.bqkv_rp,
+                kv_cache, li, positions, slot_ids,
+                self.rope_cos, self.rope_sin, nl, nkl,
+                rstd=sq_a, rstd_nt=nt, inv_h=inv_h, eps=eps)
+            q = qkv[:, :nl * d].view(-1, nl, d)
             attn = ops.attention(q, kv_cache, li, attn_meta)
             ops.linear_fused(attn.view(-1, nl * d), layer.wo_pk,
                              cfg.hidden, res=res, sq_out=sq_b, y=res)

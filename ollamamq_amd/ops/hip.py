@@ -62,6 +62,9 @@ def _try_load():
     lib.row_sumsq_bf16.argtypes = [vp, vp, i, i, vp]
     lib.sample_gumbel_bf16.argtypes = [vp, vp, vp, vp, vp, vp, i, i,
                                        vp, vp, i, vp]
+    lib.wstream_qkv_rope_bf16.argtypes = [vp, vp, vp, vp, i, i, i, i64,
+                                          vp, i, f, f, vp, vp, vp, vp,
+                                          vp, vp, vp, i, i, i, i, vp]
     for fn in ("rmsnorm_residual_bf16", "rope_bf16", "kv_append_bf16",
                "paged_attn_bf16", "swiglu_bf16", "argmax_bf16",
                "decode_attn_bf16", "skinny_gemm_bf16",
@@ -463,6 +466,51 @@ def linear_gu(x, packed, N, rstd=None, rstd_nt=0, inv_h=0.0, eps=0.0):
         _p(rstd), rstd_nt, float(inv_h), float(eps),
         _stream()), "wstream_gu")
     return act
+
+
+def _rope_pair_order(heads, d=128):
+    """Output-row order pairing each head's RoPE halves: 16 lo dims then
+    their 16 hi (d+64) dims, so every 32-row tile holds full rotation
+    pairs for the fused qkv epilogue."""
+    idx = []
+    for h in range(heads):
+        base = h * d
+        for d0 in (0, 16, 32, 48):
+            idx.extend(range(base + d0, base + d0 + 16))
+            idx.extend(range(base + 64 + d0, base + 64 + d0 + 16))
+    return idx
+
+
+def qkv_rope_order(nl, nkl, d=128):
+    """Row permutation for the whole fused qkv weight [q|k|v]."""
+    q = _rope_pair_order(nl, d)
+    k = [nl * d + i for i in _rope_pair_order(nkl, d)]
+    v = [(nl + nkl) * d + i for i in range(nkl * d)]
+    return q + k + v
+
+
+def pack_weight_qkv_rope(w, nl, nkl):
+    """Pair-ordered pack of the fused qkv weight for the RoPE epilogue."""
+    order = torch.tensor(qkv_rope_order(nl, nkl), device=w.device)
+    return pack_weight(w.index_select(0, order).contiguous())
+
+
+def linear_qkv_rope(x, packed, N, bias_rp, cache, layer, positions,
+                    slots, cos, sin, nl, nkl, rstd, rstd_nt, inv_h, eps):
+    """Fused rmsnorm -> qkv GEMM -> RoPE -> paged KV append: q returns
+    rotated in the output's standard layout; k/v land in the pool."""
+    M, K = x.shape
+    y = torch.empty((M, N), dtype=x.dtype, device=x.device)
+    kp, vp = _layer_ptrs(cache, layer)
+    pos32 = positions if positions.dtype == torch.int32 else positions.int()
+    slot32 = slots if slots.dtype == torch.int32 else slots.int()
+    _check(_lib.wstream_qkv_rope_bf16(
+        _p(y), _p(x), _p(packed), _p(bias_rp), M, N, K, x.stride(0),
+        _p(rstd), rstd_nt, float(inv_h), float(eps),
+        _p(cos), _p(sin), _p(pos32), _p(slot32), _p(cache.page_table),
+        kp, vp, nl, nkl, cache.page_size, cache.page_table.shape[1],
+        _stream()), "wstream_qkv_rope")
+    return y
 
 
 USE_WSTREAM = os.environ.get("OLLAMAMQ_NO_WSTREAM") != "1"

@@ -188,6 +188,31 @@ def gu_fused(x, packed, N, rstd, rstd_nt, inv_h, eps):
                          inv_h=inv_h, eps=eps)
 
 
+def pack_weight_qkv_rope(w, nl, nkl):
+    """Pair-ordered qkv pack for the fused RoPE/KV-append epilogue."""
+    if w.is_cuda and w.shape[1] % 64 == 0:
+        from . import hip
+        if hip.available():
+            return hip.pack_weight_qkv_rope(w, nl, nkl)
+    return None
+
+
+def qkv_rope_bias_order(bias, nl, nkl):
+    """Reorder a qkv bias to match the pair-ordered pack."""
+    from . import hip
+    order = torch.tensor(hip.qkv_rope_order(nl, nkl), device=bias.device)
+    return bias.index_select(0, order).contiguous()
+
+
+def qkv_rope_fused(x, packed, N, bias_rp, cache, layer, positions, slots,
+                   cos, sin, nl, nkl, rstd, rstd_nt, inv_h, eps):
+    """rmsnorm -> qkv GEMM -> RoPE -> paged KV append, one kernel."""
+    from . import hip
+    return hip.linear_qkv_rope(x, packed, N, bias_rp, cache, layer,
+                               positions, slots, cos, sin, nl, nkl,
+                               rstd, rstd_nt, inv_h, eps)
+
+
 def sample(logits, temperature, top_k, top_p, generator=None):
     # Sampler: torch ops compose on-GPU; custom fused kernel in ops/hip.py
     # handles the greedy + temperature paths.
