@@ -30,6 +30,14 @@ DEV void load8f_lds(const bf16* p, float* out) {
     out[4] = f2.x; out[5] = f2.y; out[6] = f3.x; out[7] = f3.y;
 }
 
+// fused-chain frag layout for the attention OUTPUT (o-GEMM streams it
+// linearly; see wstream_gemm.hip frag_off): element (m, kcol) ->
+// 16 B unit ((b*4+j)*64 + h*32 + m), kcol = b*64 + j*16 + h*8 + e
+DEV int64_t attn_frag_off(int m, int kcol) {
+    const int b = kcol >> 6, j = (kcol >> 4) & 3, h = (kcol >> 3) & 1;
+    return ((((int64_t)b * 4 + j) * 64) + h * 32 + m) * 8 + (kcol & 7);
+}
+
 DEV float wave_max(float x) {
     #pragma unroll
     for (int off = 32; off > 0; off >>= 1)
@@ -68,7 +76,8 @@ __global__ __launch_bounds__(GT * 64, 1) void k_decode_attn(
     const int* __restrict__ seq_lens,     // [S] (kv length incl. this tok)
     int Hq, int KVH, int page, int max_pages, float scale,
     int64_t qs, int split, int window,    // window 0 = full causal
-    unsigned* __restrict__ sem)           // [S*KVH] tickets (fused combine)
+    unsigned* __restrict__ sem,           // [S*KVH] tickets (fused combine)
+    int fragout)                          // out in fused-chain frag layout
 {
     const int S_idx = blockIdx.x / split;
     const int seg = blockIdx.x % split;
@@ -274,9 +283,14 @@ __global__ __launch_bounds__(GT * 64, 1) void k_decode_attn(
     // ---- emit ----
     if (split == 1) {
         const float linv = l > 0.f ? 1.f / l : 0.f;
-        bf16* orow = out + ((int64_t)S_idx * Hq + qh) * DHEAD;
-        reinterpret_cast<bf162*>(orow)[lane] =
+        const bf162 ov =
             __float22bfloat162_rn(make_float2(o0 * linv, o1 * linv));
+        if (fragout)
+            *reinterpret_cast<bf162*>(
+                out + attn_frag_off(S_idx, qh * DHEAD + 2 * lane)) = ov;
+        else
+            reinterpret_cast<bf162*>(
+                out + ((int64_t)S_idx * Hq + qh) * DHEAD)[lane] = ov;
         return;
     }
     {
@@ -332,6 +346,13 @@ __global__ __launch_bounds__(GT * 64, 1) void k_decode_attn(
             a1 = fmaf(w, op[2 * lane + 1], a1);
         }
         const float linv = ll > 0.f ? 1.f / ll : 0.f;
+        if (fragout) {
+            *reinterpret_cast<bf162*>(
+                out + attn_frag_off(S_idx, qh * DHEAD + 2 * lane)) =
+                __float22bfloat162_rn(
+                    make_float2(a0 * linv, a1 * linv));
+            return;
+        }
         reinterpret_cast<bf162*>(out + sh * DHEAD)[lane] =
             __float22bfloat162_rn(make_float2(a0 * linv, a1 * linv));
     }
@@ -341,7 +362,8 @@ __global__ __launch_bounds__(GT * 64, 1) void k_decode_attn(
 // partials exactly: m* = max m_i; o = Σ e^{m_i-m*} o_i; l = Σ e^{m_i-m*} l_i
 __global__ __launch_bounds__(256) void k_decode_combine(
     bf16* __restrict__ out, const float* __restrict__ o_part,
-    const float* __restrict__ ml_part, int total, int split)
+    const float* __restrict__ ml_part, int total, int split,
+    int Hq, int fragout)
 {
     const int sh = blockIdx.x * 4 + (threadIdx.x >> 6);  // seq*Hq + head
     const int lane = threadIdx.x & 63;
@@ -361,8 +383,14 @@ __global__ __launch_bounds__(256) void k_decode_combine(
         a1 = fmaf(w, op[2 * lane + 1], a1);
     }
     const float linv = l > 0.f ? 1.f / l : 0.f;
-    reinterpret_cast<bf162*>(out + (int64_t)sh * DHEAD)[lane] =
+    const bf162 ov =
         __float22bfloat162_rn(make_float2(a0 * linv, a1 * linv));
+    if (fragout)
+        *reinterpret_cast<bf162*>(
+            out + attn_frag_off(sh / Hq, (sh % Hq) * DHEAD + 2 * lane))
+            = ov;
+    else
+        reinterpret_cast<bf162*>(out + (int64_t)sh * DHEAD)[lane] = ov;
 }
 
 extern "C" int decode_attn_bf16(
@@ -370,7 +398,7 @@ extern "C" int decode_attn_bf16(
     const void* kpool, const void* vpool, const void* page_table,
     const void* slot_ids, const void* seq_lens, int S, int Hq, int KVH,
     int page, int max_pages, float scale, int64_t q_stride, int split,
-    int window, int chunk, void* sem, hipStream_t stream)
+    int window, int chunk, void* sem, int fragout, hipStream_t stream)
 {
     const int G = Hq / KVH;
     const int lds = 2 * chunk * (DHEAD + DKPAD) * 2 + G * DHEAD * 4
@@ -381,7 +409,8 @@ extern "C" int decode_attn_bf16(
         (bf16*)out, (float*)o_part, (float*)ml_part, (const bf16*)q,      \
         (const bf16*)kpool, (const bf16*)vpool, (const int*)page_table,   \
         (const int*)slot_ids, (const int*)seq_lens, Hq, KVH, page,        \
-        max_pages, scale, q_stride, split, window, (unsigned*)sem)
+        max_pages, scale, q_stride, split, window, (unsigned*)sem,     \
+        fragout)
 // chunk is capped at 64: the score phase assigns ONE key per lane of a
 // wave64, so a larger staged chunk would silently drop keys 64+ (a 128
 // arm measured wrong before this guard)
@@ -407,7 +436,7 @@ extern "C" int decode_attn_bf16(
         const int waves = S * Hq;
         k_decode_combine<<<(waves + 3) / 4, 256, 0, stream>>>(
             (bf16*)out, (const float*)o_part, (const float*)ml_part, waves,
-            split);
+            split, Hq, fragout);
     }
     return (int)hipGetLastError();
 }

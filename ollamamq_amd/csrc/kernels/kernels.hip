@@ -890,3 +890,44 @@ extern "C" int sample_gumbel_bf16(
     }
     return (int)hipGetLastError();
 }
+
+// ---------------------------------------------------------------------------
+// Fragify + row sum-of-squares: convert a standard [M, H] activation into
+// the 32-row MFMA fragment layout the fused decode chain streams
+// (wstream_gemm.hip frag_off) AND emit sq[m] = sum(x[m]^2) — one read of
+// the embedding output seeds the whole chain.
+// ---------------------------------------------------------------------------
+__global__ __launch_bounds__(256) void k_fragify_sumsq(
+    const bf16* __restrict__ x, bf16* __restrict__ xf,
+    float* __restrict__ sq, int H)
+{
+    const int m = blockIdx.x;
+    const bf16* row = x + (int64_t)m * H;
+    float s = 0.f;
+    // 16-byte chunk c covers k = c*8..c*8+8: frag unit
+    // ((c>>3)*4 + ((c>>1)&3))*64 + (c&1)*32 + m
+    for (int c = threadIdx.x; c < H / 8; c += 256) {
+        const uint4 v = *reinterpret_cast<const uint4*>(row + c * 8);
+        float f[8];
+        load8f(row + c * 8, f);
+        #pragma unroll
+        for (int e = 0; e < 8; e++) s += f[e] * f[e];
+        const int64_t u = ((int64_t)(c >> 3) * 4 + ((c >> 1) & 3)) * 64
+                          + (c & 1) * 32 + m;
+        *reinterpret_cast<uint4*>(xf + u * 8) = v;
+    }
+    __shared__ float red[4];
+    s = wave_reduce_sum(s);
+    if ((threadIdx.x & 63) == 0) red[threadIdx.x >> 6] = s;
+    __syncthreads();
+    if (threadIdx.x == 0)
+        sq[m] = red[0] + red[1] + red[2] + red[3];
+}
+
+extern "C" int fragify_sumsq_bf16(void* xf, void* sq, const void* x,
+                                  int M, int H, hipStream_t stream)
+{
+    k_fragify_sumsq<<<M, 256, 0, stream>>>(
+        (const bf16*)x, (bf16*)xf, (float*)sq, H);
+    return (int)hipGetLastError();
+}

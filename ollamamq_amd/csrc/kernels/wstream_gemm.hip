@@ -38,6 +38,20 @@ typedef __attribute__((ext_vector_type(4))) unsigned int u32x4;
 
 union u4bf8 { u32x4 u; bf16x8 v; };
 
+// Activation-fragment layout ("frag"): the SAME MFMA fragment order the
+// weight packs use, applied to a [32, K] activation: element (m, k) with
+// k = b*64 + j*16 + h*8 + e lives at 16-byte unit ((b*4 + j)*64 + h*32
+// + m), elem e.  Producers (GEMM epilogues, decode-attention combine,
+// the embedding fragify kernel) write it; consumers stream it LINEARLY
+// exactly like packed weights — the per-iteration x machinery (32-line
+// scattered loads or LDS staging, measured 15-30% on top of the pure
+// stream) disappears.  Buffers are always 32 rows; rows >= M hold
+// garbage that MFMA carries in dead accumulator rows (never stored).
+static __device__ __forceinline__ int64_t frag_off(int m, int kcol) {
+    const int b = kcol >> 6, j = (kcol >> 4) & 3, h = (kcol >> 3) & 1;
+    return ((((int64_t)b * 4 + j) * 64) + h * 32 + m) * 8 + (kcol & 7);
+}
+
 // Pure-stream diagnostic: same grid/geometry/addressing as the GEMM but
 // only the nt weight loads (no x, no MFMA) — measures this geometry's
 // load-path ceiling so kernel iterations know what they are chasing.
@@ -131,7 +145,7 @@ __global__ __launch_bounds__(512) void k_wstream_gemm(
     int M, int N, int K, int64_t xs, int ksplit,
     const float* __restrict__ rstd_parts, int rstd_nt, float inv_h,
     float eps, const bf16* __restrict__ res_in,
-    float* __restrict__ sq_parts, RopeEpi rp = {})
+    float* __restrict__ sq_parts, RopeEpi rp = {}, int yfrag = 0)
 {
     const int t = blockIdx.x;              // n-tile (32 cols of y)
     const int ks = blockIdx.y;
@@ -168,7 +182,12 @@ __global__ __launch_bounds__(512) void k_wstream_gemm(
         const u32x4* wbase = wp + ((int64_t)t * NB * 4) * 64 + lane;
         u4bf8 br[DEPTH][4];
         bf16x8 a0[2], a1[2], a2[2], a3[2];
+        const u4bf8* xf = reinterpret_cast<const u4bf8*>(x);
         auto lda16 = [&](bf16x8* d, int b, int off) {
+            if (XLDS == 2) {   // frag input: linear like the weights
+                d[0] = xf[((int64_t)b * 4 + (off >> 4)) * 64 + lane].v;
+                return;
+            }
             const int64_t k = ((int64_t)b << 6) + off;
             d[0] = *reinterpret_cast<const bf16x8*>(xr0 + k);
             if (MT == 2) d[1] = *reinterpret_cast<const bf16x8*>(xr1 + k);
@@ -211,7 +230,7 @@ __global__ __launch_bounds__(512) void k_wstream_gemm(
                     a3[1], vb3, acc1, 0, 0, 0);
             }
         };
-        if (XLDS) {
+        if (XLDS == 1) {
             // --- LDS-staged x fragments (MT==1 only) ---
             bf16x8 xs4[4];
             auto glbx = [&](int b) {       // 4 coalesced reads: 8 rows
@@ -340,7 +359,9 @@ __global__ __launch_bounds__(512) void k_wstream_gemm(
                 su *= rs;
             }
             const float act = (sg / (1.f + __expf(-sg))) * su;
-            y[(int64_t)m * F + t * 16 + c] = __float2bfloat16(act);
+            const int fc = t * 16 + c;
+            y[yfrag ? frag_off(m, fc) : (int64_t)m * F + fc] =
+                __float2bfloat16(act);
         }
         return;
     }
@@ -447,12 +468,13 @@ __global__ __launch_bounds__(512) void k_wstream_gemm(
                 if (ksplit == 1) {
                     if (rstd_parts) s *= rstd_sh[m];
                     if (bias) s += __bfloat162float(bias[n0 + n]);
+                    const int64_t yo = yfrag ? frag_off(m, n0 + n)
+                                             : (int64_t)m * N + n0 + n;
                     if (res_in) {
-                        s += __bfloat162float(res_in[(int64_t)m * N
-                                                     + n0 + n]);
+                        s += __bfloat162float(res_in[yo]);
                         rsq[ee] = s * s;
                     }
-                    y[(int64_t)m * N + n0 + n] = __float2bfloat16(s);
+                    y[yo] = __float2bfloat16(s);
                 } else {
                     part[((int64_t)ks * M + m) * N + n0 + n] = s;
                 }
@@ -484,22 +506,28 @@ __global__ __launch_bounds__(512) void k_wstream_gemm(
 extern "C" int wstream_gu_bf16(
     void* act, const void* x, const void* wp, int M, int N, int K,
     int64_t xs, int xlds, const void* rstd_parts, int rstd_nt,
-    float inv_h, float eps, hipStream_t stream)
+    float inv_h, float eps, int yfrag, hipStream_t stream)
 {
     dim3 grid(N / 32, 1);
-    const int lds = (xlds ? 16 * 4096 : 8 * 32 * 32 * 4) + 128;
-    if (xlds)
+    const int lds = (xlds == 1 ? 16 * 4096 : 8 * 32 * 32 * 4) + 128;
+    if (xlds == 2)
+        k_wstream_gemm<1, 1, 2, 1><<<grid, 512, lds, stream>>>(
+            (bf16*)act, nullptr, (const bf16*)x, (const u32x4*)wp,
+            nullptr, M, N, K, xs, 1, (const float*)rstd_parts, rstd_nt,
+            inv_h, eps, nullptr, nullptr, {}, yfrag);
+    else if (xlds == 1)
         k_wstream_gemm<1, 1, 1, 1><<<grid, 512, lds, stream>>>(
             (bf16*)act, nullptr, (const bf16*)x, (const u32x4*)wp,
             nullptr, M, N, K, xs, 1, (const float*)rstd_parts, rstd_nt,
-            inv_h, eps, nullptr, nullptr);
+            inv_h, eps, nullptr, nullptr, {}, yfrag);
     else
         k_wstream_gemm<1, 1, 0, 1><<<grid, 512, lds, stream>>>(
             (bf16*)act, nullptr, (const bf16*)x, (const u32x4*)wp,
             nullptr, M, N, K, xs, 1, (const float*)rstd_parts, rstd_nt,
-            inv_h, eps, nullptr, nullptr);
+            inv_h, eps, nullptr, nullptr, {}, yfrag);
     return (int)hipGetLastError();
 }
+
 
 __global__ __launch_bounds__(256) void k_wstream_combine(
     bf16* __restrict__ y, const float* __restrict__ part,
@@ -521,7 +549,7 @@ __global__ __launch_bounds__(256) void k_wstream_combine(
 __global__ __launch_bounds__(256) void k_wstream_combine_tiles(
     bf16* __restrict__ y, const float* __restrict__ part,
     const bf16* __restrict__ res_in, float* __restrict__ sq_parts,
-    int M, int N, int ksplit)
+    int M, int N, int ksplit, int yfrag)
 {
     const int t = blockIdx.x;
     const int tid = threadIdx.x;
@@ -536,9 +564,11 @@ __global__ __launch_bounds__(256) void k_wstream_combine_tiles(
             float s = 0.f;
             for (int k = 0; k < ksplit; k++)
                 s += part[((int64_t)k * M + m) * N + n0 + n];
+            const int64_t yo = yfrag ? frag_off(m, n0 + n)
+                                     : (int64_t)m * N + n0 + n;
             if (res_in)
-                s += __bfloat162float(res_in[(int64_t)m * N + n0 + n]);
-            y[(int64_t)m * N + n0 + n] = __float2bfloat16(s);
+                s += __bfloat162float(res_in[yo]);
+            y[yo] = __float2bfloat16(s);
             rsq = s * s;
         }
         sqt[m][n] = rsq;
@@ -558,7 +588,7 @@ extern "C" int wstream_gemm_bf16(
     void* y, void* part, const void* x, const void* wp, const void* bias,
     int M, int N, int K, int64_t xs, int ksplit, int depth, int xlds,
     const void* rstd_parts, int rstd_nt, float inv_h, float eps,
-    const void* res_in, void* sq_parts, hipStream_t stream)
+    const void* res_in, void* sq_parts, int yfrag, hipStream_t stream)
 {
     dim3 grid(N / 32, ksplit);
     const int lds_red = 8 * 32 * 32 * 4 + 128;     // reduce + rstd tail
@@ -568,23 +598,28 @@ extern "C" int wstream_gemm_bf16(
     float* sq = ksplit == 1 ? (float*)sq_parts : nullptr;
     if (ksplit > 1 && rstd_parts) return -100;     // fused rstd needs ks==1
     if (M <= 32) {
-        if (xlds)
+        if (xlds == 2)
+            k_wstream_gemm<1, 1, 2><<<grid, 512, lds_red, stream>>>(
+                (bf16*)y, (float*)part, (const bf16*)x, (const u32x4*)wp,
+                (const bf16*)bias, M, N, K, xs, ksplit, rp, rstd_nt,
+                inv_h, eps, ri, sq, {}, yfrag);
+        else if (xlds == 1)
             k_wstream_gemm<1, 1, 1><<<grid, 512, lds_x, stream>>>(
                 (bf16*)y, (float*)part, (const bf16*)x, (const u32x4*)wp,
                 (const bf16*)bias, M, N, K, xs, ksplit, rp, rstd_nt,
-                inv_h, eps, ri, sq);
+                inv_h, eps, ri, sq, {}, yfrag);
         else if (depth == 2)
             k_wstream_gemm<1, 2><<<grid, 512, lds_red, stream>>>(
                 (bf16*)y, (float*)part, (const bf16*)x, (const u32x4*)wp,
                 (const bf16*)bias, M, N, K, xs, ksplit, rp, rstd_nt,
-                inv_h, eps, ri, sq);
+                inv_h, eps, ri, sq, {}, yfrag);
         else
             k_wstream_gemm<1, 1><<<grid, 512, lds_red, stream>>>(
                 (bf16*)y, (float*)part, (const bf16*)x, (const u32x4*)wp,
                 (const bf16*)bias, M, N, K, xs, ksplit, rp, rstd_nt,
-                inv_h, eps, ri, sq);
+                inv_h, eps, ri, sq, {}, yfrag);
     } else {
-        if (rstd_parts || res_in || sq_parts) return -101;  // M<=32 only
+        if (rstd_parts || res_in || sq_parts || yfrag) return -101;
         if (depth == 2)
             k_wstream_gemm<2, 2><<<grid, 512, lds_red, stream>>>(
                 (bf16*)y, (float*)part, (const bf16*)x, (const u32x4*)wp,
@@ -600,7 +635,7 @@ extern "C" int wstream_gemm_bf16(
         if (res_in || sq_parts) {
             k_wstream_combine_tiles<<<N / 32, 256, 0, stream>>>(
                 (bf16*)y, (const float*)part, (const bf16*)res_in,
-                (float*)sq_parts, M, N, ksplit);
+                (float*)sq_parts, M, N, ksplit, yfrag);
         } else {
             const int64_t mn = (int64_t)M * N;
             const int64_t want = (mn + 255) / 256;
@@ -614,6 +649,7 @@ extern "C" int wstream_gemm_bf16(
 }
 
 
+
 // qkv GEMM with fused RoPE + paged KV append (fused decode chain):
 // q lands rotated in y's standard layout; k/v land in the paged pool.
 extern "C" int wstream_qkv_rope_bf16(
@@ -622,16 +658,23 @@ extern "C" int wstream_qkv_rope_bf16(
     const void* rstd_parts, int rstd_nt, float inv_h, float eps,
     const void* rcos, const void* rsin, const void* pos,
     const void* slot, const void* ptab, void* kp, void* vp,
-    int nl, int nkl, int psz, int maxp, hipStream_t stream)
+    int nl, int nkl, int psz, int maxp, int xf, hipStream_t stream)
 {
     dim3 grid(N / 32, 1);
     const int lds = 8 * 32 * 32 * 4 + 128;
     RopeEpi rp{(const float*)rcos, (const float*)rsin, (const int*)pos,
                (const int*)slot, (const int*)ptab, (bf16*)kp, (bf16*)vp,
                nl, nkl, psz, maxp};
-    k_wstream_gemm<1, 1, 0, 0, 1><<<grid, 512, lds, stream>>>(
-        (bf16*)y, nullptr, (const bf16*)x, (const u32x4*)wp,
-        (const bf16*)bias, M, N, K, xs, 1, (const float*)rstd_parts,
-        rstd_nt, inv_h, eps, nullptr, nullptr, rp);
+    if (xf)
+        k_wstream_gemm<1, 1, 2, 0, 1><<<grid, 512, lds, stream>>>(
+            (bf16*)y, nullptr, (const bf16*)x, (const u32x4*)wp,
+            (const bf16*)bias, M, N, K, xs, 1, (const float*)rstd_parts,
+            rstd_nt, inv_h, eps, nullptr, nullptr, rp);
+    else
+        k_wstream_gemm<1, 1, 0, 0, 1><<<grid, 512, lds, stream>>>(
+            (bf16*)y, nullptr, (const bf16*)x, (const u32x4*)wp,
+            (const bf16*)bias, M, N, K, xs, 1, (const float*)rstd_parts,
+            rstd_nt, inv_h, eps, nullptr, nullptr, rp);
     return (int)hipGetLastError();
 }
+

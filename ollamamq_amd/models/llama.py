@@ -321,6 +321,11 @@ class LlamaModel:
             self._sq_a = torch.zeros(max(th, 1) * 32, dtype=torch.float32,
                                      device=self.device)
             self._sq_b = torch.zeros_like(self._sq_a)
+            # frag-layout residual stream: fixed 32 rows (rows >= batch
+            # hold garbage that only dead accumulator rows ever see)
+            self._res_frag = torch.zeros(32 * self.cfg.hidden,
+                                         dtype=self.dtype,
+                                         device=self.device)
 
     # -- helpers -----------------------------------------------------------
     def _allreduce(self, x: torch.Tensor) -> torch.Tensor:
@@ -415,16 +420,21 @@ class LlamaModel:
         the MFMA-native decode path)."""
         cfg = self.cfg
         B = tokens.shape[0]
+        H = cfg.hidden
         nl, nkl, d = self.n_local_heads, self.n_local_kv_heads, cfg.head_dim
-        res = ops.embedding(tokens, self.embed)
-        if not res.is_contiguous():
-            res = res.contiguous()
+        emb = ops.embedding(tokens, self.embed)
+        if not emb.is_contiguous():
+            emb = emb.contiguous()
         sq_a, sq_b = self._sq_a, self._sq_b
-        ops.row_sumsq(res, out=sq_a[:B])
+        # res lives in the 32-row frag layout for the WHOLE chain: every
+        # consumer GEMM streams its activation input linearly exactly
+        # like packed weights (the scattered x machinery measured 15-30%
+        # on top of the pure stream rate — profiles/r02_gemm_sweep.md)
+        res = ops.fragify_sumsq(emb, xf=self._res_frag, sq=sq_a[:B])[0]
         nt = 1                      # embed stats = one partial tile
-        inv_h = 1.0 / cfg.hidden
+        inv_h = 1.0 / H
         eps = cfg.norm_eps
-        tiles_h = cfg.hidden // 32
+        tiles_h = H // 32
         for li, layer in enumerate(self.layers):
             # fused rmsnorm -> qkv -> RoPE -> paged KV append: q comes
             # back rotated in standard layout, k/v land in the pool
@@ -432,17 +442,22 @@ class LlamaModel:
                 res, layer.wqkv_pk, layer.wqkv.shape[0], layer.bqkv_rp,
                 kv_cache, li, positions, slot_ids,
                 self.rope_cos, self.rope_sin, nl, nkl,
-                rstd=sq_a, rstd_nt=nt, inv_h=inv_h, eps=eps)
+                rstd=sq_a, rstd_nt=nt, inv_h=inv_h, eps=eps,
+                K=H, M_real=B)
             q = qkv[:, :nl * d].view(-1, nl, d)
-            attn = ops.attention(q, kv_cache, li, attn_meta)
-            ops.linear_fused(attn.view(-1, nl * d), layer.wo_pk,
-                             cfg.hidden, res=res, sq_out=sq_b, y=res)
+            attn = ops.attention_decode_frag(q, kv_cache, li, attn_meta)
+            ops.linear_fused(attn, layer.wo_pk, H, res=res, sq_out=sq_b,
+                             y=res, yfrag=1, K=nl * d, xlds=2)
             act = ops.gu_fused(res, layer.wgu_pk,
                                layer.wgate_up.shape[0], rstd=sq_b,
-                               rstd_nt=tiles_h, inv_h=inv_h, eps=eps)
-            ops.linear_fused(act, layer.wdown_pk, cfg.hidden,
-                             res=res, sq_out=sq_a, y=res)
+                               rstd_nt=tiles_h, inv_h=inv_h, eps=eps,
+                               K=H, yfrag=1)
+            ops.linear_fused(act, layer.wdown_pk, H, res=res,
+                             sq_out=sq_a, y=res, yfrag=1,
+                             K=layer.wdown.shape[1], xlds=2)
             nt = tiles_h
-        return ops.linear_fused(res, self.lm_head_pk,
-                                self.lm_head.shape[0], rstd=sq_a,
-                                rstd_nt=nt, inv_h=inv_h, eps=eps)
+        logits = ops.linear_fused(res, self.lm_head_pk,
+                                  self.lm_head.shape[0], rstd=sq_a,
+                                  rstd_nt=nt, inv_h=inv_h, eps=eps,
+                                  K=H, xlds=2)
+        return logits[:B]

@@ -46,7 +46,7 @@ def _try_load():
     lib.argmax_bf16.argtypes = [vp, vp, i, i, vp, vp, i, vp]
     lib.decode_attn_bf16.argtypes = [vp, vp, vp, vp, vp, vp, vp, vp, vp,
                                      i, i, i, i, i, f, i64, i, i, i, vp,
-                                     vp]
+                                     i, vp]
     lib.skinny_gemm_bf16.argtypes = [vp, vp, vp, vp, i, i, i, i64, i, vp]
     lib.skinny_direct_bf16.argtypes = [vp, vp, vp, vp, i, i, i, i64, i, vp]
     lib.rope_append_bf16.argtypes = [vp, vp, vp, vp, vp, vp, vp, vp, vp, vp,
@@ -55,16 +55,17 @@ def _try_load():
                                       i, i, i, i, i, f, i64, i, vp]
     lib.embed_gather_bf16.argtypes = [vp, vp, vp, i, i, vp]
     lib.wstream_gemm_bf16.argtypes = [vp, vp, vp, vp, vp, i, i, i, i64, i,
-                                      i, i, vp, i, f, f, vp, vp, vp]
+                                      i, i, vp, i, f, f, vp, vp, i, vp]
     lib.wstream_pure_bf16.argtypes = [vp, vp, i, i, i, vp]
     lib.wstream_gu_bf16.argtypes = [vp, vp, vp, i, i, i, i64, i,
-                                    vp, i, f, f, vp]
+                                    vp, i, f, f, i, vp]
     lib.row_sumsq_bf16.argtypes = [vp, vp, i, i, vp]
+    lib.fragify_sumsq_bf16.argtypes = [vp, vp, vp, i, i, vp]
     lib.sample_gumbel_bf16.argtypes = [vp, vp, vp, vp, vp, vp, i, i,
                                        vp, vp, i, vp]
     lib.wstream_qkv_rope_bf16.argtypes = [vp, vp, vp, vp, i, i, i, i64,
                                           vp, i, f, f, vp, vp, vp, vp,
-                                          vp, vp, vp, i, i, i, i, vp]
+                                          vp, vp, vp, i, i, i, i, i, vp]
     for fn in ("rmsnorm_residual_bf16", "rope_bf16", "kv_append_bf16",
                "paged_attn_bf16", "swiglu_bf16", "argmax_bf16",
                "decode_attn_bf16", "skinny_gemm_bf16",
@@ -277,15 +278,18 @@ def _decode_chunk(G):
     return 32 if G == 1 else 64
 
 
-def attention_decode(q, cache, layer, meta):
+def attention_decode(q, cache, layer, meta, fragout=False):
     """Flash-decoding: KV-split partials + exact online-softmax combine.
 
     The split factor targets ≥1024 workgroups so the memory-bound KV sweep
     fills the 256-CU chip (a bare (seq, kv-head) grid at batch 32 is 1
-    workgroup/CU and runs at ~10% of HBM bandwidth)."""
+    workgroup/CU and runs at ~10% of HBM bandwidth).
+    fragout: emit the output in the fused-chain 32-row frag layout (the
+    o GEMM streams it linearly like packed weights)."""
     S, Hq, D = q.shape
     assert D == 128
-    out = torch.empty((S, Hq, D), dtype=q.dtype, device=q.device)
+    out = torch.empty((32 * Hq * D,) if fragout else (S, Hq, D),
+                      dtype=q.dtype, device=q.device)
     kp, vp = _layer_ptrs(cache, layer)
     kvh = cache.n_kv_heads
     # target ~1024 blocks; under graph capture max_kv is the pool's
@@ -318,7 +322,8 @@ def attention_decode(q, cache, layer, meta):
         _p(out), op, mp, _p(q), kp, vp, _p(cache.page_table),
         _p(slot32), _p(len32), S, Hq, kvh, cache.page_size,
         cache.page_table.shape[1], 1.0 / (D ** 0.5), _row_stride(q, D),
-        split, meta.window, _decode_chunk(Hq // kvh), sem, _stream()),
+        split, meta.window, _decode_chunk(Hq // kvh), sem,
+        1 if fragout else 0, _stream()),
         "decode_attn")
     return out
 
@@ -453,17 +458,25 @@ def pack_weight_gu(w):
     return p.view(-1)
 
 
-def linear_gu(x, packed, N, rstd=None, rstd_nt=0, inv_h=0.0, eps=0.0):
+def linear_gu(x, packed, N, rstd=None, rstd_nt=0, inv_h=0.0, eps=0.0,
+              K=None, yfrag=0):
     """act = swiglu(x @ Wgu^T) fused; N = 2F total weight rows.
     With rstd: x is the raw residual and the epilogue applies the
-    rmsnorm scale before SwiGLU (norm weight folded into the pack)."""
-    M, K = x.shape
+    rmsnorm scale before SwiGLU (norm weight folded into the pack).
+    K given => x is a frag-layout buffer (xlds=2); yfrag => the
+    activation is emitted in frag layout for the down GEMM."""
     F = N // 2
-    act = torch.empty((M, F), dtype=x.dtype, device=x.device)
-    xlds = 1 if N * K * 2 > (64 << 20) else 0
+    if K is None:
+        M, K = x.shape
+        xs = x.stride(0)
+        xlds = 1 if N * K * 2 > (64 << 20) else 0
+    else:
+        M, xs, xlds = 32, 0, 2
+    act = torch.empty((M, F) if not yfrag else (32 * F,),
+                      dtype=x.dtype, device=x.device)
     _check(_lib.wstream_gu_bf16(
-        _p(act), _p(x), _p(packed), M, N, K, x.stride(0), xlds,
-        _p(rstd), rstd_nt, float(inv_h), float(eps),
+        _p(act), _p(x), _p(packed), M, N, K, xs, xlds,
+        _p(rstd), rstd_nt, float(inv_h), float(eps), yfrag,
         _stream()), "wstream_gu")
     return act
 
@@ -496,20 +509,25 @@ def pack_weight_qkv_rope(w, nl, nkl):
 
 
 def linear_qkv_rope(x, packed, N, bias_rp, cache, layer, positions,
-                    slots, cos, sin, nl, nkl, rstd, rstd_nt, inv_h, eps):
+                    slots, cos, sin, nl, nkl, rstd, rstd_nt, inv_h, eps,
+                    K=None, M_real=32):
     """Fused rmsnorm -> qkv GEMM -> RoPE -> paged KV append: q returns
     rotated in the output's standard layout; k/v land in the pool."""
-    M, K = x.shape
+    if K is None:
+        M, K = x.shape
+        xs, xf = x.stride(0), 0
+    else:
+        M, xs, xf = M_real, 0, 1
     y = torch.empty((M, N), dtype=x.dtype, device=x.device)
     kp, vp = _layer_ptrs(cache, layer)
     pos32 = positions if positions.dtype == torch.int32 else positions.int()
     slot32 = slots if slots.dtype == torch.int32 else slots.int()
     _check(_lib.wstream_qkv_rope_bf16(
-        _p(y), _p(x), _p(packed), _p(bias_rp), M, N, K, x.stride(0),
+        _p(y), _p(x), _p(packed), _p(bias_rp), M, N, K, xs,
         _p(rstd), rstd_nt, float(inv_h), float(eps),
         _p(cos), _p(sin), _p(pos32), _p(slot32), _p(cache.page_table),
         kp, vp, nl, nkl, cache.page_size, cache.page_table.shape[1],
-        _stream()), "wstream_qkv_rope")
+        xf, _stream()), "wstream_qkv_rope")
     return y
 
 
@@ -560,7 +578,7 @@ _WS_MAX_N = int(os.environ.get("OLLAMAMQ_WS_MAX_N", "16384"))
 
 def linear_packed(x, packed, bias, N, ks=None, depth=None, xlds=None,
                   rstd=None, rstd_nt=0, inv_h=0.0, eps=0.0,
-                  res=None, sq_out=None, y=None):
+                  res=None, sq_out=None, y=None, yfrag=0, K=None):
     """y = x @ W^T via the weight-streaming kernel over pre-packed W.
 
     Fused-chain extras (decode, M<=32): `rstd`/`rstd_nt`/`inv_h`/`eps`
@@ -568,7 +586,12 @@ def linear_packed(x, packed, bias, N, ks=None, depth=None, xlds=None,
     be folded into the pack); `res` adds the residual stream into the
     output (pass y=res for the in-place residual update) and `sq_out`
     emits per-tile sum-of-squares partials for the next GEMM."""
-    M, K = x.shape
+    if K is None:
+        M, K = x.shape
+        xs = x.stride(0)
+    else:                       # frag-layout x: flat buffer, xlds==2
+        M = 32
+        xs = 0
     if y is None:
         y = torch.empty((M, N), dtype=x.dtype, device=x.device)
     if ks is None:
@@ -595,11 +618,24 @@ def linear_packed(x, packed, bias, N, ks=None, depth=None, xlds=None,
         # wins on the short ramp-bound ones (qkv/o)
         xlds = (_WS_XLDS if M <= 32 and N * K * 2 > (64 << 20) else 0)
     _check(_lib.wstream_gemm_bf16(
-        _p(y), part, _p(x), _p(packed), _p(bias), M, N, K, x.stride(0),
+        _p(y), part, _p(x), _p(packed), _p(bias), M, N, K, xs,
         ks, depth if depth is not None else _WS_DEPTH, xlds,
         _p(rstd), rstd_nt, float(inv_h), float(eps), _p(res), _p(sq_out),
-        _stream()), "wstream_gemm")
+        yfrag, _stream()), "wstream_gemm")
     return y
+
+
+def fragify_sumsq(x, xf=None, sq=None):
+    """Standard [M, H] -> 32-row frag layout + per-row sum of squares
+    (seeds the fused decode chain after the embedding gather)."""
+    M, H = x.shape
+    if xf is None:
+        xf = torch.empty(32 * H, dtype=x.dtype, device=x.device)
+    if sq is None:
+        sq = torch.empty(M, dtype=torch.float32, device=x.device)
+    _check(_lib.fragify_sumsq_bf16(_p(xf), _p(sq), _p(x), M, H,
+                                   _stream()), "fragify_sumsq")
+    return xf, sq
 
 
 def row_sumsq(x, out=None):

@@ -170,22 +170,38 @@ def row_sumsq(x, out=None):
 
 
 def linear_fused(x, packed, N, bias=None, rstd=None, rstd_nt=0,
-                 inv_h=0.0, eps=0.0, res=None, sq_out=None, y=None):
+                 inv_h=0.0, eps=0.0, res=None, sq_out=None, y=None,
+                 yfrag=0, K=None, xlds=None):
     """Weight-streaming GEMM with the fused-chain epilogue/prologue:
     optional rmsnorm scaling of the raw-residual input (norm weight
     folded into the pack), in-place residual add (pass y=res) and
-    sum-of-squares partial emission for the next GEMM's rstd."""
+    sum-of-squares partial emission for the next GEMM's rstd.
+    K given => x is a 32-row frag-layout buffer streamed linearly;
+    yfrag => the output is emitted in frag layout."""
     from . import hip
     return hip.linear_packed(x, packed, bias, N, rstd=rstd,
                              rstd_nt=rstd_nt, inv_h=inv_h, eps=eps,
-                             res=res, sq_out=sq_out, y=y)
+                             res=res, sq_out=sq_out, y=y, yfrag=yfrag,
+                             K=K, xlds=xlds)
 
 
-def gu_fused(x, packed, N, rstd, rstd_nt, inv_h, eps):
+def fragify_sumsq(x, xf=None, sq=None):
+    """Standard [M, H] -> 32-row frag layout + per-row sum of squares."""
+    from . import hip
+    return hip.fragify_sumsq(x, xf, sq)
+
+
+def attention_decode_frag(q, cache, layer, meta):
+    """Decode attention emitting its output in frag layout."""
+    from . import hip
+    return hip.attention_decode(q, cache, layer, meta, fragout=True)
+
+
+def gu_fused(x, packed, N, rstd, rstd_nt, inv_h, eps, K=None, yfrag=0):
     """rmsnorm -> gate_up GEMM -> SwiGLU, one kernel."""
     from . import hip
     return hip.linear_gu(x, packed, N, rstd=rstd, rstd_nt=rstd_nt,
-                         inv_h=inv_h, eps=eps)
+                         inv_h=inv_h, eps=eps, K=K, yfrag=yfrag)
 
 
 def pack_weight_qkv_rope(w, nl, nkl):
@@ -205,12 +221,14 @@ def qkv_rope_bias_order(bias, nl, nkl):
 
 
 def qkv_rope_fused(x, packed, N, bias_rp, cache, layer, positions, slots,
-                   cos, sin, nl, nkl, rstd, rstd_nt, inv_h, eps):
+                   cos, sin, nl, nkl, rstd, rstd_nt, inv_h, eps,
+                   K=None, M_real=32):
     """rmsnorm -> qkv GEMM -> RoPE -> paged KV append, one kernel."""
     from . import hip
     return hip.linear_qkv_rope(x, packed, N, bias_rp, cache, layer,
                                positions, slots, cos, sin, nl, nkl,
-                               rstd, rstd_nt, inv_h, eps)
+                               rstd, rstd_nt, inv_h, eps, K=K,
+                               M_real=M_real)
 
 
 def sample(logits, temperature, top_k, top_p, generator=None):
