@@ -320,7 +320,7 @@ __global__ __launch_bounds__(512) void k_wstream_gemm(
     // partials from the producing GEMM (L2-hot, overlapped with other
     // waves' stream drain), publish rstd through the smem tail
     float* rstd_sh = reinterpret_cast<float*>(
-        smem + (XLDS ? 16 * 4096 : 8 * 32 * 32 * 4));
+        smem + (XLDS == 1 ? 16 * 4096 : 8 * 32 * 32 * 4));
     if (rstd_parts) {
         // parallel partial fold: 16 lanes per row (each <= nt/16 loads,
         // independent), group-reduced with wave shuffles — a serial
