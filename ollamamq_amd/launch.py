@@ -119,7 +119,21 @@ def main():
                    "--socket", sock, "--gpu-base", str(gpu_base),
                    "--model", model, "--max-ctx", str(args.max_ctx),
                    "--max-batch", str(args.max_batch)]
-            procs.append(subprocess.Popen(cmd, env=env))
+            # RCCL small-message tuning for the TP decode all-reduces
+            # (16-64 KiB latency-bound messages over the
+            # 7-link point-to-point xGMI mesh).  OLLAMAMQ_RCCL_ALGO /
+            # _PROTO map onto NCCL_ALGO / NCCL_PROTO for the worker
+            # group only; unset, RCCL's own size-based tuning applies
+            # (forcing LL globally would hurt the prefill-size
+            # all-reduces, so there is no hard default here).
+            tenv = dict(env)
+            for src, dst in (("OLLAMAMQ_RCCL_ALGO", "NCCL_ALGO"),
+                             ("OLLAMAMQ_RCCL_PROTO", "NCCL_PROTO"),
+                             ("OLLAMAMQ_RCCL_NCHANNELS",
+                              "NCCL_MIN_NCHANNELS")):
+                if os.environ.get(src):
+                    tenv.setdefault(dst, os.environ[src])
+            procs.append(subprocess.Popen(cmd, env=tenv))
         gpu_base += tp
     n_backends = len(plan)
 
