@@ -488,3 +488,79 @@ def test_wstream_gu_fused():
         expect = torch.nn.functional.silu(g) * u
         torch.testing.assert_close(act.float(), expect,
                                    atol=8e-2, rtol=8e-2)
+
+
+def _defrag(flat, M, K):
+    """Read a 32-row frag-layout buffer back to [M, K] (test helper)."""
+    got = torch.empty(M, K)
+    fv = flat.reshape(-1).float().cpu()
+    for m in range(M):
+        for n in range(K):
+            b, j, h, e = n // 64, (n // 16) % 4, (n // 8) % 2, n % 8
+            u = ((b * 4 + j) * 64) + h * 32 + m
+            got[m, n] = fv[u * 8 + e]
+    return got
+
+
+def test_frag_layout_roundtrip():
+    """The frag-chain primitives agree on ONE layout: fragify output,
+    XF GEMM input, yfrag GEMM output, fused-GU output and the decode
+    attention fragout all use frag_off(m, k)."""
+    hip = _hip()
+    g = torch.Generator().manual_seed(3)
+    M, K, N = 5, 512, 1024
+    x = (torch.randn(M, K, generator=g) * 0.5).bfloat16().cuda()
+    xf, sq = hip.fragify_sumsq(x)
+    torch.testing.assert_close(sq.cpu(), (x.float().cpu() ** 2).sum(-1),
+                               rtol=1e-2, atol=1e-2)
+    w = (torch.randn(N, K, generator=g) * 0.1).bfloat16().cuda()
+    pk = hip.pack_weight(w)
+    y_std = hip.linear_packed(x, pk, None, N)
+    # frag input == std input
+    y_xf = hip.linear_packed(xf, pk, None, N, K=K, xlds=2)
+    torch.testing.assert_close(y_xf[:M].float(), y_std.float(),
+                               atol=5e-2, rtol=5e-2)
+    # frag output round-trips
+    y_fr = hip.linear_packed(xf, pk, None, N, K=K, xlds=2, yfrag=1)
+    torch.testing.assert_close(_defrag(y_fr, M, N), y_std.float().cpu(),
+                               atol=5e-2, rtol=5e-2)
+    # fused GU with frag in/out
+    F = 512
+    wg = (torch.randn(2 * F, K, generator=g) * 0.1).bfloat16().cuda()
+    gpk = hip.pack_weight_gu(wg)
+    act_std = hip.linear_gu(x, gpk, 2 * F)
+    act_fr = hip.linear_gu(xf, gpk, 2 * F, K=K, yfrag=1)
+    torch.testing.assert_close(_defrag(act_fr, M, F),
+                               act_std.float().cpu(), atol=5e-2,
+                               rtol=5e-2)
+
+
+def test_decode_attn_fragout():
+    """attention_decode(fragout=True) matches the standard output."""
+    hip = _hip()
+    from ollamamq_amd.models import PRESETS
+    from ollamamq_amd.engine.kvcache import PagedKVCache
+    cfg = PRESETS["tiny"]
+    kv = PagedKVCache.for_model(cfg, n_pages=64, max_slots=4, max_ctx=128,
+                                device="cuda", dtype=torch.bfloat16)
+    g = torch.Generator().manual_seed(5)
+    B, L = 3, 9
+    slots = [kv.alloc_slot() for _ in range(B)]
+    for s_ in slots:
+        kv.ensure(s_, L)
+    kv.k_pool.normal_(0, 0.3)
+    kv.v_pool.normal_(0, 0.3)
+    q = (torch.randn(B, cfg.n_heads, 128, generator=g) * 0.3) \
+        .bfloat16().cuda()
+    meta = AttnMeta(
+        mode="decode",
+        slot_ids=torch.tensor(slots, dtype=torch.int32, device="cuda"),
+        seq_lens=torch.tensor([L] * B, dtype=torch.int32, device="cuda"),
+        cu_q=torch.arange(B + 1, dtype=torch.int32, device="cuda"),
+        logits_idx=None, max_q=1, max_kv=L, window=0)
+    a_std = hip.attention_decode(q, kv, 0, meta)
+    a_fr = hip.attention_decode(q, kv, 0, meta, fragout=True)
+    Kh = cfg.n_heads * 128
+    torch.testing.assert_close(_defrag(a_fr, B, Kh),
+                               a_std.view(B, Kh).float().cpu(),
+                               atol=5e-2, rtol=5e-2)

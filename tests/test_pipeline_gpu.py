@@ -109,7 +109,8 @@ def test_pipelined_kv_pages_all_freed():
 def test_warm_graphs_precapture():
     eng = make_engine(pipelined=True)
     eng.warm_graphs()
-    assert set(eng._graphs) >= {1, 2, 4, 8}
+    # graphs are keyed (batch, sampling-class); warm pre-captures class 0
+    assert set(eng._graphs) >= {(1, 0), (2, 0), (4, 0), (8, 0)}
     assert all(e["graph"] is not None for e in eng._graphs.values())
     # throwaway slots and pages all returned
     assert len(eng.kv._free_slots) == eng.kv.max_slots
