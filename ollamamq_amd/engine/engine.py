@@ -374,8 +374,11 @@ class LlamaEngine:
 
     def _samp_class(self, seqs) -> int:
         """0 = all greedy (argmax tail), 1 = temperature-only mix (Gumbel
-        tail — exact, still one captured graph), 2 = top-k/top-p present
-        (host torch sampling; pipelined replay not captured yet)."""
+        tail — exact), 2 = top-k/top-p present (in-graph candidate
+        filtering over the top-256 logits + exact Gumbel-max on the
+        filtered set; nucleus/top-k capped at 256 candidates on this
+        path — the non-graph host path stays exact over the full
+        vocabulary)."""
         k = 0
         for s in seqs:
             p = s.params
@@ -401,6 +404,8 @@ class LlamaEngine:
             "lens": torch.zeros(B, dtype=torch.int32, device=dev),
             "temps": torch.zeros(B, dtype=torch.float32, device=dev),
             "seeds": torch.zeros(B, dtype=torch.int64, device=dev),
+            "topk": torch.zeros(B, dtype=torch.int64, device=dev),
+            "topp": torch.ones(B, dtype=torch.float32, device=dev),
         }
         meta = AttnMeta(
             mode="decode", slot_ids=bufs["slot"], seq_lens=bufs["lens"],
@@ -426,13 +431,21 @@ class LlamaEngine:
         bufs["pos"].copy_(staged[1])
         bufs["slot"].copy_(staged[2])
         bufs["lens"].copy_(staged[3])
-        if entry["klass"] == 1:
+        if entry["klass"] >= 1:
             bufs["temps"].copy_(torch.tensor(
                 [s.params.temperature for s in seqs],
                 dtype=torch.float32).to(self.dev, non_blocking=True))
             bufs["seeds"].copy_(torch.tensor(
                 [getattr(s, "noise_seed", 1234) for s in seqs],
                 dtype=torch.int64).to(self.dev, non_blocking=True))
+        if entry["klass"] == 2:
+            bufs["topk"].copy_(torch.tensor(
+                [s.params.top_k or 0 for s in seqs],
+                dtype=torch.int64).to(self.dev, non_blocking=True))
+            bufs["topp"].copy_(torch.tensor(
+                [s.params.top_p if s.params.top_p else 1.0
+                 for s in seqs],
+                dtype=torch.float32).to(self.dev, non_blocking=True))
 
     def _graph_replay(self, entry):
         bufs, meta = entry["bufs"], entry["meta"]
@@ -472,6 +485,36 @@ class LlamaEngine:
                         toks32 = _hip.sample_gumbel(
                             entry["logits"], bufs["temps"], bufs["seeds"],
                             bufs["lens"])
+                        toks = toks32.long()
+                        bufs["tok"].copy_(toks32)
+                    elif entry["klass"] == 2:
+                        # top-k/top-p over the top-256 candidates, then
+                        # exact Gumbel-max on the filtered set — all
+                        # capturable ops, no host RNG state in the graph
+                        from ..ops import hip as _hip
+                        lg = entry["logits"]
+                        C = min(256, lg.shape[1])
+                        v, idx = torch.topk(lg.float(), C, dim=-1)
+                        t = bufs["temps"].clamp(min=1e-6).unsqueeze(1)
+                        p = torch.softmax(v / t, dim=-1)
+                        ar = torch.arange(C, device=lg.device)
+                        kk = torch.where(
+                            bufs["topk"] > 0,
+                            bufs["topk"].clamp(max=C),
+                            torch.full_like(bufs["topk"], C))
+                        keep = ar.unsqueeze(0) < kk.unsqueeze(1)
+                        cum = p.cumsum(dim=-1)
+                        keep &= (cum - p) < bufs["topp"].unsqueeze(1)
+                        keep[:, 0] = True
+                        # greedy rows bypass the (garbage) filter math
+                        keep |= (bufs["temps"] <= 0).unsqueeze(1)
+                        vm = v.masked_fill(~keep,
+                                           float("-inf")).bfloat16()
+                        ci = _hip.sample_gumbel(
+                            vm.contiguous(), bufs["temps"],
+                            bufs["seeds"], bufs["lens"])
+                        toks32 = idx.gather(
+                            1, ci.long().unsqueeze(1)).squeeze(1).int()
                         toks = toks32.long()
                         bufs["tok"].copy_(toks32)
                     else:
@@ -525,7 +568,6 @@ class LlamaEngine:
     def _decode_step(self):
         seqs = self.running
         if (self.use_pipeline and self.use_graphs
-                and self._samp_class(seqs) <= 1
                 and all(self.kv.seq_lens[s.slot] + 1 <= self.kv.max_ctx
                         for s in seqs)):
             self._decode_step_pipelined(seqs)

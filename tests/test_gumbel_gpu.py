@@ -109,3 +109,48 @@ def test_sampled_decode_replays_graphs():
     torch.cuda.synchronize()
     assert list(solo.generated) == toks_a[0], \
         (solo.generated, toks_a[0])
+
+
+def test_topk_topp_decode_in_graph():
+    """Class-2 sampled workloads (top-k/top-p) replay graphs too: the
+    captured tail filters the top-256 candidates per row's own params
+    and Gumbel-samples the filtered set."""
+    from ollamamq_amd.models import LlamaModel, PRESETS
+    from ollamamq_amd.engine import GenParams, LlamaEngine, PagedKVCache
+    cfg = PRESETS["tiny"]
+    model = LlamaModel(cfg, device="cuda:0", dtype=torch.bfloat16,
+                       seed=1234)
+    kv = PagedKVCache.for_model(cfg, n_pages=128, max_slots=6,
+                                max_ctx=256, device="cuda:0",
+                                dtype=torch.bfloat16)
+    eng = LlamaEngine(model, kv, max_batch=4)
+    n = 14
+    params = [GenParams(max_tokens=n, temperature=0.8, top_k=1),
+              GenParams(max_tokens=n, temperature=0.9, top_k=5,
+                        seed=42),
+              GenParams(max_tokens=n)]                     # greedy row
+    sids = [eng.submit([2, 7, 1, 8, 2, 8], p) for p in params]
+    seqs = [eng.seqs[s] for s in sids]
+    for _ in range(80):
+        eng.step()
+        if not eng.has_work():
+            break
+    torch.cuda.synchronize()
+    toks = [list(s.generated) for s in seqs]
+    assert all(len(t) == n for t in toks), toks
+    assert any(k[1] == 2 for k in eng._graphs), eng._graphs.keys()
+    # row 0: top_k=1 == greedy; row 2 greedy: compare to a pure-greedy run
+    eng2 = LlamaEngine(model, kv, max_batch=4)
+    g_sid = eng2.submit([2, 7, 1, 8, 2, 8], GenParams(max_tokens=n))
+    gseq = eng2.seqs[g_sid]
+    for _ in range(80):
+        eng2.step()
+        if not eng2.has_work():
+            break
+    torch.cuda.synchronize()
+    assert toks[0] == list(gseq.generated), "top_k=1 must equal greedy"
+    assert toks[2] == list(gseq.generated), "greedy row drifted"
+    # row 1 top_k=5: every decode token must be in its step's top-5.
+    # (weak check: diversity only — exact support check needs per-step
+    # logits; covered by sampler unit tests)
+    assert len(set(toks[1])) >= 1
