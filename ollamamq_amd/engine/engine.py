@@ -100,6 +100,17 @@ class LlamaEngine:
                              and _os.environ.get("OLLAMAMQ_NO_PIPELINE")
                              != "1")
         self._pipe = None   # {"key": seq-id tuple, "inflight": (ev,buf,seqs)}
+        # Decode batches are PADDED UP to a bucket size so the graph set
+        # is small and fixed: lazy mid-serving captures cost ~75 MB and
+        # ~150 ms each (the r02 soak caught a 328 MB tail drift from
+        # stragglers).  Padding is nearly free — the step is
+        # weight-streaming-bound, pad rows attend over a 1-token dummy
+        # slot and their lens/pos are advance-masked in-graph.
+        self._buckets = sorted({b for b in
+                                (1, 2, 3, 4, 6, 8, 12, 16, 24, 32,
+                                 48, 64, 96, 128, max_batch)
+                                if b <= max_batch})
+        self._dummy_slot = -1
 
     # -- submission --------------------------------------------------------
     def submit(self, prompt: List[int], params: GenParams,
@@ -389,6 +400,18 @@ class LlamaEngine:
                 k = 1
         return k
 
+    def _bucket(self, n: int) -> int:
+        for b in self._buckets:
+            if b >= n:
+                return b
+        return self._buckets[-1]
+
+    def _ensure_dummy(self) -> int:
+        if self._dummy_slot < 0:
+            self._dummy_slot = self.kv.alloc_slot()
+            self.kv.ensure(self._dummy_slot, 1)
+        return self._dummy_slot
+
     def _graph_entry(self, B: int, klass: int = 0):
         """Capture (once per (batch size, sampling class)) a full decode
         forward as a hipGraph reading its inputs from static device
@@ -406,6 +429,9 @@ class LlamaEngine:
             "seeds": torch.zeros(B, dtype=torch.int64, device=dev),
             "topk": torch.zeros(B, dtype=torch.int64, device=dev),
             "topp": torch.ones(B, dtype=torch.float32, device=dev),
+            # 1 for live rows, 0 for bucket padding: the in-graph
+            # advance applies it so pad rows never grow their dummy slot
+            "adv": torch.ones(B, dtype=torch.int32, device=dev),
         }
         meta = AttnMeta(
             mode="decode", slot_ids=bufs["slot"], seq_lens=bufs["lens"],
@@ -423,28 +449,36 @@ class LlamaEngine:
 
     def _fill_bufs(self, entry, seqs, token_list, pos_list):
         bufs = entry["bufs"]
+        B = bufs["tok"].shape[0]
+        pad = B - len(seqs)
+        dummy = self._ensure_dummy() if pad else 0
         host = torch.tensor(
-            [token_list, pos_list, [s.slot for s in seqs],
-             [self.kv.seq_lens[s.slot] for s in seqs]], dtype=torch.int32)
+            [token_list + [0] * pad,
+             pos_list + [0] * pad,
+             [s.slot for s in seqs] + [dummy] * pad,
+             [self.kv.seq_lens[s.slot] for s in seqs] + [1] * pad,
+             [1] * len(seqs) + [0] * pad], dtype=torch.int32)
         staged = host.to(self.dev, non_blocking=True)
         bufs["tok"].copy_(staged[0])
         bufs["pos"].copy_(staged[1])
         bufs["slot"].copy_(staged[2])
         bufs["lens"].copy_(staged[3])
+        bufs["adv"].copy_(staged[4])
         if entry["klass"] >= 1:
             bufs["temps"].copy_(torch.tensor(
-                [s.params.temperature for s in seqs],
+                [s.params.temperature for s in seqs] + [0.0] * pad,
                 dtype=torch.float32).to(self.dev, non_blocking=True))
             bufs["seeds"].copy_(torch.tensor(
-                [getattr(s, "noise_seed", 1234) for s in seqs],
+                [getattr(s, "noise_seed", 1234) for s in seqs]
+                + [0] * pad,
                 dtype=torch.int64).to(self.dev, non_blocking=True))
         if entry["klass"] == 2:
             bufs["topk"].copy_(torch.tensor(
-                [s.params.top_k or 0 for s in seqs],
+                [s.params.top_k or 0 for s in seqs] + [0] * pad,
                 dtype=torch.int64).to(self.dev, non_blocking=True))
             bufs["topp"].copy_(torch.tensor(
                 [s.params.top_p if s.params.top_p else 1.0
-                 for s in seqs],
+                 for s in seqs] + [1.0] * pad,
                 dtype=torch.float32).to(self.dev, non_blocking=True))
 
     def _graph_replay(self, entry):
@@ -522,8 +556,8 @@ class LlamaEngine:
                                           None)
                         bufs["tok"].copy_(toks)
                     entry["out"] = toks
-                    bufs["pos"] += 1
-                    bufs["lens"] += 1
+                    bufs["pos"] += bufs["adv"]
+                    bufs["lens"] += bufs["adv"]
             finally:
                 gc.enable()
             entry["graph"] = g
@@ -531,17 +565,18 @@ class LlamaEngine:
         return entry["logits"]
 
     def _decode_forward_graphed(self, seqs, token_list, pos_list):
-        entry = self._graph_entry(len(seqs))   # class-0 graph: the host
-        self._fill_bufs(entry, seqs, token_list, pos_list)   # samples
-        return self._graph_replay(entry)
+        entry = self._graph_entry(self._bucket(len(seqs)))  # class-0:
+        self._fill_bufs(entry, seqs, token_list, pos_list)  # host samples
+        return self._graph_replay(entry)[:len(seqs)]
 
     def warm_graphs(self, sizes=None):
-        """Pre-capture decode graphs for common batch sizes at load time
-        (capture costs ~150 ms each; lazily hitting them mid-serving adds
-        first-request jitter).  Uses throwaway KV slots with one token."""
+        """Pre-capture the class-0 decode graphs for EVERY bucket at load
+        time (capture costs ~150 ms each; with bucketing this covers all
+        greedy workloads — no mid-serving capture, bounded VRAM).
+        Sampled classes (1, 2) still capture lazily on first use."""
         if not self.use_graphs:
             return
-        sizes = [b for b in (sizes or (1, 2, 4, 8, 16, self.max_batch))
+        sizes = [b for b in (sizes or self._buckets)
                  if 0 < b <= self.max_batch]
         for b in sorted(set(sizes)):
             slots = []
@@ -608,7 +643,7 @@ class LlamaEngine:
         # host/device page bookkeeping for the token this replay appends
         for s in seqs:
             self.kv.ensure(s.slot, self.kv.seq_lens[s.slot] + 1)
-        entry = self._graph_entry(len(seqs), key[1])
+        entry = self._graph_entry(self._bucket(len(seqs)), key[1])
         if self._pipe is None:
             # (re)sync device input buffers from host truth
             self._fill_bufs(entry, seqs,
