@@ -250,6 +250,35 @@ void run_tui(Server& server) {
                 << " done:" << u.processed << " drop:" << u.dropped
                 << "\r\n";
         }
+        {   // Queues panel: per-user load bars (queued + in flight as a
+            // share of the total — reference tui.rs:1124-1163; VIP '*',
+            // boost '+', processing shown cyan)
+            int64_t total_q = 0;
+            for (const auto& u : s.users)
+                total_q += (int64_t)u.queued + u.processing;
+            out << "\x1b[7m" << pad(" Queues (" +
+                                    std::to_string(total_q) +
+                                    " waiting/in-flight)", 90)
+                << "\x1b[0m\r\n";
+            for (const auto& u : s.users) {
+                const int64_t ql = (int64_t)u.queued + u.processing;
+                const int barw = (int)(std::min<double>(ql / 20.0, 1.0)
+                                       * 36.0);
+                const double pct = total_q > 0
+                                       ? 100.0 * (double)ql / total_q
+                                       : 0.0;
+                std::string bar(barw, '#');
+                const char* col = u.vip ? "\x1b[35m"
+                                  : u.boost ? "\x1b[33m"
+                                  : u.processing > 0 ? "\x1b[36m"
+                                                     : "\x1b[32m";
+                char pbuf[32];
+                snprintf(pbuf, sizeof pbuf, "%lld (%.0f%%)",
+                         (long long)ql, pct);
+                out << "  " << pad(u.name, 18) << col << pad(bar, 38)
+                    << "\x1b[0m " << pbuf << "\r\n";
+            }
+        }
         if (!s.blocked.empty()) {
             out << "\x1b[7m" << pad(" Blocked (u unblock)", 90)
                 << "\x1b[0m\r\n";

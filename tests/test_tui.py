@@ -116,3 +116,31 @@ def test_tui_model_cursor_direct_load(tmp_path):
             "cursor-load never hit the backend"
     finally:
         fleet.stop()
+
+
+def test_tui_queues_panel(tmp_path):
+    """The Queues panel renders per-user load bars (reference
+    tui.rs:1124-1163)."""
+    master, slave = pty.openpty()
+    p = subprocess.Popen(
+        [BIN, "-p", "0", "-c", os.path.join(str(tmp_path), "absent.yaml")],
+        stdin=slave, stdout=slave, stderr=subprocess.DEVNULL,
+        cwd=str(tmp_path))
+    os.close(slave)
+    out = b""
+    deadline = time.time() + 20
+    try:
+        while time.time() < deadline:
+            try:
+                out += os.read(master, 65536)
+            except OSError:
+                break
+            if b"Queues" in out:
+                os.write(master, b"q")
+                break
+        p.wait(timeout=10)
+    finally:
+        os.close(master)
+        if p.poll() is None:
+            p.terminate()
+    assert b"Queues" in out, out[-400:]
