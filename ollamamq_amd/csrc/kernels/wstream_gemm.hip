@@ -337,31 +337,35 @@ __global__ __launch_bounds__(512) void k_wstream_gemm(
             rstd_sh[m2] = rsqrtf(s * inv_h + eps);
     }
     if (GU) {
-        __syncthreads();
-        #pragma unroll
-        for (int r = 0; r < 16; r++) {
-            const int crow = (r & 3) + 8 * (r >> 2) + 4 * (lane >> 5);
-            red8[wid][crow][lane & 31] = acc0[r];
-        }
-        __syncthreads();
         const int F = N >> 1;                  // ffn width
-        const int m = tid >> 4, c = tid & 15;  // 512 threads = 32 x 16
-        if (m < M) {
-            float sg = 0.f, su = 0.f;
+        for (int mt = 0; mt < MT; ++mt) {
+            const f32x16& acc = mt ? acc1 : acc0;
+            __syncthreads();
             #pragma unroll
-            for (int wv = 0; wv < 8; wv++) {
-                sg += red8[wv][m][c];
-                su += red8[wv][m][c + 16];
+            for (int r = 0; r < 16; r++) {
+                const int crow = (r & 3) + 8 * (r >> 2) + 4 * (lane >> 5);
+                red8[wid][crow][lane & 31] = acc[r];
             }
-            if (rstd_parts) {
-                const float rs = rstd_sh[m];   // pre-silu: nonlinear
-                sg *= rs;
-                su *= rs;
+            __syncthreads();
+            const int m = tid >> 4, c = tid & 15;  // 512 thr = 32 x 16
+            const int gm = m + mt * 32;
+            if (gm < M) {
+                float sg = 0.f, su = 0.f;
+                #pragma unroll
+                for (int wv = 0; wv < 8; wv++) {
+                    sg += red8[wv][m][c];
+                    su += red8[wv][m][c + 16];
+                }
+                if (rstd_parts) {              // MT==1 chain only
+                    const float rs = rstd_sh[gm];  // pre-silu: nonlinear
+                    sg *= rs;
+                    su *= rs;
+                }
+                const float act = (sg / (1.f + __expf(-sg))) * su;
+                const int fc = t * 16 + c;
+                y[yfrag ? frag_off(gm, fc) : (int64_t)gm * F + fc] =
+                    __float2bfloat16(act);
             }
-            const float act = (sg / (1.f + __expf(-sg))) * su;
-            const int fc = t * 16 + c;
-            y[yfrag ? frag_off(m, fc) : (int64_t)m * F + fc] =
-                __float2bfloat16(act);
         }
         return;
     }
@@ -510,6 +514,16 @@ extern "C" int wstream_gu_bf16(
 {
     dim3 grid(N / 32, 1);
     const int lds = (xlds == 1 ? 16 * 4096 : 8 * 32 * 32 * 4) + 128;
+    if (M > 32) {
+        // generic-path fused gate_up+SwiGLU for decode batches 33..64
+        // (direct x loads; the frag chain stays batch<=32)
+        if (rstd_parts || yfrag || xlds) return -102;
+        k_wstream_gemm<2, 1, 0, 1><<<grid, 512, lds, stream>>>(
+            (bf16*)act, nullptr, (const bf16*)x, (const u32x4*)wp,
+            nullptr, M, N, K, xs, 1, nullptr, 0, 0.f, 0.f,
+            nullptr, nullptr, {}, 0);
+        return (int)hipGetLastError();
+    }
     if (xlds == 2)
         k_wstream_gemm<1, 1, 2, 1><<<grid, 512, lds, stream>>>(
             (bf16*)act, nullptr, (const bf16*)x, (const u32x4*)wp,

@@ -185,6 +185,16 @@ class LlamaLayer:
         wd_g = _randn((h, F), dev, dtype, gen, 1.0 / math.sqrt(cfg.ffn))
         self.wdown = wd_g[:, rank * f:(rank + 1) * f].contiguous()
         del wd_g
+        # rmsnorm weights are FOLDED into the projections at build
+        # (rmsnorm(x)*w @ W^T == rmsnorm_unit(x) @ (W diag w)^T — a pure
+        # reparameterization), so every execution path (lib GEMM, packed
+        # stream, fused chain) shares one consistent weight set and the
+        # runtime norm weights are identically ones.  A checkpoint
+        # loader must apply the same fold.
+        attn_w = torch.ones(h, device=dev, dtype=dtype)
+        mlp_w = torch.ones(h, device=dev, dtype=dtype)
+        self.wqkv = (self.wqkv * attn_w).contiguous()
+        self.wgate_up = (self.wgate_up * mlp_w).contiguous()
         self.attn_norm = torch.ones(h, device=dev, dtype=dtype)
         self.mlp_norm = torch.ones(h, device=dev, dtype=dtype)
         # packed decode copies filled by LlamaModel._pack_weights()
@@ -233,10 +243,11 @@ class LlamaModel:
         ]
         self.final_norm = torch.ones(cfg.hidden, device=dev, dtype=dtype)
         # Vocab-parallel LM head shard: rows of the global [vocab, hidden]
+        # (final_norm folds in like the per-layer norms — see LlamaLayer)
         vshard = cfg.vocab // tp_size
         lm_g = _randn((cfg.vocab, cfg.hidden), dev, dtype, gen, scale)
-        self.lm_head = lm_g[tp_rank * vshard:(tp_rank + 1) * vshard]\
-            .contiguous()
+        self.lm_head = (lm_g[tp_rank * vshard:(tp_rank + 1) * vshard]
+                        * self.final_norm).contiguous()
         del lm_g
         # RoPE cos/sin tables precomputed on host (guide: trig on device
         # turns memory-bound RoPE into VALU-bound).
@@ -286,11 +297,13 @@ class LlamaModel:
         try:
             for l in self.layers:
                 # qkv pack is PAIR-ORDERED for the fused RoPE/KV-append
-                # epilogue (head_dim==128 models; the chain requires it)
+                # epilogue (head_dim==128 models; the chain requires it).
+                # Norm weights are already folded at build, so the packs
+                # are pure layout transforms of the live weights.
                 if self.cfg.head_dim == 128:
                     l.wqkv_pk = ops.pack_weight_qkv_rope(
-                        (l.wqkv * l.attn_norm).contiguous(),
-                        self.n_local_heads, self.n_local_kv_heads)
+                        l.wqkv, self.n_local_heads,
+                        self.n_local_kv_heads)
                     if l.bqkv is not None:
                         l.bqkv_rp = ops.qkv_rope_bias_order(
                             l.bqkv, self.n_local_heads,
@@ -298,11 +311,9 @@ class LlamaModel:
                 else:
                     l.wqkv_pk = None
                 l.wo_pk = pk(l.wo)
-                l.wgu_pk = ops.pack_weight_gu(
-                    (l.wgate_up * l.mlp_norm).contiguous())
+                l.wgu_pk = ops.pack_weight_gu(l.wgate_up)
                 l.wdown_pk = pk(l.wdown)
-            self.lm_head_pk = pk((self.lm_head * self.final_norm)
-                                 .contiguous())
+            self.lm_head_pk = pk(self.lm_head)
         except torch.cuda.OutOfMemoryError:
             for l in self.layers:
                 l.wqkv_pk = l.wo_pk = l.wgu_pk = l.wdown_pk = None
@@ -388,7 +399,8 @@ class LlamaModel:
             normed, residual = ops.rmsnorm_residual(
                 x, residual, layer.mlp_norm, cfg.norm_eps
             )
-            act = ops.gateup_swiglu(normed, layer.wgate_up)
+            act = ops.gateup_swiglu(normed, layer.wgate_up,
+                                    layer.wgu_pk)
             x = ops.linear(act, layer.wdown, packed=layer.wdown_pk)
             self._allreduce(x)  # RCCL all-reduce #2 (TP)
 
@@ -401,7 +413,7 @@ class LlamaModel:
         h, _ = ops.rmsnorm_residual(h, None, self.final_norm, cfg.norm_eps)
         if return_hidden:
             return h
-        logits = ops.linear(h, self.lm_head)
+        logits = ops.linear(h, self.lm_head, packed=self.lm_head_pk)
         if self.tp_size > 1:
             # vocab-parallel logits: all-gather shards on the last dim
             shards = [torch.empty_like(logits) for _ in range(self.tp_size)]

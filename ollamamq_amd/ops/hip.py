@@ -469,7 +469,7 @@ def linear_gu(x, packed, N, rstd=None, rstd_nt=0, inv_h=0.0, eps=0.0,
     if K is None:
         M, K = x.shape
         xs = x.stride(0)
-        xlds = 1 if N * K * 2 > (64 << 20) else 0
+        xlds = (1 if N * K * 2 > (64 << 20) else 0) if M <= 32 else 0
     else:
         M, xs, xlds = 32, 0, 2
     act = torch.empty((M, F) if not yfrag else (32 * F,),

@@ -477,7 +477,8 @@ def test_wstream_gu_fused():
     """Fused gate_up+SwiGLU kernel vs F.linear + fp32-reference swiglu."""
     hip = _hip()
     for M, F, K in [(32, 14336, 4096), (1, 14336, 4096), (7, 512, 512),
-                    (32, 9472, 3584)]:
+                    (32, 9472, 3584), (64, 14336, 4096),
+                    (47, 512, 512)]:
         x = rnd(M, K, seed=M + F)
         w = rnd(2 * F, K, seed=K + F)
         pk = hip.pack_weight_gu(w)
