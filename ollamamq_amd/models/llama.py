@@ -413,7 +413,9 @@ class LlamaModel:
         h, _ = ops.rmsnorm_residual(h, None, self.final_norm, cfg.norm_eps)
         if return_hidden:
             return h
-        logits = ops.linear(h, self.lm_head, packed=self.lm_head_pk)
+        # NB: no packed= — the MT2 streaming GEMM loses to the library
+        # above batch 32 (doubled x scatter); the fused chain covers <=32
+        logits = ops.linear(h, self.lm_head)
         if self.tp_size > 1:
             # vocab-parallel logits: all-gather shards on the last dim
             shards = [torch.empty_like(logits) for _ in range(self.tp_size)]

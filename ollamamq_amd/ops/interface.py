@@ -134,10 +134,12 @@ def swiglu(gate_up):
 
 
 def gateup_swiglu(x, weight, packed_gu=None):
-    """act = swiglu(x @ Wgu^T).  On the decode path (M<=64) with a
+    """act = swiglu(x @ Wgu^T).  On the decode path (M<=32) with a
     GU-packed copy this is ONE fused weight-streaming kernel; otherwise
-    the library GEMM + the swiglu kernel."""
-    if packed_gu is not None and x.shape[0] <= 64 and _use_hip(x):
+    the library GEMM + the swiglu kernel.  (An MT2 variant for 33..64
+    exists but measured BELOW the library at 64 users — its direct
+    fragment x-loads double the scattered line traffic; see NOTES.)"""
+    if packed_gu is not None and x.shape[0] <= 32 and _use_hip(x):
         from . import hip
         if hip.USE_WSTREAM and x.dtype == torch.bfloat16 \
                 and x.stride(1) == 1:
