@@ -90,7 +90,7 @@ class LlamaEngine:
         import os as _os
         self.use_graphs = (self.dev.type == "cuda"
                            and _os.environ.get("OLLAMAMQ_NO_GRAPH") != "1")
-        self._graphs: Dict[int, tuple] = {}
+        self._graphs: Dict[tuple, dict] = {}   # (batch bucket, class)
         # Pipelined decode: the graph samples (greedy argmax) and advances
         # tok/pos/lens on-device, so steady-state steps skip the H2D input
         # fill AND the host token sync — the host consumes tokens one step

@@ -601,8 +601,7 @@ def linear_packed(x, packed, bias, N, ks=None, depth=None, xlds=None,
         else:
             ks = int(_WS_KS) if _WS_KS else _wstream_ksplit(N, K)
     if rstd is not None:
-        ks_ok = ks        # rstd fusion requires ks == 1
-        ks = 1
+        ks = 1            # rstd fusion requires ks == 1
     part = ctypes.c_void_p(0)
     if ks > 1:
         key = ("ws", M, N, ks, str(x.device))
