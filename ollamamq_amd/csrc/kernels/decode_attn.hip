@@ -440,3 +440,56 @@ extern "C" int decode_attn_bf16(
     }
     return (int)hipGetLastError();
 }
+
+// Pure-stage diagnostic: identical grid/segmentation/addressing to
+// k_decode_attn but ONLY the KV loads (no LDS, no score/PV, no
+// barriers) — the staging stream ceiling of this geometry.  If this
+// matches the real kernel's time, the kernel is geometry/bandwidth
+// capped and compute-side pipelining cannot help.
+__global__ __launch_bounds__(256) void k_decode_pure(
+    float* __restrict__ sink, const bf16* __restrict__ kpool,
+    const bf16* __restrict__ vpool, const int* __restrict__ page_table,
+    const int* __restrict__ slot_ids, const int* __restrict__ seq_lens,
+    int KVH, int page, int max_pages, int split, int chunk)
+{
+    const int S_idx = blockIdx.x / split;
+    const int seg = blockIdx.x % split;
+    const int kvh = blockIdx.y;
+    const int slot = slot_ids[S_idx];
+    const int kv_len = seq_lens[S_idx];
+    const int n_chunks = (kv_len + chunk - 1) / chunk;
+    const int c0 = (int)(((int64_t)n_chunks * seg) / split);
+    const int c1 = (int)(((int64_t)n_chunks * (seg + 1)) / split);
+    const int nslot = chunk * 16;
+    unsigned acc = 0;
+    for (int ch = c0; ch < c1; ch++) {
+        const int base = ch * chunk;
+        for (int u = threadIdx.x; u < nslot; u += 256) {
+            const int tok = u >> 4, dv = u & 15;
+            const int tk = base + tok < kv_len ? base + tok : kv_len - 1;
+            const int gp = page_table[(int64_t)slot * max_pages
+                                      + tk / page];
+            const int64_t src =
+                (((int64_t)gp * KVH + kvh) * page + tk % page) * 128
+                + dv * 8;
+            const uint4 k = *reinterpret_cast<const uint4*>(kpool + src);
+            const uint4 v = *reinterpret_cast<const uint4*>(vpool + src);
+            acc ^= k.x ^ k.w ^ v.x ^ v.w;
+        }
+    }
+    if (acc == 0xDEADBEEFu) sink[0] = 1.f;
+}
+
+extern "C" int decode_pure_bf16(
+    void* sink, const void* kpool, const void* vpool,
+    const void* page_table, const void* slot_ids, const void* seq_lens,
+    int S, int KVH, int page, int max_pages, int split, int chunk,
+    hipStream_t stream)
+{
+    dim3 grid(S * split, KVH);
+    k_decode_pure<<<grid, 256, 0, stream>>>(
+        (float*)sink, (const bf16*)kpool, (const bf16*)vpool,
+        (const int*)page_table, (const int*)slot_ids,
+        (const int*)seq_lens, KVH, page, max_pages, split, chunk);
+    return (int)hipGetLastError();
+}

@@ -63,6 +63,8 @@ def _try_load():
     lib.fragify_sumsq_bf16.argtypes = [vp, vp, vp, i, i, vp]
     lib.sample_gumbel_bf16.argtypes = [vp, vp, vp, vp, vp, vp, i, i,
                                        vp, vp, i, vp]
+    lib.decode_pure_bf16.argtypes = [vp, vp, vp, vp, vp, vp, i, i, i, i,
+                                     i, i, vp]
     lib.wstream_qkv_rope_bf16.argtypes = [vp, vp, vp, vp, i, i, i, i64,
                                           vp, i, f, f, vp, vp, vp, vp,
                                           vp, vp, vp, i, i, i, i, i, vp]
@@ -759,3 +761,17 @@ def sample(logits, temperature, top_k, top_p, generator=None):
         return out.long()
     # stochastic paths compose on-GPU torch ops (sort/softmax/multinomial)
     return ref.sample(logits, temperature, top_k, top_p, generator)
+
+
+def decode_pure(cache, layer, meta, split, chunk=64):
+    """Staging-only diagnostic for decode attention (see decode_attn.hip
+    k_decode_pure)."""
+    sink = torch.zeros(1, dtype=torch.float32, device=cache.k_pool.device)
+    kp, vp = _layer_ptrs(cache, layer)
+    S = meta.slot_ids.shape[0]
+    _check(_lib.decode_pure_bf16(
+        _p(sink), kp, vp, _p(cache.page_table), _p(meta.slot_ids.int()),
+        _p(meta.seq_lens.int()), S, cache.n_kv_heads, cache.page_size,
+        cache.page_table.shape[1], split, chunk, _stream()),
+        "decode_pure")
+    return sink

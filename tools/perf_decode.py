@@ -60,6 +60,9 @@ def main():
 
     o_new = bench(lambda l: hip.attention_decode(q, cache, l, meta),
                   "split ")
+    for sp in (1, 2, 4, 8):
+        bench(lambda l, sp=sp: hip.decode_pure(cache, l, meta, sp),
+              f"pure{sp}")
     o_old = bench(lambda l: hip._attention(q, cache, l, meta, 1), "legacy")
     diff = (o_new.float() - o_old.float()).abs().max().item()
     print("max |new-old| =", diff)
