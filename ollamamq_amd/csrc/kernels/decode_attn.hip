@@ -287,7 +287,9 @@ __global__ __launch_bounds__(GT * 64, 1) void k_decode_attn(
             __float22bfloat162_rn(make_float2(o0 * linv, o1 * linv));
         if (fragout)
             *reinterpret_cast<bf162*>(
-                out + attn_frag_off(S_idx, qh * DHEAD + 2 * lane)) = ov;
+                out + (int64_t)(S_idx >> 5) * 32 * Hq * DHEAD
+                    + attn_frag_off(S_idx & 31, qh * DHEAD + 2 * lane))
+                = ov;
         else
             reinterpret_cast<bf162*>(
                 out + ((int64_t)S_idx * Hq + qh) * DHEAD)[lane] = ov;
@@ -348,7 +350,8 @@ __global__ __launch_bounds__(GT * 64, 1) void k_decode_attn(
         const float linv = ll > 0.f ? 1.f / ll : 0.f;
         if (fragout) {
             *reinterpret_cast<bf162*>(
-                out + attn_frag_off(S_idx, qh * DHEAD + 2 * lane)) =
+                out + (int64_t)(S_idx >> 5) * 32 * Hq * DHEAD
+                    + attn_frag_off(S_idx & 31, qh * DHEAD + 2 * lane)) =
                 __float22bfloat162_rn(
                     make_float2(a0 * linv, a1 * linv));
             return;
@@ -387,8 +390,9 @@ __global__ __launch_bounds__(256) void k_decode_combine(
         __float22bfloat162_rn(make_float2(a0 * linv, a1 * linv));
     if (fragout)
         *reinterpret_cast<bf162*>(
-            out + attn_frag_off(sh / Hq, (sh % Hq) * DHEAD + 2 * lane))
-            = ov;
+            out + (int64_t)((sh / Hq) >> 5) * 32 * Hq * DHEAD
+                + attn_frag_off((sh / Hq) & 31,
+                                (sh % Hq) * DHEAD + 2 * lane)) = ov;
     else
         reinterpret_cast<bf162*>(out + (int64_t)sh * DHEAD)[lane] = ov;
 }

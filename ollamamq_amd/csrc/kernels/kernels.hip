@@ -903,9 +903,12 @@ __global__ __launch_bounds__(256) void k_fragify_sumsq(
 {
     const int m = blockIdx.x;
     const bf16* row = x + (int64_t)m * H;
+    // rows 32-63 land in the second 32-row frag half (MT2 chain)
+    bf16* xfh = xf + (int64_t)(m >> 5) * 32 * H;
+    const int mr = m & 31;
     float s = 0.f;
     // 16-byte chunk c covers k = c*8..c*8+8: frag unit
-    // ((c>>3)*4 + ((c>>1)&3))*64 + (c&1)*32 + m
+    // ((c>>3)*4 + ((c>>1)&3))*64 + (c&1)*32 + mr
     for (int c = threadIdx.x; c < H / 8; c += 256) {
         const uint4 v = *reinterpret_cast<const uint4*>(row + c * 8);
         float f[8];
@@ -913,8 +916,8 @@ __global__ __launch_bounds__(256) void k_fragify_sumsq(
         #pragma unroll
         for (int e = 0; e < 8; e++) s += f[e] * f[e];
         const int64_t u = ((int64_t)(c >> 3) * 4 + ((c >> 1) & 3)) * 64
-                          + (c & 1) * 32 + m;
-        *reinterpret_cast<uint4*>(xf + u * 8) = v;
+                          + (c & 1) * 32 + mr;
+        *reinterpret_cast<uint4*>(xfh + u * 8) = v;
     }
     __shared__ float red[4];
     s = wave_reduce_sum(s);

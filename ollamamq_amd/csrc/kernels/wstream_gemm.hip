@@ -185,7 +185,11 @@ __global__ __launch_bounds__(512) void k_wstream_gemm(
         const u4bf8* xf = reinterpret_cast<const u4bf8*>(x);
         auto lda16 = [&](bf16x8* d, int b, int off) {
             if (XLDS == 2) {   // frag input: linear like the weights
-                d[0] = xf[((int64_t)b * 4 + (off >> 4)) * 64 + lane].v;
+                const int64_t u = ((int64_t)b * 4 + (off >> 4)) * 64
+                                  + lane;
+                d[0] = xf[u].v;
+                if (MT == 2)   // second 32-row frag half
+                    d[1] = xf[u + (int64_t)4 * K].v;
                 return;
             }
             const int64_t k = ((int64_t)b << 6) + off;
@@ -322,18 +326,20 @@ __global__ __launch_bounds__(512) void k_wstream_gemm(
     float* rstd_sh = reinterpret_cast<float*>(
         smem + (XLDS == 1 ? 16 * 4096 : 8 * 32 * 32 * 4));
     if (rstd_parts) {
-        // parallel partial fold: 16 lanes per row (each <= nt/16 loads,
+        // parallel partial fold: 8-16 lanes per row (each <= nt/L loads,
         // independent), group-reduced with wave shuffles — a serial
-        // 32-thread loop here measurably stalled short-stream blocks
-        const int m2 = tid >> 4, c2 = tid & 15;
+        // 32-thread loop here measurably stalled short-stream blocks.
+        // MT==2 folds 64 rows with 8 lanes each (512 threads exactly).
+        const int L = MT == 2 ? 8 : 16;
+        const int m2 = tid / L, c2 = tid % L;
         float s = 0.f;
-        if (m2 < 32)
-            for (int i = c2; i < rstd_nt; i += 16)
+        if (m2 < MT * 32)
+            for (int i = c2; i < rstd_nt; i += L)
                 s += rstd_parts[(int64_t)m2 * rstd_nt + i];
         #pragma unroll
         for (int off = 8; off; off >>= 1)
-            s += __shfl_down(s, off, 64);
-        if (m2 < 32 && c2 == 0)
+            if (off < L) s += __shfl_down(s, off, 64);
+        if (m2 < MT * 32 && c2 == 0)
             rstd_sh[m2] = rsqrtf(s * inv_h + eps);
     }
     if (GU) {
@@ -363,86 +369,91 @@ __global__ __launch_bounds__(512) void k_wstream_gemm(
                 }
                 const float act = (sg / (1.f + __expf(-sg))) * su;
                 const int fc = t * 16 + c;
-                y[yfrag ? frag_off(gm, fc) : (int64_t)gm * F + fc] =
-                    __float2bfloat16(act);
+                y[yfrag ? (int64_t)mt * 32 * F + frag_off(m, fc)
+                        : (int64_t)gm * F + fc] = __float2bfloat16(act);
             }
         }
         return;
     }
     if (RP) {
-        __syncthreads();
-        #pragma unroll
-        for (int r = 0; r < 16; r++) {
-            const int crow = (r & 3) + 8 * (r >> 2) + 4 * (lane >> 5);
-            red8[wid][crow][lane & 31] = acc0[r];
-        }
-        __syncthreads();
         const int qt = rp.nl * 4, kt = rp.nkl * 4;   // tiles per section
-        if (t < qt + kt) {
-            // pair-packed q/k tile: cols {d0+c, d0+64+c}, c<16
-            const int m = tid >> 4, c = tid & 15;
-            if (m < M) {
-                const int sec_t = t < qt ? t : t - qt;
-                const int head = sec_t >> 2;
-                const int d0 = (sec_t & 3) * 16;
-                float lo = 0.f, hi = 0.f;
-                #pragma unroll
-                for (int wv = 0; wv < 8; wv++) {
-                    lo += red8[wv][m][c];
-                    hi += red8[wv][m][c + 16];
-                }
-                if (rstd_parts) {
-                    const float rs = rstd_sh[m];
-                    lo *= rs;
-                    hi *= rs;
-                }
-                if (bias) {   // bias is pair-reordered like the pack
-                    lo += __bfloat162float(bias[t * 32 + c]);
-                    hi += __bfloat162float(bias[t * 32 + 16 + c]);
-                }
-                const int p = rp.pos[m];
-                const float co = rp.cos_t[p * 64 + d0 + c];
-                const float si = rp.sin_t[p * 64 + d0 + c];
-                const float rlo = lo * co - hi * si;
-                const float rhi = hi * co + lo * si;
-                if (t < qt) {          // q: standard layout in y
-                    bf16* qr = y + (int64_t)m * N + head * 128;
-                    qr[d0 + c] = __float2bfloat16(rlo);
-                    qr[64 + d0 + c] = __float2bfloat16(rhi);
-                } else {               // k: rotated, straight to pool
-                    const int pg = rp.ptab[rp.slot[m] * rp.maxp
-                                           + p / rp.psz];
-                    bf16* dst = rp.kp
-                        + (((int64_t)pg * rp.nkl + head) * rp.psz
-                           + p % rp.psz) * 128;
-                    dst[d0 + c] = __float2bfloat16(rlo);
-                    dst[64 + d0 + c] = __float2bfloat16(rhi);
-                }
-            }
-        } else {
-            // v tile (plain 32-col order): straight to pool, no rope
-            const int sec_t = t - qt - kt;
-            const int head = sec_t >> 2;
-            const int d0 = (sec_t & 3) * 32;
+        for (int mt = 0; mt < MT; ++mt) {
+            const f32x16& acc = mt ? acc1 : acc0;
+            __syncthreads();
             #pragma unroll
-            for (int ee = 0; ee < 2; ee++) {
-                const int e = tid + ee * 512;
-                const int m = e >> 5, n = e & 31;
-                if (m < M) {
-                    float sv = 0.f;
+            for (int r = 0; r < 16; r++) {
+                const int crow = (r & 3) + 8 * (r >> 2) + 4 * (lane >> 5);
+                red8[wid][crow][lane & 31] = acc[r];
+            }
+            __syncthreads();
+            if (t < qt + kt) {
+                // pair-packed q/k tile: cols {d0+c, d0+64+c}, c<16
+                const int m = tid >> 4, c = tid & 15;
+                const int gm = m + mt * 32;
+                if (gm < M) {
+                    const int sec_t = t < qt ? t : t - qt;
+                    const int head = sec_t >> 2;
+                    const int d0 = (sec_t & 3) * 16;
+                    float lo = 0.f, hi = 0.f;
                     #pragma unroll
-                    for (int wv = 0; wv < 8; wv++)
-                        sv += red8[wv][m][n];
-                    if (rstd_parts) sv *= rstd_sh[m];
-                    if (bias)
-                        sv += __bfloat162float(bias[t * 32 + n]);
-                    const int p = rp.pos[m];
-                    const int pg = rp.ptab[rp.slot[m] * rp.maxp
-                                           + p / rp.psz];
-                    bf16* dst = rp.vp
-                        + (((int64_t)pg * rp.nkl + head) * rp.psz
-                           + p % rp.psz) * 128;
-                    dst[d0 + n] = __float2bfloat16(sv);
+                    for (int wv = 0; wv < 8; wv++) {
+                        lo += red8[wv][m][c];
+                        hi += red8[wv][m][c + 16];
+                    }
+                    if (rstd_parts) {
+                        const float rs = rstd_sh[gm];
+                        lo *= rs;
+                        hi *= rs;
+                    }
+                    if (bias) {  // bias is pair-reordered like the pack
+                        lo += __bfloat162float(bias[t * 32 + c]);
+                        hi += __bfloat162float(bias[t * 32 + 16 + c]);
+                    }
+                    const int p = rp.pos[gm];
+                    const float co = rp.cos_t[p * 64 + d0 + c];
+                    const float si = rp.sin_t[p * 64 + d0 + c];
+                    const float rlo = lo * co - hi * si;
+                    const float rhi = hi * co + lo * si;
+                    if (t < qt) {          // q: standard layout in y
+                        bf16* qr = y + (int64_t)gm * N + head * 128;
+                        qr[d0 + c] = __float2bfloat16(rlo);
+                        qr[64 + d0 + c] = __float2bfloat16(rhi);
+                    } else {               // k: rotated, straight to pool
+                        const int pg = rp.ptab[rp.slot[gm] * rp.maxp
+                                               + p / rp.psz];
+                        bf16* dst = rp.kp
+                            + (((int64_t)pg * rp.nkl + head) * rp.psz
+                               + p % rp.psz) * 128;
+                        dst[d0 + c] = __float2bfloat16(rlo);
+                        dst[64 + d0 + c] = __float2bfloat16(rhi);
+                    }
+                }
+            } else {
+                // v tile (plain 32-col order): straight to pool, no rope
+                const int sec_t = t - qt - kt;
+                const int head = sec_t >> 2;
+                const int d0 = (sec_t & 3) * 32;
+                #pragma unroll
+                for (int ee = 0; ee < 2; ee++) {
+                    const int e = tid + ee * 512;
+                    const int mr = e >> 5, n = e & 31;
+                    const int gm = mr + mt * 32;
+                    if (gm < M) {
+                        float sv = 0.f;
+                        #pragma unroll
+                        for (int wv = 0; wv < 8; wv++)
+                            sv += red8[wv][mr][n];
+                        if (rstd_parts) sv *= rstd_sh[gm];
+                        if (bias)
+                            sv += __bfloat162float(bias[t * 32 + n]);
+                        const int p = rp.pos[gm];
+                        const int pg = rp.ptab[rp.slot[gm] * rp.maxp
+                                               + p / rp.psz];
+                        bf16* dst = rp.vp
+                            + (((int64_t)pg * rp.nkl + head) * rp.psz
+                               + p % rp.psz) * 128;
+                        dst[d0 + n] = __float2bfloat16(sv);
+                    }
                 }
             }
         }
@@ -463,17 +474,19 @@ __global__ __launch_bounds__(512) void k_wstream_gemm(
         #pragma unroll
         for (int ee = 0; ee < 2; ee++) {
             const int e = tid + ee * 512;
-            const int m = (e >> 5) + mt * 32, n = e & 31;
+            const int mr = e >> 5, n = e & 31;
+            const int m = mr + mt * 32;
             if (m < M) {
                 float s = 0.f;
                 #pragma unroll
                 for (int wv = 0; wv < 8; wv++)
-                    s += red8[wv][m - mt * 32][n];
+                    s += red8[wv][mr][n];
                 if (ksplit == 1) {
                     if (rstd_parts) s *= rstd_sh[m];
                     if (bias) s += __bfloat162float(bias[n0 + n]);
-                    const int64_t yo = yfrag ? frag_off(m, n0 + n)
-                                             : (int64_t)m * N + n0 + n;
+                    const int64_t yo =
+                        yfrag ? (int64_t)mt * 32 * N + frag_off(mr, n0 + n)
+                              : (int64_t)m * N + n0 + n;
                     if (res_in) {
                         s += __bfloat162float(res_in[yo]);
                         rsq[ee] = s * s;
@@ -484,22 +497,23 @@ __global__ __launch_bounds__(512) void k_wstream_gemm(
                 }
             }
         }
-        // emit this block's sum-of-squares partials (M<=32, MT==1 only)
-        if (MT == 1 && sq_parts && ksplit == 1) {
+        // emit this block's sum-of-squares partials for the mt rows
+        if (sq_parts && ksplit == 1) {
             __syncthreads();
             float (*sqt)[32] = reinterpret_cast<float (*)[32]>(smem);
-            sqt[tid >> 5][tid & 31] = rsq[0];          // m 0..15
-            sqt[(tid >> 5) + 16][tid & 31] = rsq[1];   // m 16..31
+            sqt[tid >> 5][tid & 31] = rsq[0];          // rows 0..15
+            sqt[(tid >> 5) + 16][tid & 31] = rsq[1];   // rows 16..31
             __syncthreads();
             if (tid < 32) {
                 float s = 0.f;
                 #pragma unroll
                 for (int n2 = 0; n2 < 32; n2++) s += sqt[tid][n2];
-                // m-major [32][n_tiles]: the consumer's fold walks tiles
+                // m-major [M][n_tiles]: the consumer's fold walks tiles
                 // CONTIGUOUSLY (a tile-major layout cost 1 float per
                 // 128 B line and ~3 us per consumer block)
-                sq_parts[(int64_t)tid * gridDim.x + t] = s;
+                sq_parts[(int64_t)(tid + mt * 32) * gridDim.x + t] = s;
             }
+            if (MT == 2 && mt == 0) __syncthreads();  // sqt reused
         }
     }
 }
@@ -513,11 +527,18 @@ extern "C" int wstream_gu_bf16(
     float inv_h, float eps, int yfrag, hipStream_t stream)
 {
     dim3 grid(N / 32, 1);
-    const int lds = (xlds == 1 ? 16 * 4096 : 8 * 32 * 32 * 4) + 128;
+    const int lds = (xlds == 1 ? 16 * 4096 : 8 * 32 * 32 * 4) + 256;
     if (M > 32) {
-        // generic-path fused gate_up+SwiGLU for decode batches 33..64
-        // (direct x loads; the frag chain stays batch<=32)
-        if (rstd_parts || yfrag || xlds) return -102;
+        if (xlds == 2) {       // frag chain at batches 33..64
+            k_wstream_gemm<2, 1, 2, 1><<<grid, 512, lds, stream>>>(
+                (bf16*)act, nullptr, (const bf16*)x, (const u32x4*)wp,
+                nullptr, M, N, K, xs, 1, (const float*)rstd_parts,
+                rstd_nt, inv_h, eps, nullptr, nullptr, {}, yfrag);
+            return (int)hipGetLastError();
+        }
+        // generic-path fused gate_up+SwiGLU, direct x loads (unused by
+        // default: measured below lib at 64 users)
+        if (rstd_parts || yfrag) return -102;
         k_wstream_gemm<2, 1, 0, 1><<<grid, 512, lds, stream>>>(
             (bf16*)act, nullptr, (const bf16*)x, (const u32x4*)wp,
             nullptr, M, N, K, xs, 1, nullptr, 0, 0.f, 0.f,
@@ -569,31 +590,37 @@ __global__ __launch_bounds__(256) void k_wstream_combine_tiles(
     const int tid = threadIdx.x;
     __shared__ float sqt[32][32];
     const int n0 = t * 32;
-    #pragma unroll
-    for (int ee = 0; ee < 4; ee++) {
-        const int e = tid + ee * 256;
-        const int m = e >> 5, n = e & 31;
-        float rsq = 0.f;
-        if (m < M) {
-            float s = 0.f;
-            for (int k = 0; k < ksplit; k++)
-                s += part[((int64_t)k * M + m) * N + n0 + n];
-            const int64_t yo = yfrag ? frag_off(m, n0 + n)
-                                     : (int64_t)m * N + n0 + n;
-            if (res_in)
-                s += __bfloat162float(res_in[yo]);
-            y[yo] = __float2bfloat16(s);
-            rsq = s * s;
+    const int mts = (M + 31) >> 5;          // 1 or 2 m-halves
+    for (int mt = 0; mt < mts; ++mt) {
+        if (mt) __syncthreads();
+        #pragma unroll
+        for (int ee = 0; ee < 4; ee++) {
+            const int e = tid + ee * 256;
+            const int mr = e >> 5, n = e & 31;
+            const int m = mr + mt * 32;
+            float rsq = 0.f;
+            if (m < M) {
+                float s = 0.f;
+                for (int k = 0; k < ksplit; k++)
+                    s += part[((int64_t)k * M + m) * N + n0 + n];
+                const int64_t yo =
+                    yfrag ? (int64_t)mt * 32 * N + frag_off(mr, n0 + n)
+                          : (int64_t)m * N + n0 + n;
+                if (res_in)
+                    s += __bfloat162float(res_in[yo]);
+                y[yo] = __float2bfloat16(s);
+                rsq = s * s;
+            }
+            sqt[mr][n] = rsq;
         }
-        sqt[m][n] = rsq;
-    }
-    if (sq_parts) {
-        __syncthreads();
-        if (tid < 32) {
-            float s = 0.f;
-            #pragma unroll
-            for (int n2 = 0; n2 < 32; n2++) s += sqt[tid][n2];
-            sq_parts[(int64_t)tid * gridDim.x + t] = s;   // m-major
+        if (sq_parts) {
+            __syncthreads();
+            if (tid < 32) {
+                float s = 0.f;
+                #pragma unroll
+                for (int n2 = 0; n2 < 32; n2++) s += sqt[tid][n2];
+                sq_parts[(int64_t)(tid + mt * 32) * gridDim.x + t] = s;
+            }
         }
     }
 }
@@ -605,8 +632,8 @@ extern "C" int wstream_gemm_bf16(
     const void* res_in, void* sq_parts, int yfrag, hipStream_t stream)
 {
     dim3 grid(N / 32, ksplit);
-    const int lds_red = 8 * 32 * 32 * 4 + 128;     // reduce + rstd tail
-    const int lds_x = 16 * 4096 + 128;             // + 8x2 x tiles
+    const int lds_red = 8 * 32 * 32 * 4 + 256;     // reduce + rstd tail
+    const int lds_x = 16 * 4096 + 256;             // + 8x2 x tiles
     const float* rp = (const float*)rstd_parts;
     const bf16* ri = ksplit == 1 ? (const bf16*)res_in : nullptr;
     float* sq = ksplit == 1 ? (float*)sq_parts : nullptr;
@@ -633,8 +660,14 @@ extern "C" int wstream_gemm_bf16(
                 (const bf16*)bias, M, N, K, xs, ksplit, rp, rstd_nt,
                 inv_h, eps, ri, sq, {}, yfrag);
     } else {
-        if (rstd_parts || res_in || sq_parts || yfrag) return -101;
-        if (depth == 2)
+        if (xlds == 2) {
+            k_wstream_gemm<2, 1, 2><<<grid, 512, lds_red, stream>>>(
+                (bf16*)y, (float*)part, (const bf16*)x, (const u32x4*)wp,
+                (const bf16*)bias, M, N, K, xs, ksplit, rp, rstd_nt,
+                inv_h, eps, ri, sq, {}, yfrag);
+        } else if (rstd_parts || res_in || sq_parts || yfrag) {
+            return -101;           // fusion needs the frag path at MT2
+        } else if (depth == 2)
             k_wstream_gemm<2, 2><<<grid, 512, lds_red, stream>>>(
                 (bf16*)y, (float*)part, (const bf16*)x, (const u32x4*)wp,
                 (const bf16*)bias, M, N, K, xs, ksplit, nullptr, 0,
@@ -675,11 +708,16 @@ extern "C" int wstream_qkv_rope_bf16(
     int nl, int nkl, int psz, int maxp, int xf, hipStream_t stream)
 {
     dim3 grid(N / 32, 1);
-    const int lds = 8 * 32 * 32 * 4 + 128;
+    const int lds = 8 * 32 * 32 * 4 + 256;
     RopeEpi rp{(const float*)rcos, (const float*)rsin, (const int*)pos,
                (const int*)slot, (const int*)ptab, (bf16*)kp, (bf16*)vp,
                nl, nkl, psz, maxp};
-    if (xf)
+    if (xf && M > 32)
+        k_wstream_gemm<2, 1, 2, 0, 1><<<grid, 512, lds, stream>>>(
+            (bf16*)y, nullptr, (const bf16*)x, (const u32x4*)wp,
+            (const bf16*)bias, M, N, K, xs, 1, (const float*)rstd_parts,
+            rstd_nt, inv_h, eps, nullptr, nullptr, rp);
+    else if (xf)
         k_wstream_gemm<1, 1, 2, 0, 1><<<grid, 512, lds, stream>>>(
             (bf16*)y, nullptr, (const bf16*)x, (const u32x4*)wp,
             (const bf16*)bias, M, N, K, xs, 1, (const float*)rstd_parts,

@@ -329,12 +329,13 @@ class LlamaModel:
                     for l in self.layers))
         if self.fused_chain:
             th = self.cfg.hidden // 32
-            self._sq_a = torch.zeros(max(th, 1) * 32, dtype=torch.float32,
+            self._sq_a = torch.zeros(max(th, 1) * 64, dtype=torch.float32,
                                      device=self.device)
             self._sq_b = torch.zeros_like(self._sq_a)
-            # frag-layout residual stream: fixed 32 rows (rows >= batch
-            # hold garbage that only dead accumulator rows ever see)
-            self._res_frag = torch.zeros(32 * self.cfg.hidden,
+            # frag-layout residual stream: two 32-row halves (the MT2
+            # chain serves batches up to 64; rows >= batch hold garbage
+            # that only dead accumulator rows ever see)
+            self._res_frag = torch.zeros(64 * self.cfg.hidden,
                                          dtype=self.dtype,
                                          device=self.device)
 
@@ -373,7 +374,7 @@ class LlamaModel:
         selects as "last" (decode: all; prefill: final token per seq)."""
         cfg = self.cfg
         if (self.fused_chain and attn_meta.mode == "decode"
-                and tokens.shape[0] <= 32 and not return_hidden):
+                and tokens.shape[0] <= 64 and not return_hidden):
             return self._forward_decode_fused(tokens, positions, kv_cache,
                                               slot_ids, attn_meta)
         x = ops.embedding(tokens, self.embed)
@@ -461,17 +462,17 @@ class LlamaModel:
             q = qkv[:, :nl * d].view(-1, nl, d)
             attn = ops.attention_decode_frag(q, kv_cache, li, attn_meta)
             ops.linear_fused(attn, layer.wo_pk, H, res=res, sq_out=sq_b,
-                             y=res, yfrag=1, K=nl * d, xlds=2)
+                             y=res, yfrag=1, K=nl * d, xlds=2, M_frag=B)
             act = ops.gu_fused(res, layer.wgu_pk,
                                layer.wgate_up.shape[0], rstd=sq_b,
                                rstd_nt=tiles_h, inv_h=inv_h, eps=eps,
-                               K=H, yfrag=1)
+                               K=H, yfrag=1, M_frag=B)
             ops.linear_fused(act, layer.wdown_pk, H, res=res,
                              sq_out=sq_a, y=res, yfrag=1,
-                             K=layer.wdown.shape[1], xlds=2)
+                             K=layer.wdown.shape[1], xlds=2, M_frag=B)
             nt = tiles_h
         logits = ops.linear_fused(res, self.lm_head_pk,
                                   self.lm_head.shape[0], rstd=sq_a,
                                   rstd_nt=nt, inv_h=inv_h, eps=eps,
-                                  K=H, xlds=2)
+                                  K=H, xlds=2, M_frag=B)
         return logits[:B]

@@ -290,7 +290,8 @@ def attention_decode(q, cache, layer, meta, fragout=False):
     o GEMM streams it linearly like packed weights)."""
     S, Hq, D = q.shape
     assert D == 128
-    out = torch.empty((32 * Hq * D,) if fragout else (S, Hq, D),
+    out = torch.empty((((S + 31) // 32) * 32 * Hq * D,)
+                      if fragout else (S, Hq, D),
                       dtype=q.dtype, device=q.device)
     kp, vp = _layer_ptrs(cache, layer)
     kvh = cache.n_kv_heads
@@ -461,7 +462,7 @@ def pack_weight_gu(w):
 
 
 def linear_gu(x, packed, N, rstd=None, rstd_nt=0, inv_h=0.0, eps=0.0,
-              K=None, yfrag=0):
+              K=None, yfrag=0, M_frag=32):
     """act = swiglu(x @ Wgu^T) fused; N = 2F total weight rows.
     With rstd: x is the raw residual and the epilogue applies the
     rmsnorm scale before SwiGLU (norm weight folded into the pack).
@@ -473,8 +474,9 @@ def linear_gu(x, packed, N, rstd=None, rstd_nt=0, inv_h=0.0, eps=0.0,
         xs = x.stride(0)
         xlds = (1 if N * K * 2 > (64 << 20) else 0) if M <= 32 else 0
     else:
-        M, xs, xlds = 32, 0, 2
-    act = torch.empty((M, F) if not yfrag else (32 * F,),
+        M, xs, xlds = M_frag, 0, 2
+    act = torch.empty((M, F) if not yfrag
+                      else (((M + 31) // 32) * 32 * F,),
                       dtype=x.dtype, device=x.device)
     _check(_lib.wstream_gu_bf16(
         _p(act), _p(x), _p(packed), M, N, K, xs, xlds,
@@ -580,7 +582,8 @@ _WS_MAX_N = int(os.environ.get("OLLAMAMQ_WS_MAX_N", "16384"))
 
 def linear_packed(x, packed, bias, N, ks=None, depth=None, xlds=None,
                   rstd=None, rstd_nt=0, inv_h=0.0, eps=0.0,
-                  res=None, sq_out=None, y=None, yfrag=0, K=None):
+                  res=None, sq_out=None, y=None, yfrag=0, K=None,
+                  M_frag=32):
     """y = x @ W^T via the weight-streaming kernel over pre-packed W.
 
     Fused-chain extras (decode, M<=32): `rstd`/`rstd_nt`/`inv_h`/`eps`
@@ -592,7 +595,7 @@ def linear_packed(x, packed, bias, N, ks=None, depth=None, xlds=None,
         M, K = x.shape
         xs = x.stride(0)
     else:                       # frag-layout x: flat buffer, xlds==2
-        M = 32
+        M = M_frag
         xs = 0
     if y is None:
         y = torch.empty((M, N), dtype=x.dtype, device=x.device)

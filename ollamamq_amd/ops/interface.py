@@ -173,7 +173,7 @@ def row_sumsq(x, out=None):
 
 def linear_fused(x, packed, N, bias=None, rstd=None, rstd_nt=0,
                  inv_h=0.0, eps=0.0, res=None, sq_out=None, y=None,
-                 yfrag=0, K=None, xlds=None):
+                 yfrag=0, K=None, xlds=None, M_frag=32):
     """Weight-streaming GEMM with the fused-chain epilogue/prologue:
     optional rmsnorm scaling of the raw-residual input (norm weight
     folded into the pack), in-place residual add (pass y=res) and
@@ -184,7 +184,7 @@ def linear_fused(x, packed, N, bias=None, rstd=None, rstd_nt=0,
     return hip.linear_packed(x, packed, bias, N, rstd=rstd,
                              rstd_nt=rstd_nt, inv_h=inv_h, eps=eps,
                              res=res, sq_out=sq_out, y=y, yfrag=yfrag,
-                             K=K, xlds=xlds)
+                             K=K, xlds=xlds, M_frag=M_frag)
 
 
 def fragify_sumsq(x, xf=None, sq=None):
@@ -199,11 +199,13 @@ def attention_decode_frag(q, cache, layer, meta):
     return hip.attention_decode(q, cache, layer, meta, fragout=True)
 
 
-def gu_fused(x, packed, N, rstd, rstd_nt, inv_h, eps, K=None, yfrag=0):
+def gu_fused(x, packed, N, rstd, rstd_nt, inv_h, eps, K=None, yfrag=0,
+             M_frag=32):
     """rmsnorm -> gate_up GEMM -> SwiGLU, one kernel."""
     from . import hip
     return hip.linear_gu(x, packed, N, rstd=rstd, rstd_nt=rstd_nt,
-                         inv_h=inv_h, eps=eps, K=K, yfrag=yfrag)
+                         inv_h=inv_h, eps=eps, K=K, yfrag=yfrag,
+                         M_frag=M_frag)
 
 
 def pack_weight_qkv_rope(w, nl, nkl):

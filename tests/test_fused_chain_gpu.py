@@ -112,3 +112,38 @@ def test_fused_chain_logits_close():
     torch.testing.assert_close(lf.float(), lg.float(), atol=1e-1,
                                rtol=1e-1)
     assert torch.equal(lf.argmax(-1), lg.argmax(-1))
+
+
+def test_fused_chain_mt2_batch_over_32():
+    """Batches 33-64 run the MT2 frag chain (two 32-row halves) and must
+    match the generic path token-for-token."""
+    from ollamamq_amd.models import LlamaModel, PRESETS
+    from ollamamq_amd.engine import GenParams, LlamaEngine, PagedKVCache
+    cfg = PRESETS["tiny"]
+    model = LlamaModel(cfg, device="cuda:0", dtype=torch.bfloat16,
+                       seed=1234)
+    B = 40
+
+    def run(fused):
+        kv = PagedKVCache.for_model(cfg, n_pages=1024, max_slots=B + 2,
+                                    max_ctx=128, device="cuda:0",
+                                    dtype=torch.bfloat16)
+        model.fused_chain = fused
+        eng = LlamaEngine(model, kv, max_batch=B)
+        eng.use_graphs = False
+        eng.use_pipeline = False
+        seqs = [eng.seqs[eng.submit(
+                    [(i * 13 + j * 7) % cfg.vocab for j in range(6)],
+                    GenParams(max_tokens=10))]
+                for i in range(B)]
+        for _ in range(80):
+            eng.step()
+            if not eng.has_work():
+                break
+        torch.cuda.synchronize()
+        return [list(s.generated) for s in seqs]
+
+    fused = run(True)
+    generic = run(False)
+    mism = sum(1 for a, b in zip(fused, generic) if a != b)
+    assert mism == 0, f"{mism}/{B} sequences diverged"
