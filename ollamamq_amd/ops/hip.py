@@ -598,7 +598,8 @@ def linear_packed(x, packed, bias, N, ks=None, depth=None, xlds=None,
         M = M_frag
         xs = 0
     if y is None:
-        y = torch.empty((M, N), dtype=x.dtype, device=x.device)
+        y = torch.empty((((M + 31) // 32) * 32 * N,) if yfrag
+                        else (M, N), dtype=x.dtype, device=x.device)
     if ks is None:
         per_k = os.environ.get(f"OLLAMAMQ_WS_KS_{K}")   # per-shape tuning
         if per_k:
@@ -634,7 +635,8 @@ def fragify_sumsq(x, xf=None, sq=None):
     (seeds the fused decode chain after the embedding gather)."""
     M, H = x.shape
     if xf is None:
-        xf = torch.empty(32 * H, dtype=x.dtype, device=x.device)
+        xf = torch.empty(((M + 31) // 32) * 32 * H, dtype=x.dtype,
+                         device=x.device)
     if sq is None:
         sq = torch.empty(M, dtype=torch.float32, device=x.device)
     _check(_lib.fragify_sumsq_bf16(_p(xf), _p(sq), _p(x), M, H,
