@@ -78,6 +78,7 @@ def run_stack(args):
 
         stop = threading.Event()
         token_ts = []          # (time, 1) per streamed token
+        ttfts = []             # client-side submit -> first token (ms)
         ts_lock = threading.Lock()
 
         def user_loop(uid):
@@ -92,13 +93,20 @@ def run_stack(args):
                         headers={"Content-Type": "application/json",
                                  "X-User-ID": f"bench-user-{uid}"})
                     local = []
+                    t_sub = time.monotonic()
+                    first = None
                     with urllib.request.urlopen(req, timeout=120) as r:
                         for line in r:
                             if b'"done": false' in line or \
                                     b'"done":false' in line:
-                                local.append(time.monotonic())
+                                now = time.monotonic()
+                                if first is None:
+                                    first = (now - t_sub) * 1e3
+                                local.append(now)
                     with ts_lock:
                         token_ts.extend(local)
+                        if first is not None:
+                            ttfts.append(first)
                 except OSError:
                     time.sleep(0.2)
 
@@ -144,6 +152,8 @@ def run_stack(args):
                     stats.get("queue_wait", {}).get("p50_ms"),
                 "p99_queue_wait_ms":
                     stats.get("queue_wait", {}).get("p99_ms"),
+                "client_ttft_p50_ms":
+                    round(statistics.median(ttfts), 1) if ttfts else None,
                 "requests_processed": stats.get("processed"),
                 "measure_s": round(t1 - t0, 1),
             },
