@@ -25,7 +25,8 @@ ENDPOINTS = ["/api/generate", "/api/chat", "/v1/chat/completions",
              "/v1/completions"]
 
 
-def one_request(base, user, model, rng, cancel_pct, max_tokens):
+def one_request(base, user, model, rng, cancel_pct, max_tokens,
+                sampled_pct=0.0):
     ep = rng.choice(ENDPOINTS)
     openai = ep.startswith("/v1/")
     prompt = "".join(rng.choice("abcdefghij ") for _ in range(rng.randint(8, 200)))
@@ -45,6 +46,19 @@ def one_request(base, user, model, rng, cancel_pct, max_tokens):
         body["stream"] = True
     else:
         body["options"] = {"num_predict": max_tokens}
+    # sampled share: exercises the in-graph Gumbel (temperature-only)
+    # and capped top-k/top-p decode tails under real serving churn
+    if rng.random() < sampled_pct:
+        opts = body.setdefault("options", {})
+        opts["temperature"] = round(rng.uniform(0.4, 1.2), 2)
+        if rng.random() < 0.5:
+            opts["top_k"] = rng.choice([5, 20, 40])
+        if rng.random() < 0.5:
+            opts["top_p"] = round(rng.uniform(0.7, 0.98), 2)
+        if rng.random() < 0.3:
+            opts["seed"] = rng.randint(1, 10 ** 6)
+        if openai:
+            body["temperature"] = opts["temperature"]
     cancel = rng.random() < cancel_pct
     t0 = time.monotonic()
     ttft = None
@@ -79,6 +93,9 @@ def main():
     ap.add_argument("--max-tokens", type=int, default=24)
     ap.add_argument("--cancel-pct", type=float, default=0.10)
     ap.add_argument("--par", type=int, default=32)
+    ap.add_argument("--sampled-pct", type=float, default=0.0,
+                    help="fraction of requests with stochastic sampling "
+                         "params (temperature/top-k/top-p/seed)")
     args = ap.parse_args()
     models = args.models.split(",")
 
@@ -96,7 +113,8 @@ def main():
     results = []
     with cf.ThreadPoolExecutor(args.par) as ex:
         futs = [ex.submit(one_request, args.base, u, m, rng,
-                          args.cancel_pct, args.max_tokens)
+                          args.cancel_pct, args.max_tokens,
+                          args.sampled_pct)
                 for u, m, rng in jobs]
         for f in cf.as_completed(futs):
             results.append(f.result())
