@@ -569,11 +569,13 @@ class LlamaEngine:
         self._fill_bufs(entry, seqs, token_list, pos_list)  # host samples
         return self._graph_replay(entry)[:len(seqs)]
 
-    def warm_graphs(self, sizes=None):
-        """Pre-capture the class-0 decode graphs for EVERY bucket at load
-        time (capture costs ~150 ms each; with bucketing this covers all
-        greedy workloads — no mid-serving capture, bounded VRAM).
-        Sampled classes (1, 2) still capture lazily on first use."""
+    def warm_graphs(self, sizes=None, classes=(0, 1, 2)):
+        """Pre-capture the decode graphs for EVERY (bucket, sampling
+        class) at load time — a capture costs ~150 ms and ~75 MB, and a
+        LAZY capture mid-serving stalls every in-flight stream for that
+        long (a 50%-sampled stress wave against lazy class-1/2 capture
+        timed out 189 of 303 requests before this).  With bucketing the
+        set is small and fixed; load time pays ~5-10 s once."""
         if not self.use_graphs:
             return
         sizes = [b for b in (sizes or self._buckets)
@@ -589,14 +591,15 @@ class LlamaEngine:
                 for s in slots:
                     self.kv.free_slot(s)
                 continue
-            entry = self._graph_entry(b)
             host = torch.tensor(
                 [[0] * b, [0] * b, slots, [1] * b], dtype=torch.int32)
             staged = host.to(self.dev, non_blocking=True)
-            for i, k in enumerate(("tok", "pos", "slot", "lens")):
-                entry["bufs"][k].copy_(staged[i])
-            self._graph_replay(entry)
-            torch.cuda.synchronize()
+            for kl in classes:
+                entry = self._graph_entry(b, kl)
+                for i, k in enumerate(("tok", "pos", "slot", "lens")):
+                    entry["bufs"][k].copy_(staged[i])
+                self._graph_replay(entry)
+                torch.cuda.synchronize()
             for s in slots:
                 self.kv.free_slot(s)
 
