@@ -481,6 +481,48 @@ class LlamaEngine:
                  for s in seqs] + [1.0] * pad,
                 dtype=torch.float32).to(self.dev, non_blocking=True))
 
+    def _graph_tail(self, entry, logits):
+        """The in-graph sampling tail: class 0 = greedy argmax; class 1 =
+        exact Gumbel-max temperature sampling (kernels.hip sample_gumbel,
+        greedy rows ride the same kernel with T<=0), noise keyed by
+        (per-seq seed, kv length) so replays draw fresh randomness with
+        no host RNG; class 2 = per-row top-k/top-p over the top-256
+        candidates (capturable torch ops), then exact Gumbel-max on the
+        filtered set.
+
+        This MUST also run once EAGERLY in the pre-capture warmup: the
+        samplers cache split-scratch buffers (ops/hip.py _scratch), and a
+        first allocation INSIDE capture would live in one graph's private
+        memory pool while every other graph and the eager path reuse the
+        cached tensor — undefined, and it produced real memory faults
+        under mixed sampled serving (r02 stress)."""
+        bufs = entry["bufs"]
+        if entry["klass"] == 1:
+            from ..ops import hip as _hip
+            return _hip.sample_gumbel(logits, bufs["temps"],
+                                      bufs["seeds"], bufs["lens"])
+        if entry["klass"] == 2:
+            from ..ops import hip as _hip
+            C = min(256, logits.shape[1])
+            v, idx = torch.topk(logits.float(), C, dim=-1)
+            t = bufs["temps"].clamp(min=1e-6).unsqueeze(1)
+            p = torch.softmax(v / t, dim=-1)
+            ar = torch.arange(C, device=logits.device)
+            kk = torch.where(bufs["topk"] > 0,
+                             bufs["topk"].clamp(max=C),
+                             torch.full_like(bufs["topk"], C))
+            keep = ar.unsqueeze(0) < kk.unsqueeze(1)
+            cum = p.cumsum(dim=-1)
+            keep &= (cum - p) < bufs["topp"].unsqueeze(1)
+            keep[:, 0] = True
+            # greedy rows bypass the (garbage) filter math
+            keep |= (bufs["temps"] <= 0).unsqueeze(1)
+            vm = v.masked_fill(~keep, float("-inf")).bfloat16()
+            ci = _hip.sample_gumbel(vm.contiguous(), bufs["temps"],
+                                    bufs["seeds"], bufs["lens"])
+            return idx.gather(1, ci.long().unsqueeze(1)).squeeze(1).int()
+        return ops.sample(logits, 0.0, 0, 1.0, None).int()
+
     def _graph_replay(self, entry):
         bufs, meta = entry["bufs"], entry["meta"]
         if entry["graph"] is None:
@@ -495,8 +537,11 @@ class LlamaEngine:
             s.wait_stream(torch.cuda.current_stream())
             with torch.cuda.stream(s):
                 for _ in range(2):
-                    self.model.forward(bufs["tok"], bufs["pos"], self.kv,
-                                       bufs["slot"], meta)
+                    warm_logits = self.model.forward(
+                        bufs["tok"], bufs["pos"], self.kv, bufs["slot"],
+                        meta)
+                    # eager tail warmup — see _graph_tail docstring
+                    self._graph_tail(entry, warm_logits)
             torch.cuda.current_stream().wait_stream(s)
             torch.cuda.synchronize()
             gc.disable()
@@ -508,53 +553,10 @@ class LlamaEngine:
                         meta)
                     # self-advancing tail: sample in-graph and stage the
                     # NEXT step's inputs on-device, so a steady decode
-                    # batch replays back-to-back with no host round-trip.
-                    # Class 0 = greedy argmax; class 1 = exact Gumbel-max
-                    # temperature sampling (kernels.hip sample_gumbel:
-                    # greedy rows ride the same kernel with T<=0), noise
-                    # keyed by (per-seq seed, kv length) so replays draw
-                    # fresh randomness with no host RNG.
-                    if entry["klass"] == 1:
-                        from ..ops import hip as _hip
-                        toks32 = _hip.sample_gumbel(
-                            entry["logits"], bufs["temps"], bufs["seeds"],
-                            bufs["lens"])
-                        toks = toks32.long()
-                        bufs["tok"].copy_(toks32)
-                    elif entry["klass"] == 2:
-                        # top-k/top-p over the top-256 candidates, then
-                        # exact Gumbel-max on the filtered set — all
-                        # capturable ops, no host RNG state in the graph
-                        from ..ops import hip as _hip
-                        lg = entry["logits"]
-                        C = min(256, lg.shape[1])
-                        v, idx = torch.topk(lg.float(), C, dim=-1)
-                        t = bufs["temps"].clamp(min=1e-6).unsqueeze(1)
-                        p = torch.softmax(v / t, dim=-1)
-                        ar = torch.arange(C, device=lg.device)
-                        kk = torch.where(
-                            bufs["topk"] > 0,
-                            bufs["topk"].clamp(max=C),
-                            torch.full_like(bufs["topk"], C))
-                        keep = ar.unsqueeze(0) < kk.unsqueeze(1)
-                        cum = p.cumsum(dim=-1)
-                        keep &= (cum - p) < bufs["topp"].unsqueeze(1)
-                        keep[:, 0] = True
-                        # greedy rows bypass the (garbage) filter math
-                        keep |= (bufs["temps"] <= 0).unsqueeze(1)
-                        vm = v.masked_fill(~keep,
-                                           float("-inf")).bfloat16()
-                        ci = _hip.sample_gumbel(
-                            vm.contiguous(), bufs["temps"],
-                            bufs["seeds"], bufs["lens"])
-                        toks32 = idx.gather(
-                            1, ci.long().unsqueeze(1)).squeeze(1).int()
-                        toks = toks32.long()
-                        bufs["tok"].copy_(toks32)
-                    else:
-                        toks = ops.sample(entry["logits"], 0.0, 0, 1.0,
-                                          None)
-                        bufs["tok"].copy_(toks)
+                    # batch replays back-to-back with no host round-trip
+                    toks32 = self._graph_tail(entry, entry["logits"])
+                    toks = toks32.long()
+                    bufs["tok"].copy_(toks32)
                     entry["out"] = toks
                     bufs["pos"] += bufs["adv"]
                     bufs["lens"] += bufs["adv"]
