@@ -111,6 +111,8 @@ class LlamaEngine:
                                  48, 64, 96, 128, max_batch)
                                 if b <= max_batch})
         self._dummy_slot = -1
+        # opt-in: capture the top-k/top-p tail in-graph (see _samp_class)
+        self._GRAPH_TOPK = _os.environ.get("OLLAMAMQ_GRAPH_TOPK") == "1"
 
     # -- submission --------------------------------------------------------
     def submit(self, prompt: List[int], params: GenParams,
@@ -385,18 +387,23 @@ class LlamaEngine:
 
     def _samp_class(self, seqs) -> int:
         """0 = all greedy (argmax tail), 1 = temperature-only mix (Gumbel
-        tail — exact), 2 = top-k/top-p present (in-graph candidate
-        filtering over the top-256 logits + exact Gumbel-max on the
-        filtered set; nucleus/top-k capped at 256 candidates on this
-        path — the non-graph host path stays exact over the full
-        vocabulary)."""
+        tail — exact), 2 = top-k/top-p present.
+
+        Class 2 is captured ONLY behind OLLAMAMQ_GRAPH_TOPK=1: the
+        candidate tail (torch.topk over [B, 128256] + filters) is clean
+        in isolation (tools/repro_tail.py: 3000 mutated replays) and on
+        tiny models, but faults under real multi-graph 8B serving with
+        or without the fused chain (tools/repro_sampled.py WORKLOAD=topk)
+        — an unresolved interaction between the captured topk and the
+        serving graph set.  Until root-caused, top-k/p rows sample on the
+        host (exact, full vocabulary) with the forward still graphed."""
         k = 0
         for s in seqs:
             p = s.params
             if p.temperature > 0:
                 if (p.top_k or 0) > 0 or (p.top_p if p.top_p else 1.0) \
                         < 1.0:
-                    return 2
+                    return 2 if self._GRAPH_TOPK else 3
                 k = 1
         return k
 
@@ -571,7 +578,7 @@ class LlamaEngine:
         self._fill_bufs(entry, seqs, token_list, pos_list)  # host samples
         return self._graph_replay(entry)[:len(seqs)]
 
-    def warm_graphs(self, sizes=None, classes=(0, 1, 2)):
+    def warm_graphs(self, sizes=None, classes=None):
         """Pre-capture the decode graphs for EVERY (bucket, sampling
         class) at load time — a capture costs ~150 ms and ~75 MB, and a
         LAZY capture mid-serving stalls every in-flight stream for that
@@ -580,6 +587,8 @@ class LlamaEngine:
         set is small and fixed; load time pays ~5-10 s once."""
         if not self.use_graphs:
             return
+        if classes is None:
+            classes = (0, 1, 2) if self._GRAPH_TOPK else (0, 1)
         sizes = [b for b in (sizes or self._buckets)
                  if 0 < b <= self.max_batch]
         for b in sorted(set(sizes)):
@@ -608,6 +617,7 @@ class LlamaEngine:
     def _decode_step(self):
         seqs = self.running
         if (self.use_pipeline and self.use_graphs
+                and self._samp_class(seqs) <= 2
                 and all(self.kv.seq_lens[s.slot] + 1 <= self.kv.max_ctx
                         for s in seqs)):
             self._decode_step_pipelined(seqs)

@@ -111,10 +111,14 @@ def test_sampled_decode_replays_graphs():
         (solo.generated, toks_a[0])
 
 
-def test_topk_topp_decode_in_graph():
-    """Class-2 sampled workloads (top-k/top-p) replay graphs too: the
-    captured tail filters the top-256 candidates per row's own params
-    and Gumbel-samples the filtered set."""
+def test_topk_topp_decode_in_graph(monkeypatch):
+    """Opt-in class-2 graphs (OLLAMAMQ_GRAPH_TOPK=1): the captured tail
+    filters the top-256 candidates per row's own params and
+    Gumbel-samples the filtered set.  (Off by default: the captured
+    torch.topk faults under real multi-graph 8B serving — see
+    engine._samp_class; this test pins the tiny-model mechanics so the
+    path stays alive for the root-cause hunt.)"""
+    monkeypatch.setenv("OLLAMAMQ_GRAPH_TOPK", "1")
     from ollamamq_amd.models import LlamaModel, PRESETS
     from ollamamq_amd.engine import GenParams, LlamaEngine, PagedKVCache
     cfg = PRESETS["tiny"]
