@@ -764,8 +764,46 @@ def sample(logits, temperature, top_k, top_p, generator=None):
         _check(_lib.argmax_bf16(_p(out), _p(l), B, V, pb, pi, sp,
                                 _stream()), "argmax")
         return out.long()
-    # stochastic paths compose on-GPU torch ops (sort/softmax/multinomial)
-    return ref.sample(logits, temperature, top_k, top_p, generator)
+    # stochastic host path.  When every stochastic row carries
+    # 0 < top_k <= 256 the sampling is EXACT over the top-256 candidates
+    # (its nucleus is a subset of its top-k; probabilities use a
+    # full-vocabulary logsumexp, so the top-p mass cut is the true one)
+    # and costs a topk + tiny multinomial instead of the full-vocab sort
+    # (~0.5 ms/step at V=128256).  Anything else falls back to the
+    # fp32 reference (exact, sort-based).
+    B, V = logits.shape
+    dev = logits.device
+    if not torch.is_tensor(temperature):
+        temperature = torch.full((B,), float(temperature), device=dev)
+    tk = top_k if torch.is_tensor(top_k) else torch.full(
+        (B,), int(top_k or 0), dtype=torch.long, device=dev)
+    tp = top_p if torch.is_tensor(top_p) else torch.full(
+        (B,), float(top_p if top_p else 1.0), device=dev)
+    tk = tk.long()
+    stoch = temperature > 0
+    C = min(256, V)
+    capped_ok = bool(((tk > 0) & (tk <= C) | ~stoch).all())
+    if not capped_ok:
+        return ref.sample(logits, temperature, tk, tp, generator)
+    l = logits.float()
+    t = torch.where(stoch, temperature, torch.ones_like(temperature))
+    lt = l / t.unsqueeze(1)
+    v, idx = torch.topk(lt, C, dim=-1)
+    lse = torch.logsumexp(lt, dim=-1, keepdim=True)
+    p = torch.exp(v - lse)                      # exact full-vocab probs
+    ar = torch.arange(C, device=dev)
+    keep = ar.unsqueeze(0) < tk.clamp(min=1, max=C).unsqueeze(1)
+    cum = p.cumsum(dim=-1)
+    keep &= (cum - p) < tp.unsqueeze(1)
+    keep[:, 0] = True
+    pm = p * keep
+    pick = torch.multinomial(pm / pm.sum(-1, keepdim=True), 1,
+                             generator=generator).squeeze(1)
+    out_s = idx.gather(1, pick.unsqueeze(1)).squeeze(1)
+    if bool(stoch.all()):
+        return out_s
+    # greedy rows override with the true argmax
+    return torch.where(stoch, out_s, l.argmax(-1))
 
 
 def decode_pure(cache, layer, meta, split, chunk=64):

@@ -158,3 +158,35 @@ def test_topk_topp_decode_in_graph(monkeypatch):
     # (weak check: diversity only — exact support check needs per-step
     # logits; covered by sampler unit tests)
     assert len(set(toks[1])) >= 1
+
+
+def test_host_capped_sampling_exact_support():
+    """hip.sample's fast host path (top_k <= 256): every draw inside its
+    row's top-k, top_k=1 equals argmax, greedy rows exact, deterministic
+    under a fixed generator, and falls back to the sort path for pure
+    nucleus rows."""
+    hip = _hip()
+    g = torch.Generator().manual_seed(3)
+    B, V = 12, 128256
+    logits = (torch.randn(B, V, generator=g) * 2).bfloat16().cuda()
+    temps = torch.tensor([0.0, 1.0] + [0.8] * (B - 2), device="cuda")
+    tk = torch.tensor([0, 1] + [7] * (B - 2), dtype=torch.long,
+                      device="cuda")
+    tp = torch.full((B,), 0.95, device="cuda")
+    gen = torch.Generator(device="cuda")
+    gen.manual_seed(9)
+    out = hip.sample(logits, temps, tk, tp, gen)
+    am = logits.float().argmax(-1)
+    assert out[0] == am[0], "greedy row must be argmax"
+    assert out[1] == am[1], "top_k=1 must be argmax"
+    kth = logits.float().topk(7, dim=-1).values[:, -1]
+    picked = logits.float().gather(1, out.unsqueeze(1)).squeeze(1)
+    assert (picked[2:] >= kth[2:] - 1e-3).all(), "escaped top-7"
+    gen2 = torch.Generator(device="cuda")
+    gen2.manual_seed(9)
+    out2 = hip.sample(logits, temps, tk, tp, gen2)
+    assert torch.equal(out.cpu(), out2.cpu())
+    # pure-nucleus rows (top_k=0, top_p<1) take the exact sort fallback
+    tk0 = torch.zeros(B, dtype=torch.long, device="cuda")
+    out3 = hip.sample(logits, temps, tk0, tp, gen)
+    assert out3.shape[0] == B
