@@ -37,7 +37,8 @@ static void usage() {
         "      --probe-interval-ms <M> health probe cadence (default 10000)\n"
         "      --allow-all-routes    proxy unknown routes too\n"
         "      --no-tui              headless (logs to stderr)\n"
-        "  -c, --model-config <F>    config file (default appconf.yaml)\n";
+        "  -c, --model-config <F>    config file (default appconf.yaml)\n"
+        "  -V, --version             print version\n";
 }
 
 int main(int argc, char** argv) {
@@ -66,6 +67,10 @@ int main(int argc, char** argv) {
         else if (a == "--no-tui") no_tui = true;
         else if (a == "-c" || a == "--model-config") config_path = next();
         else if (a == "-h" || a == "--help") { usage(); return 0; }
+        else if (a == "-V" || a == "--version") {
+            std::cout << "ollamamq-amd 0.2.0\n";
+            return 0;
+        }
         else {
             std::cerr << "unknown flag " << a << "\n";
             usage();
