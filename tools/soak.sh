@@ -26,7 +26,7 @@ rss() { ps -o rss= -p $(pgrep -P $LAUNCH -f ollamamq-server | head -1) 2>/dev/nu
 
 # one warmup wave first: decode graphs are captured lazily per batch
 # size (bounded by --max-batch, ~75 MB each); baseline AFTER they exist
-timeout 200 python tools/stress.py --base http://127.0.0.1:$PORT \
+timeout 200 python tools/stress.py --base http://127.0.0.1:$PORT --sampled-pct ${SOAK_SAMPLED:-0} \
     --users 24 --models llama3-8b --max-tokens 24 2>&1 | tail -1
 V0=$(vram); R0=$(rss)
 echo "post-warmup baseline vram=$V0 rss=$R0"
@@ -35,7 +35,7 @@ WAVE=0
 VLOG=""
 while [ "$(date +%s)" -lt "$END" ]; do
   WAVE=$((WAVE+1))
-  timeout 200 python tools/stress.py --base http://127.0.0.1:$PORT \
+  timeout 200 python tools/stress.py --base http://127.0.0.1:$PORT --sampled-pct ${SOAK_SAMPLED:-0} \
       --users 24 --models llama3-8b --max-tokens 24 2>&1 | tail -1
   VW=$(vram); VLOG="$VLOG $VW"
   echo "wave $WAVE vram=$VW"
