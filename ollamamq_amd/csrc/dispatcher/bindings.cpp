@@ -59,6 +59,7 @@ public:
     void set_vip(const std::string& u) { st_.vip_user = u; }
     void set_boost(const std::string& u) { st_.boost_user = u; }
     void set_stuck_timeout(int64_t s) { st_.settings.stuck_timeout_s = s; }
+    void set_debug_log(bool on) { st_.log.set_debug(on); }
 
     void enqueue(const std::string& user, const std::string& model,
                  const std::string& path) {
@@ -200,6 +201,7 @@ PYBIND11_MODULE(_dispatch, m) {
         .def("set_vip", &Harness::set_vip)
         .def("set_boost", &Harness::set_boost)
         .def("set_stuck_timeout", &Harness::set_stuck_timeout)
+        .def("set_debug_log", &Harness::set_debug_log)
         .def("enqueue", &Harness::enqueue, py::arg("user"),
              py::arg("model") = "", py::arg("path") = "/api/chat")
         .def("enqueue_aged", &Harness::enqueue_aged)

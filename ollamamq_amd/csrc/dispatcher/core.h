@@ -46,6 +46,14 @@ public:
     // ./ollamamq.log; headless logs to stderr.
     void set_file_sink(const std::string& path);
     void set_stderr_sink(bool on) { stderr_sink_ = on; }
+    // Debug level (reference RUST_LOG env-filter, src/main.rs:208,215):
+    // the scheduler's per-candidate rejection reasons are emitted only
+    // when enabled (reference logs them at debug, src/dispatcher.rs:579-615).
+    void set_debug(bool on) { debug_ = on; }
+    bool debug_on() const { return debug_; }
+    void debug(const std::string& text) {
+        if (debug_) push("DBG", text);
+    }
 
 private:
     mutable std::mutex mu_;
@@ -53,6 +61,7 @@ private:
     std::deque<LogEvent> ring_;
     std::string file_path_;
     bool stderr_sink_ = false;
+    std::atomic<bool> debug_{false};
 };
 
 // ------------------------------------------------------------------ request

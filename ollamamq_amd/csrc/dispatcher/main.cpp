@@ -124,6 +124,12 @@ int main(int argc, char** argv) {
         st.log.set_stderr_sink(true);
     else
         st.log.set_file_sink("ollamamq.log");
+    // log-level env filter (reference RUST_LOG, src/main.rs:208,215);
+    // OLLAMAMQ_LOG is ours, RUST_LOG honored for migration friendliness
+    for (const char* var : {"OLLAMAMQ_LOG", "RUST_LOG"})
+        if (const char* lv = getenv(var))
+            if (std::string(lv).find("debug") != std::string::npos)
+                st.log.set_debug(true);
 
     if (!server.start(&err)) {
         std::cerr << "failed to start: " << err << "\n";

@@ -74,19 +74,28 @@ static bool family_supported(ApiType t, const std::string& path) {
     return openai ? t == ApiType::OpenAi : t == ApiType::Ollama;
 }
 
-bool backend_eligible(const BackendStatus& b, bool has_control_op,
-                      const std::string& requested_model,
-                      const std::string& path) {
-    if (!b.is_online) return false;
-    if (b.active_requests >= b.max_concurrency) return false;
-    if (has_control_op) return false;
+const char* backend_reject_reason(const BackendStatus& b, bool has_control_op,
+                                  const std::string& requested_model,
+                                  const std::string& path) {
+    if (!b.is_online) return "offline";
+    if (b.active_requests >= b.max_concurrency) return "busy";
+    if (has_control_op) return "control-op";
     // A specific model is the HARD gate when present; the API-family check
     // applies only to model-less requests (reference src/dispatcher.rs:599-616:
     // "If a specific model is requested, backend MUST have it. If no model is
     // requested, fall back to API family check.")
     if (!requested_model.empty())
-        return model_routable(requested_model, b.available_models);
-    return family_supported(b.api_type, path);
+        return model_routable(requested_model, b.available_models)
+                   ? nullptr
+                   : "model-not-available";
+    return family_supported(b.api_type, path) ? nullptr : "api-family";
+}
+
+bool backend_eligible(const BackendStatus& b, bool has_control_op,
+                      const std::string& requested_model,
+                      const std::string& path) {
+    return backend_reject_reason(b, has_control_op, requested_model, path) ==
+           nullptr;
 }
 
 size_t pick_backend(const std::vector<BackendStatus>& backends,
@@ -148,11 +157,18 @@ bool schedule_once(AppState& st, Dispatch* out) {
         // scan for the FIRST routable task in this user's queue
         for (auto it = us.queue.begin(); it != us.queue.end(); ++it) {
             std::vector<size_t> eligible;
+            const bool dbg = st.log.debug_on();
             for (size_t bi = 0; bi < st.backends.size(); bi++) {
                 const bool op = st.control_ops.count(bi) > 0;
-                if (backend_eligible(st.backends[bi], op,
-                                     it->requested_model, it->path))
+                const char* why = backend_reject_reason(
+                    st.backends[bi], op, it->requested_model, it->path);
+                if (!why)
                     eligible.push_back(bi);
+                else if (dbg)
+                    st.log.debug("sched: user=" + user + " model=" +
+                                 it->requested_model + " backend[" +
+                                 std::to_string(bi) + "] " +
+                                 st.backends[bi].url + " rejected: " + why);
             }
             if (eligible.empty()) {
                 if (!it->stuck_warned) {

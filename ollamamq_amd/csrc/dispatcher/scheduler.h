@@ -43,6 +43,13 @@ bool backend_eligible(const BackendStatus& b, bool has_control_op,
                       const std::string& requested_model,
                       const std::string& path);
 
+// Why a backend is NOT eligible (nullptr = eligible).  Reasons mirror the
+// reference's debug-level candidate log (src/dispatcher.rs:579-615):
+// "offline" | "busy" | "control-op" | "model-not-available" | "api-family".
+const char* backend_reject_reason(const BackendStatus& b, bool has_control_op,
+                                  const std::string& requested_model,
+                                  const std::string& path);
+
 // Least-connections + rotating-index pick among eligible indices.
 size_t pick_backend(const std::vector<BackendStatus>& backends,
                     const std::vector<size_t>& eligible, size_t last_idx);

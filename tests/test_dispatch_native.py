@@ -182,6 +182,26 @@ def test_stuck_timeout_503():
     assert h.queue_len("u") == 0
 
 
+def test_debug_rejection_log():
+    """Per-candidate rejection reasons at debug level (reference logs them
+    under RUST_LOG=debug, src/dispatcher.rs:579-615); silent when off."""
+    h = d.Harness()
+    b0 = h.add_backend("b0", available=["llama3"])
+    h.add_backend("b1", available=["qwen2"])
+    h.set_online(b0, False)
+    h.enqueue("u", "llama3")
+    assert h.schedule() is None          # b0 offline, b1 lacks the model
+    assert not any(k == "DBG" for k, _ in h.log_events())  # off by default
+
+    h.set_debug_log(True)
+    h.enqueue("u", "llama3")
+    assert h.schedule() is None
+    dbg = [t for k, t in h.log_events() if k == "DBG"]
+    assert any("rejected: offline" in t and "backend[0]" in t for t in dbg)
+    assert any("rejected: model-not-available" in t and "backend[1]" in t
+               for t in dbg)
+
+
 def test_json_model_extraction():
     assert d.json_get_model('{"model": "llama3", "messages": []}') == "llama3"
     assert d.json_get_model('{"x": 1}') == ""
