@@ -167,6 +167,30 @@ void run_tui(Server& server) {
     bool help = false;
     std::string status_msg;
 
+    // One key from stdin.  CSI sequences (arrow keys, Shift-Tab — reference
+    // tui.rs supports ↑/↓ nav and S-Tab panel cycle via crossterm) are
+    // translated to 256+vi-equivalent so bare Esc (27) still means quit and
+    // input mode can ignore navigation keys; unknown sequences return 0.
+    auto read_key = [&]() -> int {
+        char c;
+        if (read(STDIN_FILENO, &c, 1) != 1) return 0;
+        if (c != 27) return (unsigned char)c;
+        pollfd p2{STDIN_FILENO, POLLIN, 0};
+        if (poll(&p2, 1, 10) <= 0) return 27;  // bare Esc
+        char b = 0;
+        if (read(STDIN_FILENO, &b, 1) != 1 || b != '[') return 27;
+        char f = 0;
+        if (poll(&p2, 1, 10) > 0 && read(STDIN_FILENO, &f, 1) != 1) f = 0;
+        switch (f) {
+            case 'A': return 256 + 'k';  // up
+            case 'B': return 256 + 'j';  // down
+            case 'C': return 256 + 'l';  // right
+            case 'D': return 256 + 'h';  // left
+            case 'Z': return 256 + 'h';  // Shift-Tab: panel cycle
+        }
+        return 0;
+    };
+
     while (true) {
         Snapshot s = capture(st);
         std::ostringstream out;
@@ -181,8 +205,8 @@ void run_tui(Server& server) {
             << "  sched:" << s.counter << "\r\n";
         if (help) {
             out << "\r\n \x1b[1mKeys\x1b[0m\r\n"
-                   "  Tab     switch panel focus (Backends/Users)\r\n"
-                   "  j/k     move selection in the focused panel\r\n"
+                   "  Tab     model cursor / panel focus (S-Tab, h/l: panel)\r\n"
+                   "  j/k/arrows  move selection in the focused panel\r\n"
                    "  a       toggle available-models listing\r\n"
                    "  p / b   toggle VIP / Boost for selected user\r\n"
                    "  x / X   block selected user / their IP\r\n"
@@ -195,11 +219,9 @@ void run_tui(Server& server) {
             fflush(stdout);
             pollfd pfd{STDIN_FILENO, POLLIN, 0};
             if (poll(&pfd, 1, 100) > 0) {
-                char c;
-                if (read(STDIN_FILENO, &c, 1) == 1) {
-                    if (c == 'q' || c == 27) return;
-                    help = false;
-                }
+                const int c = read_key();
+                if (c == 'q' || c == 27) return;
+                if (c) help = false;
             }
             continue;
         }
@@ -301,8 +323,10 @@ void run_tui(Server& server) {
 
         pollfd pfd{STDIN_FILENO, POLLIN, 0};
         if (poll(&pfd, 1, 100) > 0) {
-            char c;
-            if (read(STDIN_FILENO, &c, 1) == 1) {
+            int c = read_key();
+            if (c) {
+                if (c >= 256 && input_mode) continue;  // nav keys ignored
+                if (c >= 256) c -= 256;                // arrows → j/k/h/l
                 if (input_mode) {
                     if (c == 27) {  // Esc
                         input_mode = 0;

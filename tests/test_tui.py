@@ -118,6 +118,43 @@ def test_tui_model_cursor_direct_load(tmp_path):
         fleet.stop()
 
 
+def test_tui_arrow_keys_navigate_not_quit(tmp_path):
+    """Arrow keys send ESC [ A/B: they must navigate (reference tui.rs
+    j/k/up-down parity), NOT be mistaken for a bare Esc quit; bare Esc
+    still quits."""
+    master, slave = pty.openpty()
+    p = subprocess.Popen(
+        [BIN, "-p", "0", "-c", os.path.join(str(tmp_path), "absent.yaml")],
+        stdin=slave, stdout=slave, stderr=subprocess.DEVNULL,
+        cwd=str(tmp_path))
+    os.close(slave)
+    out = b""
+    deadline = time.time() + 30
+    stage = 0
+    try:
+        while time.time() < deadline:
+            try:
+                out += os.read(master, 65536)
+            except OSError:
+                break
+            if stage == 0 and b"Backends" in out:
+                os.write(master, b"\x1b[B\x1b[A\x1b[Z")  # down, up, S-Tab
+                stage, out = 1, b""
+            elif stage == 1 and b"Backends" in out:
+                # still alive and rendering after the arrow keys
+                assert p.poll() is None, "arrow keys quit the TUI"
+                os.write(master, b"\x1b")               # bare Esc quits
+                stage = 2
+            elif stage == 2 and p.poll() is not None:
+                break
+        assert stage == 2, f"stage={stage}: {out[-300:]!r}"
+        assert p.wait(timeout=10) == 0, "Esc did not exit cleanly"
+    finally:
+        os.close(master)
+        if p.poll() is None:
+            p.terminate()
+
+
 def test_tui_queues_panel(tmp_path):
     """The Queues panel renders per-user load bars (reference
     tui.rs:1124-1163)."""
