@@ -2,6 +2,7 @@
 
 #include <cmath>
 #include <cstdio>
+#include <cstdlib>
 #include <cstring>
 
 namespace omq {
@@ -34,11 +35,16 @@ static void dump_to(const Json& j, std::string& out) {
         case Json::Type::Null: out += "null"; break;
         case Json::Type::Bool: out += j.b ? "true" : "false"; break;
         case Json::Type::Num: {
-            char buf[32];
-            if (std::floor(j.num) == j.num && std::fabs(j.num) < 1e15)
+            char buf[40];
+            if (std::floor(j.num) == j.num && std::fabs(j.num) < 1e15) {
                 snprintf(buf, sizeof buf, "%lld", (long long)j.num);
-            else
-                snprintf(buf, sizeof buf, "%g", j.num);
+            } else {
+                // shortest representation that round-trips the double
+                // (plain %g keeps 6 digits and corrupts seeds/timestamps)
+                snprintf(buf, sizeof buf, "%.15g", j.num);
+                if (strtod(buf, nullptr) != j.num)
+                    snprintf(buf, sizeof buf, "%.17g", j.num);
+            }
             out += buf;
             break;
         }
@@ -108,24 +114,53 @@ struct Parser {
                     case 'r': out += '\r'; break;
                     case 't': out += '\t'; break;
                     case 'u': {
+                        auto hex4 = [](const char* q, unsigned* v) -> bool {
+                            unsigned x = 0;
+                            for (int i = 0; i < 4; i++) {
+                                char c = q[i];
+                                x <<= 4;
+                                if (c >= '0' && c <= '9') x |= c - '0';
+                                else if (c >= 'a' && c <= 'f')
+                                    x |= c - 'a' + 10;
+                                else if (c >= 'A' && c <= 'F')
+                                    x |= c - 'A' + 10;
+                                else return false;
+                            }
+                            *v = x;
+                            return true;
+                        };
                         if (end - p < 5) return false;
-                        unsigned cp = 0;
-                        for (int i = 1; i <= 4; i++) {
-                            char c = p[i];
-                            cp <<= 4;
-                            if (c >= '0' && c <= '9') cp |= c - '0';
-                            else if (c >= 'a' && c <= 'f') cp |= c - 'a' + 10;
-                            else if (c >= 'A' && c <= 'F') cp |= c - 'A' + 10;
-                            else return false;
-                        }
+                        unsigned cp;
+                        if (!hex4(p + 1, &cp)) return false;
                         p += 4;
-                        // UTF-8 encode (surrogate pairs folded to U+FFFD)
+                        if (cp >= 0xD800 && cp <= 0xDBFF) {
+                            // UTF-16 surrogate pair (how JSON escapes
+                            // astral chars, e.g. emoji from Python's
+                            // json.dumps); unpaired halves fold to U+FFFD
+                            unsigned lo;
+                            if (end - p >= 7 && p[1] == '\\' &&
+                                p[2] == 'u' && hex4(p + 3, &lo) &&
+                                lo >= 0xDC00 && lo <= 0xDFFF) {
+                                cp = 0x10000 + ((cp - 0xD800) << 10) +
+                                     (lo - 0xDC00);
+                                p += 6;
+                            } else {
+                                cp = 0xFFFD;
+                            }
+                        } else if (cp >= 0xDC00 && cp <= 0xDFFF) {
+                            cp = 0xFFFD;  // lone low surrogate
+                        }
                         if (cp < 0x80) out += (char)cp;
                         else if (cp < 0x800) {
                             out += (char)(0xC0 | (cp >> 6));
                             out += (char)(0x80 | (cp & 0x3F));
-                        } else {
+                        } else if (cp < 0x10000) {
                             out += (char)(0xE0 | (cp >> 12));
+                            out += (char)(0x80 | ((cp >> 6) & 0x3F));
+                            out += (char)(0x80 | (cp & 0x3F));
+                        } else {
+                            out += (char)(0xF0 | (cp >> 18));
+                            out += (char)(0x80 | ((cp >> 12) & 0x3F));
                             out += (char)(0x80 | ((cp >> 6) & 0x3F));
                             out += (char)(0x80 | (cp & 0x3F));
                         }
