@@ -12,6 +12,7 @@
 #include <sys/un.h>
 #include <unistd.h>
 
+#include <algorithm>
 #include <cstring>
 
 #include "control.h"
@@ -122,7 +123,10 @@ public:
                 if (const Json* cx = j->find("ctx"))
                     for (const auto& [k, v] : cx->obj)
                         if (v.is_num()) r.loaded_ctx[k] = (int64_t)v.num;
-                max_conc_ = (int)j->get_num("max_concurrency", 1);
+                // clamp: a buggy worker reporting 0/negative would make
+                // this backend permanently unschedulable
+                max_conc_ =
+                    std::max(1, (int)j->get_num("max_concurrency", 1));
             }
         }
         ::close(fd);
@@ -155,6 +159,7 @@ public:
             auto j = Json::parse(line);
             if (j) {
                 status = (int)j->get_num("status", 200);
+                if (status < 100 || status > 599) status = 502;
                 std::string ct = j->get_str("content_type",
                                             "application/x-ndjson");
                 if (resp)
