@@ -254,12 +254,20 @@ void HttpServer::handle_conn(int fd, std::string peer_ip) {
             for (;;) {
                 size_t eol;
                 while ((eol = buf.find("\r\n")) == std::string::npos) {
-                    if (!fill(buf.size() + 1)) { ok = false; break; }
+                    // size line is tiny; cap so a client streaming
+                    // CRLF-less bytes can't grow the buffer unboundedly
+                    if (buf.size() > 4096 || !fill(buf.size() + 1)) {
+                        ok = false;
+                        break;
+                    }
                 }
                 if (!ok) break;
                 const size_t csz = strtoull(buf.c_str(), nullptr, 16);
                 buf.erase(0, eol + 2);
-                if (body.size() + csz > BODY_CAP) { ok = false; break; }
+                if (csz > BODY_CAP || body.size() + csz > BODY_CAP) {
+                    ok = false;  // csz check first: near-SIZE_MAX csz
+                    break;       // would overflow the sum
+                }
                 if (csz == 0) {
                     // consume optional trailers up to the blank line
                     size_t tend;
@@ -456,15 +464,23 @@ HttpResponse http_request(
             return true;
         };
         while (true) {
-            // read chunk-size line
+            // read chunk-size line (bounded: a backend streaming bytes
+            // with no CRLF must not grow the buffer forever)
             size_t eol;
             while ((eol = pending.find("\r\n", off)) == std::string::npos) {
-                if (!need(pending.size() - off + 1)) { eol = std::string::npos; break; }
+                if (pending.size() - off > 4096 ||
+                    !need(pending.size() - off + 1)) {
+                    eol = std::string::npos;
+                    break;
+                }
             }
             if (eol == std::string::npos) break;
             const size_t csz = strtoull(pending.c_str() + off, nullptr, 16);
             off = eol + 2;
             if (csz == 0) break;
+            // reject insane sizes: csz near SIZE_MAX would overflow
+            // csz + 2 below and deliver() would read out of bounds
+            if (csz > BODY_CAP) break;
             if (!need(csz + 2)) break;
             if (!deliver(pending.data() + off, csz)) break;
             off += csz + 2;  // skip data + CRLF

@@ -112,6 +112,72 @@ def test_chunked_request_body_garbage_size(srv):
     assert alive(srv)
 
 
+def test_chunked_overflow_size_ingress(srv):
+    """Chunk size ffffffffffffffff: csz+2 would wrap to 1 — must be a
+    clean 400, not a corrupted parse."""
+    r = raw(srv, b"POST /admin/models/load HTTP/1.1\r\nHost: x\r\n"
+                 b"Transfer-Encoding: chunked\r\n\r\n"
+                 b"ffffffffffffffff\r\nXX\r\n0\r\n\r\n",
+            expect_reply=True)
+    assert r.startswith(b"HTTP/1.1 400"), r[:80]
+    assert alive(srv)
+
+
+def test_rogue_backend_chunked_overflow(tmp_path):
+    """A malicious/buggy BACKEND answering probes with an overflowing
+    chunk size (ffffffffffffffff) must not crash the dispatcher's HTTP
+    client decoder (pre-fix: out-of-bounds deliver of SIZE_MAX bytes)."""
+    import threading
+
+    hits = []
+
+    def backend():
+        ls = socket.create_server(("127.0.0.1", 0))
+        ports.append(ls.getsockname()[1])
+        ls.settimeout(30)
+        started.set()
+        while not stop.is_set():
+            try:
+                c, _ = ls.accept()
+            except socket.timeout:
+                continue
+            with c:
+                try:
+                    c.settimeout(5)
+                    c.recv(4096)
+                    hits.append(1)
+                    c.sendall(b"HTTP/1.1 200 OK\r\n"
+                              b"Transfer-Encoding: chunked\r\n\r\n"
+                              b"ffffffffffffffff\r\n"
+                              b'{"models":[]}\r\n0\r\n\r\n')
+                except OSError:
+                    pass
+        ls.close()
+
+    ports, started, stop = [], threading.Event(), threading.Event()
+    t = threading.Thread(target=backend, daemon=True)
+    t.start()
+    assert started.wait(10)
+    p = subprocess.Popen(
+        [BIN, "--no-tui", "-p", "0", "-o",
+         f"http://127.0.0.1:{ports[0]}", "--probe-interval-ms", "200",
+         "-c", os.path.join(str(tmp_path), "absent.yaml")],
+        stderr=subprocess.PIPE, cwd=str(tmp_path), text=True)
+    try:
+        port = int(p.stderr.readline().rsplit(":", 1)[1].split()[0])
+        deadline = time.time() + 20
+        while time.time() < deadline and len(hits) < 3:
+            assert p.poll() is None, "dispatcher crashed on rogue backend"
+            time.sleep(0.2)
+        assert len(hits) >= 3, "backend was never probed"
+        assert p.poll() is None
+        assert alive(port)
+    finally:
+        stop.set()
+        p.terminate()
+        p.wait(timeout=10)
+
+
 def test_deeply_nested_json_body(srv):
     """A 100k-deep JSON nesting bomb must be rejected by the parser's
     depth guard, not blow the stack."""
