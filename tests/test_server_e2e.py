@@ -195,6 +195,47 @@ def test_any_selector_picks_resolvable(proxy, fleet):
     time.sleep(0.5)
 
 
+def test_busy_backend_409(fleet, tmp_path_factory):
+    """A backend mid-request refuses control ops with 409 (reference
+    run.sh busy-window check via the 12 s hanging /api/chat mock)."""
+    import threading
+    slow = MockFleet(chat_delay=6.0)
+    p = Proxy([slow.ollama_url], tmp_path_factory.mktemp("busy"))
+    try:
+        deadline = time.time() + 15
+        while time.time() < deadline:
+            b = httpx.get(p.base + "/admin/models").json()["backends"]
+            if b and b[0]["online"] and b[0]["api"] != "unknown":
+                break
+            time.sleep(0.3)
+        t = threading.Thread(
+            target=lambda: httpx.post(
+                p.base + "/api/chat",
+                json={"model": "llama3:latest", "messages": []},
+                timeout=30),
+            daemon=True)
+        t.start()
+        # wait until the request is actually dispatched onto the backend
+        deadline = time.time() + 10
+        busy = False
+        while time.time() < deadline:
+            b = httpx.get(p.base + "/admin/models").json()["backends"][0]
+            if b.get("active_requests", 0) > 0:
+                busy = True
+                break
+            time.sleep(0.1)
+        assert busy, "chat never occupied the backend"
+        r = httpx.post(p.base + "/admin/models/load",
+                       json={"model": "llama3:latest", "backend": 0},
+                       timeout=10)
+        assert r.status_code == 409
+        assert "busy" in r.json()["error"]
+        t.join(timeout=20)
+    finally:
+        p.stop()
+        slow.stop()
+
+
 def test_duplicate_op_409(proxy, fleet):
     # the 1 s slow LM Studio load opens the race window
     r1 = httpx.post(proxy.base + "/admin/models/load",
