@@ -119,6 +119,12 @@ class LlamaEngine:
                on_token=None) -> int:
         if len(prompt) == 0:
             prompt = [0]
+        # Ollama-parity: a prompt longer than the context window is
+        # truncated from the front (the tail conditions generation) so the
+        # request degrades instead of faulting the batch at prefill
+        limit = max(1, self.kv.max_ctx - 1)
+        if len(prompt) > limit:
+            prompt = prompt[-limit:]
         sid = next(self._ids)
         seq = Sequence(sid, list(prompt), params, on_token)
         # per-sequence noise seed for the in-graph Gumbel sampler:
@@ -279,12 +285,16 @@ class LlamaEngine:
         for s in batch:
             s.prefill_done += s._chunk
             if s.prefill_done == len(s.prompt):
-                tok = ptoks[done_prefill.index(s)]
-                s.generated.append(int(tok))
                 s.first_token_at = now
                 s.state = "running"
                 self.waiting.remove(s)
                 self.running.append(s)
+                if s.params.max_tokens <= 0:
+                    # zero-token request (Ollama num_predict: 0): admit
+                    # without emitting; _postprocess finishes it this step
+                    continue
+                tok = ptoks[done_prefill.index(s)]
+                s.generated.append(int(tok))
                 self.tokens_out += 1
                 if s.on_token:
                     s.on_token(int(tok), False)
@@ -375,12 +385,16 @@ class LlamaEngine:
         for s in batch:
             s.prefill_done += s._chunk
             if s.prefill_done == len(s.prompt):
-                tok = toks[done_prefill.index(s)]
-                s.generated.append(int(tok))
                 s.first_token_at = time.monotonic()
                 s.state = "running"
                 self.waiting.remove(s)
                 self.running.append(s)
+                if s.params.max_tokens <= 0:
+                    # zero-token request (Ollama num_predict: 0): admit
+                    # without emitting; _postprocess finishes it this step
+                    continue
+                tok = toks[done_prefill.index(s)]
+                s.generated.append(int(tok))
                 self.tokens_out += 1
                 if s.on_token:
                     s.on_token(int(tok), False)
@@ -710,6 +724,11 @@ class LlamaEngine:
             elif p.stop_token is not None and s.generated \
                     and s.generated[-1] == p.stop_token:
                 s.finish_reason = "stop"
+                finished.append(s)
+            elif s.total_len >= self.kv.max_ctx:
+                # context window exhausted: finish gracefully (another
+                # decode would have no KV slot to append into)
+                s.finish_reason = "length"
                 finished.append(s)
         return finished
 

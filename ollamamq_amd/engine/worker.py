@@ -197,10 +197,18 @@ class Worker:
 def _params_from_body(body: dict, tok: ByteTokenizer,
                       openai: bool) -> GenParams:
     opts = body.get("options") or {}
-    max_t = (body.get("max_tokens") or body.get("max_completion_tokens")
-             or opts.get("num_predict") or 64)
+    max_t = body.get("max_tokens")
+    if max_t is None:
+        max_t = body.get("max_completion_tokens")
+    if max_t is None:
+        max_t = opts.get("num_predict")
+    max_t = 64 if max_t is None else int(max_t)
+    if max_t < 0:
+        # Ollama semantics: -1 = infinite, -2 = fill context; both are
+        # bounded by the context window in the engine
+        max_t = 1 << 30
     return GenParams(
-        max_tokens=int(max_t),
+        max_tokens=max_t,
         temperature=float(body.get("temperature",
                                    opts.get("temperature", 0.0)) or 0.0),
         top_k=int(opts.get("top_k", body.get("top_k", 0)) or 0),

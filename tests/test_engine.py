@@ -193,3 +193,55 @@ def test_long_run_resource_stability():
         assert len(eng.kv._free_slots) == eng.kv.max_slots
         assert eng.kv.free_page_count() == eng.kv.n_pages
         assert all(s not in eng.seqs for s in sids)
+
+
+def make_small_ctx_engine(max_ctx=64):
+    cfg = PRESETS["tiny-cpu"]
+    model = LlamaModel(cfg, device="cpu", dtype=torch.float32, seed=7)
+    kv = PagedKVCache.for_model(cfg, n_pages=64, max_slots=4,
+                                max_ctx=max_ctx)
+    return LlamaEngine(model, kv, max_batch=2)
+
+
+def test_context_exhaustion_finishes_gracefully():
+    """A sequence reaching the context window must finish with reason
+    "length", not fault the whole batch step (Ollama parity: generation
+    is bounded by num_ctx)."""
+    eng = make_small_ctx_engine()
+    sid = eng.submit(list(range(40)), GenParams(max_tokens=1 << 30))
+    seq = eng.seqs[sid]
+    done = run_all(eng, max_steps=100)
+    assert done and done[0].finish_reason == "length"
+    assert seq.total_len == eng.kv.max_ctx
+
+
+def test_oversized_prompt_truncated_front():
+    """Prompts longer than the context window keep their TAIL (what
+    conditions generation) instead of raising at prefill."""
+    eng = make_small_ctx_engine()
+    sid = eng.submit(list(range(200)), GenParams(max_tokens=4))
+    seq = eng.seqs[sid]
+    assert len(seq.prompt) == eng.kv.max_ctx - 1
+    assert seq.prompt[-1] == 199          # tail survives
+    done = run_all(eng, max_steps=100)
+    assert done[0].finish_reason == "length"
+
+
+def test_zero_token_request():
+    """max_tokens=0 (Ollama num_predict: 0) emits NO content tokens —
+    only the done marker — in both the pure-prefill and mixed-batch
+    admission paths."""
+    for mixed in (False, True):
+        eng = make_small_ctx_engine()
+        if mixed:
+            eng.submit([9, 9, 9], GenParams(max_tokens=6))
+            eng.step()                    # one sequence already decoding
+        emitted = []
+        sid = eng.submit([1, 2, 3], GenParams(max_tokens=0),
+                         on_token=lambda t, d: emitted.append((t, d)))
+        for _ in range(20):
+            done = eng.step()
+            if any(s.seq_id == sid for s in done):
+                break
+        assert emitted == [(-1, True)], (mixed, emitted)
+        run_all(eng)
