@@ -349,9 +349,17 @@ class Conn:
 
         tok = w.tokenizers.get(model) or ByteTokenizer(
             PRESETS[model].vocab if model in PRESETS else 512)
-        params = _params_from_body(body, tok, openai)
+        try:
+            params = _params_from_body(body, tok, openai)
+            prompt = tok.encode(_prompt_text(body, path))
+        except (TypeError, ValueError) as e:
+            # malformed option types (e.g. temperature: "hot") are a
+            # client error, not a worker fault: clean 400
+            self._line({"status": 400, "content_type": "application/json"})
+            self.sock.sendall(json.dumps(
+                {"error": f"invalid options: {e}"}).encode())
+            return
         stream = body.get("stream", not openai)
-        prompt = tok.encode(_prompt_text(body, path))
 
         stops = body.get("stop") or (body.get("options") or {}).get("stop")
         if isinstance(stops, str):

@@ -147,6 +147,45 @@ def test_concurrent_users_fair_share(stack):
     assert codes == [200] * 8
 
 
+def test_malformed_options_400(stack):
+    """Garbage option types (temperature: "hot") are a client error:
+    clean 400 through the whole stack, and the worker keeps serving."""
+    r = httpx.post(stack + "/api/generate",
+                   json={"model": "tiny-cpu", "prompt": "x",
+                         "stream": False,
+                         "options": {"temperature": "hot"}},
+                   timeout=60.0)
+    assert r.status_code == 400
+    assert "invalid options" in r.json()["error"]
+    r = httpx.post(stack + "/api/generate",
+                   json={"model": "tiny-cpu", "prompt": "x",
+                         "stream": False, "options": {"num_predict": 2}},
+                   timeout=60.0)
+    assert r.status_code == 200
+
+
+def test_oversized_prompt_truncates_not_500(stack):
+    """A prompt longer than the context window degrades (front-truncated
+    in the engine) instead of erroring."""
+    r = httpx.post(stack + "/api/generate",
+                   json={"model": "tiny-cpu", "prompt": "y" * 5000,
+                         "stream": False, "options": {"num_predict": 2}},
+                   timeout=120.0)
+    assert r.status_code == 200
+
+
+def test_num_predict_negative_bounded_by_ctx(stack):
+    """num_predict: -1 (Ollama infinite) must terminate — bounded by the
+    context window — and return a non-empty response."""
+    r = httpx.post(stack + "/api/generate",
+                   json={"model": "tiny-cpu", "prompt": "z",
+                         "stream": False,
+                         "options": {"num_predict": -1}},
+                   timeout=180.0)
+    assert r.status_code == 200
+    assert r.json()["done"] is True
+
+
 def test_admin_load_unload_on_worker(stack):
     r = httpx.post(stack + "/admin/models/load",
                    json={"model": "tiny", "backend": 0, "num_ctx": 256})
