@@ -214,7 +214,10 @@ def _params_from_body(body: dict, tok: ByteTokenizer,
         top_k=int(opts.get("top_k", body.get("top_k", 0)) or 0),
         top_p=float(body.get("top_p", opts.get("top_p", 1.0)) or 1.0),
         stop_token=tok.stop_token,
-        seed=opts.get("seed") or body.get("seed"),
+        # coerce here so a junk seed is caught by the caller's 400 path,
+        # not deep inside engine.submit
+        seed=(int(seed) if (seed := opts.get("seed", body.get("seed")))
+              is not None else None),
     )
 
 
