@@ -295,6 +295,15 @@ class Conn:
     def _line(self, obj):
         self.sock.sendall((json.dumps(obj) + "\n").encode())
 
+    def _send_err(self, status, msg, openai=False):
+        # OpenAI-family clients expect {"error": {"message", "type"}};
+        # Ollama-family expects {"error": "<string>"}
+        self._line({"status": status, "content_type": "application/json"})
+        payload = ({"error": {"message": msg,
+                              "type": "invalid_request_error"}}
+                   if openai else {"error": msg})
+        self.sock.sendall(json.dumps(payload).encode())
+
     def _probe(self):
         w = self.worker
         with w.lock:
@@ -332,11 +341,10 @@ class Conn:
             return self._embed(path, body)
         if path in ("/api/create", "/api/copy", "/api/delete", "/api/pull",
                     "/api/push") or path.startswith("/api/blobs"):
-            self._line({"status": 501, "content_type": "application/json"})
-            self.sock.sendall(json.dumps(
-                {"error": f"{path} is not supported by the in-process GPU "
-                          "worker (models are resident presets; use "
-                          "/admin/models/load)"}).encode())
+            self._send_err(
+                501, f"{path} is not supported by the in-process GPU "
+                     "worker (models are resident presets; use "
+                     "/admin/models/load)")
             return
 
         openai = path.startswith("/v1/")
@@ -345,9 +353,7 @@ class Conn:
             (next(iter(w.engines), None) or
              ("tiny" if w.device == "cpu" else "llama3-8b"))
         if model is None:
-            self._line({"status": 404, "content_type": "application/json"})
-            self.sock.sendall(json.dumps(
-                {"error": f"model not found: {model_req}"}).encode())
+            self._send_err(404, f"model not found: {model_req}", openai)
             return
 
         tok = w.tokenizers.get(model) or ByteTokenizer(
@@ -358,9 +364,7 @@ class Conn:
         except (TypeError, ValueError) as e:
             # malformed option types (e.g. temperature: "hot") are a
             # client error, not a worker fault: clean 400
-            self._line({"status": 400, "content_type": "application/json"})
-            self.sock.sendall(json.dumps(
-                {"error": f"invalid options: {e}"}).encode())
+            self._send_err(400, f"invalid options: {e}", openai)
             return
         stream = body.get("stream", not openai)
 

@@ -164,6 +164,24 @@ def test_malformed_options_400(stack):
     assert r.status_code == 200
 
 
+def test_openai_error_object_format(stack):
+    """/v1/* errors use the OpenAI error-object shape so SDKs can parse
+    them; /api/* errors stay Ollama's string form."""
+    r = httpx.post(stack + "/v1/chat/completions",
+                   json={"model": "tiny-cpu", "temperature": "hot",
+                         "messages": [{"role": "user", "content": "x"}]},
+                   timeout=60.0)
+    assert r.status_code == 400
+    err = r.json()["error"]
+    assert isinstance(err, dict) and "message" in err and "type" in err
+    r = httpx.post(stack + "/api/generate",
+                   json={"model": "tiny-cpu", "prompt": "x",
+                         "options": {"temperature": "hot"}},
+                   timeout=60.0)
+    assert r.status_code == 400
+    assert isinstance(r.json()["error"], str)
+
+
 def test_oversized_prompt_truncates_not_500(stack):
     """A prompt longer than the context window degrades (front-truncated
     in the engine) instead of erroring."""
