@@ -257,7 +257,10 @@ class LlamaEngine:
 
     def _mixed_step(self, batch: List[Sequence]):
         self._drain_pipe()   # decode rows need current host tokens
-        dec = list(self.running)
+        # context-full sequences can't take another token; skip them this
+        # step — _postprocess finishes them with reason "length"
+        dec = [s for s in self.running
+               if s.total_len < self.kv.max_ctx]
         token_list = [s.generated[-1] for s in dec]
         pos_list = [s.total_len - 1 for s in dec]
         q_lens = [1] * len(dec)
@@ -637,6 +640,11 @@ class LlamaEngine:
             self._decode_step_pipelined(seqs)
             return
         self._drain_pipe()
+        # after the drain, total_len is host truth: context-full sequences
+        # can't take another token — skip them; _postprocess finishes them
+        seqs = [s for s in seqs if s.total_len < self.kv.max_ctx]
+        if not seqs:
+            return
         token_list = [s.generated[-1] for s in seqs]
         pos_list = [s.total_len - 1 for s in seqs]
         q_lens = [1] * len(seqs)

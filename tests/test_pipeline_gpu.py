@@ -119,3 +119,23 @@ def test_warm_graphs_precapture():
     sids = _workload(eng)
     run_all(eng)
     assert all(s.finish_reason for s in sids)
+
+
+def test_context_exhaustion_pipelined():
+    """Sequences hitting the context window under the pipelined/graphed
+    decode path must finish with reason "length", not fault the step
+    (the host's generated list lags the device by one token, so the
+    capacity check must run on post-drain truth)."""
+    cfg = PRESETS["tiny"]
+    model = LlamaModel(cfg, device="cuda", dtype=torch.bfloat16, seed=7)
+    kv = PagedKVCache.for_model(cfg, n_pages=64, max_slots=4,
+                                max_ctx=64, device="cuda",
+                                dtype=torch.bfloat16)
+    eng = LlamaEngine(model, kv, max_batch=2)
+    sids = [eng.submit(list(range(30 + i)), GenParams(max_tokens=1 << 30))
+            for i in range(2)]
+    seqs = [eng.seqs[s] for s in sids]
+    run_all(eng, max_steps=200)
+    for s in seqs:
+        assert s.finish_reason == "length"
+        assert s.total_len == 64
