@@ -136,6 +136,14 @@ public:
     int execute(const Task& task) override {
         auto resp = task.resp;
         int fd = uds_connect(sock_, 5000);
+        if (fd >= 0) {
+            // the 5 s connect-time SO_RCVTIMEO must NOT govern the token
+            // stream: admission behind a full batch or a long prefill can
+            // legally pause it longer.  Use the configured request
+            // timeout (reference -t semantics, default 300 s).
+            timeval tv{(time_t)(timeout_s_ > 0 ? timeout_s_ : 300), 0};
+            setsockopt(fd, SOL_SOCKET, SO_RCVTIMEO, &tv, sizeof tv);
+        }
         if (fd < 0) {
             if (resp) {
                 resp->send_status(502,

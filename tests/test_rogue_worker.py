@@ -55,6 +55,75 @@ class RogueWorker(socketserver.ThreadingUnixStreamServer):
                 self.wfile.write(b'{"done": true}\n')
 
 
+class SlowStreamWorker(socketserver.ThreadingUnixStreamServer):
+    """Valid worker whose stream pauses 7 s mid-generation (legal under
+    load: admission behind a full batch)."""
+    daemon_threads = True
+    allow_reuse_address = True
+
+    def __init__(self, path):
+        super().__init__(path, self._Handler)
+
+    class _Handler(socketserver.StreamRequestHandler):
+        def handle(self):
+            line = self.rfile.readline()
+            try:
+                cmd = json.loads(line).get("cmd")
+            except (json.JSONDecodeError, AttributeError):
+                return
+            if cmd == "probe":
+                self.wfile.write((json.dumps({
+                    "online": True, "models": ["slow-model"],
+                    "loaded": ["slow-model"],
+                    "max_concurrency": 1}) + "\n").encode())
+            elif cmd == "request":
+                self.wfile.write((json.dumps({
+                    "status": 200,
+                    "content_type": "application/x-ndjson"}) +
+                    "\n").encode())
+                self.wfile.write(b'{"response": "a", "done": false}\n')
+                self.wfile.flush()
+                time.sleep(7)   # > the old 5 s connect-time SO_RCVTIMEO
+                self.wfile.write(b'{"response": "b", "done": true}\n')
+
+
+def test_stream_survives_long_token_gap(tmp_path):
+    """A >5 s pause between stream chunks must not truncate the response:
+    the stream phase runs under the request timeout (-t), not the 5 s
+    connect timeout."""
+    sock = os.path.join(str(tmp_path), "slow.sock")
+    sw = SlowStreamWorker(sock)
+    t = threading.Thread(target=sw.serve_forever, daemon=True)
+    t.start()
+    server = subprocess.Popen(
+        [BIN, "--no-tui", "-p", "0", "-w", sock,
+         "--probe-interval-ms", "300",
+         "-c", os.path.join(str(tmp_path), "absent.yaml")],
+        stderr=subprocess.PIPE, cwd=str(tmp_path), text=True)
+    try:
+        port = int(server.stderr.readline().rsplit(":", 1)[1].split()[0])
+        base = f"http://127.0.0.1:{port}"
+        deadline = time.time() + 20
+        while time.time() < deadline:
+            try:
+                b = httpx.get(base + "/admin/models", timeout=5).json()
+                if b["backends"] and b["backends"][0]["online"]:
+                    break
+            except httpx.HTTPError:
+                pass
+            time.sleep(0.2)
+        r = httpx.post(base + "/api/generate",
+                       json={"model": "slow-model", "prompt": "x"},
+                       timeout=60)
+        assert r.status_code == 200
+        assert '"done": true' in r.text, f"stream truncated: {r.text!r}"
+    finally:
+        server.terminate()
+        server.wait(timeout=10)
+        sw.shutdown()
+        sw.server_close()
+
+
 @pytest.fixture
 def rogue(tmp_path):
     sock = os.path.join(str(tmp_path), "rogue.sock")
