@@ -148,3 +148,58 @@ def test_tp_worker_uds_wire_and_keepalive_unload():
         pass
     assert tag == "ok"
     assert unloaded, "keep_alive:0 did not unload the idle model"
+
+
+def _rank_storm(rank, world, port, q):
+    """Lockstep under concurrent load: 4 parallel generates on one model
+    interleaved with a load+unload of a SECOND model mid-flight.  The
+    op-stream broadcast and SYNC_EVERY cadence must neither deadlock nor
+    diverge; every generate completes with its full token budget."""
+    os.environ["MASTER_ADDR"] = "127.0.0.1"
+    os.environ["MASTER_PORT"] = str(port)
+    import threading
+    import time
+    import torch.distributed as dist
+    from ollamamq_amd.engine.tp_worker import TPWorker
+    dist.init_process_group("gloo", rank=rank, world_size=world)
+    w = TPWorker(rank, world, 0, max_batch=4, default_ctx=128)
+    if rank != 0:
+        time.sleep(180)
+        return
+    assert w.load("tiny", 128) is None
+    counts = [0] * 4
+    dones = [threading.Event() for _ in range(4)]
+
+    def fire(i):
+        def on_token(t, fin):
+            if fin:
+                dones[i].set()
+            else:
+                counts[i] += 1
+        w.generate("tiny", [i + 1, 2, 3], GenParams(max_tokens=6 + i),
+                   on_token)
+
+    for i in range(2):
+        fire(i)
+    # mid-flight control ops on another model
+    assert w.load("nano", 64) is None
+    for i in range(2, 4):
+        fire(i)
+    assert w.unload("nano") is None
+    for i, d in enumerate(dones):
+        assert d.wait(timeout=120), f"generate {i} never finished"
+    q.put(counts)
+
+
+def test_tp_lockstep_concurrent_ops_storm():
+    ctx = mp.get_context("spawn")
+    q = ctx.Queue()
+    procs = [ctx.Process(target=_rank_storm, args=(r, 2, 29591, q),
+                         daemon=True) for r in range(2)]
+    for p in procs:
+        p.start()
+    counts = q.get(timeout=300)
+    for p in procs:
+        p.terminate()
+        p.join(timeout=30)
+    assert counts == [6, 7, 8, 9], counts
