@@ -194,7 +194,7 @@ def _rank_storm(rank, world, port, q):
 def test_tp_lockstep_concurrent_ops_storm():
     ctx = mp.get_context("spawn")
     q = ctx.Queue()
-    procs = [ctx.Process(target=_rank_storm, args=(r, 2, 29591, q),
+    procs = [ctx.Process(target=_rank_storm, args=(r, 2, 29593, q),
                          daemon=True) for r in range(2)]
     for p in procs:
         p.start()
