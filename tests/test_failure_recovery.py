@@ -108,3 +108,54 @@ def test_worker_death_and_rejoin(tmp_path):
     finally:
         server.terminate()
         worker.terminate()
+
+
+def test_midstream_worker_death_truncates_cleanly(tmp_path):
+    """Worker dies while a response is STREAMING: the client's stream
+    ends (truncated, no done marker), the dispatcher survives, and the
+    health loop flips the backend offline (SURVEY §5 mid-stream
+    failure semantics)."""
+    if not os.path.exists(BIN):
+        subprocess.run([sys.executable, "-m", "ollamamq_amd.build"],
+                       check=True)
+    sock = os.path.join(str(tmp_path), "w.sock")
+    worker = _spawn_worker(sock)
+    assert _wait_socket(sock)
+    server = subprocess.Popen(
+        [BIN, "--no-tui", "-p", "0", "-w", sock, "--probe-interval-ms",
+         "300", "-c", os.path.join(str(tmp_path), "absent.yaml")],
+        stderr=subprocess.PIPE, cwd=str(tmp_path), text=True)
+    try:
+        port = int(server.stderr.readline().rsplit(":", 1)[1].split()[0])
+        base = f"http://127.0.0.1:{port}"
+        deadline = time.time() + 15
+        while time.time() < deadline and _online(base) is not True:
+            time.sleep(0.2)
+
+        chunks = []
+        saw_done = False
+        with httpx.stream(
+                "POST", base + "/api/generate",
+                json={"model": "tiny-cpu", "prompt": "a", "stream": True,
+                      "options": {"num_predict": 2000}},
+                timeout=60.0) as r:
+            assert r.status_code == 200
+            for line in r.iter_lines():
+                chunks.append(line)
+                if '"done": true' in line or '"done":true' in line:
+                    saw_done = True
+                    break
+                if len(chunks) == 3:
+                    worker.kill()        # mid-stream death
+        assert chunks and not saw_done, \
+            f"stream should truncate without done: {chunks[-1:]}"
+        assert server.poll() is None, "dispatcher died with the worker"
+        assert httpx.get(base + "/health", timeout=5).text == "OK"
+        deadline = time.time() + 15
+        while time.time() < deadline and _online(base) is not False:
+            time.sleep(0.2)
+        assert _online(base) is False
+    finally:
+        server.terminate()
+        if worker.poll() is None:
+            worker.terminate()
