@@ -147,6 +147,35 @@ def test_concurrent_users_fair_share(stack):
     assert codes == [200] * 8
 
 
+def test_client_disconnect_frees_backend(stack):
+    """Client hangs up mid-stream: the hangup propagates through the
+    dispatcher to the worker (engine cancel, KV freed) and the backend's
+    active_requests returns to 0 so new requests dispatch immediately."""
+    with httpx.stream(
+            "POST", stack + "/api/generate",
+            json={"model": "tiny-cpu", "prompt": "c", "stream": True,
+                  "options": {"num_predict": 5000}},
+            headers={"X-User-ID": "canceller"}, timeout=60.0) as r:
+        assert r.status_code == 200
+        for i, _ in enumerate(r.iter_lines()):
+            if i >= 2:
+                break               # context exit closes the connection
+    deadline = time.time() + 20
+    freed = False
+    while time.time() < deadline:
+        stats = httpx.get(stack + "/admin/stats", timeout=5).json()
+        if stats.get("processing", 1) == 0:
+            freed = True
+            break
+        time.sleep(0.2)
+    assert freed, f"backend never freed after disconnect: {stats}"
+    r = httpx.post(stack + "/api/generate",
+                   json={"model": "tiny-cpu", "prompt": "d",
+                         "stream": False, "options": {"num_predict": 2}},
+                   timeout=60.0)
+    assert r.status_code == 200
+
+
 def test_malformed_options_400(stack):
     """Garbage option types (temperature: "hot") are a client error:
     clean 400 through the whole stack, and the worker keeps serving."""
