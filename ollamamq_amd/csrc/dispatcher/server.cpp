@@ -158,10 +158,14 @@ void Server::proxy_handler(const HttpRequest& req, HttpConn& conn) {
     if (user.empty()) user = "anonymous";
 
     {
+        // IP check first, distinct bodies (reference dispatcher.rs:857-865)
         std::lock_guard<std::mutex> g(st_.blocked_mu);
-        if (st_.blocked_users.count(user) ||
-            st_.blocked_ips.count(req.client_ip)) {
-            conn.send(403, {}, "{\"error\":\"blocked\"}");
+        if (st_.blocked_ips.count(req.client_ip)) {
+            conn.send(403, {}, "{\"error\":\"IP blocked\"}");
+            return;
+        }
+        if (st_.blocked_users.count(user)) {
+            conn.send(403, {}, "{\"error\":\"User blocked\"}");
             return;
         }
     }
