@@ -48,7 +48,7 @@ public:
     void set_control_op(size_t i, bool active) {
         std::lock_guard<std::mutex> g(st_.control_mu);
         if (active)
-            st_.control_ops[i] = {ControlAction::Load, "x", now_ms()};
+            st_.control_ops[i] = {ControlAction::Load, "x", "x", now_ms()};
         else
             st_.control_ops.erase(i);
     }

@@ -1,5 +1,7 @@
 #include "control.h"
 
+#include <algorithm>
+
 #include <sstream>
 
 #include <thread>
@@ -112,7 +114,7 @@ ControlOutcome start_model_control(AppState& st, const ControlRequest& req) {
                 return {400, control_err("model is not loaded: " + canonical)};
         }
         st.control_ops[req.backend_idx] =
-            ControlOp{req.action, canonical, now_ms()};
+            ControlOp{req.action, canonical, req.model, now_ms()};
         snap = b;
     }
     st.log.push("CTL", std::string(req.action == ControlAction::Load
@@ -231,7 +233,8 @@ ControlOutcome admin_stats(AppState& st) {
 }
 
 ControlOutcome admin_models_state(AppState& st) {
-    Json out = Json::object();
+    // bare JSON array, sorted model lists, operation with requested +
+    // elapsed_secs — exact reference shape (control.rs:1320-1350)
     Json arr = Json::array();
     std::scoped_lock lk(st.control_mu, st.backends_mu);
     for (size_t i = 0; i < st.backends.size(); i++) {
@@ -243,14 +246,14 @@ ControlOutcome admin_models_state(AppState& st) {
         e.set("api", Json::string(api_type_name(b.api_type)));
         e.set("lmstudio", Json::boolean(b.lmstudio));
         e.set("active_requests", Json::number(b.active_requests));
-        Json avail = Json::array();
-        for (const auto& m : b.available_models)
-            avail.arr.push_back(Json::string(m));
-        e.set("available_models", std::move(avail));
-        Json loaded = Json::array();
-        for (const auto& m : b.loaded_models)
-            loaded.arr.push_back(Json::string(m));
-        e.set("loaded_models", std::move(loaded));
+        auto sorted_list = [](std::vector<std::string> v) {
+            std::sort(v.begin(), v.end());
+            Json a = Json::array();
+            for (auto& m : v) a.arr.push_back(Json::string(m));
+            return a;
+        };
+        e.set("available_models", sorted_list(b.available_models));
+        e.set("loaded_models", sorted_list(b.loaded_models));
         auto op = st.control_ops.find(i);
         if (op != st.control_ops.end()) {
             Json o = Json::object();
@@ -259,14 +262,17 @@ ControlOutcome admin_models_state(AppState& st) {
                                              ? "load"
                                              : "unload"));
             o.set("model", Json::string(op->second.model));
+            o.set("requested", Json::string(op->second.requested));
+            o.set("elapsed_secs",
+                  Json::number((double)((now_ms() -
+                                         op->second.started_ms) / 1000)));
             e.set("operation", std::move(o));
         } else {
             e.set("operation", Json::null());
         }
         arr.arr.push_back(std::move(e));
     }
-    out.set("backends", std::move(arr));
-    return {200, std::move(out)};
+    return {200, std::move(arr)};
 }
 
 // backend selector: index | "any" | URL substring (control.rs:1179-1240)

@@ -162,7 +162,8 @@ enum class ControlAction { Load, Unload };
 
 struct ControlOp {
     ControlAction action;
-    std::string model;
+    std::string model;      // canonical (post-resolution)
+    std::string requested;  // as the client asked (admin "operation")
     int64_t started_ms;
 };
 

@@ -104,7 +104,7 @@ int main() {
                     st.control_ops.erase(bi);
                 else if (st.backends[bi].active_requests == 0)
                     st.control_ops[bi] =
-                        ControlOp{ControlAction::Load, "llama3", now_ms()};
+                        ControlOp{ControlAction::Load, "llama3", "llama3", now_ms()};
             }
             st.notify();
             i++;

@@ -41,7 +41,7 @@ def _spawn_worker(sock):
 def _online(base):
     try:
         r = httpx.get(base + "/admin/models", timeout=2.0).json()
-        return r["backends"][0]["online"]
+        return r[0]["online"]
     except Exception:
         return None
 

@@ -61,7 +61,7 @@ def node(tmp_path_factory):
     while time.time() < deadline:
         try:
             r = httpx.get(base + "/admin/models", timeout=2.0).json()
-            if all(b["online"] for b in r["backends"]):
+            if all(b["online"] for b in r):
                 break
         except Exception:
             pass
@@ -74,8 +74,8 @@ def node(tmp_path_factory):
 
 def test_both_backends_online(node):
     r = httpx.get(node + "/admin/models").json()
-    assert len(r["backends"]) == 2
-    assert all(b["online"] for b in r["backends"])
+    assert len(r) == 2
+    assert all(b["online"] for b in r)
 
 
 def test_load_spreads_across_workers(node):
@@ -132,8 +132,8 @@ def hetero_node(tmp_path_factory):
     while time.time() < deadline:
         try:
             r = httpx.get(base + "/admin/models", timeout=2.0).json()
-            if all(b["online"] for b in r["backends"]) and \
-                    len(r["backends"]) == 3:
+            if all(b["online"] for b in r) and \
+                    len(r) == 3:
                 break
         except Exception:
             pass
@@ -146,7 +146,7 @@ def hetero_node(tmp_path_factory):
 
 def test_hetero_advertises_per_worker_models(hetero_node):
     r = httpx.get(hetero_node + "/admin/models").json()
-    avail = [set(b["available_models"]) for b in r["backends"]]
+    avail = [set(b["available_models"]) for b in r]
     assert avail[0] == {"tiny"} and avail[1] == {"tiny"}
     assert avail[2] == {"nano"}
 
@@ -189,7 +189,7 @@ def test_hetero_unknown_model_parks_then_503(hetero_node):
     """A model nobody carries must never dispatch (stuck-timeout 503 is
     covered in dispatch tests; here assert immediate non-routing)."""
     r = httpx.get(hetero_node + "/admin/models").json()
-    names = {m for b in r["backends"] for m in b["available_models"]}
+    names = {m for b in r for m in b["available_models"]}
     assert "gemma" not in names
 
 

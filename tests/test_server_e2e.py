@@ -79,8 +79,8 @@ def proxy(fleet, tmp_path_factory):
     deadline = time.time() + 15
     while time.time() < deadline:
         r = httpx.get(p.base + "/admin/models").json()
-        if all(b["online"] for b in r["backends"]) and \
-                all(b["api"] != "unknown" for b in r["backends"]):
+        if all(b["online"] for b in r) and \
+                all(b["api"] != "unknown" for b in r):
             break
         time.sleep(0.3)
     yield p
@@ -94,8 +94,8 @@ def test_health(proxy):
 
 def test_admin_inventory(proxy):
     r = httpx.get(proxy.base + "/admin/models").json()
-    assert len(r["backends"]) == 2
-    b0, b1 = r["backends"]
+    assert len(r) == 2
+    b0, b1 = r
     assert b0["online"] and b1["online"]
     assert b0["api"] == "ollama"
     assert "llama3:latest" in b0["available_models"]
@@ -128,7 +128,7 @@ def test_unload_ollama_keepalive_zero(proxy, fleet):
     deadline = time.time() + 10
     while time.time() < deadline:
         r = httpx.get(proxy.base + "/admin/models").json()
-        if "llama3:latest" in r["backends"][0]["loaded_models"]:
+        if "llama3:latest" in r[0]["loaded_models"]:
             break
         time.sleep(0.2)
     r = httpx.post(proxy.base + "/admin/models/unload",
@@ -204,7 +204,7 @@ def test_busy_backend_409(fleet, tmp_path_factory):
     try:
         deadline = time.time() + 15
         while time.time() < deadline:
-            b = httpx.get(p.base + "/admin/models").json()["backends"]
+            b = httpx.get(p.base + "/admin/models").json()
             if b and b[0]["online"] and b[0]["api"] != "unknown":
                 break
             time.sleep(0.3)
@@ -219,7 +219,7 @@ def test_busy_backend_409(fleet, tmp_path_factory):
         deadline = time.time() + 10
         busy = False
         while time.time() < deadline:
-            b = httpx.get(p.base + "/admin/models").json()["backends"][0]
+            b = httpx.get(p.base + "/admin/models").json()[0]
             if b.get("active_requests", 0) > 0:
                 busy = True
                 break
@@ -246,7 +246,7 @@ def test_duplicate_op_409(proxy, fleet):
     assert r2.status_code == 409
     # op visible in admin state during the window
     r = httpx.get(proxy.base + "/admin/models").json()
-    op = r["backends"][1]["operation"]
+    op = r[1]["operation"]
     assert op and op["action"] == "load"
     time.sleep(1.5)
 

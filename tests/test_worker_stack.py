@@ -56,7 +56,7 @@ def stack(tmp_path_factory):
     while time.time() < deadline:
         try:
             r = httpx.get(base + "/admin/models").json()
-            if r["backends"] and r["backends"][0]["online"]:
+            if r and r[0]["online"]:
                 break
         except Exception:
             pass
@@ -67,8 +67,7 @@ def stack(tmp_path_factory):
 
 
 def test_worker_probed(stack):
-    r = httpx.get(stack + "/admin/models").json()
-    b = r["backends"][0]
+    b = httpx.get(stack + "/admin/models").json()[0]
     assert b["online"]
     assert b["api"] == "both"
     assert "tiny-cpu" in b["available_models"]
@@ -240,7 +239,7 @@ def test_admin_load_unload_on_worker(stack):
     deadline = time.time() + 60
     ok = False
     while time.time() < deadline:
-        st = httpx.get(stack + "/admin/models").json()["backends"][0]
+        st = httpx.get(stack + "/admin/models").json()[0]
         if "tiny" in st["loaded_models"] and st["operation"] is None:
             ok = True
             break
@@ -251,7 +250,7 @@ def test_admin_load_unload_on_worker(stack):
     assert r.status_code == 202, r.text
     deadline = time.time() + 30
     while time.time() < deadline:
-        st = httpx.get(stack + "/admin/models").json()["backends"][0]
+        st = httpx.get(stack + "/admin/models").json()[0]
         if "tiny" not in st["loaded_models"]:
             return
         time.sleep(0.3)

@@ -56,7 +56,7 @@ def gpu_stack(tmp_path_factory):
     while time.time() < deadline:
         try:
             r = httpx.get(base + "/admin/models", timeout=2.0).json()
-            if r["backends"] and r["backends"][0]["online"]:
+            if r and r[0]["online"]:
                 break
         except Exception:
             pass
@@ -76,7 +76,7 @@ def test_gpu_worker_serves_ollama_chat(gpu_stack):
     lines = [json.loads(l) for l in r.text.strip().split("\n")]
     assert lines[-1]["done"] is True
     assert lines[-1]["eval_count"] >= 1
-    st = httpx.get(gpu_stack + "/admin/models").json()["backends"][0]
+    st = httpx.get(gpu_stack + "/admin/models").json()[0]
     assert st["api"] == "both" and "tiny" in st["loaded_models"]
 
 
