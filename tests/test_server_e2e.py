@@ -268,6 +268,27 @@ def test_proxy_unknown_route_404(proxy):
     assert r.status_code == 404
 
 
+def test_allow_all_routes_fallback_proxy(fleet, tmp_path_factory):
+    """--allow-all-routes: unknown paths are PROXIED to a backend instead
+    of 404ing locally (reference main.rs:294-296 fallback router)."""
+    p = Proxy([fleet.ollama_url], tmp_path_factory.mktemp("allr"),
+              extra=["--allow-all-routes"])
+    try:
+        deadline = time.time() + 15
+        while time.time() < deadline:
+            r = httpx.get(p.base + "/admin/models").json()
+            if r and r[0]["online"]:
+                break
+            time.sleep(0.2)
+        r = httpx.get(p.base + "/definitely/not/a/route", timeout=30)
+        # the mock backend answered (its own 404 JSON), so the request
+        # was proxied rather than rejected by the dispatcher's router
+        assert r.status_code == 404
+        assert r.json() == {"error": "not found"}
+    finally:
+        p.stop()
+
+
 def test_model_aware_routing(proxy):
     # unroutable model parks in queue, then stuck-times-out; use a fresh
     # user so counters don't interfere
