@@ -186,7 +186,7 @@ void run_tui(Server& server) {
             case 'B': return 256 + 'j';  // down
             case 'C': return 256 + 'l';  // right
             case 'D': return 256 + 'h';  // left
-            case 'Z': return 256 + 'h';  // Shift-Tab: panel cycle
+            case 'Z': return 256 + 'Z';  // Shift-Tab
         }
         return 0;
     };
@@ -375,6 +375,26 @@ void run_tui(Server& server) {
                                                  : (it->second + 1) % len;
                             mcur[sel_backend] = next;
                             if (next >= 5) unfolded.insert(sel_backend);
+                            break;
+                        }
+                        focus ^= 1;
+                        break;
+                    case 'Z':
+                        // Shift-Tab: model cursor BACKWARD when the
+                        // selected backend is expanded (reference
+                        // tui.rs:288-295 forward=false), else panels
+                        if (focus == 0 && expanded.count(sel_backend) &&
+                            sel_backend < (int)s.backends.size() &&
+                            !s.backends[sel_backend].models.empty()) {
+                            const int len =
+                                (int)s.backends[sel_backend].models.size();
+                            auto it = mcur.find(sel_backend);
+                            const int prev =
+                                it == mcur.end()
+                                    ? len - 1
+                                    : (it->second + len - 1) % len;
+                            mcur[sel_backend] = prev;
+                            if (prev >= 5) unfolded.insert(sel_backend);
                             break;
                         }
                         focus ^= 1;
