@@ -292,9 +292,25 @@ static std::optional<size_t> parse_backend_selector(
             return idx;
         }
         sel = bsel->str;
+        // trim (reference control.rs:1192)
+        const auto b0 = sel.find_first_not_of(" \t");
+        if (b0 == std::string::npos) sel.clear();
+        else sel = sel.substr(b0, sel.find_last_not_of(" \t") - b0 + 1);
+    }
+    // numeric STRING is an index too (reference control.rs:1210-1211);
+    // without this, "2" would substring-match any URL containing a 2
+    if (!sel.empty() &&
+        sel.find_first_not_of("0123456789") == std::string::npos) {
+        const size_t idx = (size_t)strtoull(sel.c_str(), nullptr, 10);
+        std::lock_guard<std::mutex> g(st.backends_mu);
+        if (idx >= st.backends.size()) {
+            *err = "backend index out of range";
+            return std::nullopt;
+        }
+        return idx;
     }
     std::scoped_lock lk(st.control_mu, st.backends_mu);
-    if (sel.empty() || sel == "any") {
+    if (sel.empty() || lower(sel) == "any") {
         // first online, idle, control-capable backend with the model
         // resolvable (load) or loaded (unload)
         for (size_t i = 0; i < st.backends.size(); i++) {
@@ -315,9 +331,12 @@ static std::optional<size_t> parse_backend_selector(
         *err = "no suitable backend found for model " + model;
         return std::nullopt;
     }
-    // URL substring
-    for (size_t i = 0; i < st.backends.size(); i++)
-        if (st.backends[i].url.find(sel) != std::string::npos) return i;
+    // URL substring, case-insensitive, first match (control.rs:1213-1219)
+    const std::string sl = lower(sel);
+    for (size_t i = 0; i < st.backends.size(); i++) {
+        const std::string u = lower(st.backends[i].url);
+        if (u == sl || u.find(sl) != std::string::npos) return i;
+    }
     *err = "no backend matches selector " + sel;
     return std::nullopt;
 }

@@ -195,6 +195,30 @@ def test_any_selector_picks_resolvable(proxy, fleet):
     time.sleep(0.5)
 
 
+def test_backend_selector_semantics(proxy, fleet):
+    """Reference selector rules (control.rs:1179-1240): a numeric STRING
+    is an index (not a URL substring — "0" must not match "127.0.0.1"),
+    substring match is case-insensitive, "ANY" works."""
+    # "1" as a string indexes backend 1 (the LM Studio mock), which does
+    # not have qwen2.5 — proving it was treated as an index, because as a
+    # substring it would match backend 0's URL (digits in 127.0.0.1)
+    r = httpx.post(proxy.base + "/admin/models/load",
+                   json={"model": "qwen2.5", "backend": "1"})
+    assert r.status_code == 404, r.text          # not on backend 1
+    r = httpx.post(proxy.base + "/admin/models/load",
+                   json={"model": "qwen2.5", "backend": "ANY"})
+    assert r.status_code == 202, r.text
+    assert r.json()["backend"] == 0
+    time.sleep(0.5)
+    r = httpx.post(proxy.base + "/admin/models/load",
+                   json={"model": "qwen2.5", "backend": "127.0.0.1"})
+    assert r.status_code in (202, 409), r.text   # substring hit backend 0
+    time.sleep(0.5)
+    r = httpx.post(proxy.base + "/admin/models/load",
+                   json={"model": "qwen2.5", "backend": "no-such-url"})
+    assert r.status_code == 404
+
+
 def test_busy_backend_409(fleet, tmp_path_factory):
     """A backend mid-request refuses control ops with 409 (reference
     run.sh busy-window check via the 12 s hanging /api/chat mock)."""
