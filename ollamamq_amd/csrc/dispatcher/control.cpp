@@ -403,8 +403,9 @@ void apply_model_config(AppState& st) {
                     }
             } else {
                 for (const auto& sel : e.backends) {
+                    const std::string sl = lower(sel);
                     for (size_t i = 0; i < st.backends.size(); i++) {
-                        if (st.backends[i].url.find(sel) !=
+                        if (lower(st.backends[i].url).find(sl) !=
                                 std::string::npos ||
                             sel == std::to_string(i)) {
                             per_backend[i].push_back(e);
