@@ -49,6 +49,8 @@ def main():
     ap.add_argument("--max-batch", type=int, default=64)
     ap.add_argument("--port", type=int, default=11435)
     ap.add_argument("--host", type=str, default="127.0.0.1")
+    ap.add_argument("-t", "--timeout", type=int, default=0,
+                    help="request timeout seconds (dispatcher -t)")
     ap.add_argument("--no-tui", action="store_true")
     ap.add_argument("--sock-dir", type=str, default="/tmp")
     ap.add_argument("-c", "--model-config", type=str, default="appconf.yaml")
@@ -147,6 +149,8 @@ def main():
                   "-w", ",".join(socks), "-c", args.model_config]
     if args.extra_backends:
         server_cmd += ["-o", args.extra_backends]
+    if args.timeout > 0:
+        server_cmd += ["-t", str(args.timeout)]
     if args.no_tui:
         server_cmd.append("--no-tui")
     server = subprocess.Popen(server_cmd)
