@@ -41,7 +41,15 @@ void probe_all(AppState& st, bool full_reprobe) {
             it.idx = i;
             it.impl = st.impls[i];
             it.was_online = st.backends[i].is_online;
-            if (!full_reprobe) it.skip = st.backends[i].known_bad_endpoints;
+            // offline backends get a clean full probe every tick, and a
+            // skip set that would drop BOTH primary probes is cleared
+            // entirely (reference dispatcher.rs:427-439)
+            if (!full_reprobe && it.was_online) {
+                it.skip = st.backends[i].known_bad_endpoints;
+                if (it.skip.count("/api/tags") &&
+                    it.skip.count("/v1/models"))
+                    it.skip.clear();
+            }
             items.push_back(std::move(it));
         }
     }
