@@ -71,12 +71,17 @@ void AppState::load_blocked() {
     auto j = Json::parse(text);
     if (!j) return;
     std::lock_guard<std::mutex> g(blocked_mu);
-    if (const Json* users = j->find("blocked_users"))
-        for (const auto& u : users->arr)
-            if (u.is_str()) blocked_users.insert(u.str);
-    if (const Json* ips = j->find("blocked_ips"))
-        for (const auto& ip : ips->arr)
-            if (ip.is_str()) blocked_ips.insert(ip.str);
+    // the reference's serde field names are "users"/"ips"
+    // (src/dispatcher.rs:22-25) — a migrated blocked_items.json uses
+    // those; also accept our earlier "blocked_*" spelling
+    for (const char* key : {"users", "blocked_users"})
+        if (const Json* users = j->find(key))
+            for (const auto& u : users->arr)
+                if (u.is_str()) blocked_users.insert(u.str);
+    for (const char* key : {"ips", "blocked_ips"})
+        if (const Json* ips = j->find(key))
+            for (const auto& ip : ips->arr)
+                if (ip.is_str()) blocked_ips.insert(ip.str);
 }
 
 void AppState::save_blocked() const {
@@ -89,8 +94,9 @@ void AppState::save_blocked() const {
         Json ips = Json::array();
         for (const auto& ip : blocked_ips)
             ips.arr.push_back(Json::string(ip));
-        j.set("blocked_users", std::move(users));
-        j.set("blocked_ips", std::move(ips));
+        // write the reference's field names (src/dispatcher.rs:22-25)
+        j.set("users", std::move(users));
+        j.set("ips", std::move(ips));
     }
     std::ofstream f(blocked_path, std::ios::trunc);
     f << j.dump();

@@ -356,20 +356,25 @@ def test_stuck_timeout_503(fleet, tmp_path_factory):
 
 
 def test_blocklist_persistence(fleet, tmp_path_factory):
-    tmp = tmp_path_factory.mktemp("block")
-    with open(os.path.join(str(tmp), "blocked_items.json"), "w") as f:
-        json.dump({"blocked_users": ["evil"], "blocked_ips": []}, f)
-    p = Proxy([fleet.ollama_url], tmp)
-    try:
-        r = httpx.post(p.base + "/api/chat", json={"model": "llama3"},
-                       headers={"X-User-ID": "evil"})
-        assert r.status_code == 403
-        r = httpx.post(p.base + "/api/chat",
-                       json={"model": "llama3", "messages": []},
-                       headers={"X-User-ID": "good"}, timeout=10.0)
-        assert r.status_code == 200
-    finally:
-        p.stop()
+    """blocked_items.json uses the REFERENCE's serde field names
+    ("users"/"ips", src/dispatcher.rs:22-25) so a migrated file works
+    unchanged; our earlier "blocked_*" spelling is still accepted."""
+    for fields in ({"users": ["evil"], "ips": []},
+                   {"blocked_users": ["evil"], "blocked_ips": []}):
+        tmp = tmp_path_factory.mktemp("block")
+        with open(os.path.join(str(tmp), "blocked_items.json"), "w") as f:
+            json.dump(fields, f)
+        p = Proxy([fleet.ollama_url], tmp)
+        try:
+            r = httpx.post(p.base + "/api/chat", json={"model": "llama3"},
+                           headers={"X-User-ID": "evil"})
+            assert r.status_code == 403, fields
+            r = httpx.post(p.base + "/api/chat",
+                           json={"model": "llama3", "messages": []},
+                           headers={"X-User-ID": "good"}, timeout=10.0)
+            assert r.status_code == 200
+        finally:
+            p.stop()
 
 
 def test_admin_stats_surface(proxy):
