@@ -168,8 +168,12 @@ ControlOutcome start_model_control(AppState& st, const ControlRequest& req) {
     body.set("action", Json::string(req.action == ControlAction::Load
                                         ? "load"
                                         : "unload"));
-    body.set("backend", Json::number((double)req.backend_idx));
+    // reference echoes the backend URL, not the index (control.rs:1283)
+    body.set("backend", Json::string(snap.url));
+    body.set("backend_index", Json::number((double)req.backend_idx));
     body.set("model", Json::string(canonical));
+    if (!req.identifier.empty())
+        body.set("identifier", Json::string(req.identifier));
     if (req.num_ctx > 0) body.set("num_ctx", Json::number((double)req.num_ctx));
     if (req.action == ControlAction::Load)
         body.set("keep_alive",
@@ -366,6 +370,7 @@ static ControlOutcome admin_control(AppState& st, const std::string& body,
     req.backend_idx = *idx;
     req.num_ctx = (int64_t)j->get_num("num_ctx", 0);
     req.keep_alive = (int64_t)j->get_num("keep_alive", 0);
+    req.identifier = j->get_str("identifier");
     // apply per-model config defaults (max_ctx/keep_alive) when present
     {
         std::lock_guard<std::mutex> g(st.models_mu);

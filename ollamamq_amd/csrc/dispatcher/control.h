@@ -19,6 +19,7 @@ struct ControlRequest {
     size_t backend_idx;
     int64_t num_ctx = 0;       // 0 = unspecified
     int64_t keep_alive = 0;    // 0 = default
+    std::string identifier;    // optional friendly name (echoed in 202)
 };
 
 struct ControlOutcome {
