@@ -129,7 +129,10 @@ ControlOutcome start_model_control(AppState& st, const ControlRequest& req) {
                                        ? "load "
                                        : "unload ") +
                            canonical + " on b" +
-                           std::to_string(req.backend_idx));
+                           std::to_string(req.backend_idx) +
+                           (req.identifier.empty()
+                                ? ""
+                                : " [id: " + req.identifier + "]"));
     st.notify();  // backend is now scheduler-busy
 
     const ControlRequest r = req;
@@ -485,7 +488,7 @@ void apply_model_config(AppState& st) {
                     std::this_thread::sleep_for(std::chrono::seconds(1));
                 }
                 ControlRequest lr{ControlAction::Load, canonical, idx,
-                                  e.max_ctx, e.keep_alive};
+                                  e.max_ctx, e.keep_alive, e.identifier};
                 start_model_control(st, lr);
                 // wait for completion before next entry on this backend
                 for (int i = 0; i < 6000; i++) {
