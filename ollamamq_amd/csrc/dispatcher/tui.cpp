@@ -75,10 +75,22 @@ Snapshot capture(AppState& st) {
             }
             std::sort(e.models.begin(), e.models.end());
             auto op = st.control_ops.find(i);
-            if (op != st.control_ops.end())
+            if (op != st.control_ops.end()) {
                 e.op = (op->second.action == ControlAction::Load ? "load "
                                                                  : "unload ") +
                        op->second.model;
+            } else {
+                // flash the latest completed result for 10 s (reference
+                // control.rs:33-37,91-101 history visibility)
+                for (auto it = st.control_history.rbegin();
+                     it != st.control_history.rend(); ++it) {
+                    if (it->backend != i) continue;
+                    if (now_ms() - it->finished_ms < 10000)
+                        e.op = std::string(it->ok ? "done: " : "FAILED: ") +
+                               it->model;
+                    break;
+                }
+            }
             s.backends.push_back(std::move(e));
         }
     }
