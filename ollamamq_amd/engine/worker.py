@@ -286,11 +286,13 @@ class Conn:
     def _recv_line(self):
         buf = b""
         while b"\n" not in buf:
+            if len(buf) > (1 << 30):  # dispatcher enforces a 1 GB body cap
+                return None
             chunk = self.sock.recv(65536)
             if not chunk:
                 return None
             buf += chunk
-        return buf.split(b"\n", 1)[0].decode()
+        return buf.split(b"\n", 1)[0].decode(errors="replace")
 
     def _line(self, obj):
         self.sock.sendall((json.dumps(obj) + "\n").encode())
