@@ -551,11 +551,28 @@ class Conn:
                 self.sock.sendall(json.dumps(
                     {"error": f"model not found: {m}"}).encode())
                 return
-            obj = {"details": {"family": "llama",
-                               "parameter_size": cfg.name},
-                   "model_info": {"n_layers": cfg.n_layers,
-                                  "hidden": cfg.hidden,
-                                  "context_length": cfg.max_ctx}}
+            # Ollama show schema: SDKs read details.* and the
+            # "<arch>.<key>" model_info namespace
+            obj = {
+                "modelfile": "",
+                "parameters": f"num_ctx {cfg.max_ctx}",
+                "template": "{{ .Prompt }}",
+                "details": {"format": "safetensors", "family": "llama",
+                            "families": ["llama"],
+                            "parameter_size": cfg.name,
+                            "quantization_level": "BF16"},
+                "model_info": {
+                    "general.architecture": "llama",
+                    "llama.block_count": cfg.n_layers,
+                    "llama.embedding_length": cfg.hidden,
+                    "llama.context_length": cfg.max_ctx,
+                    "llama.feed_forward_length": cfg.ffn,
+                    "llama.attention.head_count": cfg.n_heads,
+                    "llama.attention.head_count_kv": cfg.n_kv_heads,
+                    "llama.vocab_size": cfg.vocab,
+                },
+                "capabilities": ["completion"],
+            }
         else:
             obj = {"status": "ollamamq-amd worker",
                    "device": w.device, "loaded": loaded}

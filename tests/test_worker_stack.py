@@ -370,7 +370,10 @@ def test_api_show_known_and_unknown(stack):
     r = httpx.post(stack + "/api/show", json={"model": "tiny-cpu"},
                    timeout=30.0)
     assert r.status_code == 200
-    assert r.json()["model_info"]["hidden"] == 256
+    obj = r.json()
+    assert obj["model_info"]["llama.embedding_length"] == 256
+    assert obj["details"]["family"] == "llama"
+    assert "template" in obj and "capabilities" in obj
     # an UNROUTABLE model parks in the dispatcher queue (reference
     # semantics) — the worker-level 404 needs a name that routes (fuzzy
     # substring) but is ambiguous at the worker resolver ("t" matches
