@@ -518,17 +518,35 @@ class Conn:
         with w.lock:
             loaded = list(w.engines.keys())
             ctx = dict(w.loaded_ctx)
+        def entry(m, res=False):
+            # Ollama tags/ps entry schema (name/model/size/digest/details)
+            import hashlib
+            cfg = PRESETS.get(m)
+            qd = cfg.n_heads * cfg.head_dim if cfg else 0
+            kvd = cfg.n_kv_heads * cfg.head_dim if cfg else 0
+            size = 2 * (cfg.vocab * cfg.hidden * 2 + cfg.n_layers *
+                        (cfg.hidden * (qd + 2 * kvd) + qd * cfg.hidden +
+                         3 * cfg.hidden * cfg.ffn)) if cfg else 0
+            e = {"name": m, "model": m, "size": size,
+                 "digest": hashlib.sha256(m.encode()).hexdigest(),
+                 "modified_at": _now_iso(),
+                 "details": {"format": "safetensors", "family": "llama",
+                             "families": ["llama"], "parameter_size": m,
+                             "quantization_level": "BF16"}}
+            if res:
+                e["size_vram"] = size
+                e["expires_at"] = _now_iso()
+                e["context_length"] = ctx.get(m, 0)
+            return e
+
         if path == "/api/tags":
-            obj = {"models": [{"name": m, "model": m}
-                              for m in w.available_models()]}
+            obj = {"models": [entry(m) for m in w.available_models()]}
         elif path == "/v1/models":
             obj = {"object": "list",
                    "data": [{"id": m, "object": "model"}
                             for m in w.available_models()]}
         elif path == "/api/ps":
-            obj = {"models": [{"name": m, "model": m,
-                               "context_length": ctx.get(m, 0)}
-                              for m in loaded]}
+            obj = {"models": [entry(m, res=True) for m in loaded]}
         elif path == "/api/version":
             obj = {"version": VERSION}
         elif path.startswith("/v1/models/"):
