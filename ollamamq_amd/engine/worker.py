@@ -390,12 +390,15 @@ class Conn:
 
         pieces = []
         n_out = 0
+        t_first = None
         try:
             stopped = False
             while True:
                 t, done = q.get(timeout=600)
                 if done:
                     break
+                if t_first is None:
+                    t_first = time.time()
                 n_out += 1
                 piece = tok.decode_one(t)
                 pieces.append(piece)
@@ -408,8 +411,10 @@ class Conn:
                         w.cancel(model, sid)
                 if stream and not stopped:
                     self._stream_piece(path, model, piece, openai)
+            reason = "stop" if (stopped or n_out < params.max_tokens) \
+                else "length"
             self._final(path, model, pieces, n_out, t0, openai, stream,
-                        len(prompt))
+                        len(prompt), reason, t_first)
             # Ollama parity: "keep_alive": 0 on the request frees the
             # model after the response (only if the engine is idle —
             # other users' in-flight sequences always win)
@@ -442,19 +447,24 @@ class Conn:
             self.sock.sendall((json.dumps(obj) + "\n").encode())
 
     def _final(self, path, model, pieces, n_out, t0, openai, stream,
-               n_prompt):
-        dur_ns = int((time.time() - t0) * 1e9)
+               n_prompt, reason="stop", t_first=None):
+        t_end = time.time()
+        dur_ns = int((t_end - t0) * 1e9)
+        if t_first is None:
+            t_first = t_end
+        prompt_ns = int((t_first - t0) * 1e9)
+        eval_ns = max(0, dur_ns - prompt_ns)
         text = "".join(pieces)
         if openai:
             if stream:
                 obj = {"object": "chat.completion.chunk", "model": model,
                        "choices": [{"index": 0, "delta": {},
-                                    "finish_reason": "stop"}]}
+                                    "finish_reason": reason}]}
                 self.sock.sendall(f"data: {json.dumps(obj)}\n\n".encode())
                 self.sock.sendall(b"data: [DONE]\n\n")
             else:
                 key = ("text" if path == "/v1/completions" else "message")
-                choice = {"index": 0, "finish_reason": "stop"}
+                choice = {"index": 0, "finish_reason": reason}
                 if key == "text":
                     choice["text"] = text
                 else:
@@ -470,10 +480,13 @@ class Conn:
                 self.sock.sendall(json.dumps(obj).encode())
         else:
             obj = {"model": model, "created_at": _now_iso(),
-                   "done": True, "done_reason": "stop",
+                   "done": True, "done_reason": reason,
                    "total_duration": dur_ns,
+                   "load_duration": 0,
                    "prompt_eval_count": n_prompt,
-                   "eval_count": n_out}
+                   "prompt_eval_duration": prompt_ns,
+                   "eval_count": n_out,
+                   "eval_duration": eval_ns}
             if not stream:
                 if path == "/api/generate":
                     obj["response"] = text

@@ -96,6 +96,12 @@ def test_ollama_generate_nonstream(stack):
     obj = json.loads(r.text.strip())
     assert obj["done"] is True
     assert isinstance(obj["response"], str)
+    # Ollama final-chunk schema: counts, durations, done_reason (here
+    # "length": generation ran to num_predict)
+    assert obj["done_reason"] == "length"
+    assert obj["eval_count"] == 4 and obj["prompt_eval_count"] == 3
+    assert obj["total_duration"] >= obj["eval_duration"] >= 0
+    assert "prompt_eval_duration" in obj and "load_duration" in obj
 
 
 def test_openai_chat_completion(stack):
