@@ -377,6 +377,10 @@ class Conn:
 
         q: "queue.Queue" = queue.Queue()
         t0 = time.time()
+        # OpenAI SDKs validate id/created on every chunk
+        self._oai_id = ("cmpl-" if path == "/v1/completions"
+                        else "chatcmpl-") + f"{int(t0 * 1e6):x}"
+        self._oai_created = int(t0)
         sid = w.generate(model, prompt, params,
                          lambda t, done: q.put((t, done)),
                          num_ctx=int((body.get("options") or {})
@@ -429,12 +433,16 @@ class Conn:
     def _stream_piece(self, path, model, piece, openai):
         if openai:
             if path == "/v1/completions":
-                obj = {"object": "text_completion", "model": model,
-                       "choices": [{"index": 0, "text": piece}]}
+                obj = {"id": self._oai_id, "created": self._oai_created,
+                       "object": "text_completion", "model": model,
+                       "choices": [{"index": 0, "text": piece,
+                                    "finish_reason": None}]}
             else:
-                obj = {"object": "chat.completion.chunk", "model": model,
+                obj = {"id": self._oai_id, "created": self._oai_created,
+                       "object": "chat.completion.chunk", "model": model,
                        "choices": [{"index": 0,
-                                    "delta": {"content": piece}}]}
+                                    "delta": {"content": piece},
+                                    "finish_reason": None}]}
             self.sock.sendall(f"data: {json.dumps(obj)}\n\n".encode())
         else:
             if path == "/api/generate":
@@ -457,7 +465,9 @@ class Conn:
         text = "".join(pieces)
         if openai:
             if stream:
-                obj = {"object": "chat.completion.chunk", "model": model,
+                obj = {"id": self._oai_id,
+                       "created": self._oai_created,
+                       "object": "chat.completion.chunk", "model": model,
                        "choices": [{"index": 0, "delta": {},
                                     "finish_reason": reason}]}
                 self.sock.sendall(f"data: {json.dumps(obj)}\n\n".encode())
@@ -470,7 +480,9 @@ class Conn:
                 else:
                     choice["message"] = {"role": "assistant",
                                          "content": text}
-                obj = {"object": ("text_completion"
+                obj = {"id": self._oai_id,
+                       "created": self._oai_created,
+                       "object": ("text_completion"
                                   if path == "/v1/completions"
                                   else "chat.completion"),
                        "model": model, "choices": [choice],

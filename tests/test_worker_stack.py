@@ -113,6 +113,9 @@ def test_openai_chat_completion(stack):
     obj = r.json()
     assert obj["choices"][0]["message"]["role"] == "assistant"
     assert obj["usage"]["completion_tokens"] >= 1
+    # OpenAI SDK pydantic models require these
+    assert obj["id"].startswith("chatcmpl-") and obj["created"] > 0
+    assert obj["choices"][0]["finish_reason"] in ("stop", "length")
 
 
 def test_openai_stream_sse(stack):
