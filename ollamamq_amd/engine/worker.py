@@ -522,18 +522,27 @@ class Conn:
         raw = body.get("input", body.get("prompt", ""))
         inputs = raw if isinstance(raw, list) else [raw]
         tok = w.tokenizers[model]
+        t0 = time.time()
         vecs = []
+        n_tok = 0
         with w.lock:
             eng = w.engines[model]
             for text in inputs:
-                vecs.append(eng.embed(tok.encode(str(text))))
+                ids = tok.encode(str(text))
+                n_tok += len(ids)
+                vecs.append(eng.embed(ids))
         self._line({"status": 200, "content_type": "application/json"})
         if path == "/v1/embeddings":
             obj = {"object": "list", "model": model,
                    "data": [{"object": "embedding", "index": i,
-                             "embedding": v} for i, v in enumerate(vecs)]}
+                             "embedding": v} for i, v in enumerate(vecs)],
+                   "usage": {"prompt_tokens": n_tok,
+                             "total_tokens": n_tok}}
         elif path == "/api/embed":
-            obj = {"model": model, "embeddings": vecs}
+            obj = {"model": model, "embeddings": vecs,
+                   "total_duration": int((time.time() - t0) * 1e9),
+                   "load_duration": 0,
+                   "prompt_eval_count": n_tok}
         else:  # legacy /api/embeddings
             obj = {"model": model, "embedding": vecs[0] if vecs else []}
         self.sock.sendall(json.dumps(obj).encode())
