@@ -577,7 +577,9 @@ class Conn:
             obj = {"models": [entry(m) for m in w.available_models()]}
         elif path == "/v1/models":
             obj = {"object": "list",
-                   "data": [{"id": m, "object": "model"}
+                   "data": [{"id": m, "object": "model",
+                             "created": int(self.worker.started),
+                             "owned_by": "ollamamq-amd"}
                             for m in w.available_models()]}
         elif path == "/api/ps":
             obj = {"models": [entry(m, res=True) for m in loaded]}
@@ -592,7 +594,9 @@ class Conn:
                 self.sock.sendall(json.dumps(
                     {"error": f"model not found: {m}"}).encode())
                 return
-            obj = {"id": r, "object": "model"}
+            obj = {"id": r, "object": "model",
+                   "created": int(self.worker.started),
+                   "owned_by": "ollamamq-amd"}
         elif path == "/api/show":
             m = body.get("model", "")
             cfg = PRESETS.get(m) or PRESETS.get(w.resolve(m) or "")
