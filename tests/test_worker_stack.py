@@ -395,4 +395,6 @@ def test_api_show_known_and_unknown(stack):
 def test_v1_models_item_route(stack):
     r = httpx.get(stack + "/v1/models/tiny-cpu", timeout=30.0)
     assert r.status_code == 200
-    assert r.json() == {"id": "tiny-cpu", "object": "model"}
+    obj = r.json()
+    assert obj["id"] == "tiny-cpu" and obj["object"] == "model"
+    assert obj["created"] > 0 and obj["owned_by"]
