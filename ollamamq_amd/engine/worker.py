@@ -417,8 +417,10 @@ class Conn:
                     self._stream_piece(path, model, piece, openai)
             reason = "stop" if (stopped or n_out < params.max_tokens) \
                 else "length"
+            want_usage = bool((body.get("stream_options") or {})
+                              .get("include_usage"))
             self._final(path, model, pieces, n_out, t0, openai, stream,
-                        len(prompt), reason, t_first)
+                        len(prompt), reason, t_first, want_usage)
             # Ollama parity: "keep_alive": 0 on the request frees the
             # model after the response (only if the engine is idle —
             # other users' in-flight sequences always win)
@@ -455,7 +457,7 @@ class Conn:
             self.sock.sendall((json.dumps(obj) + "\n").encode())
 
     def _final(self, path, model, pieces, n_out, t0, openai, stream,
-               n_prompt, reason="stop", t_first=None):
+               n_prompt, reason="stop", t_first=None, want_usage=False):
         t_end = time.time()
         dur_ns = int((t_end - t0) * 1e9)
         if t_first is None:
@@ -471,6 +473,18 @@ class Conn:
                        "choices": [{"index": 0, "delta": {},
                                     "finish_reason": reason}]}
                 self.sock.sendall(f"data: {json.dumps(obj)}\n\n".encode())
+                if want_usage:
+                    # OpenAI stream_options.include_usage: one final
+                    # usage-only chunk before [DONE]
+                    u = {"id": self._oai_id,
+                         "created": self._oai_created,
+                         "object": "chat.completion.chunk",
+                         "model": model, "choices": [],
+                         "usage": {"prompt_tokens": n_prompt,
+                                   "completion_tokens": n_out,
+                                   "total_tokens": n_prompt + n_out}}
+                    self.sock.sendall(
+                        f"data: {json.dumps(u)}\n\n".encode())
                 self.sock.sendall(b"data: [DONE]\n\n")
             else:
                 key = ("text" if path == "/v1/completions" else "message")

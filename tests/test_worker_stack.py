@@ -130,6 +130,23 @@ def test_openai_stream_sse(stack):
     assert body.strip().endswith("data: [DONE]")
 
 
+def test_openai_stream_include_usage(stack):
+    """stream_options.include_usage: a usage-only chunk arrives before
+    [DONE] (OpenAI streaming spec)."""
+    with httpx.stream(
+            "POST", stack + "/v1/chat/completions",
+            json={"model": "tiny-cpu", "max_tokens": 4, "stream": True,
+                  "stream_options": {"include_usage": True},
+                  "messages": [{"role": "user", "content": "yo"}]},
+            timeout=120.0) as r:
+        assert r.status_code == 200
+        body = "".join(r.iter_text())
+    chunks = [json.loads(l[6:]) for l in body.splitlines()
+              if l.startswith("data: ") and l != "data: [DONE]"]
+    assert chunks[-1]["choices"] == []
+    assert chunks[-1]["usage"]["completion_tokens"] == 4
+
+
 def test_tags_and_ps_served_by_worker(stack):
     r = httpx.get(stack + "/api/tags", timeout=30.0)
     assert r.status_code == 200
