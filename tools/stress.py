@@ -27,6 +27,20 @@ ENDPOINTS = ["/api/generate", "/api/chat", "/v1/chat/completions",
 
 def one_request(base, user, model, rng, cancel_pct, max_tokens,
                 sampled_pct=0.0):
+    # ~5% of traffic hits the embedding endpoints (mixed workload)
+    if rng.random() < 0.05:
+        ep = rng.choice(["/api/embed", "/v1/embeddings"])
+        body = {"model": model,
+                "input": ["".join(rng.choice("abcde ") for _ in range(20))]}
+        t0 = time.monotonic()
+        try:
+            r = httpx.post(base + ep, json=body,
+                           headers={"X-User-ID": user}, timeout=60)
+            return {"user": user, "status": r.status_code, "tokens": 0,
+                    "ttft": None, "dur": time.monotonic() - t0}
+        except Exception as e:
+            return {"user": user, "status": -1, "error": str(e),
+                    "tokens": 0, "ttft": None, "dur": 0.0}
     ep = rng.choice(ENDPOINTS)
     openai = ep.startswith("/v1/")
     prompt = "".join(rng.choice("abcdefghij ") for _ in range(rng.randint(8, 200)))
