@@ -177,8 +177,22 @@ void HttpServer::accept_loop() {
         }
         char ip[64] = "unknown";
         inet_ntop(AF_INET, &peer.sin_addr, ip, sizeof ip);
-        std::thread(&HttpServer::handle_conn, this, fd, std::string(ip))
-            .detach();
+        // shed load above the cap: closing immediately keeps accept
+        // responsive for well-behaved clients (serving needs ≪1k
+        // concurrent streams; half-open floods go far beyond that)
+        if (live_conns_.load() >= 2048) {
+            ::close(fd);
+            continue;
+        }
+        live_conns_.fetch_add(1);
+        try {
+            std::thread(&HttpServer::handle_conn, this, fd,
+                        std::string(ip))
+                .detach();
+        } catch (const std::system_error&) {
+            live_conns_.fetch_sub(1);  // thread exhaustion: shed
+            ::close(fd);
+        }
     }
 }
 
@@ -199,12 +213,17 @@ static bool read_headers(int fd, std::string& buf, size_t& header_end) {
 }
 
 void HttpServer::handle_conn(int fd, std::string peer_ip) {
+    struct Live {                      // RAII: cap accounting
+        std::atomic<int>& n;
+        ~Live() { n.fetch_sub(1); }
+    } live{live_conns_};
     int one = 1;
     setsockopt(fd, IPPROTO_TCP, TCP_NODELAY, &one, sizeof one);
     // read timeout: a client trickling headers/body (slowloris) must not
-    // pin a connection thread forever; writes stay unbounded so slow
-    // CONSUMERS of long token streams are unaffected
-    timeval rto{120, 0};
+    // pin a connection thread for long — this is a per-recv IDLE bound,
+    // not a total-request bound, so legit slow bodies still flow; 30 s
+    // (not 120) keeps the half-open thread population small under floods
+    timeval rto{30, 0};
     setsockopt(fd, SOL_SOCKET, SO_RCVTIMEO, &rto, sizeof rto);
     std::string buf;
     while (!stopping_) {

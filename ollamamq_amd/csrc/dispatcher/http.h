@@ -72,6 +72,9 @@ private:
     int listen_fd_ = -1;
     std::thread accept_thread_;
     std::atomic<bool> stopping_{false};
+    // liveness guard: thread-per-connection needs a hard cap — a storm
+    // of half-open connections must shed load instead of starving accept
+    std::atomic<int> live_conns_{0};
 };
 
 // ------------------------------------------------------------- client side

@@ -178,6 +178,36 @@ def test_rogue_backend_chunked_overflow(tmp_path):
         p.wait(timeout=10)
 
 
+def test_half_open_flood_no_wedge(tmp_path):
+    """Hundreds of half-open connections (headers never finished) must
+    not starve the accept loop: health stays served during an under-cap
+    flood, and the server recovers immediately once sockets close
+    (pre-fix: thread pile-up with a 120 s idle timeout eventually made
+    connect() itself time out)."""
+    p = subprocess.Popen(
+        [BIN, "--no-tui", "-p", "0",
+         "-c", os.path.join(str(tmp_path), "absent.yaml")],
+        stderr=subprocess.PIPE, cwd=str(tmp_path), text=True)
+    try:
+        port = int(p.stderr.readline().rsplit(":", 1)[1].split()[0])
+        held = []
+        for _ in range(300):            # well under the 2048 cap
+            s = socket.create_connection(("127.0.0.1", port), timeout=5)
+            s.sendall(b"POST /api/chat HTTP/1.1\r\nContent-Length: 9\r\n")
+            held.append(s)
+        # health must answer WHILE 300 threads sit in read_headers
+        for _ in range(3):
+            assert alive(port)
+        for s in held:
+            s.close()
+        time.sleep(0.5)
+        assert alive(port)
+        assert p.poll() is None
+    finally:
+        p.terminate()
+        p.wait(timeout=10)
+
+
 def test_deeply_nested_json_body(srv):
     """A 100k-deep JSON nesting bomb must be rejected by the parser's
     depth guard, not blow the stack."""
